@@ -1,0 +1,358 @@
+/*
+ * oracle.c — CPU restatement of the reference's distributed-join algorithm.
+ *
+ * *** TEST INFRASTRUCTURE ONLY ***
+ * Only tests/, __graft_entry__.smoke() and bench.py's cpu_baseline leg may
+ * call, link or execute this code, and there only as the checker / reported
+ * CPU baseline — never as the thing measured as the product or shipped. The
+ * product path (distributed_join_amd) must fail loudly if its HIP extension
+ * is missing; it never routes through this file.
+ *
+ * What is restated, and from where (reference = /root/reference, read-only):
+ *  - inner-join semantics: cudf::inner_join(left, right, {0}, {0}) as the
+ *    reference calls it (src/distributed_join.cpp:71-83): output rows are all
+ *    (i, j) pairs with left_key[i] == right_key[j]; columns are all left
+ *    columns then all right columns (join key duplicated — pinned by
+ *    test/compare_against_single_gpu.cu:163-165); row order unspecified
+ *    (the reference's tests sort before comparing, :167-174); joining with an
+ *    empty side yields an empty table (distributed_join.cpp:76-83).
+ *  - partition placement: hash % nparts per dj_hash.h (the reference's
+ *    MurmurHash3 placement is parity-unpinned — SURVEY.md §8c).
+ *  - input distributions: generate_dataset.cuh:137-162 via the deterministic
+ *    counter-based restatement in dj_rng.h (shared with the HIP path).
+ *  - the end-to-end pipeline (partition -> exchange -> local join -> concat,
+ *    src/distributed_join.cpp:134-340) is join-result-invariant, so the
+ *    oracle computes the global join directly; per-stage parity (placement,
+ *    per-rank slices) is exercised by oracle_partition.
+ *
+ * Oracle pinning: the reference cannot be built here (CUDA 11 + cuDF 0.19 +
+ * UCX + MPI absent — SURVEY.md §8c). The oracle is pinned instead against the
+ * reference's own analytical known-answer tests, restated in tests/:
+ * multiples-of-3/5 join invariants (test/compare_against_analytical.cu:44-54,
+ * 152) and the string-payload invariants (test/string_payload.cu:96-106,
+ * 141-163), plus committed golden fixtures under tests/golden/.
+ *
+ * Build: oracle/Makefile (gcc -O2 -fopenmp). The OpenMP radix-partition join
+ * (oracle_cpu_radix_join) is the `cpu_baseline` leg of bench.py per
+ * BASELINE.md's CPU-baseline plan.
+ */
+#include "../distributed_join_amd/csrc/dj_rng.h"
+#include "../distributed_join_amd/csrc/dj_hash.h"
+
+#include <stdint.h>
+#include <stdlib.h>
+#include <string.h>
+
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+#define ORACLE_API __attribute__((visibility("default")))
+
+/* ---------------------------------------------------------------- inputs */
+
+ORACLE_API void oracle_gen_build(int64_t *keys,
+                                 int64_t *payloads,
+                                 int64_t build_n_global,
+                                 int64_t rand_max,
+                                 uint64_t seed,
+                                 int uniq,
+                                 int64_t row0,
+                                 int64_t nrows)
+{
+  for (int64_t t = 0; t < nrows; t++) {
+    int64_t i = row0 + t;
+    keys[t] = uniq ? dj_build_key((uint64_t)i, (uint64_t)build_n_global, rand_max, seed)
+                   : dj_build_key_nonuniq((uint64_t)i, rand_max, seed);
+    if (payloads) payloads[t] = i;
+  }
+}
+
+ORACLE_API void oracle_gen_probe(int64_t *keys,
+                                 int64_t *payloads,
+                                 int64_t build_n_global,
+                                 int64_t rand_max,
+                                 double selectivity,
+                                 uint64_t seed,
+                                 int64_t row0,
+                                 int64_t nrows)
+{
+  for (int64_t t = 0; t < nrows; t++) {
+    int64_t j = row0 + t;
+    keys[t]   = dj_probe_key((uint64_t)j, (uint64_t)build_n_global, rand_max, selectivity, seed);
+    if (payloads) payloads[t] = j;
+  }
+}
+
+ORACLE_API uint32_t oracle_row_hash(int64_t key, int hash_fn, uint32_t seed)
+{
+  return dj_row_hash(key, hash_fn, seed);
+}
+
+/* ------------------------------------------------------------- partition */
+/*
+ * Stable partition of (keys, payloads) into nparts contiguous ranges by
+ * dj_row_hash % nparts. offsets has nparts+1 entries (offsets[0] = 0).
+ * Mirrors the contract of cudf::hash_partition as the reference uses it
+ * (distributed_join.cpp:211-226): stable within a partition, offsets returned.
+ */
+ORACLE_API void oracle_partition(const int64_t *keys,
+                                 const int64_t *payloads,
+                                 int64_t n,
+                                 int nparts,
+                                 int hash_fn,
+                                 uint32_t hash_seed,
+                                 int64_t *out_keys,
+                                 int64_t *out_payloads,
+                                 int64_t *offsets)
+{
+  int64_t *count = (int64_t *)calloc((size_t)nparts, sizeof(int64_t));
+  for (int64_t i = 0; i < n; i++) count[dj_row_hash(keys[i], hash_fn, hash_seed) % (uint32_t)nparts]++;
+  offsets[0] = 0;
+  for (int p = 0; p < nparts; p++) offsets[p + 1] = offsets[p] + count[p];
+  int64_t *cursor = (int64_t *)malloc((size_t)nparts * sizeof(int64_t));
+  memcpy(cursor, offsets, (size_t)nparts * sizeof(int64_t));
+  for (int64_t i = 0; i < n; i++) {
+    uint32_t p   = dj_row_hash(keys[i], hash_fn, hash_seed) % (uint32_t)nparts;
+    int64_t dst  = cursor[p]++;
+    out_keys[dst] = keys[i];
+    if (out_payloads) out_payloads[dst] = payloads[i];
+  }
+  free(cursor);
+  free(count);
+}
+
+/* ------------------------------------------------------------ inner join */
+
+static uint64_t next_pow2_u64(uint64_t x)
+{
+  uint64_t p = 1;
+  while (p < x) p <<= 1;
+  return p;
+}
+
+/*
+ * Sequential open-addressing inner join (the parity checker).
+ * Builds on (lk, lp), probes with (rk, rp). Duplicate build keys each occupy
+ * their own slot; a probe walks from the home slot to the first empty slot
+ * and emits one output row per equal-key slot — the full cross product per
+ * key, matching cudf::inner_join semantics.
+ * Output columns: out_c0=lkey, out_c1=lpay, out_c2=rkey, out_c3=rpay
+ * (left columns then right columns, key duplicated).
+ * Returns the number of output rows; writes at most cap rows. If the result
+ * exceeds cap, keeps counting (so the caller can re-try with a larger buffer)
+ * but stops writing.
+ */
+ORACLE_API int64_t oracle_inner_join(const int64_t *lk,
+                                     const int64_t *lp,
+                                     int64_t ln,
+                                     const int64_t *rk,
+                                     const int64_t *rp,
+                                     int64_t rn,
+                                     int64_t *out_c0,
+                                     int64_t *out_c1,
+                                     int64_t *out_c2,
+                                     int64_t *out_c3,
+                                     int64_t cap)
+{
+  if (ln == 0 || rn == 0) return 0; /* empty side => empty result */
+  uint64_t nslots = next_pow2_u64((uint64_t)ln * 2 + 1);
+  uint64_t mask   = nslots - 1;
+  int64_t *slot_key = (int64_t *)malloc(nslots * sizeof(int64_t));
+  int64_t *slot_val = (int64_t *)malloc(nslots * sizeof(int64_t));
+  uint8_t *slot_used = (uint8_t *)calloc(nslots, 1);
+
+  for (int64_t i = 0; i < ln; i++) {
+    uint64_t s = dj_mix64((uint64_t)lk[i]) & mask;
+    while (slot_used[s]) s = (s + 1) & mask;
+    slot_used[s] = 1;
+    slot_key[s]  = lk[i];
+    slot_val[s]  = lp ? lp[i] : i;
+  }
+
+  int64_t nout = 0;
+  for (int64_t j = 0; j < rn; j++) {
+    int64_t key = rk[j];
+    uint64_t s  = dj_mix64((uint64_t)key) & mask;
+    while (slot_used[s]) {
+      if (slot_key[s] == key) {
+        if (nout < cap) {
+          out_c0[nout] = key;
+          out_c1[nout] = slot_val[s];
+          out_c2[nout] = key;
+          out_c3[nout] = rp ? rp[j] : j;
+        }
+        nout++;
+      }
+      s = (s + 1) & mask;
+    }
+  }
+  free(slot_key);
+  free(slot_val);
+  free(slot_used);
+  return nout;
+}
+
+/* ------------------------------------- OpenMP radix-partition join (CPU baseline) */
+/*
+ * The `cpu_baseline` leg per BASELINE.md: radix-partition both tables into
+ * buckets by the top bits of dj_mix64(key), then per-bucket open-addressing
+ * build+probe, buckets processed in parallel. Same output semantics as
+ * oracle_inner_join (order unspecified). Returns output row count; rows are
+ * written compacted per bucket into the caller's buffers when cap allows.
+ * nthreads <= 0 means OpenMP default.
+ */
+ORACLE_API int64_t oracle_cpu_radix_join(const int64_t *lk,
+                                         const int64_t *lp,
+                                         int64_t ln,
+                                         const int64_t *rk,
+                                         const int64_t *rp,
+                                         int64_t rn,
+                                         int64_t *out_c0,
+                                         int64_t *out_c1,
+                                         int64_t *out_c2,
+                                         int64_t *out_c3,
+                                         int64_t cap,
+                                         int nthreads)
+{
+  if (ln == 0 || rn == 0) return 0;
+#ifdef _OPENMP
+  if (nthreads > 0) omp_set_num_threads(nthreads);
+#endif
+  const int RB     = 8; /* 256 buckets */
+  const int NB     = 1 << RB;
+  int64_t *lcnt    = (int64_t *)calloc(NB, sizeof(int64_t));
+  int64_t *rcnt    = (int64_t *)calloc(NB, sizeof(int64_t));
+  int64_t *loff    = (int64_t *)malloc((NB + 1) * sizeof(int64_t));
+  int64_t *roff    = (int64_t *)malloc((NB + 1) * sizeof(int64_t));
+  int64_t *lk2     = (int64_t *)malloc((size_t)ln * sizeof(int64_t));
+  int64_t *lp2     = (int64_t *)malloc((size_t)ln * sizeof(int64_t));
+  int64_t *rk2     = (int64_t *)malloc((size_t)rn * sizeof(int64_t));
+  int64_t *rp2     = (int64_t *)malloc((size_t)rn * sizeof(int64_t));
+
+#define BUCKET(key) ((int)(dj_mix64((uint64_t)(key)) >> (64 - RB)))
+
+#pragma omp parallel
+  {
+#pragma omp for reduction(+ : lcnt[:NB])
+    for (int64_t i = 0; i < ln; i++) lcnt[BUCKET(lk[i])]++;
+#pragma omp for reduction(+ : rcnt[:NB])
+    for (int64_t j = 0; j < rn; j++) rcnt[BUCKET(rk[j])]++;
+  }
+  loff[0] = roff[0] = 0;
+  for (int b = 0; b < NB; b++) {
+    loff[b + 1] = loff[b] + lcnt[b];
+    roff[b + 1] = roff[b] + rcnt[b];
+  }
+  /* scatter (single-threaded cursor scatter; partition cost is reported as
+   * part of the baseline, matching the GPU path which also partitions) */
+  {
+    int64_t *lc = (int64_t *)malloc(NB * sizeof(int64_t));
+    int64_t *rc = (int64_t *)malloc(NB * sizeof(int64_t));
+    memcpy(lc, loff, NB * sizeof(int64_t));
+    memcpy(rc, roff, NB * sizeof(int64_t));
+    for (int64_t i = 0; i < ln; i++) {
+      int64_t d = lc[BUCKET(lk[i])]++;
+      lk2[d]    = lk[i];
+      lp2[d]    = lp ? lp[i] : i;
+    }
+    for (int64_t j = 0; j < rn; j++) {
+      int64_t d = rc[BUCKET(rk[j])]++;
+      rk2[d]    = rk[j];
+      rp2[d]    = rp ? rp[j] : j;
+    }
+    free(lc);
+    free(rc);
+  }
+
+  int64_t *bucket_out = (int64_t *)calloc(NB, sizeof(int64_t));
+#pragma omp parallel for schedule(dynamic)
+  for (int b = 0; b < NB; b++) {
+    int64_t lb = loff[b], le = loff[b + 1];
+    int64_t rb = roff[b], re = roff[b + 1];
+    if (le == lb || re == rb) continue;
+    uint64_t nslots = next_pow2_u64((uint64_t)(le - lb) * 2 + 1);
+    uint64_t mask   = nslots - 1;
+    int64_t *sk     = (int64_t *)malloc(nslots * sizeof(int64_t));
+    int64_t *sv     = (int64_t *)malloc(nslots * sizeof(int64_t));
+    uint8_t *su     = (uint8_t *)calloc(nslots, 1);
+    for (int64_t i = lb; i < le; i++) {
+      uint64_t s = (dj_mix64((uint64_t)lk2[i]) >> RB) & mask;
+      while (su[s]) s = (s + 1) & mask;
+      su[s] = 1;
+      sk[s] = lk2[i];
+      sv[s] = lp2[i];
+    }
+    int64_t cnt = 0;
+    for (int64_t j = rb; j < re; j++) {
+      int64_t key = rk2[j];
+      uint64_t s  = (dj_mix64((uint64_t)key) >> RB) & mask;
+      while (su[s]) {
+        if (sk[s] == key) cnt++;
+        s = (s + 1) & mask;
+      }
+    }
+    bucket_out[b] = cnt;
+    free(sk);
+    free(sv);
+    free(su);
+  }
+  /* prefix to place each bucket's output */
+  int64_t *ooff = (int64_t *)malloc((NB + 1) * sizeof(int64_t));
+  ooff[0]       = 0;
+  for (int b = 0; b < NB; b++) ooff[b + 1] = ooff[b] + bucket_out[b];
+  int64_t total = ooff[NB];
+
+  if (out_c0 && total <= cap) {
+#pragma omp parallel for schedule(dynamic)
+    for (int b = 0; b < NB; b++) {
+      int64_t lb = loff[b], le = loff[b + 1];
+      int64_t rb = roff[b], re = roff[b + 1];
+      if (le == lb || re == rb) continue;
+      uint64_t nslots = next_pow2_u64((uint64_t)(le - lb) * 2 + 1);
+      uint64_t mask   = nslots - 1;
+      int64_t *sk     = (int64_t *)malloc(nslots * sizeof(int64_t));
+      int64_t *sv     = (int64_t *)malloc(nslots * sizeof(int64_t));
+      uint8_t *su     = (uint8_t *)calloc(nslots, 1);
+      for (int64_t i = lb; i < le; i++) {
+        uint64_t s = (dj_mix64((uint64_t)lk2[i]) >> RB) & mask;
+        while (su[s]) s = (s + 1) & mask;
+        su[s] = 1;
+        sk[s] = lk2[i];
+        sv[s] = lp2[i];
+      }
+      int64_t w = ooff[b];
+      for (int64_t j = rb; j < re; j++) {
+        int64_t key = rk2[j];
+        uint64_t s  = (dj_mix64((uint64_t)key) >> RB) & mask;
+        while (su[s]) {
+          if (sk[s] == key) {
+            out_c0[w] = key;
+            out_c1[w] = sv[s];
+            out_c2[w] = key;
+            out_c3[w] = rp2[j];
+            w++;
+          }
+          s = (s + 1) & mask;
+        }
+      }
+      free(sk);
+      free(sv);
+      free(su);
+    }
+  }
+#undef BUCKET
+  free(lcnt); free(rcnt); free(loff); free(roff);
+  free(lk2); free(lp2); free(rk2); free(rp2);
+  free(bucket_out); free(ooff);
+  return total;
+}
+
+ORACLE_API int oracle_num_threads(void)
+{
+#ifdef _OPENMP
+  return omp_get_max_threads();
+#else
+  return 1;
+#endif
+}
