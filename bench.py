@@ -1,0 +1,298 @@
+#!/usr/bin/env python3
+"""bench.py — measurement harness for the MI355X-native distributed
+repartitioned hash join.
+
+Metric (BASELINE.json): joined rows/sec whole-node for the 800M x 800M
+int64/int64 join at selectivity 0.3 (100M rows per GPU per table — the
+reference benchmark's per-GPU default, benchmark/distributed_join.cu:96-109).
+`value` = total input rows of both tables across all ranks / join wall time
+(the README-comparable number: 0.392 s on 8xV100 => ~4.08e9 rows/s,
+README.md:73-86). One step = one full distributed inner join over the
+resident tables: hash-partition both tables + size exchange + RCCL all-to-all
+(N>1) + hash-table build + probe. Generation and warmup are excluded, timing
+is barrier-bracketed, max over ranks — mirroring the reference's timed region
+(benchmark/distributed_join.cu:264-286).
+
+Single process (N=1) by default; for N>1 the driver launches this under
+torch.distributed.run with one rank per GPU; torch.distributed (gloo) is used
+only for bootstrap (RCCL unique id exchange) and barriers — compute and
+communication run in libdistjoin.so (HIP + RCCL over xGMI).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+import numpy as np  # noqa: E402
+
+import distributed_join_amd as dj  # noqa: E402
+
+SELECTIVITY = 0.3
+
+
+def log(rank, *a):
+    if rank == 0:
+        print(*a, file=sys.stderr, flush=True)
+
+
+def cpu_baseline_leg(sample_rows):
+    """Oracle OpenMP radix-partition join (the 'port' CPU baseline) on a
+    bounded sample of the same workload, timed on this box's host cores.
+    Test-infrastructure import — allowed here per oracle/oracle.c header."""
+    import oracle
+    n = sample_rows
+    bk, bp = oracle.gen_build(n)
+    pk, pp = oracle.gen_probe(n, n, selectivity=SELECTIVITY)
+    # one untimed warm run on a small slice to fault pages / threads
+    oracle.cpu_radix_join(bk[:100_000], bp[:100_000], pk[:100_000], pp[:100_000],
+                          count_only=True)
+    t0 = time.perf_counter()
+    nout = oracle.cpu_radix_join(bk, bp, pk, pp, count_only=True)
+    t1 = time.perf_counter()
+    secs = t1 - t0
+    return {
+        "value": (2.0 * n) / secs,
+        "unit": "input rows/s",
+        "cores": oracle.num_threads(),
+        "kind": "port",
+        "sample": f"{n}x{n} int64/int64 sel {SELECTIVITY} radix-partition join, "
+                  f"{secs:.2f}s wall ({nout} output rows)",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--rows", type=int, default=100_000_000,
+                    help="rows per GPU per table (build and probe)")
+    ap.add_argument("--over-decom", type=int, default=1)
+    ap.add_argument("--cpu-baseline-rows", type=int, default=20_000_000)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world == 1 and args.gpus > 1:
+        print("ERROR: for --gpus N>1 launch under torch.distributed.run", file=sys.stderr)
+        sys.exit(2)
+    N = world
+
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+        dist = tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        tdist.init_process_group("gloo", rank=rank, world_size=world)
+
+    dj.require_gpu()
+    dj.lib().dj_set_device(local_rank)
+
+    # RCCL bootstrap: rank 0's unique id broadcast over gloo
+    if world > 1:
+        import torch
+        nbytes = dj.lib().dj_rccl_unique_id_bytes()
+        if rank == 0:
+            buf = np.zeros(nbytes, dtype=np.uint8)
+            dj.lib().dj_rccl_get_unique_id(buf.ctypes.data)
+            t = torch.from_numpy(buf)
+        else:
+            t = torch.zeros(nbytes, dtype=torch.uint8)
+        dist.broadcast(t, src=0)
+        dj.lib().dj_comm_init(rank, world, t.numpy().ctypes.data)
+
+    rows = args.rows
+    G = world  # partitions == ranks (over-decom batches later widen this)
+    build_n_global = rows * N
+    probe_n_global = rows * N
+    rand_max = 2 * build_n_global
+
+    log(rank, f"[bench] N={N} rows/GPU={rows} generating inputs...")
+    L = dj.lib()
+    row0 = rank * rows
+    bk, bp = dj.generate_build(build_n_global, rand_max, uniq=True, row0=row0, nrows=rows)
+    pk, pp = dj.generate_probe(probe_n_global, build_n_global, rand_max,
+                               selectivity=SELECTIVITY, row0=row0, nrows=rows)
+
+    # partition outputs + scratch
+    part = [dj.DeviceArray(rows) for _ in range(4)]  # bk,bp,pk,pp partitioned
+    scratch = L.dj_dmalloc(L.dj_partition_scratch_bytes(rows, max(G, 1)))
+
+    # receive buffers (N>1): capacity with skew margin
+    recv_cap = rows if N == 1 else int(rows * 1.03) + (1 << 16)
+    if N > 1:
+        recv = [dj.DeviceArray(recv_cap) for _ in range(4)]
+    else:
+        recv = part
+
+    # hash table + outputs
+    nslots = L.dj_join_table_slots(recv_cap)
+    d_slot_keys = L.dj_dmalloc(nslots * 8)
+    d_slot_vals = L.dj_dmalloc(nslots * 8)
+    d_error = L.dj_dmalloc(4)
+    d_counter = L.dj_dmalloc(8)
+    out_cap = recv_cap + (recv_cap >> 3)
+    outs = [dj.DeviceArray(out_cap) for _ in range(4)]
+
+    zero8 = np.zeros(1, dtype=np.int64)
+    off_b = np.zeros(G + 1, dtype=np.int64)
+    off_p = np.zeros(G + 1, dtype=np.int64)
+
+    state = {"matches": 0, "lrows": rows, "rrows": rows, "wire_elems": 0}
+
+    def step():
+        L.dj_memcpy_h2d(d_counter, zero8.ctypes.data, 8)
+        L.dj_memcpy_h2d(d_error, zero8.ctypes.data, 4)
+        # partition both tables into G ranges (seed 12345678 like
+        # distributed_join.cpp:211-226)
+        L.dj_hash_partition(bk.ptr, bp.ptr, rows, G, dj.HASH_MURMUR3, dj.SEED_INTRA,
+                            part[0].ptr, part[1].ptr, off_b.ctypes.data, scratch)
+        L.dj_hash_partition(pk.ptr, pp.ptr, rows, G, dj.HASH_MURMUR3, dj.SEED_INTRA,
+                            part[2].ptr, part[3].ptr, off_p.ctypes.data, scratch)
+        if N > 1:
+            # size exchange + personalized all-to-all per column buffer
+            scnt_b = np.diff(off_b).copy()
+            scnt_p = np.diff(off_p).copy()
+            rcnt_b = np.zeros(G, dtype=np.int64)
+            rcnt_p = np.zeros(G, dtype=np.int64)
+            L.dj_exchange_sizes(scnt_b.ctypes.data, rcnt_b.ctypes.data)
+            L.dj_exchange_sizes(scnt_p.ctypes.data, rcnt_p.ctypes.data)
+            roff_b = np.concatenate([[0], np.cumsum(rcnt_b)]).astype(np.int64)
+            roff_p = np.concatenate([[0], np.cumsum(rcnt_p)]).astype(np.int64)
+            if roff_b[-1] > recv_cap or roff_p[-1] > recv_cap:
+                raise RuntimeError("receive capacity exceeded; raise --rows margin")
+            for send, rbuf, soff, roff in ((part[0], recv[0], off_b, roff_b),
+                                           (part[1], recv[1], off_b, roff_b),
+                                           (part[2], recv[2], off_p, roff_p),
+                                           (part[3], recv[3], off_p, roff_p)):
+                L.dj_all_to_all_i64(send.ptr, soff.ctypes.data, rbuf.ptr, roff.ctypes.data)
+            lrows, rrows = int(roff_b[-1]), int(roff_p[-1])
+            state["wire_elems"] = int((scnt_b.sum() - scnt_b[rank]) * 2 +
+                                      (scnt_p.sum() - scnt_p[rank]) * 2)
+        else:
+            lrows, rrows = rows, rows
+        # local join: table init + build + probe-append
+        L.dj_join_table_init(d_slot_keys, nslots)
+        L.dj_join_build(recv[0].ptr, recv[1].ptr, lrows, d_slot_keys, d_slot_vals,
+                        nslots, d_error)
+        L.dj_join_probe(recv[2].ptr, recv[3].ptr, rrows, d_slot_keys, d_slot_vals, nslots,
+                        outs[0].ptr, outs[1].ptr, outs[2].ptr, outs[3].ptr,
+                        out_cap, d_counter)
+        state["lrows"], state["rrows"] = lrows, rrows
+
+    def barrier_sync():
+        L.dj_sync()
+        if dist is not None:
+            dist.barrier()
+
+    log(rank, f"[bench] warmup {args.warmup} steps...")
+    for _ in range(args.warmup):
+        step()
+    state["matches"] = L.dj_read_counter_i64(d_counter)
+    if L.dj_read_error_i32(d_error):
+        raise RuntimeError("build error: sentinel key")
+
+    L.dj_timing_enable(1)
+    L.dj_timing_reset()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier_sync()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    if dist is not None:
+        import torch
+        e = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    matches = L.dj_read_counter_i64(d_counter)
+    ms_per_step = elapsed / args.steps * 1000.0
+    input_rows = 2.0 * rows * N  # both tables, whole node
+    value = input_rows / (elapsed / args.steps)
+    out_rows_s = matches * N / (elapsed / args.steps)
+
+    # per-phase kernel timing (hipEvents on the launch stream)
+    phases = {name: {"ms": L.dj_timing_total_ms(pid) / args.steps,
+                     "launches": L.dj_timing_launches(pid) / args.steps}
+              for name, pid in dj.PHASES.items()}
+
+    # roofline for the dominant join kernel (algorithmic bytes: DESIGN.md §Measurement)
+    probe_ms = phases["probe"]["ms"]
+    scatter_ms = phases["part_scatter"]["ms"]
+    build_ms = phases["build"]["ms"]
+    # algorithmic bytes per step (per this rank):
+    alg = {
+        "part_scatter": 40.0 * rows * 2,   # both tables: 8 hist reread + 16 read + 16 write
+        "build": 32.0 * state["lrows"],    # 16 read + 16 table store
+        "probe": (16.0 + 16.0 * 1.5) * state["rrows"] + 32.0 * matches,
+    }
+    dom = max(("probe", probe_ms), ("part_scatter", scatter_ms), ("build", build_ms),
+              key=lambda kv: kv[1])[0]
+    dom_ms = phases[dom]["ms"]
+    achieved = (alg[dom] / 1e9) / (dom_ms / 1e3) if dom_ms > 0 else None
+    roofline = {
+        "bound": "hbm",
+        "kernel": dom,
+        "achieved": achieved,
+        "peak": 8000.0,
+        "unit": "GB/s",
+        "frac": (achieved / 8000.0) if achieved else None,
+        "traffic": None,
+    }
+
+    all_to_all_GBs = None
+    if N > 1 and phases["comm"]["ms"] > 0:
+        all_to_all_GBs = (state["wire_elems"] * 8 / 1e9) / (phases["comm"]["ms"] / 1e3)
+
+    cpu_baseline = None
+    if rank == 0 and N == 1 and not args.no_cpu_baseline:
+        log(rank, "[bench] cpu baseline leg...")
+        cpu_baseline = cpu_baseline_leg(args.cpu_baseline_rows)
+
+    if rank == 0:
+        rec = {
+            "metric": "joined rows/sec whole-node",
+            "value": value,
+            "unit": "input rows/s",
+            "n_gpus": N,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            # README.md:73-86: 0.392 s for 1.6e9 input rows on 8xV100 => 4.08e9
+            "vs_baseline": (value / 4.08e9) if N == 8 else None,
+            "dtype": "int64",
+            "data": "synthetic",
+            "config": {
+                "workload": f"{rows*N//1_000_000}Mx{rows*N//1_000_000}M int64/int64 join, "
+                            f"selectivity {SELECTIVITY}, unique build keys, "
+                            f"{rows//1_000_000}M rows/GPU/table, over_decom {args.over_decom}",
+                "rows_per_gpu": rows,
+                "selectivity": SELECTIVITY,
+                "output_rows_per_gpu": int(matches),
+                "output_rows_per_sec": out_rows_s,
+                "phases_ms": {k: round(v["ms"], 4) for k, v in phases.items()},
+                "all_to_all_GBs_per_gpu": all_to_all_GBs,
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(rec), flush=True)
+
+    if world > 1:
+        L.dj_comm_finalize()
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

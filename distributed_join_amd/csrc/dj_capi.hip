@@ -1,0 +1,354 @@
+/*
+ * dj_capi.hip — C-ABI implementation (see include/distributed_join.h for the
+ * contract and the reference interfaces each entry point replaces).
+ */
+#include "dj_error.hpp"
+#include "dj_kernels.hpp"
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <cstring>
+#include <mutex>
+#include <vector>
+
+#include "../../include/distributed_join.h"
+
+namespace {
+
+hipStream_t g_stream = nullptr;       // compute stream
+hipStream_t g_comm_stream = nullptr;  // communication stream (xGMI overlap)
+
+hipStream_t stream()
+{
+  if (!g_stream) DJ_HIP_CALL(hipStreamCreateWithFlags(&g_stream, hipStreamNonBlocking));
+  return g_stream;
+}
+
+hipStream_t comm_stream()
+{
+  if (!g_comm_stream) DJ_HIP_CALL(hipStreamCreateWithFlags(&g_comm_stream, hipStreamNonBlocking));
+  return g_comm_stream;
+}
+
+/* ------------- phase timing registry (hipEvent pairs per launch) ------------- */
+
+bool g_timing_on = false;
+
+struct TimedSpan {
+  int phase;
+  hipEvent_t start, stop;
+};
+std::vector<TimedSpan> g_spans;
+
+struct PhaseScope {
+  int phase;
+  hipStream_t s;
+  PhaseScope(int p, hipStream_t st) : phase(p), s(st)
+  {
+    if (!g_timing_on) return;
+    TimedSpan t;
+    t.phase = p;
+    DJ_HIP_CALL(hipEventCreate(&t.start));
+    DJ_HIP_CALL(hipEventCreate(&t.stop));
+    DJ_HIP_CALL(hipEventRecord(t.start, s));
+    g_spans.push_back(t);
+  }
+  ~PhaseScope()
+  {
+    if (!g_timing_on) return;
+    DJ_HIP_CALL(hipEventRecord(g_spans.back().stop, s));
+  }
+};
+
+/* ------------- RCCL communicator state ------------- */
+
+ncclComm_t g_comm = nullptr;
+int g_rank = 0;
+int g_size = 1;
+
+}  // namespace
+
+extern "C" {
+
+/* ---------------- device & memory ---------------- */
+
+int dj_device_count(void)
+{
+  int n = 0;
+  if (hipGetDeviceCount(&n) != hipSuccess) return 0;
+  return n;
+}
+
+void dj_set_device(int device) { DJ_HIP_CALL(hipSetDevice(device)); }
+
+void* dj_dmalloc(int64_t bytes)
+{
+  void* p = nullptr;
+  DJ_HIP_CALL(hipMalloc(&p, (size_t)bytes));
+  return p;
+}
+
+void dj_dfree(void* ptr)
+{
+  if (ptr) DJ_HIP_CALL(hipFree(ptr));
+}
+
+void dj_memcpy_h2d(void* d_dst, const void* h_src, int64_t bytes)
+{
+  DJ_HIP_CALL(hipMemcpyAsync(d_dst, h_src, (size_t)bytes, hipMemcpyHostToDevice, stream()));
+  DJ_HIP_CALL(hipStreamSynchronize(stream()));
+}
+
+void dj_memcpy_d2h(void* h_dst, const void* d_src, int64_t bytes)
+{
+  DJ_HIP_CALL(hipMemcpyAsync(h_dst, d_src, (size_t)bytes, hipMemcpyDeviceToHost, stream()));
+  DJ_HIP_CALL(hipStreamSynchronize(stream()));
+}
+
+void dj_memcpy_d2d(void* d_dst, const void* d_src, int64_t bytes)
+{
+  DJ_HIP_CALL(hipMemcpyAsync(d_dst, d_src, (size_t)bytes, hipMemcpyDeviceToDevice, stream()));
+}
+
+void dj_sync(void)
+{
+  DJ_HIP_CALL(hipStreamSynchronize(stream()));
+  if (g_comm_stream) DJ_HIP_CALL(hipStreamSynchronize(g_comm_stream));
+}
+
+/* ---------------- generator ---------------- */
+
+void dj_generate_build(int64_t* d_keys, int64_t* d_pay, int64_t n_global, int64_t rand_max,
+                       uint64_t seed, int uniq, int64_t row0, int64_t nrows)
+{
+  PhaseScope t(DJ_PHASE_GENERATE, stream());
+  dj::generate_build(d_keys, d_pay, n_global, rand_max, seed, uniq != 0, row0, nrows, stream());
+}
+
+void dj_generate_probe(int64_t* d_keys, int64_t* d_pay, int64_t build_n_global,
+                       int64_t rand_max, double selectivity, uint64_t seed, int64_t row0,
+                       int64_t nrows)
+{
+  PhaseScope t(DJ_PHASE_GENERATE, stream());
+  dj::generate_probe(d_keys, d_pay, build_n_global, rand_max, selectivity, seed, row0, nrows,
+                     stream());
+}
+
+/* ---------------- partition ---------------- */
+
+int64_t dj_partition_scratch_bytes(int64_t n, int nparts)
+{
+  return (int64_t)dj::hash_partition_scratch_bytes(n, nparts);
+}
+
+void dj_hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                       int hash_fn, uint32_t hash_seed, int64_t* d_out_keys,
+                       int64_t* d_out_pay, int64_t* h_offsets, void* d_scratch)
+{
+  static int64_t* d_offsets = nullptr;
+  static int d_offsets_cap = 0;
+  if (nparts + 1 > d_offsets_cap) {
+    if (d_offsets) DJ_HIP_CALL(hipFree(d_offsets));
+    DJ_HIP_CALL(hipMalloc(&d_offsets, (size_t)(nparts + 1) * sizeof(int64_t)));
+    d_offsets_cap = nparts + 1;
+  }
+  {
+    PhaseScope t(DJ_PHASE_PART_SCATTER, stream());
+    dj::hash_partition(d_keys, d_pay, n, nparts, hash_fn, hash_seed, d_out_keys, d_out_pay,
+                       d_offsets, d_scratch, stream());
+  }
+  if (h_offsets) {
+    DJ_HIP_CALL(hipMemcpyAsync(h_offsets, d_offsets, (size_t)(nparts + 1) * sizeof(int64_t),
+                               hipMemcpyDeviceToHost, stream()));
+    DJ_HIP_CALL(hipStreamSynchronize(stream()));
+  }
+}
+
+/* ---------------- local join ---------------- */
+
+int64_t dj_join_table_slots(int64_t ln) { return dj::join_table_slots(ln); }
+
+void dj_join_table_init(int64_t* d_slot_keys, int64_t nslots)
+{
+  PhaseScope t(DJ_PHASE_TABLE_INIT, stream());
+  dj::join_table_init(d_slot_keys, nslots, stream());
+}
+
+void dj_join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                   int64_t* d_slot_keys, int64_t* d_slot_vals, int64_t nslots, int* d_error)
+{
+  PhaseScope t(DJ_PHASE_BUILD, stream());
+  dj::join_build(d_lk, d_lp, ln, d_slot_keys, d_slot_vals, nslots, d_error, stream());
+}
+
+void dj_join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                   const int64_t* d_slot_keys, const int64_t* d_slot_vals, int64_t nslots,
+                   int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3,
+                   int64_t cap, int64_t* d_counter)
+{
+  PhaseScope t(DJ_PHASE_PROBE, stream());
+  dj::join_probe(d_rk, d_rp, rn, d_slot_keys, d_slot_vals, nslots, d_out0, d_out1, d_out2,
+                 d_out3, cap, d_counter, stream());
+}
+
+int64_t dj_read_counter_i64(const int64_t* d_counter)
+{
+  int64_t v = 0;
+  dj_memcpy_d2h(&v, d_counter, sizeof(int64_t));
+  return v;
+}
+
+int dj_read_error_i32(const int* d_error)
+{
+  int v = 0;
+  dj_memcpy_d2h(&v, d_error, sizeof(int));
+  return v;
+}
+
+int64_t dj_local_inner_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                            const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                            int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                            int64_t* d_out3, int64_t cap)
+{
+  if (ln == 0 || rn == 0) return 0;  // empty side => empty (distributed_join.cpp:76-83)
+  int64_t nslots = dj::join_table_slots(ln);
+  int64_t* d_slot_keys = (int64_t*)dj_dmalloc(nslots * sizeof(int64_t));
+  int64_t* d_slot_vals = (int64_t*)dj_dmalloc(nslots * sizeof(int64_t));
+  int* d_error = (int*)dj_dmalloc(sizeof(int));
+  int64_t* d_counter = (int64_t*)dj_dmalloc(sizeof(int64_t));
+  DJ_HIP_CALL(hipMemsetAsync(d_error, 0, sizeof(int), stream()));
+  DJ_HIP_CALL(hipMemsetAsync(d_counter, 0, sizeof(int64_t), stream()));
+  dj_join_table_init(d_slot_keys, nslots);
+  dj_join_build(d_lk, d_lp, ln, d_slot_keys, d_slot_vals, nslots, d_error);
+  dj_join_probe(d_rk, d_rp, rn, d_slot_keys, d_slot_vals, nslots, d_out0, d_out1, d_out2,
+                d_out3, cap, d_counter);
+  int64_t n = dj_read_counter_i64(d_counter);
+  DJ_CHECK_ERROR(dj_read_error_i32(d_error) == 0,
+                 "dj_join_build: key equal to the empty sentinel (-1) is unsupported");
+  dj_dfree(d_slot_keys);
+  dj_dfree(d_slot_vals);
+  dj_dfree(d_error);
+  dj_dfree(d_counter);
+  return n;
+}
+
+/* ---------------- timing ---------------- */
+
+void dj_timing_enable(int on) { g_timing_on = on != 0; }
+
+void dj_timing_reset(void)
+{
+  for (auto& t : g_spans) {
+    DJ_HIP_CALL(hipEventDestroy(t.start));
+    DJ_HIP_CALL(hipEventDestroy(t.stop));
+  }
+  g_spans.clear();
+}
+
+double dj_timing_total_ms(int phase)
+{
+  dj_sync();
+  double total = 0;
+  for (auto& t : g_spans) {
+    if (t.phase != phase) continue;
+    float ms = 0;
+    DJ_HIP_CALL(hipEventElapsedTime(&ms, t.start, t.stop));
+    total += ms;
+  }
+  return total;
+}
+
+int64_t dj_timing_launches(int phase)
+{
+  int64_t n = 0;
+  for (auto& t : g_spans)
+    if (t.phase == phase) n++;
+  return n;
+}
+
+/* ---------------- RCCL communicator ---------------- */
+
+int dj_rccl_unique_id_bytes(void) { return (int)sizeof(ncclUniqueId); }
+
+void dj_rccl_get_unique_id(void* h_id_bytes)
+{
+  ncclUniqueId id;
+  DJ_RCCL_CALL(ncclGetUniqueId(&id));
+  memcpy(h_id_bytes, &id, sizeof(id));
+}
+
+void dj_comm_init(int rank, int size, const void* h_id_bytes)
+{
+  ncclUniqueId id;
+  memcpy(&id, h_id_bytes, sizeof(id));
+  DJ_RCCL_CALL(ncclCommInitRank(&g_comm, size, id, rank));
+  g_rank = rank;
+  g_size = size;
+}
+
+void dj_comm_finalize(void)
+{
+  if (g_comm) {
+    DJ_RCCL_CALL(ncclCommDestroy(g_comm));
+    g_comm = nullptr;
+  }
+}
+
+int dj_comm_rank(void) { return g_rank; }
+int dj_comm_size(void) { return g_size; }
+
+void dj_all_to_all_i64(const int64_t* d_send, const int64_t* h_send_offsets, int64_t* d_recv,
+                       const int64_t* h_recv_offsets)
+{
+  DJ_CHECK_ERROR(g_comm != nullptr, "dj_all_to_all_i64: communicator not initialized");
+  hipStream_t cs = comm_stream();
+  PhaseScope t(DJ_PHASE_COMM, cs);
+  /* reference pattern: grouped per-peer send/recv of contiguous slices
+   * (all_to_all_comm.cpp:126-189); self-partition via explicit D2D copy
+   * (all_to_all_comm.cpp:610-653,710-726). Over xGMI each peer slice moves
+   * on its own point-to-point link, so pairwise grouped send/recv is the
+   * bandwidth-optimal all-to-all (no staging, no registration). */
+  DJ_RCCL_CALL(ncclGroupStart());
+  for (int p = 0; p < g_size; p++) {
+    if (p == g_rank) continue;
+    int64_t scount = h_send_offsets[p + 1] - h_send_offsets[p];
+    int64_t rcount = h_recv_offsets[p + 1] - h_recv_offsets[p];
+    if (scount > 0)
+      DJ_RCCL_CALL(ncclSend(d_send + h_send_offsets[p], (size_t)scount, ncclInt64, p, g_comm, cs));
+    if (rcount > 0)
+      DJ_RCCL_CALL(ncclRecv(d_recv + h_recv_offsets[p], (size_t)rcount, ncclInt64, p, g_comm, cs));
+  }
+  DJ_RCCL_CALL(ncclGroupEnd());
+  int64_t self_count = h_send_offsets[g_rank + 1] - h_send_offsets[g_rank];
+  if (self_count > 0) {
+    DJ_HIP_CALL(hipMemcpyAsync(d_recv + h_recv_offsets[g_rank],
+                               d_send + h_send_offsets[g_rank],
+                               (size_t)self_count * sizeof(int64_t), hipMemcpyDeviceToDevice,
+                               cs));
+  }
+  DJ_HIP_CALL(hipStreamSynchronize(cs));  // launch_communication blocks the host
+                                          // (all_to_all_comm.hpp:331 contract)
+}
+
+void dj_exchange_sizes(const int64_t* h_send_counts, int64_t* h_recv_counts)
+{
+  DJ_CHECK_ERROR(g_comm != nullptr, "dj_exchange_sizes: communicator not initialized");
+  /* replaces communicate_sizes (all_to_all_comm.cpp:54-100): allgather the
+   * G x G count matrix, read our column. Tiny (G*G*8 B). */
+  hipStream_t cs = comm_stream();
+  int64_t* d_mine = (int64_t*)dj_dmalloc((int64_t)g_size * sizeof(int64_t));
+  int64_t* d_all = (int64_t*)dj_dmalloc((int64_t)g_size * g_size * sizeof(int64_t));
+  DJ_HIP_CALL(hipMemcpyAsync(d_mine, h_send_counts, (size_t)g_size * sizeof(int64_t),
+                             hipMemcpyHostToDevice, cs));
+  DJ_RCCL_CALL(ncclAllGather(d_mine, d_all, (size_t)g_size, ncclInt64, g_comm, cs));
+  std::vector<int64_t> all((size_t)g_size * g_size);
+  DJ_HIP_CALL(hipMemcpyAsync(all.data(), d_all, all.size() * sizeof(int64_t),
+                             hipMemcpyDeviceToHost, cs));
+  DJ_HIP_CALL(hipStreamSynchronize(cs));
+  for (int p = 0; p < g_size; p++) h_recv_counts[p] = all[(size_t)p * g_size + g_rank];
+  dj_dfree(d_mine);
+  dj_dfree(d_all);
+}
+
+}  // extern "C"

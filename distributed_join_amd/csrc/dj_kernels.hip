@@ -1,0 +1,366 @@
+/*
+ * dj_kernels.hip — hand-written CDNA4 (gfx950) kernels for the distributed
+ * repartitioned hash-join hot path. MI355X-first design, not a CUDA port:
+ * 64-wide wavefront ballot/prefix-sum scatter, coalesced int64 loads,
+ * grid-stride launches sized for 256 CUs / 8 XCDs.
+ *
+ * Replaces (SURVEY.md §2 third-party kernel table):
+ *  - cudf::hash_partition  (reference calls: distributed_join.cpp:213-225,
+ *    shuffle_on.cpp:59-60): stable partition = per-wave histogram pass +
+ *    device scan + wave-ballot rank scatter (no LDS atomics, no sort).
+ *  - cudf::inner_join      (reference call: distributed_join.cpp:79):
+ *    open-addressing (linear probing, <=50%% fill, power-of-two slots)
+ *    atomicCAS build + probe with single-pass wave-aggregated append
+ *    (replaces cuDF's count+gather two-pass; row order is unspecified by the
+ *    API — reference tests sort before comparing,
+ *    compare_against_single_gpu.cu:167-174).
+ *  - generate_dataset.cuh:40-260: deterministic counter-based restatement
+ *    (spec in dj_rng.h) — bit-identical to the CPU oracle on any device.
+ */
+#include "dj_error.hpp"
+#include "dj_hash.h"
+#include "dj_kernels.hpp"
+#include "dj_rng.h"
+
+#include <hip/hip_runtime.h>
+
+namespace dj {
+
+constexpr int BLOCK = 256;
+constexpr int WAVE = 64;
+constexpr int WPB = BLOCK / WAVE;  // waves per block
+
+/* ------------------------------------------------------------------ misc */
+
+__global__ void fill_i64_kernel(int64_t* dst, int64_t value, int64_t n)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = value;
+}
+
+static int grid_for(int64_t n)
+{
+  int64_t blocks = (n + BLOCK - 1) / BLOCK;
+  if (blocks > 2048) blocks = 2048;  // 256 CU x 8 blocks, grid-stride the rest
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s)
+{
+  if (n <= 0) return;
+  hipLaunchKernelGGL(fill_i64_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, s, d_dst, value, n);
+}
+
+/* ------------------------------------------------------------- generator */
+
+__global__ void gen_build_kernel(int64_t* keys, int64_t* pay, int64_t n_global,
+                                 int64_t rand_max, uint64_t seed, int uniq, int64_t row0,
+                                 int64_t nrows)
+{
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; t < nrows; t += stride) {
+    int64_t i = row0 + t;
+    keys[t] = uniq ? dj_build_key((uint64_t)i, (uint64_t)n_global, rand_max, seed)
+                   : dj_build_key_nonuniq((uint64_t)i, rand_max, seed);
+    if (pay) pay[t] = i;
+  }
+}
+
+__global__ void gen_probe_kernel(int64_t* keys, int64_t* pay, int64_t build_n_global,
+                                 int64_t rand_max, double selectivity, uint64_t seed,
+                                 int64_t row0, int64_t nrows)
+{
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; t < nrows; t += stride) {
+    int64_t j = row0 + t;
+    keys[t] = dj_probe_key((uint64_t)j, (uint64_t)build_n_global, rand_max, selectivity, seed);
+    if (pay) pay[t] = j;
+  }
+}
+
+void generate_build(int64_t* d_keys, int64_t* d_pay, int64_t n_global, int64_t rand_max,
+                    uint64_t seed, bool uniq, int64_t row0, int64_t nrows, hipStream_t s)
+{
+  if (nrows <= 0) return;
+  hipLaunchKernelGGL(gen_build_kernel, dim3(grid_for(nrows)), dim3(BLOCK), 0, s, d_keys, d_pay,
+                     n_global, rand_max, seed, (int)uniq, row0, nrows);
+}
+
+void generate_probe(int64_t* d_keys, int64_t* d_pay, int64_t build_n_global, int64_t rand_max,
+                    double selectivity, uint64_t seed, int64_t row0, int64_t nrows, hipStream_t s)
+{
+  if (nrows <= 0) return;
+  hipLaunchKernelGGL(gen_probe_kernel, dim3(grid_for(nrows)), dim3(BLOCK), 0, s, d_keys, d_pay,
+                     build_n_global, rand_max, selectivity, seed, row0, nrows);
+}
+
+/* ------------------------------------------------------- stable partition */
+/*
+ * Wave-contiguous assignment: wave w owns rows [w*rpw, (w+1)*rpw). Lanes read
+ * row base+it*64+lane, so within one iteration lane order == row order and
+ * global memory loads coalesce to 512 B per instruction.
+ *
+ * Pass 1 (count): per-wave histogram held in registers — lane q accumulates
+ * the count for partition q from __ballot(p == q) (64-bit wave64 ballots;
+ * nparts <= 64).
+ * Scan: per-partition block scan over waves (exclusive), then a tiny scan of
+ * partition totals into global partition offsets.
+ * Pass 2 (scatter): lane q holds the running cursor of partition q; each row
+ * gets dst = shfl(cursor, p) + rank, rank = popcount(ballot-mask of its own
+ * partition & lanes-below mask) — conflict-free, stable, no atomics.
+ */
+
+struct PartGeom {
+  int64_t nwaves;
+  int64_t rows_per_wave;
+  int blocks;
+};
+
+static PartGeom part_geom(int64_t n)
+{
+  PartGeom g;
+  int64_t target = (n + 255) / 256;  // >=256 rows per wave
+  g.nwaves = target < 1 ? 1 : (target > 8192 ? 8192 : target);
+  g.blocks = (int)((g.nwaves + WPB - 1) / WPB);
+  g.nwaves = (int64_t)g.blocks * WPB;
+  g.rows_per_wave = (n + g.nwaves - 1) / g.nwaves;
+  return g;
+}
+
+size_t hash_partition_scratch_bytes(int64_t n, int nparts)
+{
+  PartGeom g = part_geom(n);
+  /* wave_counts/prefix [nwaves][nparts] (int64) + totals [nparts] */
+  return (size_t)(g.nwaves * nparts + nparts) * sizeof(int64_t);
+}
+
+__device__ __forceinline__ uint32_t part_of(int64_t key, int hash_fn, uint32_t seed, int nparts)
+{
+  return dj_row_hash(key, hash_fn, seed) % (uint32_t)nparts;
+}
+
+__global__ void part_count_kernel(const int64_t* __restrict__ keys, int64_t n, int nparts,
+                                  int hash_fn, uint32_t seed, int64_t rows_per_wave,
+                                  int64_t* __restrict__ wave_counts)
+{
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t w = (int64_t)blockIdx.x * WPB + (threadIdx.x >> 6);
+  const int64_t start = w * rows_per_wave;
+  const int64_t end = min(start + rows_per_wave, n);
+  int64_t my_count = 0;  // lane q counts partition q
+  for (int64_t base = start; base < end; base += WAVE) {
+    int64_t i = base + lane;
+    uint32_t p = 0xFFFFFFFFu;
+    if (i < end) p = part_of(keys[i], hash_fn, seed, nparts);
+    for (int q = 0; q < nparts; q++) {
+      uint64_t m = __ballot(p == (uint32_t)q);
+      if (lane == q) my_count += __popcll(m);
+    }
+  }
+  if (lane < nparts) wave_counts[w * nparts + lane] = my_count;
+}
+
+/* grid = nparts blocks; block p scans column p over nwaves (exclusive),
+ * in-place, and writes the partition total to totals[p]. */
+__global__ void part_scan_kernel(int64_t* wave_counts, int64_t nwaves, int nparts,
+                                 int64_t* totals)
+{
+  const int p = blockIdx.x;
+  __shared__ int64_t sh[BLOCK];
+  int64_t running = 0;
+  for (int64_t base = 0; base < nwaves; base += BLOCK) {
+    int64_t w = base + threadIdx.x;
+    int64_t v = (w < nwaves) ? wave_counts[w * nparts + p] : 0;
+    /* Hillis-Steele inclusive scan in LDS */
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < BLOCK; off <<= 1) {
+      int64_t add = (threadIdx.x >= off) ? sh[threadIdx.x - off] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += add;
+      __syncthreads();
+    }
+    int64_t incl = sh[threadIdx.x];
+    if (w < nwaves) wave_counts[w * nparts + p] = running + incl - v;  // exclusive
+    int64_t chunk_total = sh[BLOCK - 1];
+    running += chunk_total;
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) totals[p] = running;
+}
+
+/* single block: exclusive scan of totals -> offsets[nparts+1] */
+__global__ void part_offsets_kernel(const int64_t* totals, int nparts, int64_t* offsets)
+{
+  if (threadIdx.x == 0) {
+    int64_t acc = 0;
+    for (int p = 0; p < nparts; p++) {
+      offsets[p] = acc;
+      acc += totals[p];
+    }
+    offsets[nparts] = acc;
+  }
+}
+
+__global__ void part_scatter_kernel(const int64_t* __restrict__ keys,
+                                    const int64_t* __restrict__ pay, int64_t n, int nparts,
+                                    int hash_fn, uint32_t seed, int64_t rows_per_wave,
+                                    const int64_t* __restrict__ wave_prefix,
+                                    const int64_t* __restrict__ offsets,
+                                    int64_t* __restrict__ out_keys,
+                                    int64_t* __restrict__ out_pay)
+{
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t w = (int64_t)blockIdx.x * WPB + (threadIdx.x >> 6);
+  const int64_t start = w * rows_per_wave;
+  const int64_t end = min(start + rows_per_wave, n);
+  /* lane q holds the cursor for partition q */
+  int64_t cursor = 0;
+  if (lane < nparts) cursor = offsets[lane] + wave_prefix[w * nparts + lane];
+  const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+  for (int64_t base = start; base < end; base += WAVE) {
+    int64_t i = base + lane;
+    bool valid = i < end;
+    int64_t k = 0, v = 0;
+    uint32_t p = 0xFFFFFFFFu;
+    if (valid) {
+      k = keys[i];
+      v = pay ? pay[i] : 0;
+      p = part_of(k, hash_fn, seed, nparts);
+    }
+    uint64_t my_q_mask = 0;  // lane q's ballot mask for partition q
+    for (int q = 0; q < nparts; q++) {
+      uint64_t m = __ballot(p == (uint32_t)q);
+      if (lane == q) my_q_mask = m;
+    }
+    /* mask and cursor of MY partition, fetched from lane p */
+    uint64_t m_p = __shfl((unsigned long long)my_q_mask, (int)(valid ? p : 0));
+    int64_t base_dst = __shfl(cursor, (int)(valid ? p : 0));
+    if (valid) {
+      int rank = __popcll(m_p & lt_mask);
+      int64_t dst = base_dst + rank;
+      out_keys[dst] = k;
+      if (out_pay) out_pay[dst] = v;
+    }
+    cursor += __popcll(my_q_mask);
+  }
+}
+
+void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                    int hash_fn, uint32_t hash_seed, int64_t* d_out_keys, int64_t* d_out_pay,
+                    int64_t* d_offsets, void* d_scratch, hipStream_t s)
+{
+  DJ_CHECK_ERROR(nparts >= 1 && nparts <= kMaxPartitions, "nparts must be in [1,64]");
+  PartGeom g = part_geom(n);
+  int64_t* wave_counts = (int64_t*)d_scratch;
+  int64_t* totals = wave_counts + g.nwaves * nparts;
+  if (n <= 0) {
+    /* still produce zero offsets */
+    hipLaunchKernelGGL(part_offsets_kernel, dim3(1), dim3(64), 0, s, totals, 0, d_offsets);
+    return;
+  }
+  hipLaunchKernelGGL(part_count_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, n, nparts,
+                     hash_fn, hash_seed, g.rows_per_wave, wave_counts);
+  hipLaunchKernelGGL(part_scan_kernel, dim3(nparts), dim3(BLOCK), 0, s, wave_counts, g.nwaves,
+                     nparts, totals);
+  hipLaunchKernelGGL(part_offsets_kernel, dim3(1), dim3(64), 0, s, totals, nparts, d_offsets);
+  hipLaunchKernelGGL(part_scatter_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, d_pay, n,
+                     nparts, hash_fn, hash_seed, g.rows_per_wave, wave_counts, d_offsets,
+                     d_out_keys, d_out_pay);
+}
+
+/* ------------------------------------------------------------ local join */
+
+int64_t join_table_slots(int64_t ln)
+{
+  int64_t p = 1;
+  while (p < 2 * ln + 1) p <<= 1;  // <=50% fill
+  return p;
+}
+
+void join_table_init(int64_t* d_slot_keys, int64_t nslots, hipStream_t s)
+{
+  /* kEmptyKey == -1 == all bytes 0xFF: one HBM-rate memset */
+  DJ_HIP_CALL(hipMemsetAsync(d_slot_keys, 0xFF, (size_t)nslots * sizeof(int64_t), s));
+}
+
+__global__ void join_build_kernel(const int64_t* __restrict__ lk, const int64_t* __restrict__ lp,
+                                  int64_t ln, int64_t* __restrict__ slot_keys,
+                                  int64_t* __restrict__ slot_vals, uint64_t mask, int* error)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < ln; i += stride) {
+    int64_t key = lk[i];
+    if (key == kEmptyKey) {
+      *error = 1;  // reserved sentinel; loud failure, never silent wrong data
+      continue;
+    }
+    uint64_t slot = dj_mix64((uint64_t)key) & mask;
+    for (;;) {
+      unsigned long long old = atomicCAS((unsigned long long*)&slot_keys[slot],
+                                         (unsigned long long)kEmptyKey,
+                                         (unsigned long long)key);
+      if (old == (unsigned long long)kEmptyKey) break;
+      slot = (slot + 1) & mask;
+    }
+    slot_vals[slot] = lp ? lp[i] : i;
+  }
+}
+
+void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d_slot_keys,
+                int64_t* d_slot_vals, int64_t nslots, int* d_error, hipStream_t s)
+{
+  if (ln <= 0) return;
+  hipLaunchKernelGGL(join_build_kernel, dim3(grid_for(ln)), dim3(BLOCK), 0, s, d_lk, d_lp, ln,
+                     d_slot_keys, d_slot_vals, (uint64_t)(nslots - 1), d_error);
+}
+
+__global__ void join_probe_kernel(const int64_t* __restrict__ rk, const int64_t* __restrict__ rp,
+                                  int64_t rn, const int64_t* __restrict__ slot_keys,
+                                  const int64_t* __restrict__ slot_vals, uint64_t mask,
+                                  int64_t* __restrict__ out0, int64_t* __restrict__ out1,
+                                  int64_t* __restrict__ out2, int64_t* __restrict__ out3,
+                                  int64_t cap, int64_t* counter)
+{
+  int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; j < rn; j += stride) {
+    int64_t key = rk[j];
+    int64_t payload = rp ? rp[j] : j;
+    uint64_t slot = dj_mix64((uint64_t)key) & mask;
+    for (;;) {
+      int64_t sk = slot_keys[slot];
+      if (sk == kEmptyKey) break;
+      if (sk == key) {
+        /* per-lane atomicAdd(ptr,1): hipcc wave-aggregates active lanes into
+         * one global atomic, so contention is one atomic per wave-visit */
+        int64_t idx = (int64_t)atomicAdd((unsigned long long*)counter, 1ull);
+        if (idx < cap) {
+          out0[idx] = key;
+          out1[idx] = slot_vals[slot];
+          out2[idx] = key;
+          out3[idx] = payload;
+        }
+      }
+      slot = (slot + 1) & mask;
+    }
+  }
+}
+
+void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                const int64_t* d_slot_keys, const int64_t* d_slot_vals, int64_t nslots,
+                int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3,
+                int64_t cap, int64_t* d_counter, hipStream_t s)
+{
+  if (rn <= 0) return;
+  hipLaunchKernelGGL(join_probe_kernel, dim3(grid_for(rn)), dim3(BLOCK), 0, s, d_rk, d_rp, rn,
+                     d_slot_keys, d_slot_vals, (uint64_t)(nslots - 1), d_out0, d_out1, d_out2,
+                     d_out3, cap, d_counter);
+}
+
+}  // namespace dj

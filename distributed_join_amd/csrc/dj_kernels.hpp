@@ -1,0 +1,62 @@
+/*
+ * dj_kernels.hpp — host-side launchers for the gfx950 HIP kernels of the
+ * distributed repartitioned hash join hot path.
+ *
+ * These replace the third-party cuDF 0.19 kernels the reference calls on its
+ * hot path (SURVEY.md §2 third-party kernel table):
+ *   - hash_partition  <- cudf::hash_partition (distributed_join.cpp:213-225,
+ *                        shuffle_on.cpp:59-60)
+ *   - build/probe     <- cudf::inner_join (distributed_join.cpp:79)
+ *   - generate_*      <- generate_dataset.cuh:40-260 (restated deterministic)
+ *
+ * All launchers are stream-ordered on the given hipStream_t and operate on
+ * raw device pointers (columnar int64 arrays). No torch types.
+ */
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace dj {
+
+/* int64 value marking an empty hash-table slot; build rows with this key are
+ * rejected (loud error), see dj_kernels.hip. Set via 0xFF memset. */
+constexpr int64_t kEmptyKey = -1;
+
+constexpr int kMaxPartitions = 64;  // nparts = world_size x over_decom <= 64
+
+/* ----- synthetic inputs (deterministic, spec in dj_rng.h) ----- */
+void generate_build(int64_t* d_keys, int64_t* d_pay, int64_t n_global, int64_t rand_max,
+                    uint64_t seed, bool uniq, int64_t row0, int64_t nrows, hipStream_t s);
+void generate_probe(int64_t* d_keys, int64_t* d_pay, int64_t build_n_global, int64_t rand_max,
+                    double selectivity, uint64_t seed, int64_t row0, int64_t nrows, hipStream_t s);
+
+/* ----- stable hash partition (histogram + scan + wave-ballot scatter) ----- */
+size_t hash_partition_scratch_bytes(int64_t n, int nparts);
+/* d_offsets: device array of nparts+1 int64 partition offsets (offsets[0]=0).
+ * Stable: rows keep input order inside each partition. nparts <= 64. */
+void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                    int hash_fn, uint32_t hash_seed, int64_t* d_out_keys, int64_t* d_out_pay,
+                    int64_t* d_offsets, void* d_scratch, hipStream_t s);
+
+/* ----- local inner join: open-addressing build + probe-append ----- */
+/* Number of table slots for ln build rows (power of two, <=50% fill). */
+int64_t join_table_slots(int64_t ln);
+/* Initialize table key array to kEmptyKey (async memset). */
+void join_table_init(int64_t* d_slot_keys, int64_t nslots, hipStream_t s);
+/* Insert build rows. d_error (device int32) set to 1 if any key==kEmptyKey. */
+void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d_slot_keys,
+                int64_t* d_slot_vals, int64_t nslots, int* d_error, hipStream_t s);
+/* Probe rows; append matches (lkey, lpay, rkey, rpay) to the 4 output
+ * columns at positions drawn from d_counter (device int64, caller-zeroed).
+ * Rows beyond `cap` are counted but not written (caller re-runs bigger). */
+void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                const int64_t* d_slot_keys, const int64_t* d_slot_vals, int64_t nslots,
+                int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3,
+                int64_t cap, int64_t* d_counter, hipStream_t s);
+
+/* ----- small utilities ----- */
+void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s);
+
+}  // namespace dj

@@ -1,0 +1,156 @@
+"""GPU parity tests: the gfx950 HIP path against the CPU oracle on identical
+seeded inputs — the parity gate of SURVEY.md §8(c). All tests call through
+the C ABI (the product path); nothing here touches torch compute.
+
+Bar: bit-exact (order-insensitive where row order is unspecified by the API).
+"""
+import numpy as np
+import pytest
+
+import oracle
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dj():
+    import distributed_join_amd as dj
+    dj.require_gpu()
+    return dj
+
+
+# -------------------------------------------------------------- generator
+
+@pytest.mark.parametrize("n", [1000, 1_000_000])
+def test_generator_parity_build(dj, n):
+    gk, gp = dj.generate_build(n)
+    ok, op = oracle.gen_build(n)
+    assert (gk.to_numpy() == ok).all()
+    assert (gp.to_numpy() == op).all()
+
+
+@pytest.mark.parametrize("sel", [0.0, 0.3, 1.0])
+def test_generator_parity_probe(dj, sel):
+    n = 500_000
+    gk, gp = dj.generate_probe(n, n, selectivity=sel)
+    ok, op = oracle.gen_probe(n, n, selectivity=sel)
+    assert (gk.to_numpy() == ok).all()
+    assert (gp.to_numpy() == op).all()
+
+
+def test_generator_parity_slice(dj):
+    # rank-sliced generation equals the slice of the global table
+    n = 100_000
+    gk, _ = dj.generate_build(n, row0=40_000, nrows=10_000)
+    ok, _ = oracle.gen_build(n, row0=40_000, nrows=10_000)
+    assert (gk.to_numpy() == ok).all()
+
+
+# -------------------------------------------------------------- partition
+
+@pytest.mark.parametrize("nparts,hash_fn,seed", [
+    (1, 0, 12345678), (2, 0, 12345678), (8, 0, 12345678), (8, 0, 87654321),
+    (13, 0, 0), (64, 0, 12345678), (8, 1, 0),
+])
+def test_partition_parity(dj, nparts, hash_fn, seed):
+    n = 1_000_000
+    k, p = oracle.gen_probe(n, n)
+    dk = dj.DeviceArray.from_numpy(k)
+    dp = dj.DeviceArray.from_numpy(p)
+    gk, gp, goff = dj.hash_partition(dk, dp, n, nparts, hash_fn, seed)
+    ok, op, ooff = oracle.partition(k, p, nparts, hash_fn, seed)
+    assert (goff == ooff).all()
+    # stable partition is fully deterministic => bit-exact, not just set-equal
+    assert (gk.to_numpy() == ok).all()
+    assert (gp.to_numpy() == op).all()
+
+
+def test_partition_empty(dj):
+    dk = dj.DeviceArray(1)
+    dp = dj.DeviceArray(1)
+    gk, gp, goff = dj.hash_partition(dk, dp, 0, 8, 0, 0)
+    assert (goff == 0).all()
+
+
+# ------------------------------------------------------------------- join
+
+def _join_parity(dj, lk, lp, rk, rp, cap=None):
+    dlk, dlp = dj.DeviceArray.from_numpy(lk), dj.DeviceArray.from_numpy(lp)
+    drk, drp = dj.DeviceArray.from_numpy(rk), dj.DeviceArray.from_numpy(rp)
+    got = dj.local_inner_join(dlk, dlp, len(lk), drk, drp, len(rk), cap=cap)
+    want = oracle.inner_join(lk, lp, rk, rp)
+    g = oracle.sort_rows(*got)
+    w = oracle.sort_rows(*want)
+    assert len(g[0]) == len(w[0])
+    for a, b in zip(g, w):
+        assert (a == b).all()
+
+
+def test_join_analytical_multiples(dj):
+    # the reference's own KAT (compare_against_analytical.cu:44-54) on GPU
+    size = 300_000
+    lk = np.arange(size, dtype=np.int64) * 3
+    lp = np.arange(size, dtype=np.int64)
+    rk = np.arange(size, dtype=np.int64) * 5
+    rp = np.arange(size, dtype=np.int64)
+    dlk, dlp = dj.DeviceArray.from_numpy(lk), dj.DeviceArray.from_numpy(lp)
+    drk, drp = dj.DeviceArray.from_numpy(rk), dj.DeviceArray.from_numpy(rp)
+    c0, c1, c2, c3 = dj.local_inner_join(dlk, dlp, size, drk, drp, size)
+    assert len(c0) == size // 5
+    assert (c0 % 15 == 0).all() and (c1 == c0 // 3).all()
+    assert (c3 == c2 // 5).all() and (c0 == c2).all()
+
+
+def test_join_parity_config1(dj):
+    # BASELINE config 1: 1M x 1M, selectivity 0.3, unique build keys
+    n = 1_000_000
+    bk, bp = oracle.gen_build(n)
+    pk, pp = oracle.gen_probe(n, n, selectivity=0.3)
+    _join_parity(dj, bk, bp, pk, pp)
+
+
+def test_join_parity_duplicates(dj):
+    # non-unique build keys with a tiny key space => heavy duplicate chains
+    n = 100_000
+    rng = np.random.RandomState(7)
+    lk = rng.randint(0, 1000, n).astype(np.int64)
+    lp = np.arange(n, dtype=np.int64)
+    rk = rng.randint(0, 1000, n).astype(np.int64)
+    rp = np.arange(n, dtype=np.int64)
+    # ~n*n/1000 = 10M output rows
+    _join_parity(dj, lk, lp, rk, rp)
+
+
+def test_join_parity_selectivity_1(dj):
+    n = 200_000
+    bk, bp = oracle.gen_build(n)
+    pk, pp = oracle.gen_probe(n, n, selectivity=1.0)
+    _join_parity(dj, bk, bp, pk, pp)
+
+
+def test_join_empty_sides(dj):
+    n = 1000
+    bk, bp = oracle.gen_build(n)
+    d = dj.DeviceArray.from_numpy(bk)
+    dp = dj.DeviceArray.from_numpy(bp)
+    empty = dj.DeviceArray(1)
+    assert dj.lib().dj_local_inner_join(empty.ptr, empty.ptr, 0, d.ptr, dp.ptr, n,
+                                        None, None, None, None, 0) == 0
+    assert dj.lib().dj_local_inner_join(d.ptr, dp.ptr, n, empty.ptr, empty.ptr, 0,
+                                        None, None, None, None, 0) == 0
+
+
+def test_join_overflow_retry(dj):
+    # cap smaller than the result: count is still exact, retry path works
+    n = 10_000
+    bk, bp = oracle.gen_build(n)
+    pk, pp = oracle.gen_probe(n, n, selectivity=1.0)
+    _join_parity(dj, bk, bp, pk, pp, cap=16)
+
+
+def test_negative_keys(dj):
+    lk = np.array([-5, -3, 0, 7, 2**62, -2**62], dtype=np.int64)
+    lp = np.arange(6, dtype=np.int64)
+    rk = np.array([-3, 7, -5, 123, -2**62], dtype=np.int64)
+    rp = np.arange(5, dtype=np.int64)
+    _join_parity(dj, lk, lp, rk, rp)
