@@ -154,9 +154,17 @@ void dj_hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
     d_offsets_cap = nparts + 1;
   }
   {
+    PhaseScope t(DJ_PHASE_PART_COUNT, stream());
+    dj::partition_count(d_keys, n, nparts, hash_fn, hash_seed, d_scratch, stream());
+  }
+  {
+    PhaseScope t(DJ_PHASE_PART_SCAN, stream());
+    dj::partition_scan(n, nparts, d_scratch, d_offsets, stream());
+  }
+  {
     PhaseScope t(DJ_PHASE_PART_SCATTER, stream());
-    dj::hash_partition(d_keys, d_pay, n, nparts, hash_fn, hash_seed, d_out_keys, d_out_pay,
-                       d_offsets, d_scratch, stream());
+    dj::partition_scatter(d_keys, d_pay, n, nparts, hash_fn, hash_seed, d_offsets, d_scratch,
+                          d_out_keys, d_out_pay, stream());
   }
   if (h_offsets) {
     DJ_HIP_CALL(hipMemcpyAsync(h_offsets, d_offsets, (size_t)(nparts + 1) * sizeof(int64_t),

@@ -250,27 +250,49 @@ __global__ void part_scatter_kernel(const int64_t* __restrict__ keys,
   }
 }
 
+void partition_count(const int64_t* d_keys, int64_t n, int nparts, int hash_fn,
+                     uint32_t hash_seed, void* d_scratch, hipStream_t s)
+{
+  DJ_CHECK_ERROR(nparts >= 1 && nparts <= kMaxPartitions, "nparts must be in [1,64]");
+  if (n <= 0) return;
+  PartGeom g = part_geom(n);
+  hipLaunchKernelGGL(part_count_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, n, nparts,
+                     hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch);
+}
+
+void partition_scan(int64_t n, int nparts, void* d_scratch, int64_t* d_offsets, hipStream_t s)
+{
+  if (n <= 0) {
+    DJ_HIP_CALL(hipMemsetAsync(d_offsets, 0, (size_t)(nparts + 1) * sizeof(int64_t), s));
+    return;
+  }
+  PartGeom g = part_geom(n);
+  int64_t* wave_counts = (int64_t*)d_scratch;
+  int64_t* totals = wave_counts + g.nwaves * nparts;
+  hipLaunchKernelGGL(part_scan_kernel, dim3(nparts), dim3(BLOCK), 0, s, wave_counts, g.nwaves,
+                     nparts, totals);
+  hipLaunchKernelGGL(part_offsets_kernel, dim3(1), dim3(64), 0, s, totals, nparts, d_offsets);
+}
+
+void partition_scatter(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                       int hash_fn, uint32_t hash_seed, const int64_t* d_offsets,
+                       void* d_scratch, int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s)
+{
+  if (n <= 0) return;
+  PartGeom g = part_geom(n);
+  hipLaunchKernelGGL(part_scatter_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, d_pay, n,
+                     nparts, hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch,
+                     d_offsets, d_out_keys, d_out_pay);
+}
+
 void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
                     int hash_fn, uint32_t hash_seed, int64_t* d_out_keys, int64_t* d_out_pay,
                     int64_t* d_offsets, void* d_scratch, hipStream_t s)
 {
-  DJ_CHECK_ERROR(nparts >= 1 && nparts <= kMaxPartitions, "nparts must be in [1,64]");
-  PartGeom g = part_geom(n);
-  int64_t* wave_counts = (int64_t*)d_scratch;
-  int64_t* totals = wave_counts + g.nwaves * nparts;
-  if (n <= 0) {
-    /* still produce zero offsets */
-    hipLaunchKernelGGL(part_offsets_kernel, dim3(1), dim3(64), 0, s, totals, 0, d_offsets);
-    return;
-  }
-  hipLaunchKernelGGL(part_count_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, n, nparts,
-                     hash_fn, hash_seed, g.rows_per_wave, wave_counts);
-  hipLaunchKernelGGL(part_scan_kernel, dim3(nparts), dim3(BLOCK), 0, s, wave_counts, g.nwaves,
-                     nparts, totals);
-  hipLaunchKernelGGL(part_offsets_kernel, dim3(1), dim3(64), 0, s, totals, nparts, d_offsets);
-  hipLaunchKernelGGL(part_scatter_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, d_pay, n,
-                     nparts, hash_fn, hash_seed, g.rows_per_wave, wave_counts, d_offsets,
-                     d_out_keys, d_out_pay);
+  partition_count(d_keys, n, nparts, hash_fn, hash_seed, d_scratch, s);
+  partition_scan(n, nparts, d_scratch, d_offsets, s);
+  partition_scatter(d_keys, d_pay, n, nparts, hash_fn, hash_seed, d_offsets, d_scratch,
+                    d_out_keys, d_out_pay, s);
 }
 
 /* ------------------------------------------------------------ local join */

@@ -34,6 +34,14 @@ void generate_probe(int64_t* d_keys, int64_t* d_pay, int64_t build_n_global, int
 
 /* ----- stable hash partition (histogram + scan + wave-ballot scatter) ----- */
 size_t hash_partition_scratch_bytes(int64_t n, int nparts);
+/* sub-steps, so the C ABI can time each kernel separately */
+void partition_count(const int64_t* d_keys, int64_t n, int nparts, int hash_fn,
+                     uint32_t hash_seed, void* d_scratch, hipStream_t s);
+void partition_scan(int64_t n, int nparts, void* d_scratch, int64_t* d_offsets, hipStream_t s);
+void partition_scatter(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                       int hash_fn, uint32_t hash_seed, const int64_t* d_offsets,
+                       void* d_scratch, int64_t* d_out_keys, int64_t* d_out_pay,
+                       hipStream_t s);
 /* d_offsets: device array of nparts+1 int64 partition offsets (offsets[0]=0).
  * Stable: rows keep input order inside each partition. nparts <= 64. */
 void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
