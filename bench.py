@@ -132,10 +132,9 @@ def main():
     else:
         recv = part
 
-    # hash table + outputs
+    # hash table ({key,val} pairs) + outputs
     nslots = L.dj_join_table_slots(recv_cap)
-    d_slot_keys = L.dj_dmalloc(nslots * 8)
-    d_slot_vals = L.dj_dmalloc(nslots * 8)
+    d_table = L.dj_dmalloc(nslots * 16)
     d_error = L.dj_dmalloc(4)
     d_counter = L.dj_dmalloc(8)
     out_cap = recv_cap + (recv_cap >> 3)
@@ -179,10 +178,9 @@ def main():
         else:
             lrows, rrows = rows, rows
         # local join: table init + build + probe-append
-        L.dj_join_table_init(d_slot_keys, nslots)
-        L.dj_join_build(recv[0].ptr, recv[1].ptr, lrows, d_slot_keys, d_slot_vals,
-                        nslots, d_error)
-        L.dj_join_probe(recv[2].ptr, recv[3].ptr, rrows, d_slot_keys, d_slot_vals, nslots,
+        L.dj_join_table_init(d_table, nslots)
+        L.dj_join_build(recv[0].ptr, recv[1].ptr, lrows, d_table, nslots, d_error)
+        L.dj_join_probe(recv[2].ptr, recv[3].ptr, rrows, d_table, nslots,
                         outs[0].ptr, outs[1].ptr, outs[2].ptr, outs[3].ptr,
                         out_cap, d_counter)
         state["lrows"], state["rrows"] = lrows, rrows

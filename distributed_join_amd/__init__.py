@@ -58,8 +58,8 @@ def lib():
             "dj_hash_partition": ([vp, vp, i64, i32, i32, u32, vp, vp, vp, vp], None),
             "dj_join_table_slots": ([i64], i64),
             "dj_join_table_init": ([vp, i64], None),
-            "dj_join_build": ([vp, vp, i64, vp, vp, i64, vp], None),
-            "dj_join_probe": ([vp, vp, i64, vp, vp, i64, vp, vp, vp, vp, i64, vp], None),
+            "dj_join_build": ([vp, vp, i64, vp, i64, vp], None),
+            "dj_join_probe": ([vp, vp, i64, vp, i64, vp, vp, vp, vp, i64, vp], None),
             "dj_read_counter_i64": ([vp], i64),
             "dj_read_error_i32": ([vp], i32),
             "dj_local_inner_join": ([vp, vp, i64, vp, vp, i64, vp, vp, vp, vp, i64], i64),

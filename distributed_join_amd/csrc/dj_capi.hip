@@ -177,26 +177,26 @@ void dj_hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
 
 int64_t dj_join_table_slots(int64_t ln) { return dj::join_table_slots(ln); }
 
-void dj_join_table_init(int64_t* d_slot_keys, int64_t nslots)
+void dj_join_table_init(int64_t* d_table, int64_t nslots)
 {
   PhaseScope t(DJ_PHASE_TABLE_INIT, stream());
-  dj::join_table_init(d_slot_keys, nslots, stream());
+  dj::join_table_init(d_table, nslots, stream());
 }
 
 void dj_join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
-                   int64_t* d_slot_keys, int64_t* d_slot_vals, int64_t nslots, int* d_error)
+                   int64_t* d_table, int64_t nslots, int* d_error)
 {
   PhaseScope t(DJ_PHASE_BUILD, stream());
-  dj::join_build(d_lk, d_lp, ln, d_slot_keys, d_slot_vals, nslots, d_error, stream());
+  dj::join_build(d_lk, d_lp, ln, d_table, nslots, d_error, stream());
 }
 
 void dj_join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
-                   const int64_t* d_slot_keys, const int64_t* d_slot_vals, int64_t nslots,
+                   const int64_t* d_table, int64_t nslots,
                    int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3,
                    int64_t cap, int64_t* d_counter)
 {
   PhaseScope t(DJ_PHASE_PROBE, stream());
-  dj::join_probe(d_rk, d_rp, rn, d_slot_keys, d_slot_vals, nslots, d_out0, d_out1, d_out2,
+  dj::join_probe(d_rk, d_rp, rn, d_table, nslots, d_out0, d_out1, d_out2,
                  d_out3, cap, d_counter, stream());
 }
 
@@ -221,21 +221,19 @@ int64_t dj_local_inner_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln
 {
   if (ln == 0 || rn == 0) return 0;  // empty side => empty (distributed_join.cpp:76-83)
   int64_t nslots = dj::join_table_slots(ln);
-  int64_t* d_slot_keys = (int64_t*)dj_dmalloc(nslots * sizeof(int64_t));
-  int64_t* d_slot_vals = (int64_t*)dj_dmalloc(nslots * sizeof(int64_t));
+  int64_t* d_table = (int64_t*)dj_dmalloc(nslots * 2 * sizeof(int64_t));
   int* d_error = (int*)dj_dmalloc(sizeof(int));
   int64_t* d_counter = (int64_t*)dj_dmalloc(sizeof(int64_t));
   DJ_HIP_CALL(hipMemsetAsync(d_error, 0, sizeof(int), stream()));
   DJ_HIP_CALL(hipMemsetAsync(d_counter, 0, sizeof(int64_t), stream()));
-  dj_join_table_init(d_slot_keys, nslots);
-  dj_join_build(d_lk, d_lp, ln, d_slot_keys, d_slot_vals, nslots, d_error);
-  dj_join_probe(d_rk, d_rp, rn, d_slot_keys, d_slot_vals, nslots, d_out0, d_out1, d_out2,
+  dj_join_table_init(d_table, nslots);
+  dj_join_build(d_lk, d_lp, ln, d_table, nslots, d_error);
+  dj_join_probe(d_rk, d_rp, rn, d_table, nslots, d_out0, d_out1, d_out2,
                 d_out3, cap, d_counter);
   int64_t n = dj_read_counter_i64(d_counter);
   DJ_CHECK_ERROR(dj_read_error_i32(d_error) == 0,
                  "dj_join_build: key equal to the empty sentinel (-1) is unsupported");
-  dj_dfree(d_slot_keys);
-  dj_dfree(d_slot_vals);
+  dj_dfree(d_table);
   dj_dfree(d_error);
   dj_dfree(d_counter);
   return n;

@@ -304,15 +304,18 @@ int64_t join_table_slots(int64_t ln)
   return p;
 }
 
-void join_table_init(int64_t* d_slot_keys, int64_t nslots, hipStream_t s)
+void join_table_init(int64_t* d_table, int64_t nslots, hipStream_t s)
 {
-  /* kEmptyKey == -1 == all bytes 0xFF: one HBM-rate memset */
-  DJ_HIP_CALL(hipMemsetAsync(d_slot_keys, 0xFF, (size_t)nslots * sizeof(int64_t), s));
+  /* kEmptyKey == -1 == all bytes 0xFF: one HBM-rate memset over the
+   * interleaved {key,val} pairs */
+  DJ_HIP_CALL(hipMemsetAsync(d_table, 0xFF, (size_t)nslots * 2 * sizeof(int64_t), s));
 }
 
+/* Table layout: nslots interleaved 16 B {key, val} pairs — one random
+ * cache-line fetch serves both the key compare and the payload read. */
 __global__ void join_build_kernel(const int64_t* __restrict__ lk, const int64_t* __restrict__ lp,
-                                  int64_t ln, int64_t* __restrict__ slot_keys,
-                                  int64_t* __restrict__ slot_vals, uint64_t mask, int* error)
+                                  int64_t ln, longlong2* __restrict__ table, uint64_t mask,
+                                  int* error)
 {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
@@ -324,65 +327,92 @@ __global__ void join_build_kernel(const int64_t* __restrict__ lk, const int64_t*
     }
     uint64_t slot = dj_mix64((uint64_t)key) & mask;
     for (;;) {
-      unsigned long long old = atomicCAS((unsigned long long*)&slot_keys[slot],
+      unsigned long long old = atomicCAS((unsigned long long*)&table[slot].x,
                                          (unsigned long long)kEmptyKey,
                                          (unsigned long long)key);
       if (old == (unsigned long long)kEmptyKey) break;
       slot = (slot + 1) & mask;
     }
-    slot_vals[slot] = lp ? lp[i] : i;
+    table[slot].y = lp ? lp[i] : i;
   }
 }
 
-void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d_slot_keys,
-                int64_t* d_slot_vals, int64_t nslots, int* d_error, hipStream_t s)
+void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d_table,
+                int64_t nslots, int* d_error, hipStream_t s)
 {
   if (ln <= 0) return;
   hipLaunchKernelGGL(join_build_kernel, dim3(grid_for(ln)), dim3(BLOCK), 0, s, d_lk, d_lp, ln,
-                     d_slot_keys, d_slot_vals, (uint64_t)(nslots - 1), d_error);
+                     (longlong2*)d_table, (uint64_t)(nslots - 1), d_error);
 }
 
+/* Probe with wave-aggregated output append: matches are emitted via one
+ * atomicAdd per wave (ballot leader) instead of one per lane — the single
+ * global counter would otherwise serialize the whole kernel. All lanes of a
+ * wave iterate in lockstep so the ballots are well-defined. */
 __global__ void join_probe_kernel(const int64_t* __restrict__ rk, const int64_t* __restrict__ rp,
-                                  int64_t rn, const int64_t* __restrict__ slot_keys,
-                                  const int64_t* __restrict__ slot_vals, uint64_t mask,
-                                  int64_t* __restrict__ out0, int64_t* __restrict__ out1,
-                                  int64_t* __restrict__ out2, int64_t* __restrict__ out3,
-                                  int64_t cap, int64_t* counter)
+                                  int64_t rn, const longlong2* __restrict__ table,
+                                  uint64_t mask, int64_t* __restrict__ out0,
+                                  int64_t* __restrict__ out1, int64_t* __restrict__ out2,
+                                  int64_t* __restrict__ out3, int64_t cap,
+                                  unsigned long long* counter)
 {
-  int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; j < rn; j += stride) {
-    int64_t key = rk[j];
-    int64_t payload = rp ? rp[j] : j;
-    uint64_t slot = dj_mix64((uint64_t)key) & mask;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;; j += stride) {
+    const bool row_valid = j < rn;
+    if (__ballot(row_valid) == 0) break;  // whole wave done (uniform)
+    int64_t key = 0, payload = 0;
+    uint64_t slot = 0;
+    bool walking = row_valid;
+    if (row_valid) {
+      key = rk[j];
+      payload = rp ? rp[j] : j;
+      slot = dj_mix64((uint64_t)key) & mask;
+    }
     for (;;) {
-      int64_t sk = slot_keys[slot];
-      if (sk == kEmptyKey) break;
-      if (sk == key) {
-        /* per-lane atomicAdd(ptr,1): hipcc wave-aggregates active lanes into
-         * one global atomic, so contention is one atomic per wave-visit */
-        int64_t idx = (int64_t)atomicAdd((unsigned long long*)counter, 1ull);
+      int64_t mval = 0;
+      bool have = false;
+      while (walking) {
+        longlong2 e = table[slot];
+        if (e.x == kEmptyKey) {
+          walking = false;
+          break;
+        }
+        slot = (slot + 1) & mask;
+        if (e.x == key) {
+          mval = e.y;
+          have = true;
+          break;
+        }
+      }
+      uint64_t m = __ballot(have);
+      if (m == 0) break;
+      const int leader = (int)(__ffsll((unsigned long long)m) - 1);
+      unsigned long long base = 0;
+      if (lane == leader) base = atomicAdd(counter, (unsigned long long)__popcll(m));
+      base = __shfl(base, leader);
+      if (have) {
+        int64_t idx = (int64_t)base + __popcll(m & lt_mask);
         if (idx < cap) {
           out0[idx] = key;
-          out1[idx] = slot_vals[slot];
+          out1[idx] = mval;
           out2[idx] = key;
           out3[idx] = payload;
         }
       }
-      slot = (slot + 1) & mask;
     }
   }
 }
 
-void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
-                const int64_t* d_slot_keys, const int64_t* d_slot_vals, int64_t nslots,
-                int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3,
-                int64_t cap, int64_t* d_counter, hipStream_t s)
+void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int64_t* d_table,
+                int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s)
 {
   if (rn <= 0) return;
   hipLaunchKernelGGL(join_probe_kernel, dim3(grid_for(rn)), dim3(BLOCK), 0, s, d_rk, d_rp, rn,
-                     d_slot_keys, d_slot_vals, (uint64_t)(nslots - 1), d_out0, d_out1, d_out2,
-                     d_out3, cap, d_counter);
+                     (const longlong2*)d_table, (uint64_t)(nslots - 1), d_out0, d_out1, d_out2,
+                     d_out3, cap, (unsigned long long*)d_counter);
 }
 
 }  // namespace dj

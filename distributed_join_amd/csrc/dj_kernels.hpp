@@ -49,20 +49,22 @@ void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int 
                     int64_t* d_offsets, void* d_scratch, hipStream_t s);
 
 /* ----- local inner join: open-addressing build + probe-append ----- */
-/* Number of table slots for ln build rows (power of two, <=50% fill). */
+/* Number of table slots for ln build rows (power of two, <=50% fill).
+ * The table buffer holds nslots interleaved 16 B {key,val} pairs
+ * (2*nslots int64). */
 int64_t join_table_slots(int64_t ln);
-/* Initialize table key array to kEmptyKey (async memset). */
-void join_table_init(int64_t* d_slot_keys, int64_t nslots, hipStream_t s);
+/* Initialize table pairs to kEmptyKey (async memset over 16*nslots bytes). */
+void join_table_init(int64_t* d_table, int64_t nslots, hipStream_t s);
 /* Insert build rows. d_error (device int32) set to 1 if any key==kEmptyKey. */
-void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d_slot_keys,
-                int64_t* d_slot_vals, int64_t nslots, int* d_error, hipStream_t s);
+void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d_table,
+                int64_t nslots, int* d_error, hipStream_t s);
 /* Probe rows; append matches (lkey, lpay, rkey, rpay) to the 4 output
- * columns at positions drawn from d_counter (device int64, caller-zeroed).
- * Rows beyond `cap` are counted but not written (caller re-runs bigger). */
-void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
-                const int64_t* d_slot_keys, const int64_t* d_slot_vals, int64_t nslots,
-                int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3,
-                int64_t cap, int64_t* d_counter, hipStream_t s);
+ * columns at positions drawn from d_counter (device int64, caller-zeroed),
+ * one wave-aggregated atomic per emit round. Rows beyond `cap` are counted
+ * but not written (caller re-runs bigger). */
+void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int64_t* d_table,
+                int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s);
 
 /* ----- small utilities ----- */
 void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s);

@@ -56,19 +56,19 @@ void dj_hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        int64_t* d_out_pay, int64_t* h_offsets, void* d_scratch);
 
 /* ---------------- local inner join (replaces cudf::inner_join as called at
- * distributed_join.cpp:71-83). Open-addressing table: nslots =
- * dj_join_table_slots(ln) slots; caller allocates d_slot_keys/d_slot_vals
- * (nslots int64 each), d_error (1 int32, zeroed), d_counter (1 int64,
- * zeroed). Output columns are (lkey, lpay, rkey, rpay) — left columns then
- * right columns with the join key duplicated, row order unspecified
- * (reference pin: compare_against_single_gpu.cu:163-174). Build keys equal
- * to -1 (the empty-slot sentinel) are a loud error. */
+ * distributed_join.cpp:71-83). Open-addressing table of nslots =
+ * dj_join_table_slots(ln) interleaved 16 B {key,val} pairs; caller
+ * allocates d_table (2*nslots int64), d_error (1 int32, zeroed), d_counter
+ * (1 int64, zeroed). Output columns are (lkey, lpay, rkey, rpay) — left
+ * columns then right columns with the join key duplicated, row order
+ * unspecified (reference pin: compare_against_single_gpu.cu:163-174).
+ * Build keys equal to -1 (the empty-slot sentinel) are a loud error. */
 int64_t dj_join_table_slots(int64_t ln);
-void dj_join_table_init(int64_t* d_slot_keys, int64_t nslots);
+void dj_join_table_init(int64_t* d_table, int64_t nslots);
 void dj_join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
-                   int64_t* d_slot_keys, int64_t* d_slot_vals, int64_t nslots, int* d_error);
+                   int64_t* d_table, int64_t nslots, int* d_error);
 void dj_join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
-                   const int64_t* d_slot_keys, const int64_t* d_slot_vals, int64_t nslots,
+                   const int64_t* d_table, int64_t nslots,
                    int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3,
                    int64_t cap, int64_t* d_counter);
 int64_t dj_read_counter_i64(const int64_t* d_counter);
