@@ -132,9 +132,8 @@ def main():
     else:
         recv = part
 
-    # hash table ({key,val} pairs) + outputs
-    nslots = L.dj_join_table_slots(recv_cap)
-    d_table = L.dj_dmalloc(nslots * 16)
+    # bucketed-join scratch + outputs
+    bucket_scratch = L.dj_dmalloc(L.dj_bucket_join_scratch_bytes(recv_cap, recv_cap))
     d_error = L.dj_dmalloc(4)
     d_counter = L.dj_dmalloc(8)
     out_cap = recv_cap + (recv_cap >> 3)
@@ -177,12 +176,11 @@ def main():
                                       (scnt_p.sum() - scnt_p[rank]) * 2)
         else:
             lrows, rrows = rows, rows
-        # local join: table init + build + probe-append
-        L.dj_join_table_init(d_table, nslots)
-        L.dj_join_build(recv[0].ptr, recv[1].ptr, lrows, d_table, nslots, d_error)
-        L.dj_join_probe(recv[2].ptr, recv[3].ptr, rrows, d_table, nslots,
-                        outs[0].ptr, outs[1].ptr, outs[2].ptr, outs[3].ptr,
-                        out_cap, d_counter)
+        # local join: bucketed LDS build + probe (fused)
+        L.dj_bucket_local_join(recv[0].ptr, recv[1].ptr, lrows,
+                               recv[2].ptr, recv[3].ptr, rrows,
+                               outs[0].ptr, outs[1].ptr, outs[2].ptr, outs[3].ptr,
+                               out_cap, d_counter, d_error, bucket_scratch)
         state["lrows"], state["rrows"] = lrows, rrows
 
     def barrier_sync():
@@ -224,17 +222,17 @@ def main():
               for name, pid in dj.PHASES.items()}
 
     # roofline for the dominant join kernel (algorithmic bytes: DESIGN.md §Measurement)
-    probe_ms = phases["probe"]["ms"]
-    scatter_ms = phases["part_scatter"]["ms"]
-    build_ms = phases["build"]["ms"]
-    # algorithmic bytes per step (per this rank):
+    # bucket_scatter phase = full bucket partition of both tables
+    # (count 8 B read + scatter 16 B read + 16 B write per row);
+    # join_fused = streaming read of both bucketed tables + output writes
+    # (LDS table traffic is on-chip, not HBM-algorithmic);
+    # part_scatter = rank-level stable scatter of both tables.
     alg = {
-        "part_scatter": 40.0 * rows * 2,   # both tables: 8 hist reread + 16 read + 16 write
-        "build": 32.0 * state["lrows"],    # 16 read + 16 table store
-        "probe": (16.0 + 16.0 * 1.5) * state["rrows"] + 32.0 * matches,
+        "part_scatter": 32.0 * rows * 2,
+        "bucket_scatter": 40.0 * (state["lrows"] + state["rrows"]),
+        "join_fused": 16.0 * (state["lrows"] + state["rrows"]) + 32.0 * matches,
     }
-    dom = max(("probe", probe_ms), ("part_scatter", scatter_ms), ("build", build_ms),
-              key=lambda kv: kv[1])[0]
+    dom = max(alg.keys(), key=lambda k: phases[k]["ms"])
     dom_ms = phases[dom]["ms"]
     achieved = (alg[dom] / 1e9) / (dom_ms / 1e3) if dom_ms > 0 else None
     roofline = {

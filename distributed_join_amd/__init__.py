@@ -23,6 +23,7 @@ DEFAULT_SEED = 1234
 PHASES = {
     "generate": 0, "part_count": 1, "part_scan": 2, "part_scatter": 3,
     "table_init": 4, "build": 5, "probe": 6, "comm": 7, "concat": 8,
+    "bucket_count": 9, "bucket_scan": 10, "bucket_scatter": 11, "join_fused": 12,
 }
 
 _lib = None
@@ -63,6 +64,10 @@ def lib():
             "dj_read_counter_i64": ([vp], i64),
             "dj_read_error_i32": ([vp], i32),
             "dj_local_inner_join": ([vp, vp, i64, vp, vp, i64, vp, vp, vp, vp, i64], i64),
+            "dj_local_inner_join_global": ([vp, vp, i64, vp, vp, i64, vp, vp, vp, vp, i64], i64),
+            "dj_bucket_join_scratch_bytes": ([i64, i64], i64),
+            "dj_bucket_local_join": ([vp, vp, i64, vp, vp, i64, vp, vp, vp, vp, i64,
+                                      vp, vp, vp], None),
             "dj_timing_enable": ([i32], None),
             "dj_timing_reset": ([], None),
             "dj_timing_total_ms": ([i32], dbl),
@@ -175,6 +180,26 @@ def local_inner_join(d_lk, d_lp, ln, d_rk, d_rp, rn, cap=None):
         outs = [DeviceArray(cap) for _ in range(4)]
         n = L.dj_local_inner_join(d_lk.ptr, d_lp.ptr, ln, d_rk.ptr, d_rp.ptr, rn,
                                   outs[0].ptr, outs[1].ptr, outs[2].ptr, outs[3].ptr, cap)
+        if n <= cap:
+            res = tuple(o.to_numpy(n) for o in outs)
+            for o in outs:
+                o.free()
+            return res
+        for o in outs:
+            o.free()
+        cap = n
+
+
+def local_inner_join_global(d_lk, d_lp, ln, d_rk, d_rp, rn, cap=None):
+    """Global-table engine (the skew-fallback path), for cross-checking."""
+    L = lib()
+    if cap is None:
+        cap = max(int(rn) * 2, 16)
+    while True:
+        outs = [DeviceArray(cap) for _ in range(4)]
+        n = L.dj_local_inner_join_global(d_lk.ptr, d_lp.ptr, ln, d_rk.ptr, d_rp.ptr, rn,
+                                         outs[0].ptr, outs[1].ptr, outs[2].ptr, outs[3].ptr,
+                                         cap)
         if n <= cap:
             res = tuple(o.to_numpy(n) for o in outs)
             for o in outs:

@@ -220,6 +220,132 @@ int64_t dj_local_inner_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln
                             int64_t* d_out3, int64_t cap)
 {
   if (ln == 0 || rn == 0) return 0;  // empty side => empty (distributed_join.cpp:76-83)
+  void* d_scratch = dj_dmalloc(dj_bucket_join_scratch_bytes(ln, rn));
+  int* d_error = (int*)dj_dmalloc(sizeof(int));
+  int64_t* d_counter = (int64_t*)dj_dmalloc(sizeof(int64_t));
+  DJ_HIP_CALL(hipMemsetAsync(d_error, 0, sizeof(int), stream()));
+  DJ_HIP_CALL(hipMemsetAsync(d_counter, 0, sizeof(int64_t), stream()));
+  dj_bucket_local_join(d_lk, d_lp, ln, d_rk, d_rp, rn, d_out0, d_out1, d_out2, d_out3, cap,
+                       d_counter, d_error, d_scratch);
+  int64_t n = dj_read_counter_i64(d_counter);
+  DJ_CHECK_ERROR(dj_read_error_i32(d_error) == 0,
+                 "dj build: key equal to the empty sentinel (-1) is unsupported");
+  dj_dfree(d_scratch);
+  dj_dfree(d_error);
+  dj_dfree(d_counter);
+  return n;
+}
+
+/* ---------------- bucketed LDS local join ---------------- */
+
+namespace {
+
+struct BucketScratch {
+  int64_t *lbk, *lbp, *rbk, *rbp;  // bucketed copies
+  int64_t *loff, *roff;            // int64[B+1]
+  uint32_t* counts;                // u32[kBucketBlocks*B]
+  uint32_t* totals;                // u32[B]
+  uint32_t* flags;                 // u32[B]
+  int* any_overflow;               // int[1]
+};
+
+BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
+{
+  char* p = (char*)base;
+  auto take = [&](size_t bytes) {
+    void* r = p;
+    p += (bytes + 255) & ~(size_t)255;
+    return r;
+  };
+  BucketScratch s;
+  s.lbk = (int64_t*)take((size_t)ln * 8);
+  s.lbp = (int64_t*)take((size_t)ln * 8);
+  s.rbk = (int64_t*)take((size_t)rn * 8);
+  s.rbp = (int64_t*)take((size_t)rn * 8);
+  s.loff = (int64_t*)take((size_t)(B + 1) * 8);
+  s.roff = (int64_t*)take((size_t)(B + 1) * 8);
+  s.counts = (uint32_t*)take((size_t)dj::kBucketBlocks * B * 4);
+  s.totals = (uint32_t*)take((size_t)B * 4);
+  s.flags = (uint32_t*)take((size_t)B * 4);
+  s.any_overflow = (int*)take(16);
+  return s;
+}
+
+}  // namespace
+
+int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
+{
+  int B = dj::bucket_count_for(ln, rn);
+  size_t bytes = 0;
+  auto add = [&](size_t b) { bytes += (b + 255) & ~(size_t)255; };
+  add((size_t)ln * 8);
+  add((size_t)ln * 8);
+  add((size_t)rn * 8);
+  add((size_t)rn * 8);
+  add((size_t)(B + 1) * 8);
+  add((size_t)(B + 1) * 8);
+  add((size_t)dj::kBucketBlocks * B * 4);
+  add((size_t)B * 4);
+  add((size_t)B * 4);
+  add(16);
+  return (int64_t)bytes;
+}
+
+void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                          const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                          int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                          int64_t* d_out3, int64_t cap, int64_t* d_counter, int* d_error,
+                          void* d_scratch)
+{
+  if (ln == 0 || rn == 0) return;  // empty side => empty (distributed_join.cpp:76-83)
+  const int B = dj::bucket_count_for(ln, rn);
+  BucketScratch s = carve_bucket_scratch(d_scratch, ln, rn, B);
+  hipStream_t st = stream();
+  {
+    PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
+    dj::bucket_partition(d_lk, d_lp, ln, B, s.counts, s.totals, s.loff, s.lbk, s.lbp, st);
+    dj::bucket_partition(d_rk, d_rp, rn, B, s.counts, s.totals, s.roff, s.rbk, s.rbp, st);
+  }
+  DJ_HIP_CALL(hipMemsetAsync(s.any_overflow, 0, sizeof(int), st));
+  DJ_HIP_CALL(hipMemsetAsync(s.flags, 0, (size_t)B * 4, st));
+  {
+    PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
+    dj::lds_join(s.lbk, s.lbp, s.loff, s.rbk, s.rbp, s.roff, B, d_out0, d_out1, d_out2,
+                 d_out3, cap, d_counter, s.flags, s.any_overflow, d_error, st);
+  }
+  /* skew fallback: buckets whose build side exceeded the LDS row cap */
+  int any = 0;
+  DJ_HIP_CALL(hipMemcpyAsync(&any, s.any_overflow, sizeof(int), hipMemcpyDeviceToHost, st));
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  if (any) {
+    std::vector<uint32_t> flags((size_t)B);
+    std::vector<int64_t> loff((size_t)B + 1), roff((size_t)B + 1);
+    DJ_HIP_CALL(hipMemcpy(flags.data(), s.flags, (size_t)B * 4, hipMemcpyDeviceToHost));
+    DJ_HIP_CALL(hipMemcpy(loff.data(), s.loff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
+    DJ_HIP_CALL(hipMemcpy(roff.data(), s.roff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
+    for (int b = 0; b < B; b++) {
+      if (!flags[b]) continue;
+      int64_t lnb = loff[b + 1] - loff[b];
+      int64_t rnb = roff[b + 1] - roff[b];
+      if (lnb == 0 || rnb == 0) continue;
+      int64_t nslots = dj::join_table_slots(lnb);
+      int64_t* d_table = (int64_t*)dj_dmalloc(nslots * 2 * sizeof(int64_t));
+      dj_join_table_init(d_table, nslots);
+      dj_join_build(s.lbk + loff[b], s.lbp + loff[b], lnb, d_table, nslots, d_error);
+      dj_join_probe(s.rbk + roff[b], s.rbp + roff[b], rnb, d_table, nslots, d_out0, d_out1,
+                    d_out2, d_out3, cap, d_counter);
+      DJ_HIP_CALL(hipStreamSynchronize(st));
+      dj_dfree(d_table);
+    }
+  }
+}
+
+int64_t dj_local_inner_join_global(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                                   const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                                   int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                                   int64_t* d_out3, int64_t cap)
+{
+  if (ln == 0 || rn == 0) return 0;
   int64_t nslots = dj::join_table_slots(ln);
   int64_t* d_table = (int64_t*)dj_dmalloc(nslots * 2 * sizeof(int64_t));
   int* d_error = (int*)dj_dmalloc(sizeof(int));
@@ -228,8 +354,8 @@ int64_t dj_local_inner_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln
   DJ_HIP_CALL(hipMemsetAsync(d_counter, 0, sizeof(int64_t), stream()));
   dj_join_table_init(d_table, nslots);
   dj_join_build(d_lk, d_lp, ln, d_table, nslots, d_error);
-  dj_join_probe(d_rk, d_rp, rn, d_table, nslots, d_out0, d_out1, d_out2,
-                d_out3, cap, d_counter);
+  dj_join_probe(d_rk, d_rp, rn, d_table, nslots, d_out0, d_out1, d_out2, d_out3, cap,
+                d_counter);
   int64_t n = dj_read_counter_i64(d_counter);
   DJ_CHECK_ERROR(dj_read_error_i32(d_error) == 0,
                  "dj_join_build: key equal to the empty sentinel (-1) is unsupported");

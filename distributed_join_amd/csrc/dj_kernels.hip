@@ -415,4 +415,255 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
                      d_out3, cap, (unsigned long long*)d_counter);
 }
 
+/* ----------------------------------------------- bucketed LDS join ------ */
+/*
+ * MI355X-native local join: instead of one HBM-resident hash table (random
+ * 128 B line fetches + device-scope atomics bound the kernel far below the
+ * HBM roofline), partition both tables into B buckets small enough that a
+ * bucket's hash table fits in LDS (160 KiB/CU), then one fused kernel per
+ * bucket: build an 8192-slot LDS table (ds atomicCAS) and probe it — every
+ * random access is LDS traffic (~150 TB/s chip-wide), HBM sees only
+ * streaming reads/writes. The bucket decomposition is internal to the local
+ * join (row order of the join result is unspecified by the API), so the
+ * bucket scatter is NON-stable — unlike the rank-level hash_partition above,
+ * which stays stable (reference pin: SURVEY.md appendix, batch offsets).
+ *
+ * Bucket id = high 32 bits of dj_mix64(key); LDS slot = low 32 bits — the
+ * rank-level partition uses MurmurHash3%G, so bucket occupancy is
+ * independent of which rank the row landed on.
+ */
+
+constexpr int BUCKET_BLOCKS = 256;   // one per CU; each owns a contiguous chunk
+constexpr int BUCKET_THREADS = 1024;
+constexpr int JOIN_LDS_SLOTS = 8192;          // 128 KiB of longlong2 pairs
+constexpr int JOIN_BUCKET_ROW_CAP = 6144;     // 75% fill cap; above -> fallback
+
+int bucket_count_for(int64_t ln, int64_t rn)
+{
+  int64_t maxn = ln > rn ? ln : rn;
+  int B = 256;
+  while (B < 32768 && maxn / B > 3000) B <<= 1;
+  return B;
+}
+
+__device__ __forceinline__ uint32_t bucket_of(int64_t key, int B)
+{
+  return (uint32_t)(dj_mix64((uint64_t)key) >> 32) & (uint32_t)(B - 1);
+}
+
+/* grid = BUCKET_BLOCKS; dynamic LDS: u32 hist[B] */
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_count_kernel(
+  const int64_t* __restrict__ keys, int64_t n, int B, uint32_t* __restrict__ counts)
+{
+  extern __shared__ uint32_t hist[];
+  for (int p = threadIdx.x; p < B; p += blockDim.x) hist[p] = 0;
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t start = (int64_t)blockIdx.x * chunk;
+  const int64_t end = min(start + chunk, n);
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    atomicAdd(&hist[bucket_of(keys[i], B)], 1u);
+  __syncthreads();
+  for (int p = threadIdx.x; p < B; p += blockDim.x)
+    counts[(size_t)blockIdx.x * B + p] = hist[p];
+}
+
+/* grid = B blocks x BUCKET_BLOCKS threads: exclusive scan of counts[:, p]
+ * over blocks (in place), totals[p] = column sum */
+__global__ void bucket_scanA_kernel(uint32_t* counts, int B, uint32_t* totals)
+{
+  const int p = blockIdx.x;
+  __shared__ uint32_t sh[BUCKET_BLOCKS];
+  uint32_t v = counts[(size_t)threadIdx.x * B + p];
+  sh[threadIdx.x] = v;
+  __syncthreads();
+  for (int off = 1; off < BUCKET_BLOCKS; off <<= 1) {
+    uint32_t add = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
+    __syncthreads();
+    sh[threadIdx.x] += add;
+    __syncthreads();
+  }
+  counts[(size_t)threadIdx.x * B + p] = sh[threadIdx.x] - v;  // exclusive
+  if (threadIdx.x == BUCKET_BLOCKS - 1) totals[p] = sh[threadIdx.x];
+}
+
+/* single block: exclusive scan of totals[B] -> offsets[B+1] (int64) */
+__global__ void bucket_scanB_kernel(const uint32_t* totals, int B, int64_t* offsets)
+{
+  __shared__ int64_t sh[BUCKET_THREADS];
+  __shared__ int64_t running_sh;
+  if (threadIdx.x == 0) running_sh = 0;
+  __syncthreads();
+  for (int base = 0; base < B; base += BUCKET_THREADS) {
+    int p = base + threadIdx.x;
+    int64_t v = (p < B) ? (int64_t)totals[p] : 0;
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < BUCKET_THREADS; off <<= 1) {
+      int64_t add = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += add;
+      __syncthreads();
+    }
+    int64_t rbase = running_sh;
+    __syncthreads();  // all reads of running_sh precede the update below
+    if (p < B) offsets[p] = rbase + sh[threadIdx.x] - v;
+    if (threadIdx.x == BUCKET_THREADS - 1) running_sh = rbase + sh[threadIdx.x];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) offsets[B] = offsets[B - 1] + (int64_t)totals[B - 1];
+}
+
+/* grid = BUCKET_BLOCKS; dynamic LDS: u32 cursor[B] preloaded with
+ * offsets[p] + this block's exclusive prefix; non-stable scatter */
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_kernel(
+  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int B,
+  const uint32_t* __restrict__ counts, const int64_t* __restrict__ offsets,
+  int64_t* __restrict__ out_keys, int64_t* __restrict__ out_pay)
+{
+  extern __shared__ uint32_t cursor[];
+  for (int p = threadIdx.x; p < B; p += blockDim.x)
+    cursor[p] = (uint32_t)offsets[p] + counts[(size_t)blockIdx.x * B + p];
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t start = (int64_t)blockIdx.x * chunk;
+  const int64_t end = min(start + chunk, n);
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
+    int64_t k = keys[i];
+    int64_t v = pay ? pay[i] : i;
+    uint32_t dst = atomicAdd(&cursor[bucket_of(k, B)], 1u);
+    out_keys[dst] = k;
+    out_pay[dst] = v;
+  }
+}
+
+/* fused per-bucket LDS build + probe */
+__global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
+  const int64_t* __restrict__ lk, const int64_t* __restrict__ lp,
+  const int64_t* __restrict__ loff, const int64_t* __restrict__ rk,
+  const int64_t* __restrict__ rp, const int64_t* __restrict__ roff, int B,
+  int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
+  int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter,
+  uint32_t* __restrict__ overflow_flags, int* __restrict__ any_overflow,
+  int* __restrict__ error)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbl = (longlong2*)smem;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+  const uint32_t smask = JOIN_LDS_SLOTS - 1;
+
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const int64_t l0 = loff[b], l1 = loff[b + 1];
+    const int64_t r0 = roff[b], r1 = roff[b + 1];
+    const int64_t lnb = l1 - l0;
+    if (lnb == 0 || r1 == r0) continue;
+    if (lnb > JOIN_BUCKET_ROW_CAP) {  // skew overflow: host-side fallback joins it
+      if (threadIdx.x == 0) {
+        overflow_flags[b] = 1;
+        *any_overflow = 1;
+      }
+      continue;
+    }
+    /* init table */
+    for (int s = threadIdx.x; s < JOIN_LDS_SLOTS; s += blockDim.x) tbl[s].x = kEmptyKey;
+    __syncthreads();
+    /* build */
+    for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
+      int64_t key = lk[i];
+      if (key == kEmptyKey) {
+        *error = 1;
+        continue;
+      }
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
+      for (;;) {
+        unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
+                                           (unsigned long long)kEmptyKey,
+                                           (unsigned long long)key);
+        if (old == (unsigned long long)kEmptyKey) break;
+        slot = (slot + 1) & smask;
+      }
+      tbl[slot].y = lp[i];
+    }
+    __syncthreads();
+    /* probe (wave-lockstep aggregated emit, as the global probe kernel) */
+    const int64_t rows = r1 - r0;
+    const int64_t iters = (rows + blockDim.x - 1) / blockDim.x;
+    for (int64_t it = 0; it < iters; it++) {
+      const int64_t j = r0 + it * blockDim.x + threadIdx.x;
+      const bool row_valid = j < r1;
+      int64_t key = 0, payload = 0;
+      uint32_t slot = 0;
+      bool walking = row_valid;
+      if (row_valid) {
+        key = rk[j];
+        payload = rp[j];
+        slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
+      }
+      for (;;) {
+        int64_t mval = 0;
+        bool have = false;
+        while (walking) {
+          longlong2 e = tbl[slot];
+          if (e.x == kEmptyKey) {
+            walking = false;
+            break;
+          }
+          slot = (slot + 1) & smask;
+          if (e.x == key) {
+            mval = e.y;
+            have = true;
+            break;
+          }
+        }
+        uint64_t m = __ballot(have);
+        if (m == 0) break;
+        const int leader = (int)(__ffsll((unsigned long long)m) - 1);
+        unsigned long long base = 0;
+        if (lane == leader) base = atomicAdd(counter, (unsigned long long)__popcll(m));
+        base = __shfl(base, leader);
+        if (have) {
+          int64_t idx = (int64_t)base + __popcll(m & lt_mask);
+          if (idx < cap) {
+            out0[idx] = key;
+            out1[idx] = mval;
+            out2[idx] = key;
+            out3[idx] = payload;
+          }
+        }
+      }
+    }
+    __syncthreads();  // table re-initialized next bucket
+  }
+}
+
+void bucket_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
+                      uint32_t* d_counts, uint32_t* d_totals, int64_t* d_offsets,
+                      int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s)
+{
+  DJ_CHECK_ERROR(n < (int64_t)UINT32_MAX, "bucket_partition: n must be < 2^32");
+  size_t lds = (size_t)B * sizeof(uint32_t);
+  hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
+                     d_keys, n, B, d_counts);
+  hipLaunchKernelGGL(bucket_scanA_kernel, dim3(B), dim3(BUCKET_BLOCKS), 0, s, d_counts, B,
+                     d_totals);
+  hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, B,
+                     d_offsets);
+  hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
+                     d_keys, d_pay, n, B, d_counts, d_offsets, d_out_keys, d_out_pay);
+}
+
+void lds_join(const int64_t* d_lk, const int64_t* d_lp, const int64_t* d_loff,
+              const int64_t* d_rk, const int64_t* d_rp, const int64_t* d_roff, int B,
+              int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3, int64_t cap,
+              int64_t* d_counter, uint32_t* d_overflow_flags, int* d_any_overflow,
+              int* d_error, hipStream_t s)
+{
+  int grid = B < 4096 ? B : 4096;
+  hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS),
+                     JOIN_LDS_SLOTS * sizeof(longlong2), s, d_lk, d_lp, d_loff, d_rk, d_rp,
+                     d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
+                     (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow, d_error);
+}
+
 }  // namespace dj

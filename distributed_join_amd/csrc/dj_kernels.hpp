@@ -66,6 +66,25 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
                 int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
                 int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s);
 
+/* ----- bucketed LDS join (the product local-join path; see dj_kernels.hip
+ * "bucketed LDS join" comment block) ----- */
+constexpr int kBucketBlocks = 256;
+constexpr int kJoinBucketRowCap = 6144;
+int bucket_count_for(int64_t ln, int64_t rn);
+/* Non-stable partition into B buckets. d_counts: u32[kBucketBlocks*B];
+ * d_totals: u32[B]; d_offsets: int64[B+1] (device). */
+void bucket_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
+                      uint32_t* d_counts, uint32_t* d_totals, int64_t* d_offsets,
+                      int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s);
+/* Fused per-bucket LDS build+probe over bucketed tables. Buckets whose build
+ * side exceeds kJoinBucketRowCap set overflow_flags[b] and any_overflow and
+ * are skipped (host runs the global-table path on them). */
+void lds_join(const int64_t* d_lk, const int64_t* d_lp, const int64_t* d_loff,
+              const int64_t* d_rk, const int64_t* d_rp, const int64_t* d_roff, int B,
+              int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3, int64_t cap,
+              int64_t* d_counter, uint32_t* d_overflow_flags, int* d_any_overflow,
+              int* d_error, hipStream_t s);
+
 /* ----- small utilities ----- */
 void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s);
 

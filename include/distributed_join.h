@@ -74,13 +74,32 @@ void dj_join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
 int64_t dj_read_counter_i64(const int64_t* d_counter);
 int dj_read_error_i32(const int* d_error);
 
-/* Convenience one-call local join (allocates its own scratch; for tests and
- * smoke, not the bench timed region). Returns the match count; writes at
- * most cap rows. */
+/* ---------------- bucketed LDS local join (the PRODUCT local-join path;
+ * same drop-in semantics as above, different engine: both tables are
+ * bucket-partitioned so each bucket's hash table lives in LDS — no
+ * HBM-resident table, all hot random access on-chip). Buckets whose build
+ * side exceeds the LDS capacity (data skew / heavy duplicates) fall back to
+ * the global-table path transparently. d_counter/d_error as above. */
+int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn);
+void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                          const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                          int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                          int64_t* d_out3, int64_t cap, int64_t* d_counter, int* d_error,
+                          void* d_scratch);
+
+/* Convenience one-call local join over the bucketed path (allocates its own
+ * scratch; for tests and smoke, not the bench timed region). Returns the
+ * match count; writes at most cap rows. */
 int64_t dj_local_inner_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
                             const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
                             int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
                             int64_t* d_out3, int64_t cap);
+/* Same over the global-table build/probe path (the fallback engine),
+ * exposed so tests can cross-check both engines. */
+int64_t dj_local_inner_join_global(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                                   const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                                   int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                                   int64_t* d_out3, int64_t cap);
 
 /* ---------------- phase timing (hipEvent pairs around every kernel launch;
  * replaces the reference's report_timing wall-clock prints,
@@ -95,7 +114,11 @@ enum dj_phase {
   DJ_PHASE_PROBE = 6,
   DJ_PHASE_COMM = 7,
   DJ_PHASE_CONCAT = 8,
-  DJ_PHASE_COUNT_ = 9
+  DJ_PHASE_BUCKET_COUNT = 9,
+  DJ_PHASE_BUCKET_SCAN = 10,
+  DJ_PHASE_BUCKET_SCATTER = 11,
+  DJ_PHASE_JOIN_FUSED = 12,
+  DJ_PHASE_COUNT_ = 13
 };
 void dj_timing_enable(int on);
 void dj_timing_reset(void);

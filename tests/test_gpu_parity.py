@@ -148,6 +148,32 @@ def test_join_overflow_retry(dj):
     _join_parity(dj, bk, bp, pk, pp, cap=16)
 
 
+def test_bucket_vs_global_engines(dj):
+    # the bucketed-LDS engine and the global-table engine must agree
+    n = 500_000
+    bk, bp = oracle.gen_build(n)
+    pk, pp = oracle.gen_probe(n, n, selectivity=0.5)
+    dlk, dlp = dj.DeviceArray.from_numpy(bk), dj.DeviceArray.from_numpy(bp)
+    drk, drp = dj.DeviceArray.from_numpy(pk), dj.DeviceArray.from_numpy(pp)
+    a = oracle.sort_rows(*dj.local_inner_join(dlk, dlp, n, drk, drp, n))
+    b = oracle.sort_rows(*dj.local_inner_join_global(dlk, dlp, n, drk, drp, n))
+    assert len(a[0]) == len(b[0])
+    for x, y in zip(a, b):
+        assert (x == y).all()
+
+
+def test_skew_fallback_single_hot_key(dj):
+    # one key repeated 100k times on the build side => its bucket overflows
+    # the LDS row cap and must take the global-table fallback path
+    n = 100_000
+    lk = np.full(n, 42, dtype=np.int64)
+    lk[:100] = np.arange(100, dtype=np.int64) + 1000  # a few normal buckets too
+    lp = np.arange(n, dtype=np.int64)
+    rk = np.array([42, 1000, 7, 42], dtype=np.int64)
+    rp = np.arange(4, dtype=np.int64)
+    _join_parity(dj, lk, lp, rk, rp)
+
+
 def test_negative_keys(dj):
     lk = np.array([-5, -3, 0, 7, 2**62, -2**62], dtype=np.int64)
     lp = np.arange(6, dtype=np.int64)
