@@ -241,16 +241,20 @@ int64_t dj_local_inner_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln
 namespace {
 
 struct BucketScratch {
-  int64_t *lbk, *lbp, *rbk, *rbp;  // bucketed copies
+  int64_t *lbk, *lbp, *rbk, *rbp;  // bucketed copies (final)
+  int64_t *tmpk, *tmpp;            // pass-A staging (size max(ln,rn))
   int64_t *loff, *roff;            // int64[B+1]
-  uint32_t* counts;                // u32[kBucketBlocks*B]
-  uint32_t* totals;                // u32[B]
+  int64_t* segoff;                 // int64[PA+1]
+  uint32_t* counts;                // u32[kBucketBlocks*PA]
+  uint32_t* totals;                // u32[PA]
   uint32_t* flags;                 // u32[B]
   int* any_overflow;               // int[1]
 };
 
 BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
 {
+  const int PA = B / dj::kSubBuckets;
+  const int64_t maxn = ln > rn ? ln : rn;
   char* p = (char*)base;
   auto take = [&](size_t bytes) {
     void* r = p;
@@ -262,10 +266,13 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
   s.lbp = (int64_t*)take((size_t)ln * 8);
   s.rbk = (int64_t*)take((size_t)rn * 8);
   s.rbp = (int64_t*)take((size_t)rn * 8);
+  s.tmpk = (int64_t*)take((size_t)maxn * 8);
+  s.tmpp = (int64_t*)take((size_t)maxn * 8);
   s.loff = (int64_t*)take((size_t)(B + 1) * 8);
   s.roff = (int64_t*)take((size_t)(B + 1) * 8);
-  s.counts = (uint32_t*)take((size_t)dj::kBucketBlocks * B * 4);
-  s.totals = (uint32_t*)take((size_t)B * 4);
+  s.segoff = (int64_t*)take((size_t)(PA + 1) * 8);
+  s.counts = (uint32_t*)take((size_t)dj::kBucketBlocks * PA * 4);
+  s.totals = (uint32_t*)take((size_t)PA * 4);
   s.flags = (uint32_t*)take((size_t)B * 4);
   s.any_overflow = (int*)take(16);
   return s;
@@ -276,16 +283,21 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
 int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
 {
   int B = dj::bucket_count_for(ln, rn);
+  const int PA = B / dj::kSubBuckets;
+  const int64_t maxn = ln > rn ? ln : rn;
   size_t bytes = 0;
   auto add = [&](size_t b) { bytes += (b + 255) & ~(size_t)255; };
   add((size_t)ln * 8);
   add((size_t)ln * 8);
   add((size_t)rn * 8);
   add((size_t)rn * 8);
+  add((size_t)maxn * 8);
+  add((size_t)maxn * 8);
   add((size_t)(B + 1) * 8);
   add((size_t)(B + 1) * 8);
-  add((size_t)dj::kBucketBlocks * B * 4);
-  add((size_t)B * 4);
+  add((size_t)(PA + 1) * 8);
+  add((size_t)dj::kBucketBlocks * PA * 4);
+  add((size_t)PA * 4);
   add((size_t)B * 4);
   add(16);
   return (int64_t)bytes;
@@ -303,8 +315,10 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
   hipStream_t st = stream();
   {
     PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
-    dj::bucket_partition(d_lk, d_lp, ln, B, s.counts, s.totals, s.loff, s.lbk, s.lbp, st);
-    dj::bucket_partition(d_rk, d_rp, rn, B, s.counts, s.totals, s.roff, s.rbk, s.rbp, st);
+    dj::bucket_partition2(d_lk, d_lp, ln, B, s.tmpk, s.tmpp, s.counts, s.totals, s.segoff,
+                          s.loff, s.lbk, s.lbp, st);
+    dj::bucket_partition2(d_rk, d_rp, rn, B, s.tmpk, s.tmpp, s.counts, s.totals, s.segoff,
+                          s.roff, s.rbk, s.rbp, st);
   }
   DJ_HIP_CALL(hipMemsetAsync(s.any_overflow, 0, sizeof(int), st));
   DJ_HIP_CALL(hipMemsetAsync(s.flags, 0, (size_t)B * 4, st));

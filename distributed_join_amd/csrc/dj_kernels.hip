@@ -417,64 +417,84 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
 
 /* ----------------------------------------------- bucketed LDS join ------ */
 /*
- * MI355X-native local join: instead of one HBM-resident hash table (random
- * 128 B line fetches + device-scope atomics bound the kernel far below the
- * HBM roofline), partition both tables into B buckets small enough that a
- * bucket's hash table fits in LDS (160 KiB/CU), then one fused kernel per
- * bucket: build an 8192-slot LDS table (ds atomicCAS) and probe it — every
- * random access is LDS traffic (~150 TB/s chip-wide), HBM sees only
- * streaming reads/writes. The bucket decomposition is internal to the local
- * join (row order of the join result is unspecified by the API), so the
- * bucket scatter is NON-stable — unlike the rank-level hash_partition above,
- * which stays stable (reference pin: SURVEY.md appendix, batch offsets).
+ * MI355X-native local join. A single HBM-resident hash table is bound far
+ * below the HBM roofline by random 128 B line fetches, device-scope atomic
+ * throughput (~18 G scattered CAS/s measured) and, worst, the single output
+ * counter (~83 M serialized atomics/s measured — experiments/membench.hip).
+ * Instead: partition both tables into B buckets small enough that a
+ * bucket's hash table fits in LDS, then one fused kernel per bucket builds
+ * a 4096-slot LDS table (ds atomicCAS) and probes it twice (count, then
+ * write) — all random access is on-chip, HBM sees only streaming traffic,
+ * and the global output counter is touched ONCE per bucket.
  *
- * Bucket id = high 32 bits of dj_mix64(key); LDS slot = low 32 bits — the
- * rank-level partition uses MurmurHash3%G, so bucket occupancy is
- * independent of which rank the row landed on.
+ * The bucket partition is two-level so every scatter pass has at most 1024
+ * write streams alive per block (partial-line writes then merge in the
+ * per-XCD L2 — a single-pass 32768-way scatter measurably thrashes it):
+ *   pass A: P_A = B/256 groups by bits [40, 40+log2(P_A)) of dj_mix64(key)
+ *           (block-chunked count + cross-block scan + LDS-cursor scatter);
+ *   pass B: one block per group: 256 sub-buckets by bits [32,40), count +
+ *           in-block scan + scatter, emitting the final bucket offsets.
+ * The decomposition is internal to the local join (result row order is
+ * unspecified by the API), so both passes are NON-stable — unlike the
+ * rank-level hash_partition above, which stays stable (reference pin:
+ * SURVEY.md appendix, batch offsets).
+ * LDS slot hash = low 32 bits of dj_mix64; the rank-level partition uses
+ * MurmurHash3%G, so bucket occupancy is independent of rank placement.
  */
 
-constexpr int BUCKET_BLOCKS = 256;   // one per CU; each owns a contiguous chunk
+constexpr int BUCKET_BLOCKS = 256;   // chunking blocks for pass A
 constexpr int BUCKET_THREADS = 1024;
-constexpr int JOIN_LDS_SLOTS = 8192;          // 128 KiB of longlong2 pairs
-constexpr int JOIN_BUCKET_ROW_CAP = 6144;     // 75% fill cap; above -> fallback
+constexpr int JOIN_LDS_SLOTS = 4096;        // 64 KiB of longlong2 pairs -> 2 blocks/CU
+constexpr int SUB_BUCKETS = 256;            // pass-B fanout (fixed)
 
 int bucket_count_for(int64_t ln, int64_t rn)
 {
   int64_t maxn = ln > rn ? ln : rn;
-  int B = 256;
-  while (B < 32768 && maxn / B > 3000) B <<= 1;
-  return B;
+  int64_t B = 256;
+  while (B < 262144 && maxn / B > 1500) B <<= 1;
+  return (int)B;
 }
 
+/* final bucket id = bits [32, 32+log2(B)) of dj_mix64(key) */
 __device__ __forceinline__ uint32_t bucket_of(int64_t key, int B)
 {
   return (uint32_t)(dj_mix64((uint64_t)key) >> 32) & (uint32_t)(B - 1);
 }
+/* pass-A group = bucket id >> 8 ; pass-B sub-bucket = bucket id & 255 */
+__device__ __forceinline__ uint32_t groupA_of(int64_t key, int PA)
+{
+  return (uint32_t)(dj_mix64((uint64_t)key) >> 40) & (uint32_t)(PA - 1);
+}
+__device__ __forceinline__ uint32_t subB_of(int64_t key)
+{
+  return (uint32_t)(dj_mix64((uint64_t)key) >> 32) & (uint32_t)(SUB_BUCKETS - 1);
+}
 
-/* grid = BUCKET_BLOCKS; dynamic LDS: u32 hist[B] */
+/* ---- pass A kernels (P groups, P <= 1024) ---- */
+
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_count_kernel(
-  const int64_t* __restrict__ keys, int64_t n, int B, uint32_t* __restrict__ counts)
+  const int64_t* __restrict__ keys, int64_t n, int P, uint32_t* __restrict__ counts)
 {
   extern __shared__ uint32_t hist[];
-  for (int p = threadIdx.x; p < B; p += blockDim.x) hist[p] = 0;
+  for (int p = threadIdx.x; p < P; p += blockDim.x) hist[p] = 0;
   __syncthreads();
   const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
   const int64_t start = (int64_t)blockIdx.x * chunk;
   const int64_t end = min(start + chunk, n);
   for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
-    atomicAdd(&hist[bucket_of(keys[i], B)], 1u);
+    atomicAdd(&hist[groupA_of(keys[i], P)], 1u);
   __syncthreads();
-  for (int p = threadIdx.x; p < B; p += blockDim.x)
-    counts[(size_t)blockIdx.x * B + p] = hist[p];
+  for (int p = threadIdx.x; p < P; p += blockDim.x)
+    counts[(size_t)blockIdx.x * P + p] = hist[p];
 }
 
-/* grid = B blocks x BUCKET_BLOCKS threads: exclusive scan of counts[:, p]
+/* grid = P blocks x BUCKET_BLOCKS threads: exclusive scan of counts[:, p]
  * over blocks (in place), totals[p] = column sum */
-__global__ void bucket_scanA_kernel(uint32_t* counts, int B, uint32_t* totals)
+__global__ void bucket_scanA_kernel(uint32_t* counts, int P, uint32_t* totals)
 {
   const int p = blockIdx.x;
   __shared__ uint32_t sh[BUCKET_BLOCKS];
-  uint32_t v = counts[(size_t)threadIdx.x * B + p];
+  uint32_t v = counts[(size_t)threadIdx.x * P + p];
   sh[threadIdx.x] = v;
   __syncthreads();
   for (int off = 1; off < BUCKET_BLOCKS; off <<= 1) {
@@ -483,20 +503,20 @@ __global__ void bucket_scanA_kernel(uint32_t* counts, int B, uint32_t* totals)
     sh[threadIdx.x] += add;
     __syncthreads();
   }
-  counts[(size_t)threadIdx.x * B + p] = sh[threadIdx.x] - v;  // exclusive
+  counts[(size_t)threadIdx.x * P + p] = sh[threadIdx.x] - v;  // exclusive
   if (threadIdx.x == BUCKET_BLOCKS - 1) totals[p] = sh[threadIdx.x];
 }
 
-/* single block: exclusive scan of totals[B] -> offsets[B+1] (int64) */
-__global__ void bucket_scanB_kernel(const uint32_t* totals, int B, int64_t* offsets)
+/* single block: exclusive scan of totals[P] -> segoff[P+1] (int64) */
+__global__ void bucket_scanB_kernel(const uint32_t* totals, int P, int64_t* segoff)
 {
   __shared__ int64_t sh[BUCKET_THREADS];
   __shared__ int64_t running_sh;
   if (threadIdx.x == 0) running_sh = 0;
   __syncthreads();
-  for (int base = 0; base < B; base += BUCKET_THREADS) {
+  for (int base = 0; base < P; base += BUCKET_THREADS) {
     int p = base + threadIdx.x;
-    int64_t v = (p < B) ? (int64_t)totals[p] : 0;
+    int64_t v = (p < P) ? (int64_t)totals[p] : 0;
     sh[threadIdx.x] = v;
     __syncthreads();
     for (int off = 1; off < BUCKET_THREADS; off <<= 1) {
@@ -507,23 +527,21 @@ __global__ void bucket_scanB_kernel(const uint32_t* totals, int B, int64_t* offs
     }
     int64_t rbase = running_sh;
     __syncthreads();  // all reads of running_sh precede the update below
-    if (p < B) offsets[p] = rbase + sh[threadIdx.x] - v;
+    if (p < P) segoff[p] = rbase + sh[threadIdx.x] - v;
     if (threadIdx.x == BUCKET_THREADS - 1) running_sh = rbase + sh[threadIdx.x];
     __syncthreads();
   }
-  if (threadIdx.x == 0) offsets[B] = offsets[B - 1] + (int64_t)totals[B - 1];
+  if (threadIdx.x == 0) segoff[P] = running_sh;
 }
 
-/* grid = BUCKET_BLOCKS; dynamic LDS: u32 cursor[B] preloaded with
- * offsets[p] + this block's exclusive prefix; non-stable scatter */
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_kernel(
-  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int B,
-  const uint32_t* __restrict__ counts, const int64_t* __restrict__ offsets,
+  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int P,
+  const uint32_t* __restrict__ counts, const int64_t* __restrict__ segoff,
   int64_t* __restrict__ out_keys, int64_t* __restrict__ out_pay)
 {
   extern __shared__ uint32_t cursor[];
-  for (int p = threadIdx.x; p < B; p += blockDim.x)
-    cursor[p] = (uint32_t)offsets[p] + counts[(size_t)blockIdx.x * B + p];
+  for (int p = threadIdx.x; p < P; p += blockDim.x)
+    cursor[p] = (uint32_t)segoff[p] + counts[(size_t)blockIdx.x * P + p];
   __syncthreads();
   const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
   const int64_t start = (int64_t)blockIdx.x * chunk;
@@ -531,13 +549,56 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_kernel(
   for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
     int64_t k = keys[i];
     int64_t v = pay ? pay[i] : i;
-    uint32_t dst = atomicAdd(&cursor[bucket_of(k, B)], 1u);
+    uint32_t dst = atomicAdd(&cursor[groupA_of(k, P)], 1u);
     out_keys[dst] = k;
     out_pay[dst] = v;
   }
 }
 
-/* fused per-bucket LDS build + probe */
+/* trivial segoff = {0, n} for the single-group (B == 256) case */
+__global__ void set_segoff1_kernel(int64_t* segoff, int64_t n)
+{
+  segoff[0] = 0;
+  segoff[1] = n;
+}
+
+/* ---- pass B: one block per pass-A group; 256 sub-buckets in-block ---- */
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
+  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay,
+  const int64_t* __restrict__ segoff, int B,
+  int64_t* __restrict__ out_keys, int64_t* __restrict__ out_pay,
+  int64_t* __restrict__ bucket_offsets /* B+1 */)
+{
+  __shared__ uint32_t hist[SUB_BUCKETS];
+  __shared__ uint32_t cur[SUB_BUCKETS];
+  const int a = blockIdx.x;
+  const int64_t s0 = segoff[a], s1 = segoff[a + 1];
+  for (int p = threadIdx.x; p < SUB_BUCKETS; p += blockDim.x) hist[p] = 0;
+  __syncthreads();
+  for (int64_t i = s0 + threadIdx.x; i < s1; i += blockDim.x)
+    atomicAdd(&hist[subB_of(keys[i])], 1u);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    uint32_t acc = 0;
+    for (int j = 0; j < SUB_BUCKETS; j++) {
+      uint32_t c = hist[j];
+      cur[j] = acc;
+      bucket_offsets[(size_t)a * SUB_BUCKETS + j] = s0 + acc;
+      acc += c;
+    }
+    if (a == gridDim.x - 1) bucket_offsets[B] = s1;
+  }
+  __syncthreads();
+  for (int64_t i = s0 + threadIdx.x; i < s1; i += blockDim.x) {
+    int64_t k = keys[i];
+    int64_t v = pay ? pay[i] : i;
+    uint32_t dst = atomicAdd(&cur[subB_of(k)], 1u);
+    out_keys[s0 + dst] = k;
+    out_pay[s0 + dst] = v;
+  }
+}
+
+/* ---- fused per-bucket LDS build + two-phase probe (count, then write) ---- */
 __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   const int64_t* __restrict__ lk, const int64_t* __restrict__ lp,
   const int64_t* __restrict__ loff, const int64_t* __restrict__ rk,
@@ -549,8 +610,9 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbl = (longlong2*)smem;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+  long long* base_sh = (long long*)(smem + JOIN_LDS_SLOTS * sizeof(longlong2));
+  uint32_t* total_sh = (uint32_t*)(base_sh + 1);
+  uint32_t* cur_sh = total_sh + 1;
   const uint32_t smask = JOIN_LDS_SLOTS - 1;
 
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
@@ -558,15 +620,18 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
     const int64_t r0 = roff[b], r1 = roff[b + 1];
     const int64_t lnb = l1 - l0;
     if (lnb == 0 || r1 == r0) continue;
-    if (lnb > JOIN_BUCKET_ROW_CAP) {  // skew overflow: host-side fallback joins it
+    if (lnb > kJoinBucketRowCap) {  // skew overflow: host-side fallback joins it
       if (threadIdx.x == 0) {
         overflow_flags[b] = 1;
         *any_overflow = 1;
       }
       continue;
     }
-    /* init table */
     for (int s = threadIdx.x; s < JOIN_LDS_SLOTS; s += blockDim.x) tbl[s].x = kEmptyKey;
+    if (threadIdx.x == 0) {
+      *total_sh = 0;
+      *cur_sh = 0;
+    }
     __syncthreads();
     /* build */
     for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
@@ -586,71 +651,82 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
       tbl[slot].y = lp[i];
     }
     __syncthreads();
-    /* probe (wave-lockstep aggregated emit, as the global probe kernel) */
-    const int64_t rows = r1 - r0;
-    const int64_t iters = (rows + blockDim.x - 1) / blockDim.x;
-    for (int64_t it = 0; it < iters; it++) {
-      const int64_t j = r0 + it * blockDim.x + threadIdx.x;
-      const bool row_valid = j < r1;
-      int64_t key = 0, payload = 0;
-      uint32_t slot = 0;
-      bool walking = row_valid;
-      if (row_valid) {
-        key = rk[j];
-        payload = rp[j];
-        slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
-      }
+    /* probe phase 1: count my matches */
+    uint32_t my = 0;
+    for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
+      int64_t key = rk[j];
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
       for (;;) {
-        int64_t mval = 0;
-        bool have = false;
-        while (walking) {
-          longlong2 e = tbl[slot];
-          if (e.x == kEmptyKey) {
-            walking = false;
-            break;
-          }
-          slot = (slot + 1) & smask;
-          if (e.x == key) {
-            mval = e.y;
-            have = true;
-            break;
-          }
-        }
-        uint64_t m = __ballot(have);
-        if (m == 0) break;
-        const int leader = (int)(__ffsll((unsigned long long)m) - 1);
-        unsigned long long base = 0;
-        if (lane == leader) base = atomicAdd(counter, (unsigned long long)__popcll(m));
-        base = __shfl(base, leader);
-        if (have) {
-          int64_t idx = (int64_t)base + __popcll(m & lt_mask);
-          if (idx < cap) {
-            out0[idx] = key;
-            out1[idx] = mval;
-            out2[idx] = key;
-            out3[idx] = payload;
+        longlong2 e = tbl[slot];
+        if (e.x == kEmptyKey) break;
+        if (e.x == key) my++;
+        slot = (slot + 1) & smask;
+      }
+    }
+    if (my) atomicAdd(total_sh, my);
+    __syncthreads();
+    /* ONE global atomic per bucket reserves the output range */
+    if (threadIdx.x == 0 && *total_sh)
+      *base_sh = (long long)atomicAdd(counter, (unsigned long long)*total_sh);
+    __syncthreads();
+    if (*total_sh) {
+      const long long base = *base_sh;
+      uint32_t w = my ? atomicAdd(cur_sh, my) : 0;
+      /* probe phase 2: rewalk in the same order, write */
+      if (my) {
+        for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
+          int64_t key = rk[j];
+          int64_t payload = rp[j];
+          uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
+          for (;;) {
+            longlong2 e = tbl[slot];
+            if (e.x == kEmptyKey) break;
+            if (e.x == key) {
+              long long idx = base + (long long)w;
+              if (idx < cap) {
+                out0[idx] = key;
+                out1[idx] = e.y;
+                out2[idx] = key;
+                out3[idx] = payload;
+              }
+              w++;
+            }
+            slot = (slot + 1) & smask;
           }
         }
       }
     }
-    __syncthreads();  // table re-initialized next bucket
+    __syncthreads();
   }
 }
 
-void bucket_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
-                      uint32_t* d_counts, uint32_t* d_totals, int64_t* d_offsets,
-                      int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s)
+void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
+                       int64_t* d_tmpk, int64_t* d_tmpp, uint32_t* d_counts,
+                       uint32_t* d_totals, int64_t* d_segoff, int64_t* d_offsets,
+                       int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s)
 {
   DJ_CHECK_ERROR(n < (int64_t)UINT32_MAX, "bucket_partition: n must be < 2^32");
-  size_t lds = (size_t)B * sizeof(uint32_t);
-  hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
-                     d_keys, n, B, d_counts);
-  hipLaunchKernelGGL(bucket_scanA_kernel, dim3(B), dim3(BUCKET_BLOCKS), 0, s, d_counts, B,
-                     d_totals);
-  hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, B,
-                     d_offsets);
-  hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
-                     d_keys, d_pay, n, B, d_counts, d_offsets, d_out_keys, d_out_pay);
+  const int PA = B / SUB_BUCKETS;
+  DJ_CHECK_ERROR(PA >= 1 && PA <= 1024, "bucket_partition: B out of range");
+  const int64_t* src_k = d_keys;
+  const int64_t* src_p = d_pay;
+  if (PA == 1) {
+    hipLaunchKernelGGL(set_segoff1_kernel, dim3(1), dim3(1), 0, s, d_segoff, n);
+  } else {
+    size_t lds = (size_t)PA * sizeof(uint32_t);
+    hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
+                       d_keys, n, PA, d_counts);
+    hipLaunchKernelGGL(bucket_scanA_kernel, dim3(PA), dim3(BUCKET_BLOCKS), 0, s, d_counts, PA,
+                       d_totals);
+    hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, PA,
+                       d_segoff);
+    hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
+                       d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmpk, d_tmpp);
+    src_k = d_tmpk;
+    src_p = d_tmpp;
+  }
+  hipLaunchKernelGGL(bucket_subpart_kernel, dim3(PA), dim3(BUCKET_THREADS), 0, s, src_k, src_p,
+                     d_segoff, B, d_out_keys, d_out_pay, d_offsets);
 }
 
 void lds_join(const int64_t* d_lk, const int64_t* d_lp, const int64_t* d_loff,
@@ -659,11 +735,12 @@ void lds_join(const int64_t* d_lk, const int64_t* d_lp, const int64_t* d_loff,
               int64_t* d_counter, uint32_t* d_overflow_flags, int* d_any_overflow,
               int* d_error, hipStream_t s)
 {
-  int grid = B < 4096 ? B : 4096;
-  hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS),
-                     JOIN_LDS_SLOTS * sizeof(longlong2), s, d_lk, d_lp, d_loff, d_rk, d_rp,
-                     d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
-                     (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow, d_error);
+  int grid = B < 8192 ? B : 8192;
+  size_t lds = JOIN_LDS_SLOTS * sizeof(longlong2) + 16;
+  hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lk, d_lp,
+                     d_loff, d_rk, d_rp, d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
+                     (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
+                     d_error);
 }
 
 }  // namespace dj

@@ -69,13 +69,16 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
 /* ----- bucketed LDS join (the product local-join path; see dj_kernels.hip
  * "bucketed LDS join" comment block) ----- */
 constexpr int kBucketBlocks = 256;
-constexpr int kJoinBucketRowCap = 6144;
+constexpr int kSubBuckets = 256;
+constexpr int kJoinBucketRowCap = 3072;  // 75% of the 4096-slot LDS table
 int bucket_count_for(int64_t ln, int64_t rn);
-/* Non-stable partition into B buckets. d_counts: u32[kBucketBlocks*B];
- * d_totals: u32[B]; d_offsets: int64[B+1] (device). */
-void bucket_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
-                      uint32_t* d_counts, uint32_t* d_totals, int64_t* d_offsets,
-                      int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s);
+/* Two-level non-stable partition into B buckets (B = PA*256, PA<=1024).
+ * d_tmpk/d_tmpp: int64[n] pass-A staging; d_counts: u32[kBucketBlocks*PA];
+ * d_totals: u32[PA]; d_segoff: int64[PA+1]; d_offsets: int64[B+1]. */
+void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
+                       int64_t* d_tmpk, int64_t* d_tmpp, uint32_t* d_counts,
+                       uint32_t* d_totals, int64_t* d_segoff, int64_t* d_offsets,
+                       int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s);
 /* Fused per-bucket LDS build+probe over bucketed tables. Buckets whose build
  * side exceeds kJoinBucketRowCap set overflow_flags[b] and any_overflow and
  * are skipped (host runs the global-table path on them). */
