@@ -241,14 +241,14 @@ int64_t dj_local_inner_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln
 namespace {
 
 struct BucketScratch {
-  int64_t *lbk, *lbp, *rbk, *rbp;  // bucketed copies (final)
-  int64_t *tmpk, *tmpp;            // pass-A staging (size max(ln,rn))
-  int64_t *loff, *roff;            // int64[B+1]
-  int64_t* segoff;                 // int64[PA+1]
-  uint32_t* counts;                // u32[kBucketBlocks*PA]
-  uint32_t* totals;                // u32[PA]
-  uint32_t* flags;                 // u32[B]
-  int* any_overflow;               // int[1]
+  longlong2 *lpairs, *rpairs;  // bucketed {key,payload} pairs (final)
+  longlong2* tmp_pairs;        // pass-A staging (size max(ln,rn))
+  int64_t *loff, *roff;        // int64[B+1]
+  int64_t* segoff;             // int64[PA+1]
+  uint32_t* counts;            // u32[kBucketBlocks*PA]
+  uint32_t* totals;            // u32[PA]
+  uint32_t* flags;             // u32[B]
+  int* any_overflow;           // int[1]
 };
 
 BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
@@ -262,12 +262,9 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
     return r;
   };
   BucketScratch s;
-  s.lbk = (int64_t*)take((size_t)ln * 8);
-  s.lbp = (int64_t*)take((size_t)ln * 8);
-  s.rbk = (int64_t*)take((size_t)rn * 8);
-  s.rbp = (int64_t*)take((size_t)rn * 8);
-  s.tmpk = (int64_t*)take((size_t)maxn * 8);
-  s.tmpp = (int64_t*)take((size_t)maxn * 8);
+  s.lpairs = (longlong2*)take((size_t)ln * 16);
+  s.rpairs = (longlong2*)take((size_t)rn * 16);
+  s.tmp_pairs = (longlong2*)take((size_t)maxn * 16);
   s.loff = (int64_t*)take((size_t)(B + 1) * 8);
   s.roff = (int64_t*)take((size_t)(B + 1) * 8);
   s.segoff = (int64_t*)take((size_t)(PA + 1) * 8);
@@ -287,12 +284,9 @@ int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
   const int64_t maxn = ln > rn ? ln : rn;
   size_t bytes = 0;
   auto add = [&](size_t b) { bytes += (b + 255) & ~(size_t)255; };
-  add((size_t)ln * 8);
-  add((size_t)ln * 8);
-  add((size_t)rn * 8);
-  add((size_t)rn * 8);
-  add((size_t)maxn * 8);
-  add((size_t)maxn * 8);
+  add((size_t)ln * 16);
+  add((size_t)rn * 16);
+  add((size_t)maxn * 16);
   add((size_t)(B + 1) * 8);
   add((size_t)(B + 1) * 8);
   add((size_t)(PA + 1) * 8);
@@ -315,16 +309,16 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
   hipStream_t st = stream();
   {
     PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
-    dj::bucket_partition2(d_lk, d_lp, ln, B, s.tmpk, s.tmpp, s.counts, s.totals, s.segoff,
-                          s.loff, s.lbk, s.lbp, st);
-    dj::bucket_partition2(d_rk, d_rp, rn, B, s.tmpk, s.tmpp, s.counts, s.totals, s.segoff,
-                          s.roff, s.rbk, s.rbp, st);
+    dj::bucket_partition2(d_lk, d_lp, ln, B, s.tmp_pairs, s.counts, s.totals, s.segoff,
+                          s.loff, s.lpairs, st);
+    dj::bucket_partition2(d_rk, d_rp, rn, B, s.tmp_pairs, s.counts, s.totals, s.segoff,
+                          s.roff, s.rpairs, st);
   }
   DJ_HIP_CALL(hipMemsetAsync(s.any_overflow, 0, sizeof(int), st));
   DJ_HIP_CALL(hipMemsetAsync(s.flags, 0, (size_t)B * 4, st));
   {
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
-    dj::lds_join(s.lbk, s.lbp, s.loff, s.rbk, s.rbp, s.roff, B, d_out0, d_out1, d_out2,
+    dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, d_out0, d_out1, d_out2,
                  d_out3, cap, d_counter, s.flags, s.any_overflow, d_error, st);
   }
   /* skew fallback: buckets whose build side exceeded the LDS row cap */
@@ -345,9 +339,9 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
       int64_t nslots = dj::join_table_slots(lnb);
       int64_t* d_table = (int64_t*)dj_dmalloc(nslots * 2 * sizeof(int64_t));
       dj_join_table_init(d_table, nslots);
-      dj_join_build(s.lbk + loff[b], s.lbp + loff[b], lnb, d_table, nslots, d_error);
-      dj_join_probe(s.rbk + roff[b], s.rbp + roff[b], rnb, d_table, nslots, d_out0, d_out1,
-                    d_out2, d_out3, cap, d_counter);
+      dj::join_build_pairs(s.lpairs + loff[b], lnb, d_table, nslots, d_error, st);
+      dj::join_probe_pairs(s.rpairs + roff[b], rnb, d_table, nslots, d_out0, d_out1, d_out2,
+                           d_out3, cap, d_counter, st);
       DJ_HIP_CALL(hipStreamSynchronize(st));
       dj_dfree(d_table);
     }

@@ -312,15 +312,18 @@ void join_table_init(int64_t* d_table, int64_t nslots, hipStream_t s)
 }
 
 /* Table layout: nslots interleaved 16 B {key, val} pairs — one random
- * cache-line fetch serves both the key compare and the payload read. */
+ * cache-line fetch serves both the key compare and the payload read.
+ * PAIRS: input rows as interleaved longlong2 instead of two columns. */
+template <bool PAIRS>
 __global__ void join_build_kernel(const int64_t* __restrict__ lk, const int64_t* __restrict__ lp,
+                                  const longlong2* __restrict__ lrows,
                                   int64_t ln, longlong2* __restrict__ table, uint64_t mask,
                                   int* error)
 {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < ln; i += stride) {
-    int64_t key = lk[i];
+    int64_t key = PAIRS ? lrows[i].x : lk[i];
     if (key == kEmptyKey) {
       *error = 1;  // reserved sentinel; loud failure, never silent wrong data
       continue;
@@ -333,7 +336,7 @@ __global__ void join_build_kernel(const int64_t* __restrict__ lk, const int64_t*
       if (old == (unsigned long long)kEmptyKey) break;
       slot = (slot + 1) & mask;
     }
-    table[slot].y = lp ? lp[i] : i;
+    table[slot].y = PAIRS ? lrows[i].y : (lp ? lp[i] : i);
   }
 }
 
@@ -341,7 +344,17 @@ void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d
                 int64_t nslots, int* d_error, hipStream_t s)
 {
   if (ln <= 0) return;
-  hipLaunchKernelGGL(join_build_kernel, dim3(grid_for(ln)), dim3(BLOCK), 0, s, d_lk, d_lp, ln,
+  hipLaunchKernelGGL(join_build_kernel<false>, dim3(grid_for(ln)), dim3(BLOCK), 0, s, d_lk,
+                     d_lp, (const longlong2*)nullptr, ln, (longlong2*)d_table,
+                     (uint64_t)(nslots - 1), d_error);
+}
+
+void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int64_t nslots,
+                      int* d_error, hipStream_t s)
+{
+  if (ln <= 0) return;
+  hipLaunchKernelGGL(join_build_kernel<true>, dim3(grid_for(ln)), dim3(BLOCK), 0, s,
+                     (const int64_t*)nullptr, (const int64_t*)nullptr, d_rows, ln,
                      (longlong2*)d_table, (uint64_t)(nslots - 1), d_error);
 }
 
@@ -349,7 +362,9 @@ void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d
  * atomicAdd per wave (ballot leader) instead of one per lane — the single
  * global counter would otherwise serialize the whole kernel. All lanes of a
  * wave iterate in lockstep so the ballots are well-defined. */
+template <bool PAIRS>
 __global__ void join_probe_kernel(const int64_t* __restrict__ rk, const int64_t* __restrict__ rp,
+                                  const longlong2* __restrict__ rrows,
                                   int64_t rn, const longlong2* __restrict__ table,
                                   uint64_t mask, int64_t* __restrict__ out0,
                                   int64_t* __restrict__ out1, int64_t* __restrict__ out2,
@@ -366,8 +381,14 @@ __global__ void join_probe_kernel(const int64_t* __restrict__ rk, const int64_t*
     uint64_t slot = 0;
     bool walking = row_valid;
     if (row_valid) {
-      key = rk[j];
-      payload = rp ? rp[j] : j;
+      if (PAIRS) {
+        longlong2 r = rrows[j];
+        key = r.x;
+        payload = r.y;
+      } else {
+        key = rk[j];
+        payload = rp ? rp[j] : j;
+      }
       slot = dj_mix64((uint64_t)key) & mask;
     }
     for (;;) {
@@ -410,9 +431,21 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
                 int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s)
 {
   if (rn <= 0) return;
-  hipLaunchKernelGGL(join_probe_kernel, dim3(grid_for(rn)), dim3(BLOCK), 0, s, d_rk, d_rp, rn,
-                     (const longlong2*)d_table, (uint64_t)(nslots - 1), d_out0, d_out1, d_out2,
-                     d_out3, cap, (unsigned long long*)d_counter);
+  hipLaunchKernelGGL(join_probe_kernel<false>, dim3(grid_for(rn)), dim3(BLOCK), 0, s, d_rk,
+                     d_rp, (const longlong2*)nullptr, rn, (const longlong2*)d_table,
+                     (uint64_t)(nslots - 1), d_out0, d_out1, d_out2, d_out3, cap,
+                     (unsigned long long*)d_counter);
+}
+
+void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_table,
+                      int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                      int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s)
+{
+  if (rn <= 0) return;
+  hipLaunchKernelGGL(join_probe_kernel<true>, dim3(grid_for(rn)), dim3(BLOCK), 0, s,
+                     (const int64_t*)nullptr, (const int64_t*)nullptr, d_rows, rn,
+                     (const longlong2*)d_table, (uint64_t)(nslots - 1), d_out0, d_out1,
+                     d_out2, d_out3, cap, (unsigned long long*)d_counter);
 }
 
 /* ----------------------------------------------- bucketed LDS join ------ */
@@ -442,7 +475,7 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
  * MurmurHash3%G, so bucket occupancy is independent of rank placement.
  */
 
-constexpr int BUCKET_BLOCKS = 256;   // chunking blocks for pass A
+constexpr int BUCKET_BLOCKS = 512;   // chunking blocks for pass A (2/CU)
 constexpr int BUCKET_THREADS = 1024;
 constexpr int JOIN_LDS_SLOTS = 4096;        // 64 KiB of longlong2 pairs -> 2 blocks/CU
 constexpr int SUB_BUCKETS = 256;            // pass-B fanout (fixed)
@@ -451,7 +484,7 @@ int bucket_count_for(int64_t ln, int64_t rn)
 {
   int64_t maxn = ln > rn ? ln : rn;
   int64_t B = 256;
-  while (B < 262144 && maxn / B > 1500) B <<= 1;
+  while (B < 262144 && maxn / B > 1600) B <<= 1;
   return (int)B;
 }
 
@@ -537,7 +570,7 @@ __global__ void bucket_scanB_kernel(const uint32_t* totals, int P, int64_t* sego
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_kernel(
   const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int P,
   const uint32_t* __restrict__ counts, const int64_t* __restrict__ segoff,
-  int64_t* __restrict__ out_keys, int64_t* __restrict__ out_pay)
+  longlong2* __restrict__ out_pairs)
 {
   extern __shared__ uint32_t cursor[];
   for (int p = threadIdx.x; p < P; p += blockDim.x)
@@ -547,11 +580,11 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_kernel(
   const int64_t start = (int64_t)blockIdx.x * chunk;
   const int64_t end = min(start + chunk, n);
   for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
-    int64_t k = keys[i];
-    int64_t v = pay ? pay[i] : i;
-    uint32_t dst = atomicAdd(&cursor[groupA_of(k, P)], 1u);
-    out_keys[dst] = k;
-    out_pay[dst] = v;
+    longlong2 row;
+    row.x = keys[i];
+    row.y = pay ? pay[i] : i;
+    uint32_t dst = atomicAdd(&cursor[groupA_of(row.x, P)], 1u);
+    out_pairs[dst] = row;  // one 16 B store: half the write streams of 2 columns
   }
 }
 
@@ -563,11 +596,13 @@ __global__ void set_segoff1_kernel(int64_t* segoff, int64_t n)
 }
 
 /* ---- pass B: one block per pass-A group; 256 sub-buckets in-block ---- */
+/* pass B over interleaved pairs. single_level: input is the original two
+ * column arrays instead of pass-A pairs (B == 256 case). */
+template <bool SINGLE_LEVEL>
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
-  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay,
-  const int64_t* __restrict__ segoff, int B,
-  int64_t* __restrict__ out_keys, int64_t* __restrict__ out_pay,
-  int64_t* __restrict__ bucket_offsets /* B+1 */)
+  const longlong2* __restrict__ in_pairs, const int64_t* __restrict__ keys,
+  const int64_t* __restrict__ pay, const int64_t* __restrict__ segoff, int B,
+  longlong2* __restrict__ out_pairs, int64_t* __restrict__ bucket_offsets /* B+1 */)
 {
   __shared__ uint32_t hist[SUB_BUCKETS];
   __shared__ uint32_t cur[SUB_BUCKETS];
@@ -575,8 +610,10 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
   const int64_t s0 = segoff[a], s1 = segoff[a + 1];
   for (int p = threadIdx.x; p < SUB_BUCKETS; p += blockDim.x) hist[p] = 0;
   __syncthreads();
-  for (int64_t i = s0 + threadIdx.x; i < s1; i += blockDim.x)
-    atomicAdd(&hist[subB_of(keys[i])], 1u);
+  for (int64_t i = s0 + threadIdx.x; i < s1; i += blockDim.x) {
+    int64_t k = SINGLE_LEVEL ? keys[i] : in_pairs[i].x;
+    atomicAdd(&hist[subB_of(k)], 1u);
+  }
   __syncthreads();
   if (threadIdx.x == 0) {
     uint32_t acc = 0;
@@ -590,19 +627,22 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
   }
   __syncthreads();
   for (int64_t i = s0 + threadIdx.x; i < s1; i += blockDim.x) {
-    int64_t k = keys[i];
-    int64_t v = pay ? pay[i] : i;
-    uint32_t dst = atomicAdd(&cur[subB_of(k)], 1u);
-    out_keys[s0 + dst] = k;
-    out_pay[s0 + dst] = v;
+    longlong2 row;
+    if (SINGLE_LEVEL) {
+      row.x = keys[i];
+      row.y = pay ? pay[i] : i;
+    } else {
+      row = in_pairs[i];
+    }
+    uint32_t dst = atomicAdd(&cur[subB_of(row.x)], 1u);
+    out_pairs[s0 + dst] = row;
   }
 }
 
 /* ---- fused per-bucket LDS build + two-phase probe (count, then write) ---- */
 __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
-  const int64_t* __restrict__ lk, const int64_t* __restrict__ lp,
-  const int64_t* __restrict__ loff, const int64_t* __restrict__ rk,
-  const int64_t* __restrict__ rp, const int64_t* __restrict__ roff, int B,
+  const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
+  const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
   int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
   int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter,
   uint32_t* __restrict__ overflow_flags, int* __restrict__ any_overflow,
@@ -635,26 +675,26 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
     __syncthreads();
     /* build */
     for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
-      int64_t key = lk[i];
-      if (key == kEmptyKey) {
+      longlong2 row = lrows[i];
+      if (row.x == kEmptyKey) {
         *error = 1;
         continue;
       }
-      uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)row.x) & smask;
       for (;;) {
         unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
                                            (unsigned long long)kEmptyKey,
-                                           (unsigned long long)key);
+                                           (unsigned long long)row.x);
         if (old == (unsigned long long)kEmptyKey) break;
         slot = (slot + 1) & smask;
       }
-      tbl[slot].y = lp[i];
+      tbl[slot].y = row.y;
     }
     __syncthreads();
     /* probe phase 1: count my matches */
     uint32_t my = 0;
     for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
-      int64_t key = rk[j];
+      int64_t key = rrows[j].x;
       uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
       for (;;) {
         longlong2 e = tbl[slot];
@@ -675,8 +715,9 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
       /* probe phase 2: rewalk in the same order, write */
       if (my) {
         for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
-          int64_t key = rk[j];
-          int64_t payload = rp[j];
+          longlong2 prow = rrows[j];
+          int64_t key = prow.x;
+          int64_t payload = prow.y;
           uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
           for (;;) {
             longlong2 e = tbl[slot];
@@ -701,17 +742,18 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
 }
 
 void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
-                       int64_t* d_tmpk, int64_t* d_tmpp, uint32_t* d_counts,
-                       uint32_t* d_totals, int64_t* d_segoff, int64_t* d_offsets,
-                       int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s)
+                       longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
+                       int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
+                       hipStream_t s)
 {
   DJ_CHECK_ERROR(n < (int64_t)UINT32_MAX, "bucket_partition: n must be < 2^32");
   const int PA = B / SUB_BUCKETS;
   DJ_CHECK_ERROR(PA >= 1 && PA <= 1024, "bucket_partition: B out of range");
-  const int64_t* src_k = d_keys;
-  const int64_t* src_p = d_pay;
   if (PA == 1) {
     hipLaunchKernelGGL(set_segoff1_kernel, dim3(1), dim3(1), 0, s, d_segoff, n);
+    hipLaunchKernelGGL(bucket_subpart_kernel<true>, dim3(PA), dim3(BUCKET_THREADS), 0, s,
+                       (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B, d_out_pairs,
+                       d_offsets);
   } else {
     size_t lds = (size_t)PA * sizeof(uint32_t);
     hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
@@ -721,24 +763,22 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
     hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, PA,
                        d_segoff);
     hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
-                       d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmpk, d_tmpp);
-    src_k = d_tmpk;
-    src_p = d_tmpp;
+                       d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmp_pairs);
+    hipLaunchKernelGGL(bucket_subpart_kernel<false>, dim3(PA), dim3(BUCKET_THREADS), 0, s,
+                       d_tmp_pairs, (const int64_t*)nullptr, (const int64_t*)nullptr, d_segoff,
+                       B, d_out_pairs, d_offsets);
   }
-  hipLaunchKernelGGL(bucket_subpart_kernel, dim3(PA), dim3(BUCKET_THREADS), 0, s, src_k, src_p,
-                     d_segoff, B, d_out_keys, d_out_pay, d_offsets);
 }
 
-void lds_join(const int64_t* d_lk, const int64_t* d_lp, const int64_t* d_loff,
-              const int64_t* d_rk, const int64_t* d_rp, const int64_t* d_roff, int B,
-              int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3, int64_t cap,
-              int64_t* d_counter, uint32_t* d_overflow_flags, int* d_any_overflow,
-              int* d_error, hipStream_t s)
+void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
+              const int64_t* d_roff, int B, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+              int64_t* d_out3, int64_t cap, int64_t* d_counter, uint32_t* d_overflow_flags,
+              int* d_any_overflow, int* d_error, hipStream_t s)
 {
   int grid = B < 8192 ? B : 8192;
   size_t lds = JOIN_LDS_SLOTS * sizeof(longlong2) + 16;
-  hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lk, d_lp,
-                     d_loff, d_rk, d_rp, d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
+  hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lrows,
+                     d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
                      (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
                      d_error);
 }

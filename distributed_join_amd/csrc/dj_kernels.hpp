@@ -72,21 +72,27 @@ constexpr int kBucketBlocks = 256;
 constexpr int kSubBuckets = 256;
 constexpr int kJoinBucketRowCap = 3072;  // 75% of the 4096-slot LDS table
 int bucket_count_for(int64_t ln, int64_t rn);
-/* Two-level non-stable partition into B buckets (B = PA*256, PA<=1024).
- * d_tmpk/d_tmpp: int64[n] pass-A staging; d_counts: u32[kBucketBlocks*PA];
- * d_totals: u32[PA]; d_segoff: int64[PA+1]; d_offsets: int64[B+1]. */
+/* Two-level non-stable partition into B buckets (B = PA*256, PA<=1024) of
+ * interleaved 16 B {key,payload} pairs. d_tmp_pairs: longlong2[n] pass-A
+ * staging; d_counts: u32[kBucketBlocks*PA]; d_totals: u32[PA]; d_segoff:
+ * int64[PA+1]; d_offsets: int64[B+1]. */
 void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
-                       int64_t* d_tmpk, int64_t* d_tmpp, uint32_t* d_counts,
-                       uint32_t* d_totals, int64_t* d_segoff, int64_t* d_offsets,
-                       int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s);
-/* Fused per-bucket LDS build+probe over bucketed tables. Buckets whose build
- * side exceeds kJoinBucketRowCap set overflow_flags[b] and any_overflow and
- * are skipped (host runs the global-table path on them). */
-void lds_join(const int64_t* d_lk, const int64_t* d_lp, const int64_t* d_loff,
-              const int64_t* d_rk, const int64_t* d_rp, const int64_t* d_roff, int B,
-              int64_t* d_out0, int64_t* d_out1, int64_t* d_out2, int64_t* d_out3, int64_t cap,
-              int64_t* d_counter, uint32_t* d_overflow_flags, int* d_any_overflow,
-              int* d_error, hipStream_t s);
+                       longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
+                       int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
+                       hipStream_t s);
+/* Fused per-bucket LDS build+probe over bucketed pair tables. Buckets whose
+ * build side exceeds kJoinBucketRowCap set overflow_flags[b]/any_overflow
+ * and are skipped (host runs the global-table path on them). */
+void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
+              const int64_t* d_roff, int B, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+              int64_t* d_out3, int64_t cap, int64_t* d_counter, uint32_t* d_overflow_flags,
+              int* d_any_overflow, int* d_error, hipStream_t s);
+/* Global-table build/probe over interleaved pair inputs (skew fallback). */
+void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int64_t nslots,
+                      int* d_error, hipStream_t s);
+void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_table,
+                      int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                      int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s);
 
 /* ----- small utilities ----- */
 void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s);
