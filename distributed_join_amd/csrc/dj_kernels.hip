@@ -475,7 +475,7 @@ void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_tabl
  * MurmurHash3%G, so bucket occupancy is independent of rank placement.
  */
 
-constexpr int BUCKET_BLOCKS = 512;   // chunking blocks for pass A (2/CU)
+constexpr int BUCKET_BLOCKS = kBucketBlocks;  // chunking blocks for pass A (2/CU); scratch sizing uses the same constant
 constexpr int BUCKET_THREADS = 1024;
 constexpr int JOIN_LDS_SLOTS = 4096;        // 64 KiB of longlong2 pairs -> 2 blocks/CU
 constexpr int SUB_BUCKETS = 256;            // pass-B fanout (fixed)

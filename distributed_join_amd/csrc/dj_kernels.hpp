@@ -68,7 +68,7 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
 
 /* ----- bucketed LDS join (the product local-join path; see dj_kernels.hip
  * "bucketed LDS join" comment block) ----- */
-constexpr int kBucketBlocks = 256;
+constexpr int kBucketBlocks = 512;
 constexpr int kSubBuckets = 256;
 constexpr int kJoinBucketRowCap = 3072;  // 75% of the 4096-slot LDS table
 int bucket_count_for(int64_t ln, int64_t rn);
