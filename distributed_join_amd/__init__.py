@@ -80,6 +80,14 @@ def lib():
             "dj_comm_size": ([], i32),
             "dj_all_to_all_i64": ([vp, vp, vp, vp], None),
             "dj_exchange_sizes": ([vp, vp], None),
+            "dj_cpp_comm_create": ([i32, i32, vp], vp),
+            "dj_cpp_comm_destroy": ([vp], None),
+            "dj_cpp_distributed_inner_join_i64": ([vp, vp, vp, i64, vp, vp, i64, i32, i32], vp),
+            "dj_cpp_shuffle_on_i64": ([vp, vp, vp, i64, i32, u32], vp),
+            "dj_table_num_rows": ([vp], i64),
+            "dj_table_num_columns": ([vp], i32),
+            "dj_table_column_data": ([vp, i32], vp),
+            "dj_table_free": ([vp], None),
         }
         for name, (argtypes, restype) in sigs.items():
             f = getattr(_lib, name)
@@ -208,6 +216,51 @@ def local_inner_join_global(d_lk, d_lp, ln, d_rk, d_rp, rn, cap=None):
         for o in outs:
             o.free()
         cap = n
+
+
+def table_to_numpy(tbl_ptr):
+    """Copy an opaque cudf::table* (C ABI) into numpy columns and free it."""
+    L = lib()
+    n = L.dj_table_num_rows(tbl_ptr)
+    ncols = L.dj_table_num_columns(tbl_ptr)
+    cols = []
+    for c in range(ncols):
+        data = L.dj_table_column_data(tbl_ptr, c)
+        out = np.empty(n, dtype=np.int64)
+        if n:
+            L.dj_memcpy_d2h(out.ctypes.data, data, n * 8)
+        cols.append(out)
+    L.dj_table_free(tbl_ptr)
+    return cols
+
+
+class CppCommunicator:
+    """RCCLCommunicator (or single-process LocalCommunicator) handle."""
+
+    def __init__(self, rank=0, size=1, id_bytes=None):
+        L = lib()
+        ptr = None if id_bytes is None else id_bytes.ctypes.data
+        self.ptr = L.dj_cpp_comm_create(rank, size, ptr)
+
+    def destroy(self):
+        if self.ptr:
+            lib().dj_cpp_comm_destroy(self.ptr)
+            self.ptr = None
+
+
+def cpp_distributed_inner_join(comm, d_lk, d_lp, ln, d_rk, d_rp, rn, over_decom=1,
+                               report_timing=False):
+    """The C++ drop-in path (distributed_inner_join) over int64 columns.
+    Returns numpy columns (lkey, lpay, rkey, rpay) of this rank's result."""
+    t = lib().dj_cpp_distributed_inner_join_i64(comm.ptr, d_lk.ptr, d_lp.ptr, ln,
+                                                d_rk.ptr, d_rp.ptr, rn, over_decom,
+                                                int(report_timing))
+    return table_to_numpy(t)
+
+
+def cpp_shuffle_on(comm, d_keys, d_pay, n, hash_fn=HASH_MURMUR3, seed=0):
+    t = lib().dj_cpp_shuffle_on_i64(comm.ptr, d_keys.ptr, d_pay.ptr, n, hash_fn, seed)
+    return table_to_numpy(t)
 
 
 def timing(phase):

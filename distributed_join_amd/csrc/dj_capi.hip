@@ -4,6 +4,7 @@
  */
 #include "dj_error.hpp"
 #include "dj_kernels.hpp"
+#include "dj_timing.hpp"
 
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
@@ -41,25 +42,7 @@ struct TimedSpan {
 };
 std::vector<TimedSpan> g_spans;
 
-struct PhaseScope {
-  int phase;
-  hipStream_t s;
-  PhaseScope(int p, hipStream_t st) : phase(p), s(st)
-  {
-    if (!g_timing_on) return;
-    TimedSpan t;
-    t.phase = p;
-    DJ_HIP_CALL(hipEventCreate(&t.start));
-    DJ_HIP_CALL(hipEventCreate(&t.stop));
-    DJ_HIP_CALL(hipEventRecord(t.start, s));
-    g_spans.push_back(t);
-  }
-  ~PhaseScope()
-  {
-    if (!g_timing_on) return;
-    DJ_HIP_CALL(hipEventRecord(g_spans.back().stop, s));
-  }
-};
+using PhaseScope = dj_timing::Scope;
 
 /* ------------- RCCL communicator state ------------- */
 
@@ -68,6 +51,28 @@ int g_rank = 0;
 int g_size = 1;
 
 }  // namespace
+
+namespace dj_timing {
+
+bool enabled() { return g_timing_on; }
+
+void record_begin(int phase, hipStream_t s)
+{
+  TimedSpan t;
+  t.phase = phase;
+  DJ_HIP_CALL(hipEventCreate(&t.start));
+  DJ_HIP_CALL(hipEventCreate(&t.stop));
+  DJ_HIP_CALL(hipEventRecord(t.start, s));
+  g_spans.push_back(t);
+}
+
+void record_end(hipStream_t s) { DJ_HIP_CALL(hipEventRecord(g_spans.back().stop, s)); }
+
+}  // namespace dj_timing
+
+/* streams shared with the C++ orchestration layer (dj_cpp_api.hip) */
+hipStream_t dj_rt_stream() { return stream(); }
+hipStream_t dj_rt_comm_stream() { return comm_stream(); }
 
 extern "C" {
 

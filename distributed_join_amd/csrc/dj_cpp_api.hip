@@ -1,0 +1,1006 @@
+/*
+ * dj_cpp_api.hip — the C++ drop-in surface (include/distributed_join.hpp,
+ * shuffle_on.hpp, all_to_all_comm.hpp, communicator.hpp,
+ * distribute_table.hpp) over the gfx950 kernels in dj_kernels.hip and RCCL
+ * over xGMI. See each public header for the reference interface mirrored.
+ */
+#include "dj_error.hpp"
+#include "dj_kernels.hpp"
+#include "dj_runtime.hpp"
+#include "dj_timing.hpp"
+
+#include "../../include/all_to_all_comm.hpp"
+#include "../../include/communicator.hpp"
+#include "../../include/compression.hpp"
+#include "../../include/distribute_table.hpp"
+#include "../../include/distributed_join.hpp"
+#include "../../include/distributed_join.h"
+#include "../../include/shuffle_on.hpp"
+
+#include "dj_hash.h"
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <algorithm>
+#include <cstring>
+#include <vector>
+
+/* ---------------------------------------------------------------- helpers */
+
+namespace {
+
+constexpr int kBlock = 256;
+
+int grid_for_n(int64_t n)
+{
+  int64_t blocks = (n + kBlock - 1) / kBlock;
+  if (blocks > 2048) blocks = 2048;
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+__global__ void iota_i64_kernel(int64_t* dst, int64_t n)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = i;
+}
+
+__global__ void gather_i64_kernel(const int64_t* __restrict__ src,
+                                  const int64_t* __restrict__ idx, int64_t n,
+                                  int64_t* __restrict__ dst)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = src[idx[i]];
+}
+
+__global__ void gather_i32_kernel(const int32_t* __restrict__ src,
+                                  const int64_t* __restrict__ idx, int64_t n,
+                                  int32_t* __restrict__ dst)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = src[idx[i]];
+}
+
+__global__ void widen_i32_kernel(const int32_t* __restrict__ src, int64_t n,
+                                 int64_t* __restrict__ dst)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = src[i];
+}
+
+/* small owning device buffer */
+struct DBuf {
+  void* p{nullptr};
+  DBuf() = default;
+  explicit DBuf(size_t bytes)
+  {
+    if (bytes) DJ_HIP_CALL(hipMalloc(&p, bytes));
+  }
+  DBuf(const DBuf&) = delete;
+  DBuf& operator=(const DBuf&) = delete;
+  DBuf(DBuf&& o) noexcept : p(o.p) { o.p = nullptr; }
+  DBuf& operator=(DBuf&& o) noexcept
+  {
+    if (this != &o) {
+      if (p) (void)hipFree(p);
+      p = o.p;
+      o.p = nullptr;
+    }
+    return *this;
+  }
+  ~DBuf()
+  {
+    if (p) (void)hipFree(p);
+  }
+  int64_t* i64() { return (int64_t*)p; }
+};
+
+void sync_streams()
+{
+  DJ_HIP_CALL(hipStreamSynchronize(dj_rt_stream()));
+  DJ_HIP_CALL(hipStreamSynchronize(dj_rt_comm_stream()));
+}
+
+/* int64 view of a key column (widening INT32; spec: dj_hash.h operates on
+ * int64 keys — INT32 keys are widened, identity placement k % G preserved) */
+const int64_t* key_as_i64(cudf::column_view col, DBuf& tmp)
+{
+  if (col.type().id() == cudf::type_id::INT64) return col.head<int64_t>();
+  DJ_CHECK_ERROR(col.type().id() == cudf::type_id::INT32,
+                 "join/shuffle key column must be INT32 or INT64");
+  tmp = DBuf((size_t)col.size() * 8);
+  hipLaunchKernelGGL(widen_i32_kernel, dim3(grid_for_n(col.size())), dim3(kBlock), 0,
+                     dj_rt_stream(), col.head<int32_t>(), (int64_t)col.size(), tmp.i64());
+  return tmp.i64();
+}
+
+void check_no_compression(std::vector<ColumnCompressionOptions> const& opts)
+{
+  for (auto& o : opts)
+    if (o.compression_method != CompressionMethod::none)
+      throw std::runtime_error(
+        "compression is not implemented in this build (CompressionMethod::none only; "
+        "SURVEY.md §8f rank 3)");
+}
+
+Communicator* g_default_comm = nullptr;
+
+/* size-1 stand-in so the C++ API works single-process without RCCL */
+class LocalCommunicator : public Communicator {
+ public:
+  LocalCommunicator()
+  {
+    mpi_rank = 0;
+    mpi_size = 1;
+  }
+  void initialize() override {}
+  void start() override {}
+  void stop() override { sync_streams(); }
+  void send(const void*, int64_t, int, int) override
+  {
+    DJ_CHECK_ERROR(false, "LocalCommunicator cannot send (single process)");
+  }
+  void recv(void*, int64_t, int, int) override
+  {
+    DJ_CHECK_ERROR(false, "LocalCommunicator cannot recv (single process)");
+  }
+  void finalize() override {}
+  bool group_by_batch() override { return true; }
+};
+
+}  // namespace
+
+/* --------------------------------------------------------- cudf::column */
+
+namespace cudf {
+
+column::column(data_type type, size_type size) : _type(type), _size(size)
+{
+  size_t bytes = (size_t)size * size_of(type);
+  if (bytes) DJ_HIP_CALL(hipMalloc(&_data, bytes));
+}
+
+column::column(data_type type, size_type size, void* adopt) : _type(type), _size(size), _data(adopt)
+{
+}
+
+column::column(column&& o) noexcept : _type(o._type), _size(o._size), _data(o._data)
+{
+  o._data = nullptr;
+  o._size = 0;
+}
+
+column::~column()
+{
+  if (_data) (void)hipFree(_data);
+}
+
+}  // namespace cudf
+
+/* ----------------------------------------------------------- Communicator */
+
+struct RCCLCommunicatorImpl {
+  ncclComm_t comm{nullptr};
+  hipStream_t stream{nullptr};
+};
+
+int rccl_unique_id_size() { return (int)sizeof(ncclUniqueId); }
+
+void rccl_unique_id(void* out_bytes)
+{
+  ncclUniqueId id;
+  DJ_RCCL_CALL(ncclGetUniqueId(&id));
+  memcpy(out_bytes, &id, sizeof(id));
+}
+
+RCCLCommunicator::RCCLCommunicator(int rank, int size, const void* id_bytes)
+{
+  impl = new RCCLCommunicatorImpl;
+  mpi_rank = rank;
+  mpi_size = size;
+  DJ_HIP_CALL(hipGetDevice(&current_device));
+  ncclUniqueId id;
+  memcpy(&id, id_bytes, sizeof(id));
+  DJ_RCCL_CALL(ncclCommInitRank(&impl->comm, size, id, rank));
+  impl->stream = dj_rt_comm_stream();
+  g_default_comm = this;
+}
+
+void RCCLCommunicator::initialize() {}
+
+void RCCLCommunicator::start() { DJ_RCCL_CALL(ncclGroupStart()); }
+
+void RCCLCommunicator::stop()
+{
+  DJ_RCCL_CALL(ncclGroupEnd());
+  DJ_HIP_CALL(hipStreamSynchronize(impl->stream));
+}
+
+void RCCLCommunicator::send(const void* buf, int64_t count, int element_size, int dest)
+{
+  if (count <= 0) return;
+  DJ_RCCL_CALL(ncclSend(buf, (size_t)count * element_size, ncclInt8, dest, impl->comm,
+                        impl->stream));
+}
+
+void RCCLCommunicator::recv(void* buf, int64_t count, int element_size, int source)
+{
+  if (count <= 0) return;
+  DJ_RCCL_CALL(ncclRecv(buf, (size_t)count * element_size, ncclInt8, source, impl->comm,
+                        impl->stream));
+}
+
+void RCCLCommunicator::finalize()
+{
+  if (impl->comm) {
+    DJ_RCCL_CALL(ncclCommDestroy(impl->comm));
+    impl->comm = nullptr;
+  }
+}
+
+RCCLCommunicator::~RCCLCommunicator()
+{
+  if (g_default_comm == this) g_default_comm = nullptr;
+  delete impl;
+}
+
+Communicator* default_communicator()
+{
+  static LocalCommunicator local;
+  return g_default_comm ? g_default_comm : &local;
+}
+
+/* ------------------------------------------------------------ compression */
+
+std::vector<ColumnCompressionOptions> generate_compression_options_distributed(
+  cudf::table_view input, bool compression)
+{
+  if (compression)
+    throw std::runtime_error(
+      "compression is not implemented in this build (CompressionMethod::none only)");
+  return std::vector<ColumnCompressionOptions>((size_t)input.num_columns(),
+                                               ColumnCompressionOptions(CompressionMethod::none));
+}
+
+/* ------------------------------------------------------ CommunicationGroup */
+
+CommunicationGroup::CommunicationGroup(int grid_size, int stride)
+  : CommunicationGroup(grid_size, stride, default_communicator()->mpi_rank)
+{
+}
+
+CommunicationGroup::CommunicationGroup(int grid_size_, int stride_, int mpi_rank_)
+  : mpi_rank(mpi_rank_), grid_size(grid_size_), stride(stride_)
+{
+  DJ_CHECK_ERROR(stride > 0 && grid_size % stride == 0,
+                 "Group size should be a multiple of stride");
+  group_start = mpi_rank / grid_size * grid_size + mpi_rank % stride;
+}
+
+/* ------------------------------------------------------- communicate_sizes */
+
+void communicate_sizes(std::vector<int64_t> const& send_offset,
+                       std::vector<int64_t>& recv_offset,
+                       CommunicationGroup comm_group,
+                       Communicator* communicator)
+{
+  const int G = comm_group.size();
+  const int me = comm_group.get_local_idx();
+  std::vector<int64_t> send_counts(G);
+  for (int i = 0; i < G; i++) send_counts[i] = send_offset[i + 1] - send_offset[i];
+  std::vector<int64_t> recv_counts(G, 0);
+  recv_counts[me] = send_counts[me];
+  if (G > 1) {
+    /* counts move through the communicator via a small device staging
+     * buffer (the reference kept them on MPI host buffers,
+     * all_to_all_comm.cpp:68-69; RCCL wants device memory) */
+    DBuf d_send((size_t)G * 8), d_recv((size_t)G * 8);
+    DJ_HIP_CALL(hipMemcpyAsync(d_send.p, send_counts.data(), (size_t)G * 8,
+                               hipMemcpyHostToDevice, dj_rt_comm_stream()));
+    DJ_HIP_CALL(hipStreamSynchronize(dj_rt_comm_stream()));
+    communicator->start();
+    for (int i = 0; i < G; i++) {
+      if (i == me) continue;
+      int peer = comm_group.get_global_rank(i);
+      communicator->send(d_send.i64() + i, 1, 8, peer);
+      communicator->recv(d_recv.i64() + i, 1, 8, peer);
+    }
+    communicator->stop();
+    std::vector<int64_t> got(G);
+    DJ_HIP_CALL(hipMemcpyAsync(got.data(), d_recv.p, (size_t)G * 8, hipMemcpyDeviceToHost,
+                               dj_rt_comm_stream()));
+    DJ_HIP_CALL(hipStreamSynchronize(dj_rt_comm_stream()));
+    for (int i = 0; i < G; i++)
+      if (i != me) recv_counts[i] = got[i];
+  }
+  recv_offset.resize(G + 1);
+  recv_offset[0] = 0;
+  for (int i = 0; i < G; i++) recv_offset[i + 1] = recv_offset[i] + recv_counts[i];
+}
+
+void communicate_sizes(std::vector<cudf::size_type> const& send_offset,
+                       std::vector<int64_t>& recv_offset,
+                       CommunicationGroup comm_group,
+                       Communicator* communicator)
+{
+  std::vector<int64_t> wide(send_offset.begin(), send_offset.end());
+  communicate_sizes(wide, recv_offset, comm_group, communicator);
+}
+
+void warmup_all_to_all(Communicator* communicator)
+{
+  /* mirrors all_to_all_comm.cpp:191-233: a throwaway all-to-all to
+   * establish RCCL channels before the timed region */
+  const int G = communicator->mpi_size;
+  if (G <= 1) return;
+  const int64_t per_peer = 1 << 18;
+  DBuf d_send((size_t)G * per_peer * 8), d_recv((size_t)G * per_peer * 8);
+  communicator->start();
+  for (int p = 0; p < G; p++) {
+    if (p == communicator->mpi_rank) continue;
+    communicator->send(d_send.i64() + p * per_peer, per_peer, 8, p);
+    communicator->recv(d_recv.i64() + p * per_peer, per_peer, 8, p);
+  }
+  communicator->stop();
+}
+
+/* --------------------------------------------------------- all_to_all plan */
+
+void append_to_all_to_all_comm_buffers(cudf::table_view input,
+                                       cudf::mutable_table_view output,
+                                       std::vector<cudf::size_type> const& send_offsets,
+                                       std::vector<int64_t> const& recv_offsets,
+                                       std::vector<AllToAllCommBuffer>& all_to_all_comm_buffers,
+                                       std::vector<ColumnCompressionOptions> compression_options)
+{
+  check_no_compression(compression_options);
+  for (cudf::size_type c = 0; c < input.num_columns(); c++) {
+    DJ_CHECK_ERROR(input.column(c).type().id() == cudf::type_id::INT64 ||
+                     input.column(c).type().id() == cudf::type_id::INT32,
+                   "all-to-all: fixed-width INT32/INT64 columns only in this build");
+    std::vector<int64_t> soff(send_offsets.begin(), send_offsets.end());
+    all_to_all_comm_buffers.emplace_back(
+      input.column(c).head<int8_t>(), output.column(c).head<int8_t>(), soff, recv_offsets,
+      input.column(c).type(), compression_options[c].compression_method,
+      compression_options[c].cascaded_format);
+  }
+}
+
+void all_to_all_comm(std::vector<AllToAllCommBuffer>& all_to_all_comm_buffers,
+                     CommunicationGroup comm_group,
+                     Communicator* communicator,
+                     bool include_current_rank,
+                     bool report_timing,
+                     void* preallocated_pinned_buffer)
+{
+  (void)report_timing;
+  (void)preallocated_pinned_buffer;
+  const int G = comm_group.size();
+  const int me = comm_group.get_local_idx();
+  for (auto& buf : all_to_all_comm_buffers) {
+    const int esize = cudf::size_of(buf.dtype);
+    for (int i = 0; i < G; i++) {
+      int64_t scount = buf.send_offsets[i + 1] - buf.send_offsets[i];
+      int64_t rcount = buf.recv_offsets[i + 1] - buf.recv_offsets[i];
+      const int8_t* src = (const int8_t*)buf.send_buffer + buf.send_offsets[i] * esize;
+      int8_t* dst = (int8_t*)buf.recv_buffer + buf.recv_offsets[i] * esize;
+      if (i == me) {
+        if (include_current_rank && scount > 0) {
+          /* self-partition: direct D2D on the comm stream (the reference's
+           * explicit copy, all_to_all_comm.cpp:610-653) */
+          DJ_HIP_CALL(hipMemcpyAsync(dst, src, (size_t)scount * esize,
+                                     hipMemcpyDeviceToDevice, dj_rt_comm_stream()));
+        }
+        continue;
+      }
+      int peer = comm_group.get_global_rank(i);
+      communicator->send(src, scount, esize, peer);
+      communicator->recv(dst, rcount, esize, peer);
+    }
+  }
+}
+
+void postprocess_all_to_all_comm(std::vector<AllToAllCommBuffer>& all_to_all_comm_buffers,
+                                 CommunicationGroup comm_group,
+                                 Communicator* communicator,
+                                 bool include_current_rank,
+                                 bool report_timing)
+{
+  /* no compression => nothing to decompress; string offsets rebuilt by the
+   * strings path (config 4) when it lands */
+  (void)all_to_all_comm_buffers;
+  (void)comm_group;
+  (void)communicator;
+  (void)include_current_rank;
+  (void)report_timing;
+}
+
+/* ---------------------------------------------------- AllToAllCommunicator */
+
+AllToAllCommunicator::AllToAllCommunicator(
+  cudf::table_view input_table_,
+  std::vector<cudf::size_type> offsets,
+  CommunicationGroup comm_group_,
+  Communicator* communicator_,
+  std::vector<ColumnCompressionOptions> compression_options_,
+  bool explicit_copy_to_current_rank_)
+  : input_table(input_table_),
+    comm_group(comm_group_),
+    communicator(communicator_),
+    explicit_copy_to_current_rank(explicit_copy_to_current_rank_),
+    send_offsets(std::move(offsets)),
+    compression_options(std::move(compression_options_))
+{
+  check_no_compression(compression_options);
+  DJ_CHECK_ERROR((int)send_offsets.size() == comm_group.size() + 1,
+                 "AllToAllCommunicator: offsets must have comm_group.size()+1 entries");
+  communicate_sizes(send_offsets, recv_offsets, comm_group, communicator);
+}
+
+AllToAllCommunicator::AllToAllCommunicator(
+  cudf::table_view input_table_,
+  std::vector<cudf::size_type> offsets,
+  Communicator* communicator_,
+  std::vector<ColumnCompressionOptions> compression_options_,
+  bool explicit_copy_to_current_rank_)
+  : AllToAllCommunicator(input_table_,
+                         std::move(offsets),
+                         CommunicationGroup(communicator_->mpi_size, 1, communicator_->mpi_rank),
+                         communicator_,
+                         std::move(compression_options_),
+                         explicit_copy_to_current_rank_)
+{
+}
+
+std::unique_ptr<cudf::table> AllToAllCommunicator::allocate_communicated_table()
+{
+  std::vector<std::unique_ptr<cudf::column>> cols;
+  cudf::size_type nrows = (cudf::size_type)recv_offsets.back();
+  for (cudf::size_type c = 0; c < input_table.num_columns(); c++)
+    cols.push_back(std::make_unique<cudf::column>(input_table.column(c).type(), nrows));
+  auto out = std::make_unique<cudf::table>(std::move(cols));
+  if (explicit_copy_to_current_rank) {
+    /* copy the self partition now, outside launch_communication, so the
+     * comm phase moves only remote slices (all_to_all_comm.cpp:701-729) */
+    const int me = comm_group.get_local_idx();
+    for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
+      const int esize = cudf::size_of(input_table.column(c).type());
+      int64_t scount = (int64_t)send_offsets[me + 1] - send_offsets[me];
+      if (scount <= 0) continue;
+      DJ_HIP_CALL(hipMemcpyAsync(
+        (int8_t*)out->get_column(c).head() + recv_offsets[me] * esize,
+        input_table.column(c).head<int8_t>() + (int64_t)send_offsets[me] * esize,
+        (size_t)scount * esize, hipMemcpyDeviceToDevice, dj_rt_stream()));
+    }
+    DJ_HIP_CALL(hipStreamSynchronize(dj_rt_stream()));
+  }
+  return out;
+}
+
+void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communicated_table,
+                                                bool report_timing,
+                                                void* preallocated_pinned_buffer)
+{
+  std::vector<AllToAllCommBuffer> bufs;
+  append_to_all_to_all_comm_buffers(input_table, communicated_table, send_offsets, recv_offsets,
+                                    bufs, compression_options);
+  dj_timing::Scope t(DJ_PHASE_COMM, dj_rt_comm_stream());
+  communicator->start();
+  all_to_all_comm(bufs, comm_group, communicator, !explicit_copy_to_current_rank, report_timing,
+                  preallocated_pinned_buffer);
+  communicator->stop();  // blocks the host (all_to_all_comm.hpp:331 contract)
+  postprocess_all_to_all_comm(bufs, comm_group, communicator, !explicit_copy_to_current_rank,
+                              report_timing);
+}
+
+/* ----------------------------------------------------- partition + join core */
+
+namespace {
+
+/* stable hash-partition of an arbitrary table into nparts contiguous ranges:
+ * permutation computed on (key, iota) with the stable wave-ballot kernel,
+ * then one gather per column. 2-column all-INT64 tables skip the gathers
+ * (keys+payload move directly through the partition scatter). */
+struct PartitionedTable {
+  std::unique_ptr<cudf::table> tbl;
+  std::vector<cudf::size_type> offsets;  // nparts+1
+};
+
+PartitionedTable partition_table(cudf::table_view in, cudf::size_type key_col, int nparts,
+                                 int hash_fn, uint32_t seed)
+{
+  hipStream_t st = dj_rt_stream();
+  const int64_t n = in.num_rows();
+  DBuf key_tmp;
+  const int64_t* keys = key_as_i64(in.column(key_col), key_tmp);
+  DBuf scratch(dj::hash_partition_scratch_bytes(n, nparts));
+  DBuf d_off((size_t)(nparts + 1) * 8);
+
+  const bool fast2 = in.num_columns() == 2 &&
+                     in.column(0).type().id() == cudf::type_id::INT64 &&
+                     in.column(1).type().id() == cudf::type_id::INT64;
+
+  std::vector<std::unique_ptr<cudf::column>> cols;
+  for (cudf::size_type c = 0; c < in.num_columns(); c++)
+    cols.push_back(std::make_unique<cudf::column>(in.column(c).type(), (cudf::size_type)n));
+  auto out = std::make_unique<cudf::table>(std::move(cols));
+
+  {
+    dj_timing::Scope t1(DJ_PHASE_PART_COUNT, st);
+    dj::partition_count(keys, n, nparts, hash_fn, seed, scratch.p, st);
+  }
+  {
+    dj_timing::Scope t2(DJ_PHASE_PART_SCAN, st);
+    dj::partition_scan(n, nparts, scratch.p, d_off.i64(), st);
+  }
+  if (fast2) {
+    const cudf::size_type pay_col = key_col == 0 ? 1 : 0;
+    dj_timing::Scope t3(DJ_PHASE_PART_SCATTER, st);
+    dj::partition_scatter(keys, in.column(pay_col).head<int64_t>(), n, nparts, hash_fn, seed,
+                          d_off.i64(), scratch.p,
+                          (int64_t*)out->get_column(key_col).head(),
+                          (int64_t*)out->get_column(pay_col).head(), st);
+  } else {
+    DBuf iota((size_t)n * 8), perm((size_t)n * 8), keys_out((size_t)n * 8);
+    hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(n)), dim3(kBlock), 0, st, iota.i64(), n);
+    {
+      dj_timing::Scope t3(DJ_PHASE_PART_SCATTER, st);
+      dj::partition_scatter(keys, iota.i64(), n, nparts, hash_fn, seed, d_off.i64(), scratch.p,
+                            keys_out.i64(), perm.i64(), st);
+    }
+    for (cudf::size_type c = 0; c < in.num_columns(); c++) {
+      if (in.column(c).type().id() == cudf::type_id::INT64) {
+        hipLaunchKernelGGL(gather_i64_kernel, dim3(grid_for_n(n)), dim3(kBlock), 0, st,
+                           in.column(c).head<int64_t>(), perm.i64(), n,
+                           (int64_t*)out->get_column(c).head());
+      } else {
+        hipLaunchKernelGGL(gather_i32_kernel, dim3(grid_for_n(n)), dim3(kBlock), 0, st,
+                           in.column(c).head<int32_t>(), perm.i64(), n,
+                           (int32_t*)out->get_column(c).head());
+      }
+    }
+  }
+  std::vector<int64_t> off_host(nparts + 1);
+  DJ_HIP_CALL(hipMemcpyAsync(off_host.data(), d_off.p, (size_t)(nparts + 1) * 8,
+                             hipMemcpyDeviceToHost, st));
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  PartitionedTable r;
+  r.tbl = std::move(out);
+  r.offsets.assign(off_host.begin(), off_host.end());
+  return r;
+}
+
+/* local inner join of two tables via the bucketed-LDS engine; returns
+ * left-cols + right-cols with keys duplicated, row order unspecified */
+std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table_view right,
+                                              cudf::size_type left_on, cudf::size_type right_on)
+{
+  hipStream_t st = dj_rt_stream();
+  const int64_t ln = left.num_rows(), rn = right.num_rows();
+  const cudf::size_type ncl = left.num_columns(), ncr = right.num_columns();
+
+  const bool fast2 = ncl == 2 && ncr == 2 && left_on == 0 && right_on == 0 &&
+                     left.column(0).type().id() == cudf::type_id::INT64 &&
+                     left.column(1).type().id() == cudf::type_id::INT64 &&
+                     right.column(0).type().id() == cudf::type_id::INT64 &&
+                     right.column(1).type().id() == cudf::type_id::INT64;
+
+  auto make_empty = [&]() {
+    std::vector<std::unique_ptr<cudf::column>> cols;
+    for (cudf::size_type c = 0; c < ncl; c++)
+      cols.push_back(std::make_unique<cudf::column>(left.column(c).type(), 0));
+    for (cudf::size_type c = 0; c < ncr; c++)
+      cols.push_back(std::make_unique<cudf::column>(right.column(c).type(), 0));
+    return std::make_unique<cudf::table>(std::move(cols));
+  };
+  if (ln == 0 || rn == 0) return make_empty();  // distributed_join.cpp:76-83
+
+  DBuf lkey_tmp, rkey_tmp;
+  const int64_t* lk = key_as_i64(left.column(left_on), lkey_tmp);
+  const int64_t* rk = key_as_i64(right.column(right_on), rkey_tmp);
+  const int64_t* lp;
+  const int64_t* rp;
+  DBuf liota, riota;
+  if (fast2) {
+    lp = left.column(1).head<int64_t>();
+    rp = right.column(1).head<int64_t>();
+  } else {
+    liota = DBuf((size_t)ln * 8);
+    riota = DBuf((size_t)rn * 8);
+    hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(ln)), dim3(kBlock), 0, st, liota.i64(), ln);
+    hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(rn)), dim3(kBlock), 0, st, riota.i64(), rn);
+    lp = liota.i64();
+    rp = riota.i64();
+  }
+
+  DBuf scratch((size_t)dj_bucket_join_scratch_bytes(ln, rn));
+  DBuf d_err(16), d_cnt(16);
+  int64_t cap = std::max<int64_t>(rn + (rn >> 3), 1024);
+  for (;;) {
+    DBuf o0((size_t)cap * 8), o1((size_t)cap * 8), o2((size_t)cap * 8), o3((size_t)cap * 8);
+    DJ_HIP_CALL(hipMemsetAsync(d_err.p, 0, 4, st));
+    DJ_HIP_CALL(hipMemsetAsync(d_cnt.p, 0, 8, st));
+    dj_bucket_local_join(lk, lp, ln, rk, rp, rn, o0.i64(), o1.i64(), o2.i64(), o3.i64(), cap,
+                         d_cnt.i64(), (int*)d_err.p, scratch.p);
+    int64_t nout = 0;
+    int err = 0;
+    DJ_HIP_CALL(hipMemcpyAsync(&nout, d_cnt.p, 8, hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipMemcpyAsync(&err, d_err.p, 4, hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    DJ_CHECK_ERROR(err == 0, "join build: key equal to the empty sentinel (-1) is unsupported");
+    if (nout > cap) {
+      cap = nout;
+      continue;
+    }
+    /* assemble output table */
+    std::vector<std::unique_ptr<cudf::column>> cols;
+    if (fast2) {
+      cols.push_back(std::make_unique<cudf::column>(cudf::data_type(cudf::type_id::INT64),
+                                                    (cudf::size_type)nout, o0.p));
+      o0.p = nullptr;
+      cols.push_back(std::make_unique<cudf::column>(cudf::data_type(cudf::type_id::INT64),
+                                                    (cudf::size_type)nout, o1.p));
+      o1.p = nullptr;
+      cols.push_back(std::make_unique<cudf::column>(cudf::data_type(cudf::type_id::INT64),
+                                                    (cudf::size_type)nout, o2.p));
+      o2.p = nullptr;
+      cols.push_back(std::make_unique<cudf::column>(cudf::data_type(cudf::type_id::INT64),
+                                                    (cudf::size_type)nout, o3.p));
+      o3.p = nullptr;
+    } else {
+      /* general: o1/o3 hold source row indices; gather every column */
+      auto gather_col = [&](cudf::column_view src, DBuf& idx) {
+        auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
+        if (nout > 0) {
+          if (src.type().id() == cudf::type_id::INT64)
+            hipLaunchKernelGGL(gather_i64_kernel, dim3(grid_for_n(nout)), dim3(kBlock), 0, st,
+                               src.head<int64_t>(), idx.i64(), nout, (int64_t*)col->head());
+          else
+            hipLaunchKernelGGL(gather_i32_kernel, dim3(grid_for_n(nout)), dim3(kBlock), 0, st,
+                               src.head<int32_t>(), idx.i64(), nout, (int32_t*)col->head());
+        }
+        return col;
+      };
+      for (cudf::size_type c = 0; c < ncl; c++) cols.push_back(gather_col(left.column(c), o1));
+      for (cudf::size_type c = 0; c < ncr; c++) cols.push_back(gather_col(right.column(c), o3));
+      DJ_HIP_CALL(hipStreamSynchronize(st));
+    }
+    return std::make_unique<cudf::table>(std::move(cols));
+  }
+}
+
+std::unique_ptr<cudf::table> concat_tables(std::vector<std::unique_ptr<cudf::table>>& parts)
+{
+  if (parts.size() == 1) return std::move(parts[0]);
+  hipStream_t st = dj_rt_stream();
+  dj_timing::Scope t(DJ_PHASE_CONCAT, st);
+  int64_t total = 0;
+  for (auto& p : parts) total += p->num_rows();
+  std::vector<std::unique_ptr<cudf::column>> cols;
+  for (cudf::size_type c = 0; c < parts[0]->num_columns(); c++) {
+    auto type = parts[0]->get_column(c).type();
+    auto col = std::make_unique<cudf::column>(type, (cudf::size_type)total);
+    int64_t off = 0;
+    for (auto& p : parts) {
+      int64_t nrows = p->num_rows();
+      if (nrows > 0)
+        DJ_HIP_CALL(hipMemcpyAsync((int8_t*)col->head() + off * cudf::size_of(type),
+                                   p->get_column(c).head(),
+                                   (size_t)nrows * cudf::size_of(type),
+                                   hipMemcpyDeviceToDevice, st));
+      off += nrows;
+    }
+    cols.push_back(std::move(col));
+  }
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  return std::make_unique<cudf::table>(std::move(cols));
+}
+
+}  // namespace
+
+/* --------------------------------------------------- distributed_inner_join */
+
+std::unique_ptr<cudf::table> distributed_inner_join(
+  cudf::table_view left,
+  cudf::table_view right,
+  std::vector<cudf::size_type> const& left_on,
+  std::vector<cudf::size_type> const& right_on,
+  Communicator* communicator,
+  std::vector<ColumnCompressionOptions> left_compression_options,
+  std::vector<ColumnCompressionOptions> right_compression_options,
+  int over_decom_factor,
+  bool report_timing,
+  void* preallocated_pinned_buffer,
+  int nvlink_domain_size)
+{
+  (void)nvlink_domain_size;  // flat xGMI all-to-all; see distributed_join.hpp header
+  DJ_CHECK_ERROR(left_on.size() == 1 && right_on.size() == 1,
+                 "this build joins on a single key column (the hot-path shape)");
+  check_no_compression(left_compression_options);
+  check_no_compression(right_compression_options);
+
+  const int G = communicator->mpi_size;
+  CommunicationGroup group(G, 1, communicator->mpi_rank);
+  const int nparts = G * over_decom_factor;
+  DJ_CHECK_ERROR(nparts <= dj::kMaxPartitions,
+                 "world_size x over_decom_factor must be <= 64");
+
+  /* rank-level stable partition, seed 12345678 (distributed_join.cpp:211-226) */
+  PartitionedTable lpart =
+    partition_table(left, left_on[0], nparts, DJ_HASH_MURMUR3, DJ_SEED_INTRA);
+  PartitionedTable rpart =
+    partition_table(right, right_on[0], nparts, DJ_HASH_MURMUR3, DJ_SEED_INTRA);
+
+  std::vector<std::unique_ptr<cudf::table>> batch_results;
+  for (int b = 0; b < over_decom_factor; b++) {
+    /* batch b = partitions [b*G, (b+1)*G): a contiguous offset slice
+     * (distributed_join.cpp:247-266) */
+    std::vector<cudf::size_type> lslice(lpart.offsets.begin() + b * G,
+                                        lpart.offsets.begin() + b * G + G + 1);
+    std::vector<cudf::size_type> rslice(rpart.offsets.begin() + b * G,
+                                        rpart.offsets.begin() + b * G + G + 1);
+    AllToAllCommunicator latoa(lpart.tbl->view(), lslice, group, communicator,
+                               left_compression_options, true);
+    AllToAllCommunicator ratoa(rpart.tbl->view(), rslice, group, communicator,
+                               right_compression_options, true);
+    auto lrecv = latoa.allocate_communicated_table();
+    auto rrecv = ratoa.allocate_communicated_table();
+    latoa.launch_communication(lrecv->mutable_view(), report_timing,
+                               preallocated_pinned_buffer);
+    ratoa.launch_communication(rrecv->mutable_view(), report_timing,
+                               preallocated_pinned_buffer);
+    batch_results.push_back(
+      local_inner_join(lrecv->view(), rrecv->view(), left_on[0], right_on[0]));
+  }
+  return concat_tables(batch_results);
+}
+
+/* ------------------------------------------------------------- shuffle_on */
+
+std::unique_ptr<cudf::table> shuffle_on(cudf::table_view const& input,
+                                        std::vector<cudf::size_type> const& on_columns,
+                                        CommunicationGroup comm_group,
+                                        Communicator* communicator,
+                                        std::vector<ColumnCompressionOptions> compression_options,
+                                        cudf::hash_id hash_function,
+                                        uint32_t hash_seed,
+                                        bool report_timing,
+                                        void* preallocated_pinned_buffer)
+{
+  DJ_CHECK_ERROR(on_columns.size() == 1, "this build shuffles on a single key column");
+  const int hash_fn =
+    hash_function == cudf::hash_id::HASH_IDENTITY ? DJ_HASH_IDENTITY : DJ_HASH_MURMUR3;
+  PartitionedTable part =
+    partition_table(input, on_columns[0], comm_group.size(), hash_fn, hash_seed);
+  AllToAllCommunicator atoa(part.tbl->view(), part.offsets, comm_group, communicator,
+                            compression_options, false);
+  auto out = atoa.allocate_communicated_table();
+  atoa.launch_communication(out->mutable_view(), report_timing, preallocated_pinned_buffer);
+  return out;
+}
+
+std::unique_ptr<cudf::table> shuffle_on(cudf::table_view const& input,
+                                        std::vector<cudf::size_type> const& on_columns,
+                                        Communicator* communicator,
+                                        std::vector<ColumnCompressionOptions> compression_options,
+                                        cudf::hash_id hash_function,
+                                        uint32_t hash_seed,
+                                        bool report_timing,
+                                        void* preallocated_pinned_buffer)
+{
+  return shuffle_on(input, on_columns,
+                    CommunicationGroup(communicator->mpi_size, 1, communicator->mpi_rank),
+                    communicator, std::move(compression_options), hash_function, hash_seed,
+                    report_timing, preallocated_pinned_buffer);
+}
+
+/* --------------------------------------------------------- distribute_table */
+
+namespace {
+constexpr int kMaxSchemaCols = 32;
+}
+
+std::unique_ptr<cudf::table> distribute_table(cudf::table_view global_table,
+                                              Communicator* communicator)
+{
+  const int G = communicator->mpi_size;
+  const int rank = communicator->mpi_rank;
+  hipStream_t st = dj_rt_stream();
+
+  /* schema + per-rank row counts: root sends [nrows_r, ncols, dtype ids...] */
+  std::vector<int64_t> header(2 + kMaxSchemaCols, 0);
+  if (rank == 0) {
+    DJ_CHECK_ERROR(global_table.num_columns() <= kMaxSchemaCols, "too many columns");
+    header[1] = global_table.num_columns();
+    for (cudf::size_type c = 0; c < global_table.num_columns(); c++)
+      header[2 + c] = (int64_t)global_table.column(c).type().id();
+  }
+  int64_t total = rank == 0 ? global_table.num_rows() : 0;
+  DBuf d_hdr((size_t)header.size() * 8 * (rank == 0 ? G : 1));
+  if (G > 1) {
+    if (rank == 0) {
+      std::vector<int64_t> all;
+      for (int r = 0; r < G; r++) {
+        int64_t rows_r = total / G + (r < total % G ? 1 : 0);
+        header[0] = rows_r;
+        all.insert(all.end(), header.begin(), header.end());
+      }
+      DJ_HIP_CALL(hipMemcpyAsync(d_hdr.p, all.data(), all.size() * 8, hipMemcpyHostToDevice, st));
+      DJ_HIP_CALL(hipStreamSynchronize(st));
+    }
+    communicator->start();
+    if (rank == 0) {
+      for (int r = 1; r < G; r++)
+        communicator->send(d_hdr.i64() + (size_t)r * header.size(), header.size(), 8, r);
+    } else {
+      communicator->recv(d_hdr.i64(), header.size(), 8, 0);
+    }
+    communicator->stop();
+    if (rank != 0) {
+      DJ_HIP_CALL(hipMemcpyAsync(header.data(), d_hdr.p, header.size() * 8,
+                                 hipMemcpyDeviceToHost, st));
+      DJ_HIP_CALL(hipStreamSynchronize(st));
+    } else {
+      header[0] = total / G + (0 < total % G ? 1 : 0);
+    }
+  } else {
+    header[0] = total;
+  }
+
+  const int64_t my_rows = header[0];
+  const int ncols = (int)header[1];
+  std::vector<std::unique_ptr<cudf::column>> cols;
+  for (int c = 0; c < ncols; c++)
+    cols.push_back(std::make_unique<cudf::column>(
+      cudf::data_type((cudf::type_id)header[2 + c]), (cudf::size_type)my_rows));
+  auto local = std::make_unique<cudf::table>(std::move(cols));
+
+  /* row data: root sends each rank its contiguous slice */
+  communicator->start();
+  if (rank == 0) {
+    int64_t row0 = 0;
+    for (int r = 0; r < G; r++) {
+      int64_t rows_r = total / G + (r < total % G ? 1 : 0);
+      for (int c = 0; c < ncols; c++) {
+        const int esize = cudf::size_of(global_table.column(c).type());
+        const int8_t* src = global_table.column(c).head<int8_t>() + row0 * esize;
+        if (r == 0) {
+          DJ_HIP_CALL(hipMemcpyAsync(local->get_column(c).head(), src, (size_t)rows_r * esize,
+                                     hipMemcpyDeviceToDevice, dj_rt_comm_stream()));
+        } else {
+          communicator->send(src, rows_r, esize, r);
+        }
+      }
+      row0 += rows_r;
+    }
+  } else {
+    for (int c = 0; c < ncols; c++) {
+      const int esize = cudf::size_of(local->get_column(c).type());
+      communicator->recv(local->get_column(c).head(), my_rows, esize, 0);
+    }
+  }
+  communicator->stop();
+  return local;
+}
+
+std::unique_ptr<cudf::table> collect_tables(cudf::table_view table, Communicator* communicator)
+{
+  const int G = communicator->mpi_size;
+  const int rank = communicator->mpi_rank;
+  CommunicationGroup group(G, 1, rank);
+
+  /* per-rank row counts to root via communicate_sizes (send all rows to
+   * local rank 0) */
+  std::vector<int64_t> send_offset(G + 1, (int64_t)table.num_rows());
+  send_offset[0] = 0;
+  std::vector<int64_t> recv_offset;
+  communicate_sizes(send_offset, recv_offset, group, communicator);
+
+  std::unique_ptr<cudf::table> merged;
+  if (rank == 0) {
+    std::vector<std::unique_ptr<cudf::column>> cols;
+    for (cudf::size_type c = 0; c < table.num_columns(); c++)
+      cols.push_back(std::make_unique<cudf::column>(table.column(c).type(),
+                                                    (cudf::size_type)recv_offset.back()));
+    merged = std::make_unique<cudf::table>(std::move(cols));
+  }
+
+  communicator->start();
+  for (cudf::size_type c = 0; c < table.num_columns(); c++) {
+    const int esize = cudf::size_of(table.column(c).type());
+    if (rank == 0) {
+      for (int r = 0; r < G; r++) {
+        int64_t count = recv_offset[r + 1] - recv_offset[r];
+        int8_t* dst = (int8_t*)merged->get_column(c).head() + recv_offset[r] * esize;
+        if (r == 0) {
+          DJ_HIP_CALL(hipMemcpyAsync(dst, table.column(c).head<int8_t>(),
+                                     (size_t)count * esize, hipMemcpyDeviceToDevice,
+                                     dj_rt_comm_stream()));
+        } else {
+          communicator->recv(dst, count, esize, r);
+        }
+      }
+    } else {
+      communicator->send(table.column(c).head<int8_t>(), table.num_rows(), esize, 0);
+    }
+  }
+  communicator->stop();
+  return merged;  // nullptr on non-root ranks
+}
+
+/* --------------------------- C ABI over the C++ orchestration (for the
+ * Python measurement harness; additive — SURVEY.md §8b) ------------------ */
+
+extern "C" {
+
+void* dj_cpp_comm_create(int rank, int size, const void* id_bytes)
+{
+  if (size <= 1) {
+    auto* c = new LocalCommunicator();
+    g_default_comm = c;
+    return c;
+  }
+  return new RCCLCommunicator(rank, size, id_bytes);
+}
+
+void dj_cpp_comm_destroy(void* comm)
+{
+  auto* c = (Communicator*)comm;
+  c->finalize();
+  if (g_default_comm == c) g_default_comm = nullptr;
+  delete c;
+}
+
+/* full distributed_inner_join over int64 key/payload columns; returns an
+ * opaque cudf::table* */
+void* dj_cpp_distributed_inner_join_i64(void* comm, const int64_t* d_lk, const int64_t* d_lp,
+                                        int64_t ln, const int64_t* d_rk, const int64_t* d_rp,
+                                        int64_t rn, int over_decom, int report_timing)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  cudf::table_view left(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)ln, d_lk),
+     column_view(data_type(type_id::INT64), (cudf::size_type)ln, d_lp)});
+  cudf::table_view right(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)rn, d_rk),
+     column_view(data_type(type_id::INT64), (cudf::size_type)rn, d_rp)});
+  auto opts = generate_compression_options_distributed(left, false);
+  auto result = distributed_inner_join(left, right, {0}, {0}, (Communicator*)comm, opts, opts,
+                                       over_decom, report_timing != 0, nullptr, 1);
+  return result.release();
+}
+
+void* dj_cpp_shuffle_on_i64(void* comm, const int64_t* d_keys, const int64_t* d_pay, int64_t n,
+                            int hash_function, uint32_t seed)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  cudf::table_view input(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)n, d_keys),
+     column_view(data_type(type_id::INT64), (cudf::size_type)n, d_pay)});
+  auto opts = generate_compression_options_distributed(input, false);
+  auto result =
+    shuffle_on(input, {0}, (Communicator*)comm, opts,
+               hash_function == DJ_HASH_IDENTITY ? cudf::hash_id::HASH_IDENTITY
+                                                 : cudf::hash_id::HASH_MURMUR3,
+               seed, false, nullptr);
+  return result.release();
+}
+
+int64_t dj_table_num_rows(void* tbl) { return ((cudf::table*)tbl)->num_rows(); }
+int dj_table_num_columns(void* tbl) { return ((cudf::table*)tbl)->num_columns(); }
+const void* dj_table_column_data(void* tbl, int i)
+{
+  return ((cudf::table*)tbl)->get_column(i).head();
+}
+void dj_table_free(void* tbl) { delete (cudf::table*)tbl; }
+
+}  // extern "C"

@@ -1,0 +1,46 @@
+/*
+ * distributed_join.hpp — top-level distributed inner join, mirroring the
+ * reference's exact signature (reference: src/distributed_join.hpp:65-76)
+ * over our re-declared cudf types (dj_cudf_types.hpp).
+ *
+ * Semantics kept (reference pins):
+ *  - collective over all ranks; `left`/`right` are each rank's slice of the
+ *    global tables (distributed_join.hpp:30-44 doc)
+ *  - result columns: all left columns then all right columns, join keys
+ *    duplicated (compare_against_single_gpu.cu:163-165); row order
+ *    unspecified; concatenation over ranks is the global join result
+ *  - joining with an empty side yields an empty table
+ *    (distributed_join.cpp:76-83)
+ *  - over_decom_factor batches the exchange+join (distributed_join.cpp:244-329)
+ *  - partition seed 12345678 intra-node (distributed_join.cpp:211)
+ *
+ * Changed vs reference: nvlink_domain_size is accepted but the exchange is
+ * always the flat single-level xGMI all-to-all — on one MI355X node every
+ * GPU pair has a direct link, so the reference's 2-level IB x NVLink
+ * hierarchy (distributed_join.cpp:152-199) collapses;
+ * the join result is partition-invariant, so results are identical for any
+ * nvlink_domain_size. The engine below the API is the bucketed-LDS join
+ * (dj_kernels.hip), not cuDF.
+ */
+#pragma once
+
+#include "communicator.hpp"
+#include "compression.hpp"
+#include "dj_cudf_types.hpp"
+
+#include <cstdint>
+#include <memory>
+#include <vector>
+
+std::unique_ptr<cudf::table> distributed_inner_join(
+  cudf::table_view left,
+  cudf::table_view right,
+  std::vector<cudf::size_type> const& left_on,
+  std::vector<cudf::size_type> const& right_on,
+  Communicator* communicator,
+  std::vector<ColumnCompressionOptions> left_compression_options,
+  std::vector<ColumnCompressionOptions> right_compression_options,
+  int over_decom_factor            = 1,
+  bool report_timing               = false,
+  void* preallocated_pinned_buffer = nullptr,
+  int nvlink_domain_size           = 1);

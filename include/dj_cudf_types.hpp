@@ -1,0 +1,173 @@
+/*
+ * dj_cudf_types.hpp — OUR minimal re-declarations of the cuDF types the
+ * reference's public API surface uses (cudf::table, cudf::table_view,
+ * cudf::column, cudf::size_type, cudf::data_type, cudf::hash_id).
+ *
+ * This is NOT NVIDIA cuDF. Per the drop-in contract ("the
+ * reference's exact C++ signatures, our own headers re-declaring the types
+ * cuDF provided"), code written against the reference's
+ * distributed_inner_join / shuffle_on / Communicator interfaces compiles and
+ * links against these headers unchanged. Only the subset of the cuDF API the
+ * reference's interface touches is provided; column data lives in HIP device
+ * memory (MI355X HBM3E). Do not link this together with real libcudf.
+ *
+ * v1 supports fixed-width INT32/INT64 columns (the BASELINE configs 1-3
+ * types); STRING is declared for the config-4 path.
+ */
+#pragma once
+
+#include <cstdint>
+#include <memory>
+#include <stdexcept>
+#include <vector>
+
+namespace cudf {
+
+using size_type = int32_t;
+
+enum class type_id : int32_t { EMPTY = 0, INT32, INT64, STRING };
+
+struct data_type {
+  data_type() = default;
+  explicit constexpr data_type(type_id id) : _id(id) {}
+  constexpr type_id id() const { return _id; }
+  bool operator==(data_type o) const { return _id == o._id; }
+  bool operator!=(data_type o) const { return _id != o._id; }
+
+ private:
+  type_id _id{type_id::EMPTY};
+};
+
+inline constexpr size_type size_of(data_type t)
+{
+  switch (t.id()) {
+    case type_id::INT32: return 4;
+    case type_id::INT64: return 8;
+    default: return 0;  // STRING handled via children
+  }
+}
+
+/* mirrors cudf::hash_id as used by shuffle_on.hpp:49 */
+enum class hash_id : int32_t { HASH_IDENTITY = 0, HASH_MURMUR3 = 1 };
+constexpr uint32_t DEFAULT_HASH_SEED = 0;
+
+class column_view {
+ public:
+  column_view() = default;
+  column_view(data_type type, size_type size, const void* data)
+    : _type(type), _size(size), _data(data)
+  {
+  }
+  data_type type() const { return _type; }
+  size_type size() const { return _size; }
+  template <typename T>
+  const T* head() const
+  {
+    return static_cast<const T*>(_data);
+  }
+
+ private:
+  data_type _type{};
+  size_type _size{0};
+  const void* _data{nullptr};
+};
+
+class mutable_column_view {
+ public:
+  mutable_column_view() = default;
+  mutable_column_view(data_type type, size_type size, void* data)
+    : _type(type), _size(size), _data(data)
+  {
+  }
+  data_type type() const { return _type; }
+  size_type size() const { return _size; }
+  template <typename T>
+  T* head() const
+  {
+    return static_cast<T*>(_data);
+  }
+  operator column_view() const { return column_view(_type, _size, _data); }
+
+ private:
+  data_type _type{};
+  size_type _size{0};
+  void* _data{nullptr};
+};
+
+/* owning device column (HIP device memory; allocation in dj_cpp_api.hip) */
+class column {
+ public:
+  column(data_type type, size_type size);            // allocates device memory
+  column(data_type type, size_type size, void* adopt_device_ptr);
+  column(const column&) = delete;
+  column& operator=(const column&) = delete;
+  column(column&& o) noexcept;
+  ~column();
+
+  data_type type() const { return _type; }
+  size_type size() const { return _size; }
+  void* head() { return _data; }
+  const void* head() const { return _data; }
+  column_view view() const { return column_view(_type, _size, _data); }
+  mutable_column_view mutable_view() { return mutable_column_view(_type, _size, _data); }
+
+ private:
+  data_type _type{};
+  size_type _size{0};
+  void* _data{nullptr};
+};
+
+class table_view {
+ public:
+  table_view() = default;
+  explicit table_view(std::vector<column_view> cols) : _cols(std::move(cols)) {}
+  size_type num_columns() const { return (size_type)_cols.size(); }
+  size_type num_rows() const { return _cols.empty() ? 0 : _cols[0].size(); }
+  column_view column(size_type i) const { return _cols.at(i); }
+
+ private:
+  std::vector<column_view> _cols;
+};
+
+class mutable_table_view {
+ public:
+  mutable_table_view() = default;
+  explicit mutable_table_view(std::vector<mutable_column_view> cols) : _cols(std::move(cols)) {}
+  size_type num_columns() const { return (size_type)_cols.size(); }
+  size_type num_rows() const { return _cols.empty() ? 0 : _cols[0].size(); }
+  mutable_column_view column(size_type i) const { return _cols.at(i); }
+  operator table_view() const
+  {
+    std::vector<column_view> v(_cols.begin(), _cols.end());
+    return table_view(std::move(v));
+  }
+
+ private:
+  std::vector<mutable_column_view> _cols;
+};
+
+class table {
+ public:
+  explicit table(std::vector<std::unique_ptr<column>> cols) : _cols(std::move(cols)) {}
+  size_type num_columns() const { return (size_type)_cols.size(); }
+  size_type num_rows() const { return _cols.empty() ? 0 : _cols[0]->size(); }
+  column& get_column(size_type i) { return *_cols.at(i); }
+  const column& get_column(size_type i) const { return *_cols.at(i); }
+  table_view view() const
+  {
+    std::vector<column_view> v;
+    for (auto& c : _cols) v.push_back(c->view());
+    return table_view(std::move(v));
+  }
+  mutable_table_view mutable_view()
+  {
+    std::vector<mutable_column_view> v;
+    for (auto& c : _cols) v.push_back(c->mutable_view());
+    return mutable_table_view(std::move(v));
+  }
+
+ private:
+  std::vector<std::unique_ptr<column>> _cols;
+};
+
+}  // namespace cudf
