@@ -7,15 +7,17 @@ int64/int64 join at selectivity 0.3 (100M rows per GPU per table — the
 reference benchmark's per-GPU default, benchmark/distributed_join.cu:96-109).
 `value` = total input rows of both tables across all ranks / join wall time
 (the README-comparable number: 0.392 s on 8xV100 => ~4.08e9 rows/s,
-README.md:73-86). One step = one full distributed inner join over the
-resident tables: hash-partition both tables + size exchange + RCCL all-to-all
-(N>1) + hash-table build + probe. Generation and warmup are excluded, timing
-is barrier-bracketed, max over ranks — mirroring the reference's timed region
-(benchmark/distributed_join.cu:264-286).
+README.md:73-86).
+
+One step = one full `distributed_inner_join` call through the C++ drop-in
+path (include/distributed_join.hpp -> stable hash partition + RCCL all-to-all
+over xGMI + bucketed-LDS local join + concat), via its C ABI. Generation and
+warmup are excluded; timing is barrier-bracketed, max over ranks — mirroring
+the reference's timed region (benchmark/distributed_join.cu:264-286).
 
 Single process (N=1) by default; for N>1 the driver launches this under
-torch.distributed.run with one rank per GPU; torch.distributed (gloo) is used
-only for bootstrap (RCCL unique id exchange) and barriers — compute and
+torch.distributed.run with one rank per GPU; torch.distributed (gloo) is
+used only for bootstrap (RCCL unique id exchange) and barriers — compute and
 communication run in libdistjoin.so (HIP + RCCL over xGMI).
 """
 import argparse
@@ -47,7 +49,6 @@ def cpu_baseline_leg(sample_rows):
     n = sample_rows
     bk, bp = oracle.gen_build(n)
     pk, pp = oracle.gen_probe(n, n, selectivity=SELECTIVITY)
-    # one untimed warm run on a small slice to fault pages / threads
     oracle.cpu_radix_join(bk[:100_000], bp[:100_000], pk[:100_000], pp[:100_000],
                           count_only=True)
     t0 = time.perf_counter()
@@ -86,102 +87,48 @@ def main():
 
     dist = None
     if world > 1:
-        import torch
         import torch.distributed as tdist
         dist = tdist
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         tdist.init_process_group("gloo", rank=rank, world_size=world)
 
     dj.require_gpu()
-    dj.lib().dj_set_device(local_rank)
+    L = dj.lib()
+    L.dj_set_device(local_rank)
 
     # RCCL bootstrap: rank 0's unique id broadcast over gloo
+    id_bytes = None
     if world > 1:
         import torch
-        nbytes = dj.lib().dj_rccl_unique_id_bytes()
+        nbytes = L.dj_rccl_unique_id_bytes()
         if rank == 0:
             buf = np.zeros(nbytes, dtype=np.uint8)
-            dj.lib().dj_rccl_get_unique_id(buf.ctypes.data)
+            L.dj_rccl_get_unique_id(buf.ctypes.data)
             t = torch.from_numpy(buf)
         else:
             t = torch.zeros(nbytes, dtype=torch.uint8)
         dist.broadcast(t, src=0)
-        dj.lib().dj_comm_init(rank, world, t.numpy().ctypes.data)
+        id_bytes = np.ascontiguousarray(t.numpy())
+    comm = dj.CppCommunicator(rank, world, id_bytes)
 
     rows = args.rows
-    G = world  # partitions == ranks (over-decom batches later widen this)
     build_n_global = rows * N
     probe_n_global = rows * N
     rand_max = 2 * build_n_global
 
     log(rank, f"[bench] N={N} rows/GPU={rows} generating inputs...")
-    L = dj.lib()
     row0 = rank * rows
     bk, bp = dj.generate_build(build_n_global, rand_max, uniq=True, row0=row0, nrows=rows)
     pk, pp = dj.generate_probe(probe_n_global, build_n_global, rand_max,
                                selectivity=SELECTIVITY, row0=row0, nrows=rows)
 
-    # partition outputs + scratch
-    part = [dj.DeviceArray(rows) for _ in range(4)]  # bk,bp,pk,pp partitioned
-    scratch = L.dj_dmalloc(L.dj_partition_scratch_bytes(rows, max(G, 1)))
-
-    # receive buffers (N>1): capacity with skew margin
-    recv_cap = rows if N == 1 else int(rows * 1.03) + (1 << 16)
-    if N > 1:
-        recv = [dj.DeviceArray(recv_cap) for _ in range(4)]
-    else:
-        recv = part
-
-    # bucketed-join scratch + outputs
-    bucket_scratch = L.dj_dmalloc(L.dj_bucket_join_scratch_bytes(recv_cap, recv_cap))
-    d_error = L.dj_dmalloc(4)
-    d_counter = L.dj_dmalloc(8)
-    out_cap = recv_cap + (recv_cap >> 3)
-    outs = [dj.DeviceArray(out_cap) for _ in range(4)]
-
-    zero8 = np.zeros(1, dtype=np.int64)
-    off_b = np.zeros(G + 1, dtype=np.int64)
-    off_p = np.zeros(G + 1, dtype=np.int64)
-
-    state = {"matches": 0, "lrows": rows, "rrows": rows, "wire_elems": 0}
+    state = {"matches": 0}
 
     def step():
-        L.dj_memcpy_h2d(d_counter, zero8.ctypes.data, 8)
-        L.dj_memcpy_h2d(d_error, zero8.ctypes.data, 4)
-        # partition both tables into G ranges (seed 12345678 like
-        # distributed_join.cpp:211-226)
-        L.dj_hash_partition(bk.ptr, bp.ptr, rows, G, dj.HASH_MURMUR3, dj.SEED_INTRA,
-                            part[0].ptr, part[1].ptr, off_b.ctypes.data, scratch)
-        L.dj_hash_partition(pk.ptr, pp.ptr, rows, G, dj.HASH_MURMUR3, dj.SEED_INTRA,
-                            part[2].ptr, part[3].ptr, off_p.ctypes.data, scratch)
-        if N > 1:
-            # size exchange + personalized all-to-all per column buffer
-            scnt_b = np.diff(off_b).copy()
-            scnt_p = np.diff(off_p).copy()
-            rcnt_b = np.zeros(G, dtype=np.int64)
-            rcnt_p = np.zeros(G, dtype=np.int64)
-            L.dj_exchange_sizes(scnt_b.ctypes.data, rcnt_b.ctypes.data)
-            L.dj_exchange_sizes(scnt_p.ctypes.data, rcnt_p.ctypes.data)
-            roff_b = np.concatenate([[0], np.cumsum(rcnt_b)]).astype(np.int64)
-            roff_p = np.concatenate([[0], np.cumsum(rcnt_p)]).astype(np.int64)
-            if roff_b[-1] > recv_cap or roff_p[-1] > recv_cap:
-                raise RuntimeError("receive capacity exceeded; raise --rows margin")
-            for send, rbuf, soff, roff in ((part[0], recv[0], off_b, roff_b),
-                                           (part[1], recv[1], off_b, roff_b),
-                                           (part[2], recv[2], off_p, roff_p),
-                                           (part[3], recv[3], off_p, roff_p)):
-                L.dj_all_to_all_i64(send.ptr, soff.ctypes.data, rbuf.ptr, roff.ctypes.data)
-            lrows, rrows = int(roff_b[-1]), int(roff_p[-1])
-            state["wire_elems"] = int((scnt_b.sum() - scnt_b[rank]) * 2 +
-                                      (scnt_p.sum() - scnt_p[rank]) * 2)
-        else:
-            lrows, rrows = rows, rows
-        # local join: bucketed LDS build + probe (fused)
-        L.dj_bucket_local_join(recv[0].ptr, recv[1].ptr, lrows,
-                               recv[2].ptr, recv[3].ptr, rrows,
-                               outs[0].ptr, outs[1].ptr, outs[2].ptr, outs[3].ptr,
-                               out_cap, d_counter, d_error, bucket_scratch)
-        state["lrows"], state["rrows"] = lrows, rrows
+        t = L.dj_cpp_distributed_inner_join_i64(comm.ptr, bk.ptr, bp.ptr, rows,
+                                                pk.ptr, pp.ptr, rows, args.over_decom, 0)
+        state["matches"] = L.dj_table_num_rows(t)
+        L.dj_table_free(t)
 
     def barrier_sync():
         L.dj_sync()
@@ -191,9 +138,6 @@ def main():
     log(rank, f"[bench] warmup {args.warmup} steps...")
     for _ in range(args.warmup):
         step()
-    state["matches"] = L.dj_read_counter_i64(d_counter)
-    if L.dj_read_error_i32(d_error):
-        raise RuntimeError("build error: sentinel key")
 
     L.dj_timing_enable(1)
     L.dj_timing_reset()
@@ -210,27 +154,32 @@ def main():
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = float(e.item())
 
-    matches = L.dj_read_counter_i64(d_counter)
+    matches = state["matches"]
     ms_per_step = elapsed / args.steps * 1000.0
     input_rows = 2.0 * rows * N  # both tables, whole node
     value = input_rows / (elapsed / args.steps)
     out_rows_s = matches * N / (elapsed / args.steps)
 
-    # per-phase kernel timing (hipEvents on the launch stream)
+    # per-phase kernel timing (hipEvents inside the C++ orchestration)
     phases = {name: {"ms": L.dj_timing_total_ms(pid) / args.steps,
                      "launches": L.dj_timing_launches(pid) / args.steps}
               for name, pid in dj.PHASES.items()}
 
-    # roofline for the dominant join kernel (algorithmic bytes: DESIGN.md §Measurement)
-    # bucket_scatter phase = full bucket partition of both tables
-    # (count 8 B read + scatter 16 B read + 16 B write per row);
-    # join_fused = streaming read of both bucketed tables + output writes
-    # (LDS table traffic is on-chip, not HBM-algorithmic);
-    # part_scatter = rank-level stable scatter of both tables.
+    # roofline for the dominant join kernel (algorithmic bytes per step for
+    # this rank; derivation in DESIGN.md §Measurement):
+    #   part_scatter   — rank-level stable scatter, both tables:
+    #                    16 B read + 16 B write per row
+    #   bucket_scatter — full two-level bucket partition of both local
+    #                    tables: 8 B count-read + 16 B read + 16 B write per
+    #                    row per level (pass A skipped when B == 256)
+    #   join_fused     — streaming read of both bucketed tables (16 B/row) +
+    #                    32 B per output row; LDS traffic is on-chip
+    lrows = rrows = float(rows)  # hash-uniform => received ~= sent rows
+    nlevels = 2 if rows / 256 > 1600 else 1
     alg = {
         "part_scatter": 32.0 * rows * 2,
-        "bucket_scatter": 40.0 * (state["lrows"] + state["rrows"]),
-        "join_fused": 16.0 * (state["lrows"] + state["rrows"]) + 32.0 * matches,
+        "bucket_scatter": 40.0 * nlevels * (lrows + rrows),
+        "join_fused": 16.0 * (lrows + rrows) + 32.0 * matches,
     }
     dom = max(alg.keys(), key=lambda k: phases[k]["ms"])
     dom_ms = phases[dom]["ms"]
@@ -247,7 +196,9 @@ def main():
 
     all_to_all_GBs = None
     if N > 1 and phases["comm"]["ms"] > 0:
-        all_to_all_GBs = (state["wire_elems"] * 8 / 1e9) / (phases["comm"]["ms"] / 1e3)
+        # hash-uniform egress estimate: both tables, (G-1)/G of rows, 16 B/row
+        est_bytes = 2.0 * rows * 16.0 * (N - 1) / N
+        all_to_all_GBs = (est_bytes / 1e9) / (phases["comm"]["ms"] / 1e3)
 
     cpu_baseline = None
     if rank == 0 and N == 1 and not args.no_cpu_baseline:
@@ -277,6 +228,7 @@ def main():
                 "selectivity": SELECTIVITY,
                 "output_rows_per_gpu": int(matches),
                 "output_rows_per_sec": out_rows_s,
+                "engine": "C++ distributed_inner_join (drop-in path)",
                 "phases_ms": {k: round(v["ms"], 4) for k, v in phases.items()},
                 "all_to_all_GBs_per_gpu": all_to_all_GBs,
             },
@@ -285,8 +237,8 @@ def main():
         }
         print(json.dumps(rec), flush=True)
 
+    comm.destroy()
     if world > 1:
-        L.dj_comm_finalize()
         dist.destroy_process_group()
 
 

@@ -73,13 +73,23 @@ __global__ void widen_i32_kernel(const int32_t* __restrict__ src, int64_t n,
   for (; i < n; i += stride) dst[i] = src[i];
 }
 
-/* small owning device buffer */
+/* small owning device buffer (pool-backed: hipMallocAsync on the compute
+ * stream + host sync — the RMM-pool role of the reference's setup.cpp:51-67;
+ * frees via hipFree, which device-syncs, keeping cross-stream reuse safe) */
+void* pool_alloc(size_t bytes)
+{
+  void* p = nullptr;
+  DJ_HIP_CALL(hipMallocAsync(&p, bytes, dj_rt_stream()));
+  DJ_HIP_CALL(hipStreamSynchronize(dj_rt_stream()));
+  return p;
+}
+
 struct DBuf {
   void* p{nullptr};
   DBuf() = default;
   explicit DBuf(size_t bytes)
   {
-    if (bytes) DJ_HIP_CALL(hipMalloc(&p, bytes));
+    if (bytes) p = pool_alloc(bytes);
   }
   DBuf(const DBuf&) = delete;
   DBuf& operator=(const DBuf&) = delete;
@@ -95,7 +105,7 @@ struct DBuf {
   }
   ~DBuf()
   {
-    if (p) (void)hipFree(p);
+    if (p) (void)hipFreeAsync(p, dj_rt_stream());
   }
   int64_t* i64() { return (int64_t*)p; }
 };
@@ -162,7 +172,7 @@ namespace cudf {
 column::column(data_type type, size_type size) : _type(type), _size(size)
 {
   size_t bytes = (size_t)size * size_of(type);
-  if (bytes) DJ_HIP_CALL(hipMalloc(&_data, bytes));
+  if (bytes) _data = pool_alloc(bytes);
 }
 
 column::column(data_type type, size_type size, void* adopt) : _type(type), _size(size), _data(adopt)
