@@ -78,10 +78,29 @@ __global__ void widen_i32_kernel(const int32_t* __restrict__ src, int64_t n,
  * frees via hipFree, which device-syncs, keeping cross-stream reuse safe) */
 void* pool_alloc(size_t bytes)
 {
+  static bool pool_configured = false;
+  if (!pool_configured) {
+    /* keep freed memory cached in the pool (RMM-pool behavior) instead of
+     * returning it to the OS at every sync */
+    int dev = 0;
+    DJ_HIP_CALL(hipGetDevice(&dev));
+    hipMemPool_t pool;
+    DJ_HIP_CALL(hipDeviceGetDefaultMemPool(&pool, dev));
+    uint64_t threshold = UINT64_MAX;
+    DJ_HIP_CALL(hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold, &threshold));
+    pool_configured = true;
+  }
   void* p = nullptr;
   DJ_HIP_CALL(hipMallocAsync(&p, bytes, dj_rt_stream()));
   DJ_HIP_CALL(hipStreamSynchronize(dj_rt_stream()));
   return p;
+}
+
+/* async-pool memory must be freed with hipFreeAsync (plain hipFree does not
+ * return it to the pool) */
+void pool_free(void* p)
+{
+  if (p) (void)hipFreeAsync(p, dj_rt_stream());
 }
 
 struct DBuf {
@@ -97,16 +116,13 @@ struct DBuf {
   DBuf& operator=(DBuf&& o) noexcept
   {
     if (this != &o) {
-      if (p) (void)hipFree(p);
+      pool_free(p);
       p = o.p;
       o.p = nullptr;
     }
     return *this;
   }
-  ~DBuf()
-  {
-    if (p) (void)hipFreeAsync(p, dj_rt_stream());
-  }
+  ~DBuf() { pool_free(p); }
   int64_t* i64() { return (int64_t*)p; }
 };
 
@@ -187,7 +203,9 @@ column::column(column&& o) noexcept : _type(o._type), _size(o._size), _data(o._d
 
 column::~column()
 {
-  if (_data) (void)hipFree(_data);
+  /* async-pool memory must be freed with hipFreeAsync; ROCm routes plain
+   * hipMalloc pointers through it correctly too */
+  if (_data) (void)hipFreeAsync(_data, dj_rt_stream());
 }
 
 }  // namespace cudf
