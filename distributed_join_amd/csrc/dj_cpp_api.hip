@@ -24,6 +24,9 @@
 
 #include <algorithm>
 #include <cstring>
+#include <map>
+#include <mutex>
+#include <unordered_map>
 #include <vector>
 
 /* ---------------------------------------------------------------- helpers */
@@ -76,32 +79,68 @@ __global__ void widen_i32_kernel(const int32_t* __restrict__ src, int64_t n,
 /* small owning device buffer (pool-backed: hipMallocAsync on the compute
  * stream + host sync — the RMM-pool role of the reference's setup.cpp:51-67;
  * frees via hipFree, which device-syncs, keeping cross-stream reuse safe) */
-void* pool_alloc(size_t bytes)
-{
-  static bool pool_configured = false;
-  if (!pool_configured) {
-    /* keep freed memory cached in the pool (RMM-pool behavior) instead of
-     * returning it to the OS at every sync */
-    int dev = 0;
-    DJ_HIP_CALL(hipGetDevice(&dev));
-    hipMemPool_t pool;
-    DJ_HIP_CALL(hipDeviceGetDefaultMemPool(&pool, dev));
-    uint64_t threshold = UINT64_MAX;
-    DJ_HIP_CALL(hipMemPoolSetAttribute(pool, hipMemPoolAttrReleaseThreshold, &threshold));
-    pool_configured = true;
+/* Caching device allocator — the RMM-pool role of the reference's
+ * setup.cpp:51-67. hipMalloc-backed, size-binned free list; freed blocks
+ * are cached and reused (hipMalloc/hipFree per call would device-sync and
+ * dominate the step time; ROCm's default hipMallocAsync mempool failed at
+ * GB-scale blocks on gfx950). Reuse is safe because every free in this
+ * layer happens after a host-synchronized phase boundary. */
+class CachingAllocator {
+ public:
+  void* alloc(size_t bytes)
+  {
+    bytes = (bytes + 255) & ~(size_t)255;
+    {
+      std::lock_guard<std::mutex> g(m);
+      auto it = free_list.lower_bound(bytes);
+      if (it != free_list.end() && it->first <= bytes * 2 + (64 << 10)) {
+        void* p = it->second;
+        live[p] = it->first;
+        free_list.erase(it);
+        return p;
+      }
+    }
+    void* p = nullptr;
+    hipError_t e = hipMalloc(&p, bytes);
+    if (e == hipErrorOutOfMemory) {
+      trim();
+      e = hipMalloc(&p, bytes);
+    }
+    DJ_CHECK_ERROR(e == hipSuccess, "device allocator: out of memory");
+    std::lock_guard<std::mutex> g(m);
+    live[p] = bytes;
+    return p;
   }
-  void* p = nullptr;
-  DJ_HIP_CALL(hipMallocAsync(&p, bytes, dj_rt_stream()));
-  DJ_HIP_CALL(hipStreamSynchronize(dj_rt_stream()));
-  return p;
+  void dealloc(void* p)
+  {
+    if (!p) return;
+    std::lock_guard<std::mutex> g(m);
+    auto it = live.find(p);
+    DJ_CHECK_ERROR(it != live.end(), "device allocator: unknown pointer freed");
+    free_list.emplace(it->second, p);
+    live.erase(it);
+  }
+  void trim()
+  {
+    std::lock_guard<std::mutex> g(m);
+    for (auto& kv : free_list) (void)hipFree(kv.second);
+    free_list.clear();
+  }
+
+ private:
+  std::mutex m;
+  std::multimap<size_t, void*> free_list;
+  std::unordered_map<void*, size_t> live;
+};
+
+CachingAllocator& pool()
+{
+  static CachingAllocator a;
+  return a;
 }
 
-/* async-pool memory must be freed with hipFreeAsync (plain hipFree does not
- * return it to the pool) */
-void pool_free(void* p)
-{
-  if (p) (void)hipFreeAsync(p, dj_rt_stream());
-}
+void* pool_alloc(size_t bytes) { return pool().alloc(bytes); }
+void pool_free(void* p) { pool().dealloc(p); }
 
 struct DBuf {
   void* p{nullptr};
@@ -203,9 +242,7 @@ column::column(column&& o) noexcept : _type(o._type), _size(o._size), _data(o._d
 
 column::~column()
 {
-  /* async-pool memory must be freed with hipFreeAsync; ROCm routes plain
-   * hipMalloc pointers through it correctly too */
-  if (_data) (void)hipFreeAsync(_data, dj_rt_stream());
+  if (_data) pool_free(_data);
 }
 
 }  // namespace cudf
