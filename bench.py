@@ -184,6 +184,18 @@ def main():
     dom = max(alg.keys(), key=lambda k: phases[k]["ms"])
     dom_ms = phases[dom]["ms"]
     achieved = (alg[dom] / 1e9) / (dom_ms / 1e3) if dom_ms > 0 else None
+    # measured HBM traffic per step: PMC calibration committed under
+    # profiles/r01_pmc_traffic.json (rocprofv3 --pmc FETCH_SIZE / WRITE_SIZE
+    # passes; FETCH x2 gfx950 correction — see that file)
+    traffic = None
+    try:
+        with open(os.path.join(REPO, "profiles", "r01_pmc_traffic.json")) as fh:
+            coeff = json.load(fh)["phase_bytes_per_input_row_pair"]
+        traffic = coeff[dom] * (lrows + rrows)
+        if dom == "join_fused":
+            traffic += 0.0  # output writes already included in the calibration
+    except Exception:
+        pass
     roofline = {
         "bound": "hbm",
         "kernel": dom,
@@ -191,7 +203,7 @@ def main():
         "peak": 8000.0,
         "unit": "GB/s",
         "frac": (achieved / 8000.0) if achieved else None,
-        "traffic": None,
+        "traffic": traffic,
     }
 
     all_to_all_GBs = None
