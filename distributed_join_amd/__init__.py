@@ -84,6 +84,14 @@ def lib():
             "dj_cpp_comm_destroy": ([vp], None),
             "dj_cpp_distributed_inner_join_i64": ([vp, vp, vp, i64, vp, vp, i64, i32, i32], vp),
             "dj_cpp_shuffle_on_i64": ([vp, vp, vp, i64, i32, u32], vp),
+            "dj_gen_test_strings": ([vp, i64, ctypes.POINTER(ctypes.c_void_p),
+                                     ctypes.POINTER(ctypes.c_void_p),
+                                     ctypes.POINTER(ctypes.c_int64)], None),
+            "dj_cpp_distributed_inner_join_i64str": ([vp, vp, vp, vp, i64, i64,
+                                                      vp, vp, vp, i64, i64, i32, i32], vp),
+            "dj_table_column_type": ([vp, i32], i32),
+            "dj_table_column_chars": ([vp, i32], vp),
+            "dj_table_column_chars_size": ([vp, i32], i64),
             "dj_table_num_rows": ([vp], i64),
             "dj_table_num_columns": ([vp], i32),
             "dj_table_column_data": ([vp, i32], vp),
@@ -218,18 +226,38 @@ def local_inner_join_global(d_lk, d_lp, ln, d_rk, d_rp, rn, cap=None):
         cap = n
 
 
+TYPE_INT8, TYPE_INT32, TYPE_INT64, TYPE_STRING = 1, 2, 3, 4
+
+
 def table_to_numpy(tbl_ptr):
-    """Copy an opaque cudf::table* (C ABI) into numpy columns and free it."""
+    """Copy an opaque cudf::table* (C ABI) into numpy columns and free it.
+    Fixed-width columns -> int64/int32 arrays; STRING columns ->
+    (offsets int32 array, chars uint8 array) tuples."""
     L = lib()
     n = L.dj_table_num_rows(tbl_ptr)
     ncols = L.dj_table_num_columns(tbl_ptr)
     cols = []
     for c in range(ncols):
+        t = L.dj_table_column_type(tbl_ptr, c)
         data = L.dj_table_column_data(tbl_ptr, c)
-        out = np.empty(n, dtype=np.int64)
-        if n:
-            L.dj_memcpy_d2h(out.ctypes.data, data, n * 8)
-        cols.append(out)
+        if t == TYPE_STRING:
+            off = np.empty(n + 1, dtype=np.int32)
+            L.dj_memcpy_d2h(off.ctypes.data, data, (n + 1) * 4)
+            nch = L.dj_table_column_chars_size(tbl_ptr, c)
+            ch = np.empty(max(nch, 1), dtype=np.uint8)
+            if nch:
+                L.dj_memcpy_d2h(ch.ctypes.data, L.dj_table_column_chars(tbl_ptr, c), nch)
+            cols.append((off, ch[:nch]))
+        elif t == TYPE_INT32:
+            out = np.empty(n, dtype=np.int32)
+            if n:
+                L.dj_memcpy_d2h(out.ctypes.data, data, n * 4)
+            cols.append(out)
+        else:
+            out = np.empty(n, dtype=np.int64)
+            if n:
+                L.dj_memcpy_d2h(out.ctypes.data, data, n * 8)
+            cols.append(out)
     L.dj_table_free(tbl_ptr)
     return cols
 
@@ -255,6 +283,28 @@ def cpp_distributed_inner_join(comm, d_lk, d_lp, ln, d_rk, d_rp, rn, over_decom=
     t = lib().dj_cpp_distributed_inner_join_i64(comm.ptr, d_lk.ptr, d_lp.ptr, ln,
                                                 d_rk.ptr, d_rp.ptr, rn, over_decom,
                                                 int(report_timing))
+    return table_to_numpy(t)
+
+
+def gen_test_strings(d_keys, n):
+    """Deterministic string payload from keys (len=k%7+1, char='a'+k%26).
+    Returns (offsets_ptr, chars_ptr, chars_bytes); caller frees via dj_dfree."""
+    L = lib()
+    off = ctypes.c_void_p()
+    ch = ctypes.c_void_p()
+    nb = ctypes.c_int64()
+    L.dj_gen_test_strings(d_keys.ptr, n, ctypes.byref(off), ctypes.byref(ch),
+                          ctypes.byref(nb))
+    return off.value, ch.value, nb.value
+
+
+def cpp_distributed_inner_join_str(comm, d_lk, l_strings, ln, d_rk, r_strings, rn,
+                                   over_decom=1):
+    """Distributed join with int64 keys and STRING payloads (config 4)."""
+    lo, lc, lb = l_strings
+    ro, rc, rb = r_strings
+    t = lib().dj_cpp_distributed_inner_join_i64str(
+        comm.ptr, d_lk.ptr, lo, lc, lb, ln, d_rk.ptr, ro, rc, rb, rn, over_decom, 0)
     return table_to_numpy(t)
 
 

@@ -234,15 +234,44 @@ column::column(data_type type, size_type size, void* adopt) : _type(type), _size
 {
 }
 
-column::column(column&& o) noexcept : _type(o._type), _size(o._size), _data(o._data)
+column::column(size_type size, int64_t chars_bytes)
+  : _type(data_type(type_id::STRING)), _size(size), _chars_size(chars_bytes)
+{
+  _data = pool_alloc(((size_t)size + 1) * sizeof(int32_t));
+  if (chars_bytes) _chars = pool_alloc((size_t)chars_bytes);
+}
+
+column::column(column&& o) noexcept
+  : _type(o._type), _size(o._size), _data(o._data), _chars(o._chars), _chars_size(o._chars_size)
 {
   o._data = nullptr;
+  o._chars = nullptr;
   o._size = 0;
+  o._chars_size = 0;
+}
+
+column& column::operator=(column&& o) noexcept
+{
+  if (this != &o) {
+    if (_data) pool_free(_data);
+    if (_chars) pool_free(_chars);
+    _type = o._type;
+    _size = o._size;
+    _data = o._data;
+    _chars = o._chars;
+    _chars_size = o._chars_size;
+    o._data = nullptr;
+    o._chars = nullptr;
+    o._size = 0;
+    o._chars_size = 0;
+  }
+  return *this;
 }
 
 column::~column()
 {
   if (_data) pool_free(_data);
+  if (_chars) pool_free(_chars);
 }
 
 }  // namespace cudf
@@ -487,6 +516,17 @@ void postprocess_all_to_all_comm(std::vector<AllToAllCommBuffer>& all_to_all_com
 
 /* ---------------------------------------------------- AllToAllCommunicator */
 
+/* per string column: char-offset boundaries per peer + device row-size
+ * buffers (reference all_to_all_comm.hpp:349-360 members; sizes, not
+ * offsets, go on the wire — offsets rebuilt receiver-side by scan,
+ * strings_column.cu:111-131) */
+struct AllToAllCommunicator::StringsState {
+  std::vector<std::vector<int64_t>> send_char_offsets;  // [col][G+1]
+  std::vector<std::vector<int64_t>> recv_char_offsets;  // [col][G+1]
+  std::vector<DBuf> sizes_to_send;                      // [col] int32[n]
+  std::vector<DBuf> sizes_received;                     // [col] int32[n_recv]
+};
+
 AllToAllCommunicator::AllToAllCommunicator(
   cudf::table_view input_table_,
   std::vector<cudf::size_type> offsets,
@@ -505,6 +545,53 @@ AllToAllCommunicator::AllToAllCommunicator(
   DJ_CHECK_ERROR((int)send_offsets.size() == comm_group.size() + 1,
                  "AllToAllCommunicator: offsets must have comm_group.size()+1 entries");
   communicate_sizes(send_offsets, recv_offsets, comm_group, communicator);
+
+  /* strings columns: exchange per-peer char byte counts; precompute the row
+   * sizes to send (gather_string_offsets + calculate_string_sizes_from_offsets
+   * roles, strings_column.cu:39-109) */
+  const int G = comm_group.size();
+  const int64_t n = input_table.num_rows();
+  const int64_t n_recv = recv_offsets.back();
+  hipStream_t st = dj_rt_stream();
+  for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
+    if (input_table.column(c).type().id() != cudf::type_id::STRING) {
+      if (strings) {
+        strings->send_char_offsets.emplace_back();
+        strings->recv_char_offsets.emplace_back();
+        strings->sizes_to_send.emplace_back();
+        strings->sizes_received.emplace_back();
+      }
+      continue;
+    }
+    if (!strings) {
+      strings = std::make_shared<StringsState>();
+      for (cudf::size_type cc = 0; cc < c; cc++) {
+        strings->send_char_offsets.emplace_back();
+        strings->recv_char_offsets.emplace_back();
+        strings->sizes_to_send.emplace_back();
+        strings->sizes_received.emplace_back();
+      }
+    }
+    auto col = input_table.column(c);
+    /* char offsets at the G+1 send row boundaries */
+    std::vector<int64_t> soff(G + 1);
+    for (int k = 0; k <= G; k++) {
+      int32_t v = 0;
+      DJ_HIP_CALL(hipMemcpyAsync(&v, col.head<int32_t>() + send_offsets[k], 4,
+                                 hipMemcpyDeviceToHost, st));
+      DJ_HIP_CALL(hipStreamSynchronize(st));
+      soff[k] = v;
+    }
+    std::vector<int64_t> roff;
+    communicate_sizes(soff, roff, comm_group, communicator);
+    DBuf sizes((size_t)(n > 0 ? n : 1) * 4);
+    dj::sizes_from_offsets(col.head<int32_t>(), n, (int32_t*)sizes.p, st);
+    strings->send_char_offsets.push_back(std::move(soff));
+    strings->recv_char_offsets.push_back(std::move(roff));
+    strings->sizes_to_send.push_back(std::move(sizes));
+    strings->sizes_received.push_back(DBuf((size_t)(n_recv > 0 ? n_recv : 1) * 4));
+  }
+  (void)st;
 }
 
 AllToAllCommunicator::AllToAllCommunicator(
@@ -526,21 +613,42 @@ std::unique_ptr<cudf::table> AllToAllCommunicator::allocate_communicated_table()
 {
   std::vector<std::unique_ptr<cudf::column>> cols;
   cudf::size_type nrows = (cudf::size_type)recv_offsets.back();
-  for (cudf::size_type c = 0; c < input_table.num_columns(); c++)
-    cols.push_back(std::make_unique<cudf::column>(input_table.column(c).type(), nrows));
+  for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
+    if (input_table.column(c).type().id() == cudf::type_id::STRING)
+      cols.push_back(std::make_unique<cudf::column>(
+        nrows, strings->recv_char_offsets[c].back()));
+    else
+      cols.push_back(std::make_unique<cudf::column>(input_table.column(c).type(), nrows));
+  }
   auto out = std::make_unique<cudf::table>(std::move(cols));
   if (explicit_copy_to_current_rank) {
     /* copy the self partition now, outside launch_communication, so the
      * comm phase moves only remote slices (all_to_all_comm.cpp:701-729) */
     const int me = comm_group.get_local_idx();
     for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
-      const int esize = cudf::size_of(input_table.column(c).type());
       int64_t scount = (int64_t)send_offsets[me + 1] - send_offsets[me];
       if (scount <= 0) continue;
-      DJ_HIP_CALL(hipMemcpyAsync(
-        (int8_t*)out->get_column(c).head() + recv_offsets[me] * esize,
-        input_table.column(c).head<int8_t>() + (int64_t)send_offsets[me] * esize,
-        (size_t)scount * esize, hipMemcpyDeviceToDevice, dj_rt_stream()));
+      if (input_table.column(c).type().id() == cudf::type_id::STRING) {
+        /* self slice of row SIZES into sizes_received + chars slice */
+        DJ_HIP_CALL(hipMemcpyAsync(
+          (int32_t*)strings->sizes_received[c].p + recv_offsets[me],
+          (const int32_t*)strings->sizes_to_send[c].p + send_offsets[me],
+          (size_t)scount * 4, hipMemcpyDeviceToDevice, dj_rt_stream()));
+        int64_t cbytes =
+          strings->send_char_offsets[c][me + 1] - strings->send_char_offsets[c][me];
+        if (cbytes > 0)
+          DJ_HIP_CALL(hipMemcpyAsync(
+            (uint8_t*)out->get_column(c).chars() + strings->recv_char_offsets[c][me],
+            (const uint8_t*)input_table.column(c).chars() +
+              strings->send_char_offsets[c][me],
+            (size_t)cbytes, hipMemcpyDeviceToDevice, dj_rt_stream()));
+      } else {
+        const int esize = cudf::size_of(input_table.column(c).type());
+        DJ_HIP_CALL(hipMemcpyAsync(
+          (int8_t*)out->get_column(c).head() + recv_offsets[me] * esize,
+          input_table.column(c).head<int8_t>() + (int64_t)send_offsets[me] * esize,
+          (size_t)scount * esize, hipMemcpyDeviceToDevice, dj_rt_stream()));
+      }
     }
     DJ_HIP_CALL(hipStreamSynchronize(dj_rt_stream()));
   }
@@ -552,20 +660,76 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
                                                 void* preallocated_pinned_buffer)
 {
   std::vector<AllToAllCommBuffer> bufs;
-  append_to_all_to_all_comm_buffers(input_table, communicated_table, send_offsets, recv_offsets,
-                                    bufs, compression_options);
-  dj_timing::Scope t(DJ_PHASE_COMM, dj_rt_comm_stream());
-  communicator->start();
-  all_to_all_comm(bufs, comm_group, communicator, !explicit_copy_to_current_rank, report_timing,
-                  preallocated_pinned_buffer);
-  communicator->stop();  // blocks the host (all_to_all_comm.hpp:331 contract)
+  std::vector<int64_t> soff(send_offsets.begin(), send_offsets.end());
+  for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
+    auto in = input_table.column(c);
+    auto out = communicated_table.column(c);
+    if (in.type().id() == cudf::type_id::STRING) {
+      /* two wire buffers per string column: row SIZES (row offsets) and the
+       * chars bytes (char offsets) — strings_column.cu strategy */
+      bufs.emplace_back(strings->sizes_to_send[c].p, strings->sizes_received[c].p, soff,
+                        recv_offsets, cudf::data_type(cudf::type_id::INT32),
+                        CompressionMethod::none, nvcompCascadedFormatOpts{});
+      bufs.emplace_back(in.chars(), out.chars(), strings->send_char_offsets[c],
+                        strings->recv_char_offsets[c], cudf::data_type(cudf::type_id::INT8),
+                        CompressionMethod::none, nvcompCascadedFormatOpts{});
+    } else {
+      bufs.emplace_back(in.head<int8_t>(), out.head<int8_t>(), soff, recv_offsets, in.type(),
+                        compression_options[c].compression_method,
+                        compression_options[c].cascaded_format);
+    }
+  }
+  {
+    dj_timing::Scope t(DJ_PHASE_COMM, dj_rt_comm_stream());
+    communicator->start();
+    all_to_all_comm(bufs, comm_group, communicator, !explicit_copy_to_current_rank,
+                    report_timing, preallocated_pinned_buffer);
+    communicator->stop();  // blocks the host (all_to_all_comm.hpp:331 contract)
+  }
   postprocess_all_to_all_comm(bufs, comm_group, communicator, !explicit_copy_to_current_rank,
                               report_timing);
+  /* receiver-side: rebuild string offsets from the received sizes by scan
+   * with offset[0]=0 (strings_column.cu:111-131) */
+  if (strings) {
+    hipStream_t st = dj_rt_stream();
+    const int64_t n_recv = recv_offsets.back();
+    for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
+      if (input_table.column(c).type().id() != cudf::type_id::STRING) continue;
+      DBuf scr(dj::offsets_from_sizes_scratch_bytes(n_recv));
+      dj::offsets_from_sizes((const int32_t*)strings->sizes_received[c].p, n_recv,
+                             communicated_table.column(c).head<int32_t>(), scr.p, st);
+    }
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+  }
 }
 
 /* ----------------------------------------------------- partition + join core */
 
 namespace {
+
+/* gather a STRING column by a row-index permutation: sizes gather -> scan ->
+ * chars gather (the reference's thrust::gather + scan strings recipe,
+ * strings_column.cu:39-131, as one reusable step) */
+std::unique_ptr<cudf::column> gather_string_column(cudf::column_view src, const int64_t* d_idx,
+                                                   int64_t n)
+{
+  hipStream_t st = dj_rt_stream();
+  DBuf sizes((size_t)(n > 0 ? n : 1) * 4);
+  dj::gather_sizes(src.head<int32_t>(), d_idx, n, (int32_t*)sizes.p, st);
+  DBuf off(((size_t)n + 1) * 4);
+  DBuf scan_scratch(dj::offsets_from_sizes_scratch_bytes(n));
+  dj::offsets_from_sizes((const int32_t*)sizes.p, n, (int32_t*)off.p, scan_scratch.p, st);
+  int32_t total = 0;
+  DJ_HIP_CALL(hipMemcpyAsync(&total, (int32_t*)off.p + n, 4, hipMemcpyDeviceToHost, st));
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  auto col = std::make_unique<cudf::column>((cudf::size_type)n, (int64_t)total);
+  DJ_HIP_CALL(hipMemcpyAsync(col->head(), off.p, ((size_t)n + 1) * 4,
+                             hipMemcpyDeviceToDevice, st));
+  dj::gather_chars(src.head<int32_t>(), (const uint8_t*)src.chars(), d_idx, n,
+                   (const int32_t*)col->head(), (uint8_t*)col->chars(), st);
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  return col;
+}
 
 /* stable hash-partition of an arbitrary table into nparts contiguous ranges:
  * permutation computed on (key, iota) with the stable wave-ballot kernel,
@@ -591,8 +755,12 @@ PartitionedTable partition_table(cudf::table_view in, cudf::size_type key_col, i
                      in.column(1).type().id() == cudf::type_id::INT64;
 
   std::vector<std::unique_ptr<cudf::column>> cols;
-  for (cudf::size_type c = 0; c < in.num_columns(); c++)
-    cols.push_back(std::make_unique<cudf::column>(in.column(c).type(), (cudf::size_type)n));
+  for (cudf::size_type c = 0; c < in.num_columns(); c++) {
+    if (in.column(c).type().id() == cudf::type_id::STRING)
+      cols.push_back(std::make_unique<cudf::column>((cudf::size_type)0, (int64_t)0));
+    else
+      cols.push_back(std::make_unique<cudf::column>(in.column(c).type(), (cudf::size_type)n));
+  }
   auto out = std::make_unique<cudf::table>(std::move(cols));
 
   {
@@ -623,10 +791,15 @@ PartitionedTable partition_table(cudf::table_view in, cudf::size_type key_col, i
         hipLaunchKernelGGL(gather_i64_kernel, dim3(grid_for_n(n)), dim3(kBlock), 0, st,
                            in.column(c).head<int64_t>(), perm.i64(), n,
                            (int64_t*)out->get_column(c).head());
-      } else {
+      } else if (in.column(c).type().id() == cudf::type_id::INT32) {
         hipLaunchKernelGGL(gather_i32_kernel, dim3(grid_for_n(n)), dim3(kBlock), 0, st,
                            in.column(c).head<int32_t>(), perm.i64(), n,
                            (int32_t*)out->get_column(c).head());
+      } else if (in.column(c).type().id() == cudf::type_id::STRING) {
+        auto scol = gather_string_column(in.column(c), perm.i64(), n);
+        out->get_column(c) = std::move(*scol);
+      } else {
+        DJ_CHECK_ERROR(false, "partition: unsupported column type");
       }
     }
   }
@@ -655,12 +828,15 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
                      right.column(0).type().id() == cudf::type_id::INT64 &&
                      right.column(1).type().id() == cudf::type_id::INT64;
 
+  auto empty_col = [&](cudf::column_view v) {
+    if (v.type().id() == cudf::type_id::STRING)
+      return std::make_unique<cudf::column>((cudf::size_type)0, (int64_t)0);
+    return std::make_unique<cudf::column>(v.type(), (cudf::size_type)0);
+  };
   auto make_empty = [&]() {
     std::vector<std::unique_ptr<cudf::column>> cols;
-    for (cudf::size_type c = 0; c < ncl; c++)
-      cols.push_back(std::make_unique<cudf::column>(left.column(c).type(), 0));
-    for (cudf::size_type c = 0; c < ncr; c++)
-      cols.push_back(std::make_unique<cudf::column>(right.column(c).type(), 0));
+    for (cudf::size_type c = 0; c < ncl; c++) cols.push_back(empty_col(left.column(c)));
+    for (cudf::size_type c = 0; c < ncr; c++) cols.push_back(empty_col(right.column(c)));
     return std::make_unique<cudf::table>(std::move(cols));
   };
   if (ln == 0 || rn == 0) return make_empty();  // distributed_join.cpp:76-83
@@ -720,6 +896,8 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
     } else {
       /* general: o1/o3 hold source row indices; gather every column */
       auto gather_col = [&](cudf::column_view src, DBuf& idx) {
+        if (src.type().id() == cudf::type_id::STRING)
+          return gather_string_column(src, idx.i64(), nout);
         auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
         if (nout > 0) {
           if (src.type().id() == cudf::type_id::INT64)
@@ -739,6 +917,15 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
   }
 }
 
+/* rebase a part's offsets by a constant and append (STRING concat) */
+__global__ void rebase_offsets_kernel(const int32_t* __restrict__ src, int64_t n, int32_t base,
+                                      int32_t* __restrict__ dst)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = src[i] + base;
+}
+
 std::unique_ptr<cudf::table> concat_tables(std::vector<std::unique_ptr<cudf::table>>& parts)
 {
   if (parts.size() == 1) return std::move(parts[0]);
@@ -749,6 +936,28 @@ std::unique_ptr<cudf::table> concat_tables(std::vector<std::unique_ptr<cudf::tab
   std::vector<std::unique_ptr<cudf::column>> cols;
   for (cudf::size_type c = 0; c < parts[0]->num_columns(); c++) {
     auto type = parts[0]->get_column(c).type();
+    if (type.id() == cudf::type_id::STRING) {
+      int64_t total_chars = 0;
+      for (auto& p : parts) total_chars += p->get_column(c).chars_size();
+      auto col = std::make_unique<cudf::column>((cudf::size_type)total, total_chars);
+      int64_t row_off = 0, char_off = 0;
+      for (auto& p : parts) {
+        int64_t nrows = p->num_rows();
+        int64_t nchars = p->get_column(c).chars_size();
+        if (nrows > 0)
+          hipLaunchKernelGGL(rebase_offsets_kernel, dim3(grid_for_n(nrows + 1)), dim3(kBlock),
+                             0, st, (const int32_t*)p->get_column(c).head(), nrows + 1,
+                             (int32_t)char_off, (int32_t*)col->head() + row_off);
+        if (nchars > 0)
+          DJ_HIP_CALL(hipMemcpyAsync((uint8_t*)col->chars() + char_off,
+                                     p->get_column(c).chars(), (size_t)nchars,
+                                     hipMemcpyDeviceToDevice, st));
+        row_off += nrows;
+        char_off += nchars;
+      }
+      cols.push_back(std::move(col));
+      continue;
+    }
     auto col = std::make_unique<cudf::column>(type, (cudf::size_type)total);
     int64_t off = 0;
     for (auto& p : parts) {
@@ -1058,6 +1267,67 @@ void* dj_cpp_shuffle_on_i64(void* comm, const int64_t* d_keys, const int64_t* d_
                                                  : cudf::hash_id::HASH_MURMUR3,
                seed, false, nullptr);
   return result.release();
+}
+
+/* deterministic test/bench string payload from keys (string_payload.cu
+ * pattern): returns offsets (int32[n+1]) and chars device buffers via out
+ * params; caller frees with dj_dfree */
+void dj_gen_test_strings(const int64_t* d_keys, int64_t n, void** out_offsets,
+                         void** out_chars, int64_t* out_chars_bytes)
+{
+  hipStream_t st = dj_rt_stream();
+  DBuf sizes((size_t)(n > 0 ? n : 1) * 4);
+  dj::make_test_string_sizes(d_keys, n, (int32_t*)sizes.p, st);
+  int32_t* offsets = (int32_t*)dj_dmalloc((n + 1) * 4);
+  DBuf scr(dj::offsets_from_sizes_scratch_bytes(n));
+  dj::offsets_from_sizes((const int32_t*)sizes.p, n, offsets, scr.p, st);
+  int32_t total = 0;
+  DJ_HIP_CALL(hipMemcpyAsync(&total, offsets + n, 4, hipMemcpyDeviceToHost, st));
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  uint8_t* chars = (uint8_t*)dj_dmalloc(total > 0 ? total : 1);
+  dj::fill_test_strings(d_keys, n, offsets, chars, st);
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  *out_offsets = offsets;
+  *out_chars = chars;
+  *out_chars_bytes = total;
+}
+
+/* distributed join with int64 keys and STRING payloads (BASELINE config 4) */
+void* dj_cpp_distributed_inner_join_i64str(void* comm, const int64_t* d_lk,
+                                           const void* l_offsets, const void* l_chars,
+                                           int64_t l_chars_bytes, int64_t ln,
+                                           const int64_t* d_rk, const void* r_offsets,
+                                           const void* r_chars, int64_t r_chars_bytes,
+                                           int64_t rn, int over_decom, int report_timing)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  cudf::table_view left(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)ln, d_lk),
+     column_view(data_type(type_id::STRING), (cudf::size_type)ln, l_offsets, l_chars,
+                 l_chars_bytes)});
+  cudf::table_view right(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)rn, d_rk),
+     column_view(data_type(type_id::STRING), (cudf::size_type)rn, r_offsets, r_chars,
+                 r_chars_bytes)});
+  auto opts = generate_compression_options_distributed(left, false);
+  auto result = distributed_inner_join(left, right, {0}, {0}, (Communicator*)comm, opts, opts,
+                                       over_decom, report_timing != 0, nullptr, 1);
+  return result.release();
+}
+
+int dj_table_column_type(void* tbl, int i)
+{
+  return (int)((cudf::table*)tbl)->get_column(i).type().id();
+}
+const void* dj_table_column_chars(void* tbl, int i)
+{
+  return ((cudf::table*)tbl)->get_column(i).chars();
+}
+int64_t dj_table_column_chars_size(void* tbl, int i)
+{
+  return ((cudf::table*)tbl)->get_column(i).chars_size();
 }
 
 int64_t dj_table_num_rows(void* tbl) { return ((cudf::table*)tbl)->num_rows(); }

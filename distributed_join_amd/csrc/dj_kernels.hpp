@@ -94,6 +94,20 @@ void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_tabl
                       int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
                       int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s);
 
+/* ----- strings-column kernels (dj_strings.hip; see its header comment for
+ * the reference helpers each replaces) ----- */
+void sizes_from_offsets(const int32_t* d_offsets, int64_t n, int32_t* d_sizes, hipStream_t s);
+size_t offsets_from_sizes_scratch_bytes(int64_t n);
+void offsets_from_sizes(const int32_t* d_sizes, int64_t n, int32_t* d_offsets, void* d_scratch,
+                        hipStream_t s);
+void gather_sizes(const int32_t* d_src_off, const int64_t* d_idx, int64_t n, int32_t* d_sizes,
+                  hipStream_t s);
+void gather_chars(const int32_t* d_src_off, const uint8_t* d_src_chars, const int64_t* d_idx,
+                  int64_t n, const int32_t* d_dst_off, uint8_t* d_dst_chars, hipStream_t s);
+void make_test_string_sizes(const int64_t* d_keys, int64_t n, int32_t* d_sizes, hipStream_t s);
+void fill_test_strings(const int64_t* d_keys, int64_t n, const int32_t* d_offsets,
+                       uint8_t* d_chars, hipStream_t s);
+
 /* ----- small utilities ----- */
 void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s);
 

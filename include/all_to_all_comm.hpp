@@ -22,6 +22,7 @@
 #include "dj_cudf_types.hpp"
 
 #include <cstdint>
+#include <memory>
 #include <vector>
 
 enum COMM_TAGS { placeholder_tag, exchange_size_tag };
@@ -150,4 +151,11 @@ class AllToAllCommunicator {
   std::vector<cudf::size_type> send_offsets;
   std::vector<int64_t> recv_offsets;
   std::vector<ColumnCompressionOptions> compression_options;
+  /* strings machinery (reference all_to_all_comm.hpp:349-360: per string
+   * column the char-offset boundaries per peer, plus device buffers of row
+   * sizes to send / received — sizes, not offsets, go on the wire). Opaque
+   * here; defined in dj_cpp_api.hip. */
+  struct StringsState;
+  std::shared_ptr<StringsState> strings;
+  friend struct AllToAllCommunicatorAccess;
 };

@@ -25,7 +25,7 @@ namespace cudf {
 
 using size_type = int32_t;
 
-enum class type_id : int32_t { EMPTY = 0, INT32, INT64, STRING };
+enum class type_id : int32_t { EMPTY = 0, INT8, INT32, INT64, STRING };
 
 struct data_type {
   data_type() = default;
@@ -41,6 +41,7 @@ struct data_type {
 inline constexpr size_type size_of(data_type t)
 {
   switch (t.id()) {
+    case type_id::INT8: return 1;
     case type_id::INT32: return 4;
     case type_id::INT64: return 8;
     default: return 0;  // STRING handled via children
@@ -58,6 +59,15 @@ class column_view {
     : _type(type), _size(size), _data(data)
   {
   }
+  /* STRING column: `data` = int32 offsets (size+1 entries), plus a chars
+   * byte buffer — mirroring cuDF's offsets/chars children, which the
+   * reference's strings path relies on (strings_column.cu:39-145).
+   * child(0) = offsets view, child(1) = chars view. */
+  column_view(data_type type, size_type size, const void* offsets, const void* chars,
+              int64_t chars_bytes)
+    : _type(type), _size(size), _data(offsets), _chars(chars), _chars_size(chars_bytes)
+  {
+  }
   data_type type() const { return _type; }
   size_type size() const { return _size; }
   template <typename T>
@@ -65,11 +75,25 @@ class column_view {
   {
     return static_cast<const T*>(_data);
   }
+  const void* chars() const { return _chars; }
+  int64_t chars_size() const { return _chars_size; }
+  column_view child(int i) const
+  {
+    if (i == 0) return column_view(data_type(type_id::INT32), _size + 1, _data);
+    return column_view(data_type(type_id::INT8), (size_type)_chars_size, _chars);
+  }
+  template <typename T>
+  const T* begin() const
+  {
+    return head<T>();
+  }
 
  private:
   data_type _type{};
   size_type _size{0};
   const void* _data{nullptr};
+  const void* _chars{nullptr};
+  int64_t _chars_size{0};
 };
 
 class mutable_column_view {
@@ -79,6 +103,11 @@ class mutable_column_view {
     : _type(type), _size(size), _data(data)
   {
   }
+  mutable_column_view(data_type type, size_type size, void* offsets, void* chars,
+                      int64_t chars_bytes)
+    : _type(type), _size(size), _data(offsets), _chars(chars), _chars_size(chars_bytes)
+  {
+  }
   data_type type() const { return _type; }
   size_type size() const { return _size; }
   template <typename T>
@@ -86,12 +115,19 @@ class mutable_column_view {
   {
     return static_cast<T*>(_data);
   }
-  operator column_view() const { return column_view(_type, _size, _data); }
+  void* chars() const { return _chars; }
+  int64_t chars_size() const { return _chars_size; }
+  operator column_view() const
+  {
+    return column_view(_type, _size, _data, _chars, _chars_size);
+  }
 
  private:
   data_type _type{};
   size_type _size{0};
   void* _data{nullptr};
+  void* _chars{nullptr};
+  int64_t _chars_size{0};
 };
 
 /* owning device column (HIP device memory; allocation in dj_cpp_api.hip) */
@@ -99,22 +135,33 @@ class column {
  public:
   column(data_type type, size_type size);            // allocates device memory
   column(data_type type, size_type size, void* adopt_device_ptr);
+  /* STRING column: allocates int32[size+1] offsets + chars_bytes chars */
+  column(size_type size, int64_t chars_bytes);
   column(const column&) = delete;
   column& operator=(const column&) = delete;
   column(column&& o) noexcept;
+  column& operator=(column&& o) noexcept;
   ~column();
 
   data_type type() const { return _type; }
   size_type size() const { return _size; }
   void* head() { return _data; }
   const void* head() const { return _data; }
-  column_view view() const { return column_view(_type, _size, _data); }
-  mutable_column_view mutable_view() { return mutable_column_view(_type, _size, _data); }
+  void* chars() { return _chars; }
+  const void* chars() const { return _chars; }
+  int64_t chars_size() const { return _chars_size; }
+  column_view view() const { return column_view(_type, _size, _data, _chars, _chars_size); }
+  mutable_column_view mutable_view()
+  {
+    return mutable_column_view(_type, _size, _data, _chars, _chars_size);
+  }
 
  private:
   data_type _type{};
   size_type _size{0};
   void* _data{nullptr};
+  void* _chars{nullptr};
+  int64_t _chars_size{0};
 };
 
 class table_view {
