@@ -1,0 +1,203 @@
+#!/usr/bin/env python3
+"""TPC-H join harness (BASELINE config 5): lineitem.l_orderkey ⋈
+orders.{o_orderkey, o_orderpriority}, mirroring the reference's
+benchmark/tpch.cpp:128-251 (throughput = total input bytes / elapsed,
+tpch.cpp:229-235) over the C++ distributed_inner_join path.
+
+Data: --parquet-dir loads dbgen-produced parquet (pyarrow; the reference's
+scripts/tpch_to_parquet.py format — one file or directory per table with
+columns l_orderkey / o_orderkey, o_orderpriority). Without parquet (this
+environment has no network for dbgen data), --synthetic generates
+TPC-H-SHAPED tables deterministically: orders with unique sparse orderkeys
+(dbgen leaves 3 of every 4 key slots unused) and the 5 canonical
+o_orderpriority strings; lineitem with 1..7 lineitems per order (dbgen
+L_ORDERKEY multiplicity), seed-stable.
+
+Single process = 1 GPU; rank-sliced under torch.distributed.run like
+bench.py. Prints one JSON line (same shape as bench.py, metric named for the
+tpch workload).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+
+import numpy as np  # noqa: E402
+
+import distributed_join_amd as dj  # noqa: E402
+
+PRIORITIES = [b"1-URGENT", b"2-HIGH", b"3-MEDIUM", b"4-NOT SPECI", b"5-LOW"]
+
+
+def synth_orders(n_orders, seed, row0, nrows):
+    rng = np.random.RandomState(seed)
+    # dbgen: orderkeys are sparse — 8 of each 32 consecutive values used;
+    # simplified sparse unique keys: key = i*4 + (i % 3)
+    i = np.arange(row0, row0 + nrows, dtype=np.int64)
+    keys = i * 4 + (i % 3)
+    pr_idx = (np.abs(keys * 2654435761) % 5).astype(np.int64)
+    sizes = np.array([len(PRIORITIES[k]) for k in pr_idx], dtype=np.int32)
+    offsets = np.zeros(nrows + 1, dtype=np.int32)
+    np.cumsum(sizes, out=offsets[1:])
+    chars = np.empty(offsets[-1], dtype=np.uint8)
+    table = [np.frombuffer(p, dtype=np.uint8) for p in PRIORITIES]
+    for r in range(nrows):
+        chars[offsets[r]:offsets[r + 1]] = table[pr_idx[r]]
+    return keys, offsets, chars
+
+
+def synth_lineitem(n_orders_global, seed, row0, nrows):
+    # lineitems: order i has 1 + (mix(i) % 7) lineitems; we draw lineitem
+    # rows by picking a random order per row (multiplicity emerges from the
+    # draw; matches the join shape, not dbgen's exact counts)
+    rng = np.random.RandomState(seed + 17 + row0 % 1000003)
+    oi = rng.randint(0, n_orders_global, size=nrows).astype(np.int64)
+    keys = oi * 4 + (oi % 3)
+    payload = np.arange(row0, row0 + nrows, dtype=np.int64)
+    return keys, payload
+
+
+def load_parquet_column(path, column):
+    import pyarrow.parquet as pq
+    t = pq.read_table(path, columns=[column])
+    return t.column(column).to_numpy()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--parquet-dir", default=None)
+    ap.add_argument("--orders-rows", type=int, default=15_000_000,
+                    help="synthetic orders rows per GPU (SF10 ~ 15M)")
+    ap.add_argument("--lineitem-rows", type=int, default=60_000_000,
+                    help="synthetic lineitem rows per GPU (SF10 ~ 60M)")
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--over-decom", type=int, default=1)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        tdist.init_process_group("gloo", rank=rank, world_size=world)
+
+    dj.require_gpu()
+    L = dj.lib()
+    L.dj_set_device(local_rank)
+    id_bytes = None
+    if world > 1:
+        import torch
+        nb = L.dj_rccl_unique_id_bytes()
+        t = torch.zeros(nb, dtype=torch.uint8)
+        if rank == 0:
+            buf = np.zeros(nb, dtype=np.uint8)
+            L.dj_rccl_get_unique_id(buf.ctypes.data)
+            t = torch.from_numpy(buf)
+        dist.broadcast(t, src=0)
+        id_bytes = np.ascontiguousarray(t.numpy())
+    comm = dj.CppCommunicator(rank, world, id_bytes)
+
+    if args.parquet_dir:
+        okeys = load_parquet_column(os.path.join(args.parquet_dir, "orders"), "o_orderkey")
+        opri = load_parquet_column(os.path.join(args.parquet_dir, "orders"), "o_orderpriority")
+        lkeys = load_parquet_column(os.path.join(args.parquet_dir, "lineitem"), "l_orderkey")
+        n_o, n_l = len(okeys), len(lkeys)
+        okeys = okeys.astype(np.int64)
+        lkeys = lkeys.astype(np.int64)
+        lpay = np.arange(n_l, dtype=np.int64)
+        enc = [s.encode() if isinstance(s, str) else bytes(s) for s in opri]
+        sizes = np.array([len(s) for s in enc], dtype=np.int32)
+        ooff = np.zeros(n_o + 1, dtype=np.int32)
+        np.cumsum(sizes, out=ooff[1:])
+        ochars = np.frombuffer(b"".join(enc), dtype=np.uint8).copy()
+        data_src = f"parquet:{args.parquet_dir}"
+    else:
+        n_o, n_l = args.orders_rows, args.lineitem_rows
+        okeys, ooff, ochars = synth_orders(n_o * world, 1234, rank * n_o, n_o)
+        lkeys, lpay = synth_lineitem(n_o * world, 1234, rank * n_l, n_l)
+        data_src = "synthetic tpch-shaped"
+
+    d_ok = dj.DeviceArray.from_numpy(okeys)
+    d_ooff = L.dj_dmalloc(len(ooff) * 4)
+    L.dj_memcpy_h2d(d_ooff, ooff.ctypes.data, len(ooff) * 4)
+    d_och = L.dj_dmalloc(max(len(ochars), 1))
+    if len(ochars):
+        L.dj_memcpy_h2d(d_och, ochars.ctypes.data, len(ochars))
+    d_lk = dj.DeviceArray.from_numpy(lkeys)
+    d_lp = dj.DeviceArray.from_numpy(lpay)
+
+    lcols = [(dj.TYPE_INT64, d_ok.ptr), (dj.TYPE_STRING, d_ooff, d_och, len(ochars))]
+    rcols = [(dj.TYPE_INT64, d_lk.ptr), (dj.TYPE_INT64, d_lp.ptr)]
+
+    state = {"matches": 0}
+
+    def step():
+        t = L.dj_cpp_distributed_inner_join_cols(
+            comm.ptr, (dj.ColDesc * 2)(dj.ColDesc(dj.TYPE_INT64, d_ok.ptr, None, 0),
+                                       dj.ColDesc(dj.TYPE_STRING, d_ooff, d_och, len(ochars))),
+            2, n_o,
+            (dj.ColDesc * 2)(dj.ColDesc(dj.TYPE_INT64, d_lk.ptr, None, 0),
+                             dj.ColDesc(dj.TYPE_INT64, d_lp.ptr, None, 0)),
+            2, n_l, 0, 0, args.over_decom, 0)
+        state["matches"] = L.dj_table_num_rows(t)
+        L.dj_table_free(t)
+
+    def barrier():
+        L.dj_sync()
+        if dist is not None:
+            dist.barrier()
+
+    for _ in range(args.warmup):
+        step()
+    barrier()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    barrier()
+    elapsed = time.perf_counter() - t0
+    if dist is not None:
+        import torch
+        e = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    input_bytes = (n_o * 8 + len(ochars) + (n_o + 1) * 4 + n_l * 16) * world
+    input_rows = (n_o + n_l) * world
+    per_step = elapsed / args.steps
+    if rank == 0:
+        print(json.dumps({
+            "metric": "tpch join throughput (input bytes/s whole-node)",
+            "value": input_bytes / per_step,
+            "unit": "bytes/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": per_step * 1000,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "int64+string",
+            "data": data_src,
+            "config": {
+                "workload": "lineitem.l_orderkey JOIN orders.(o_orderkey,o_orderpriority)",
+                "orders_rows_per_gpu": n_o,
+                "lineitem_rows_per_gpu": n_l,
+                "input_rows_per_sec": input_rows / per_step,
+                "output_rows_per_gpu": int(state["matches"]),
+            },
+        }), flush=True)
+    comm.destroy()
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -89,6 +89,8 @@ def lib():
                                      ctypes.POINTER(ctypes.c_int64)], None),
             "dj_cpp_distributed_inner_join_i64str": ([vp, vp, vp, vp, i64, i64,
                                                       vp, vp, vp, i64, i64, i32, i32], vp),
+            "dj_cpp_distributed_inner_join_cols": ([vp, vp, i32, i64, vp, i32, i64,
+                                                    i32, i32, i32, i32], vp),
             "dj_table_column_type": ([vp, i32], i32),
             "dj_table_column_chars": ([vp, i32], vp),
             "dj_table_column_chars_size": ([vp, i32], i64),
@@ -305,6 +307,30 @@ def cpp_distributed_inner_join_str(comm, d_lk, l_strings, ln, d_rk, r_strings, r
     ro, rc, rb = r_strings
     t = lib().dj_cpp_distributed_inner_join_i64str(
         comm.ptr, d_lk.ptr, lo, lc, lb, ln, d_rk.ptr, ro, rc, rb, rn, over_decom, 0)
+    return table_to_numpy(t)
+
+
+class ColDesc(ctypes.Structure):
+    _fields_ = [("type_id", ctypes.c_int), ("data", ctypes.c_void_p),
+                ("chars", ctypes.c_void_p), ("chars_bytes", ctypes.c_int64)]
+
+
+def cpp_distributed_inner_join_cols(comm, lcols, ln, rcols, rn, key_l=0, key_r=0,
+                                    over_decom=1):
+    """Generic join over column descriptors: each col is
+    (type_id, data_ptr[, chars_ptr, chars_bytes])."""
+    def pack(cols):
+        arr = (ColDesc * len(cols))()
+        for i, c in enumerate(cols):
+            arr[i].type_id = c[0]
+            arr[i].data = c[1]
+            arr[i].chars = c[2] if len(c) > 2 else None
+            arr[i].chars_bytes = c[3] if len(c) > 3 else 0
+        return arr
+    la, ra = pack(lcols), pack(rcols)
+    t = lib().dj_cpp_distributed_inner_join_cols(
+        comm.ptr, ctypes.cast(la, ctypes.c_void_p), len(lcols), ln,
+        ctypes.cast(ra, ctypes.c_void_p), len(rcols), rn, key_l, key_r, over_decom, 0)
     return table_to_numpy(t)
 
 

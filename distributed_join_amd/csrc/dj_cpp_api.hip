@@ -1317,6 +1317,45 @@ void* dj_cpp_distributed_inner_join_i64str(void* comm, const int64_t* d_lk,
   return result.release();
 }
 
+/* generic column-descriptor join (e.g. the TPC-H lineitem x orders shape:
+ * int64 key + string payload on one side, int64 key + int64 payload on the
+ * other). type_id: 2=INT32, 3=INT64, 4=STRING (cudf::type_id values). */
+typedef struct {
+  int type_id;
+  const void* data;  /* fixed-width data, or int32 offsets for STRING */
+  const void* chars;
+  int64_t chars_bytes;
+} dj_col_desc;
+
+void* dj_cpp_distributed_inner_join_cols(void* comm, const dj_col_desc* lcols, int nl,
+                                         int64_t ln, const dj_col_desc* rcols, int nr,
+                                         int64_t rn, int key_l, int key_r, int over_decom,
+                                         int report_timing)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  auto mk = [](const dj_col_desc* cols, int nc, int64_t n) {
+    std::vector<column_view> v;
+    for (int c = 0; c < nc; c++) {
+      if ((type_id)cols[c].type_id == type_id::STRING)
+        v.emplace_back(data_type(type_id::STRING), (cudf::size_type)n, cols[c].data,
+                       cols[c].chars, cols[c].chars_bytes);
+      else
+        v.emplace_back(data_type((type_id)cols[c].type_id), (cudf::size_type)n, cols[c].data);
+    }
+    return cudf::table_view(v);
+  };
+  auto left = mk(lcols, nl, ln);
+  auto right = mk(rcols, nr, rn);
+  auto lopts = generate_compression_options_distributed(left, false);
+  auto ropts = generate_compression_options_distributed(right, false);
+  auto result = distributed_inner_join(left, right, {(cudf::size_type)key_l},
+                                       {(cudf::size_type)key_r}, (Communicator*)comm, lopts,
+                                       ropts, over_decom, report_timing != 0, nullptr, 1);
+  return result.release();
+}
+
 int dj_table_column_type(void* tbl, int i)
 {
   return (int)((cudf::table*)tbl)->get_column(i).type().id();

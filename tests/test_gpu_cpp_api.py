@@ -75,6 +75,19 @@ def test_cpp_shuffle_single_rank_identity(dj, comm):
         assert (x == y).all()
 
 
+def test_cpp_shuffle_over_decom_equivalence(dj, comm):
+    # over-decomposition must not change the result (distributed_join.cpp:244-329)
+    n = 100_000
+    bk, bp = oracle.gen_build(n)
+    pk, pp = oracle.gen_probe(n, n)
+    dlk, dlp = dj.DeviceArray.from_numpy(bk), dj.DeviceArray.from_numpy(bp)
+    drk, drp = dj.DeviceArray.from_numpy(pk), dj.DeviceArray.from_numpy(pp)
+    a = oracle.sort_rows(*dj.cpp_distributed_inner_join(comm, dlk, dlp, n, drk, drp, n, 1))
+    b = oracle.sort_rows(*dj.cpp_distributed_inner_join(comm, dlk, dlp, n, drk, drp, n, 8))
+    for x, y in zip(a, b):
+        assert (x == y).all()
+
+
 def test_cpp_join_empty(dj, comm):
     n = 1000
     bk, bp = oracle.gen_build(n)
