@@ -494,14 +494,14 @@ void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_tabl
 
 constexpr int BUCKET_BLOCKS = kBucketBlocks;  // chunking blocks for pass A (2/CU); scratch sizing uses the same constant
 constexpr int BUCKET_THREADS = 1024;
-constexpr int JOIN_LDS_SLOTS = 8192;        // 128 KiB of longlong2 pairs -> 1 block/CU
+constexpr int JOIN_LDS_SLOTS = 4096;        // 64 KiB of longlong2 pairs -> 2 blocks/CU
 constexpr int SUB_BUCKETS = 256;            // pass-B fanout (fixed)
 
 int bucket_count_for(int64_t ln, int64_t rn)
 {
   int64_t maxn = ln > rn ? ln : rn;
   int64_t B = 256;
-  while (B < 262144 && maxn / B > 3200) B <<= 1;
+  while (B < 262144 && maxn / B > 1600) B <<= 1;
   return (int)B;
 }
 
