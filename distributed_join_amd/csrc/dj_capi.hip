@@ -302,13 +302,22 @@ int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
   return (int64_t)bytes;
 }
 
-void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
-                          const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
-                          int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
-                          int64_t* d_out3, int64_t cap, int64_t* d_counter, int* d_error,
-                          void* d_scratch)
+/* enqueue-only core: all kernels stream-ordered, no host syncs. The skew
+ * any-overflow flag is written to d_any_overflow (device int) for the caller
+ * to check after its own sync; oversized buckets are SKIPPED by the fused
+ * kernel and must be re-joined by the caller (dj_bucket_local_join does this
+ * via the global-table path; the pipelined C++ orchestration redoes the
+ * whole batch). */
+void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                                  const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                                  int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                                  int64_t* d_out3, int64_t cap, int64_t* d_counter,
+                                  int* d_error, int* d_any_overflow, void* d_scratch)
 {
-  if (ln == 0 || rn == 0) return;  // empty side => empty (distributed_join.cpp:76-83)
+  if (ln == 0 || rn == 0) {
+    DJ_HIP_CALL(hipMemsetAsync(d_any_overflow, 0, sizeof(int), stream()));
+    return;  // empty side => empty (distributed_join.cpp:76-83)
+  }
   const int B = dj::bucket_count_for(ln, rn);
   BucketScratch s = carve_bucket_scratch(d_scratch, ln, rn, B);
   hipStream_t st = stream();
@@ -319,13 +328,27 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
     dj::bucket_partition2(d_rk, d_rp, rn, B, s.tmp_pairs, s.counts, s.totals, s.segoff,
                           s.roff, s.rpairs, st);
   }
-  DJ_HIP_CALL(hipMemsetAsync(s.any_overflow, 0, sizeof(int), st));
+  DJ_HIP_CALL(hipMemsetAsync(d_any_overflow, 0, sizeof(int), st));
   DJ_HIP_CALL(hipMemsetAsync(s.flags, 0, (size_t)B * 4, st));
   {
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
     dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, d_out0, d_out1, d_out2,
-                 d_out3, cap, d_counter, s.flags, s.any_overflow, d_error, st);
+                 d_out3, cap, d_counter, s.flags, d_any_overflow, d_error, st);
   }
+}
+
+void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                          const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                          int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                          int64_t* d_out3, int64_t cap, int64_t* d_counter, int* d_error,
+                          void* d_scratch)
+{
+  if (ln == 0 || rn == 0) return;  // empty side => empty (distributed_join.cpp:76-83)
+  const int B = dj::bucket_count_for(ln, rn);
+  BucketScratch s = carve_bucket_scratch(d_scratch, ln, rn, B);
+  hipStream_t st = stream();
+  dj_bucket_local_join_enqueue(d_lk, d_lp, ln, d_rk, d_rp, rn, d_out0, d_out1, d_out2,
+                               d_out3, cap, d_counter, d_error, s.any_overflow, d_scratch);
   /* skew fallback: buckets whose build side exceeded the LDS row cap */
   int any = 0;
   DJ_HIP_CALL(hipMemcpyAsync(&any, s.any_overflow, sizeof(int), hipMemcpyDeviceToHost, st));

@@ -691,6 +691,9 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
   /* receiver-side: rebuild string offsets from the received sizes by scan
    * with offset[0]=0 (strings_column.cu:111-131) */
   if (strings) {
+    /* enqueued on the compute stream without a host sync so the offsets
+     * rebuild pipelines with the local join under over-decomposition;
+     * compute-stream consumers are ordered behind it */
     hipStream_t st = dj_rt_stream();
     const int64_t n_recv = recv_offsets.back();
     for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
@@ -698,8 +701,8 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
       DBuf scr(dj::offsets_from_sizes_scratch_bytes(n_recv));
       dj::offsets_from_sizes((const int32_t*)strings->sizes_received[c].p, n_recv,
                              communicated_table.column(c).head<int32_t>(), scr.p, st);
+      DJ_HIP_CALL(hipStreamSynchronize(st));  // scr freed on return; scan must be done
     }
-    DJ_HIP_CALL(hipStreamSynchronize(st));
   }
 }
 
@@ -1010,26 +1013,176 @@ std::unique_ptr<cudf::table> distributed_inner_join(
   PartitionedTable rpart =
     partition_table(right, right_on[0], nparts, DJ_HASH_MURMUR3, DJ_SEED_INTRA);
 
-  std::vector<std::unique_ptr<cudf::table>> batch_results;
+  /* --- pre-phase: size exchanges + every allocation (pool allocs sync the
+   * compute stream, so none happen inside the pipeline) --- */
+  const bool fast2 = left.num_columns() == 2 && right.num_columns() == 2 &&
+                     left_on[0] == 0 && right_on[0] == 0 &&
+                     left.column(0).type().id() == cudf::type_id::INT64 &&
+                     left.column(1).type().id() == cudf::type_id::INT64 &&
+                     right.column(0).type().id() == cudf::type_id::INT64 &&
+                     right.column(1).type().id() == cudf::type_id::INT64;
+  struct Batch {
+    std::unique_ptr<AllToAllCommunicator> latoa, ratoa;
+    std::unique_ptr<cudf::table> lrecv, rrecv;
+    DBuf lkw, rkw, liota, riota;  // general-path key widen / row-index payloads
+    DBuf o0, o1, o2, o3;          // engine outputs
+    DBuf meta;                    // counter(8) + error(4) + any_overflow(4)
+    int64_t ln{0}, rn{0}, cap{0};
+  };
+  std::vector<Batch> batches(over_decom_factor);
+  int64_t max_ln = 1, max_rn = 1;
   for (int b = 0; b < over_decom_factor; b++) {
-    /* batch b = partitions [b*G, (b+1)*G): a contiguous offset slice
-     * (distributed_join.cpp:247-266) */
+    Batch& bt = batches[b];
     std::vector<cudf::size_type> lslice(lpart.offsets.begin() + b * G,
                                         lpart.offsets.begin() + b * G + G + 1);
     std::vector<cudf::size_type> rslice(rpart.offsets.begin() + b * G,
                                         rpart.offsets.begin() + b * G + G + 1);
-    AllToAllCommunicator latoa(lpart.tbl->view(), lslice, group, communicator,
-                               left_compression_options, true);
-    AllToAllCommunicator ratoa(rpart.tbl->view(), rslice, group, communicator,
-                               right_compression_options, true);
-    auto lrecv = latoa.allocate_communicated_table();
-    auto rrecv = ratoa.allocate_communicated_table();
-    latoa.launch_communication(lrecv->mutable_view(), report_timing,
-                               preallocated_pinned_buffer);
-    ratoa.launch_communication(rrecv->mutable_view(), report_timing,
-                               preallocated_pinned_buffer);
-    batch_results.push_back(
-      local_inner_join(lrecv->view(), rrecv->view(), left_on[0], right_on[0]));
+    bt.latoa = std::make_unique<AllToAllCommunicator>(lpart.tbl->view(), lslice, group,
+                                                      communicator, left_compression_options,
+                                                      true);
+    bt.ratoa = std::make_unique<AllToAllCommunicator>(rpart.tbl->view(), rslice, group,
+                                                      communicator, right_compression_options,
+                                                      true);
+    bt.lrecv = bt.latoa->allocate_communicated_table();
+    bt.rrecv = bt.ratoa->allocate_communicated_table();
+    bt.ln = bt.lrecv->num_rows();
+    bt.rn = bt.rrecv->num_rows();
+    max_ln = std::max(max_ln, bt.ln);
+    max_rn = std::max(max_rn, bt.rn);
+    bt.cap = std::max<int64_t>(bt.rn + (bt.rn >> 3), 1024);
+    bt.o0 = DBuf((size_t)bt.cap * 8);
+    bt.o1 = DBuf((size_t)bt.cap * 8);
+    bt.o2 = DBuf((size_t)bt.cap * 8);
+    bt.o3 = DBuf((size_t)bt.cap * 8);
+    bt.meta = DBuf(16);
+    if (!fast2) {
+      if (bt.ln && left.column(left_on[0]).type().id() == cudf::type_id::INT32)
+        bt.lkw = DBuf((size_t)bt.ln * 8);
+      if (bt.rn && right.column(right_on[0]).type().id() == cudf::type_id::INT32)
+        bt.rkw = DBuf((size_t)bt.rn * 8);
+      bt.liota = DBuf((size_t)std::max<int64_t>(bt.ln, 1) * 8);
+      bt.riota = DBuf((size_t)std::max<int64_t>(bt.rn, 1) * 8);
+    }
+  }
+  DBuf scratch((size_t)dj_bucket_join_scratch_bytes(max_ln, max_rn));
+  hipStream_t st = dj_rt_stream();
+
+  /* --- pipeline: comm of batch b on the comm stream overlaps the join of
+   * batch b-1 on the compute stream (the host blocks only in
+   * launch_communication's stop; join kernels are enqueued without syncs) —
+   * replaces the reference's join thread + atomic-flag busy wait
+   * (distributed_join.cpp:283-329) with stream ordering --- */
+  for (int b = 0; b < over_decom_factor; b++) {
+    Batch& bt = batches[b];
+    bt.latoa->launch_communication(bt.lrecv->mutable_view(), report_timing,
+                                   preallocated_pinned_buffer);
+    bt.ratoa->launch_communication(bt.rrecv->mutable_view(), report_timing,
+                                   preallocated_pinned_buffer);
+    if (bt.ln == 0 || bt.rn == 0) {
+      DJ_HIP_CALL(hipMemsetAsync(bt.meta.p, 0, 16, st));
+      continue;
+    }
+    const int64_t* lk;
+    const int64_t* rk;
+    const int64_t* lp;
+    const int64_t* rp;
+    if (fast2) {
+      lk = (const int64_t*)bt.lrecv->get_column(0).head();
+      lp = (const int64_t*)bt.lrecv->get_column(1).head();
+      rk = (const int64_t*)bt.rrecv->get_column(0).head();
+      rp = (const int64_t*)bt.rrecv->get_column(1).head();
+    } else {
+      auto widen_or_use = [&](cudf::column_view col, DBuf& w) -> const int64_t* {
+        if (col.type().id() == cudf::type_id::INT64) return col.head<int64_t>();
+        hipLaunchKernelGGL(widen_i32_kernel, dim3(grid_for_n(col.size())), dim3(kBlock), 0, st,
+                           col.head<int32_t>(), (int64_t)col.size(), w.i64());
+        return w.i64();
+      };
+      lk = widen_or_use(bt.lrecv->view().column(left_on[0]), bt.lkw);
+      rk = widen_or_use(bt.rrecv->view().column(right_on[0]), bt.rkw);
+      hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(bt.ln)), dim3(kBlock), 0, st,
+                         bt.liota.i64(), bt.ln);
+      hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(bt.rn)), dim3(kBlock), 0, st,
+                         bt.riota.i64(), bt.rn);
+      lp = bt.liota.i64();
+      rp = bt.riota.i64();
+    }
+    DJ_HIP_CALL(hipMemsetAsync(bt.meta.p, 0, 16, st));
+    dj_bucket_local_join_enqueue(lk, lp, bt.ln, rk, rp, bt.rn, bt.o0.i64(), bt.o1.i64(),
+                                 bt.o2.i64(), bt.o3.i64(), bt.cap, bt.meta.i64(),
+                                 (int*)((char*)bt.meta.p + 8),
+                                 (int*)((char*)bt.meta.p + 12), scratch.p);
+  }
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+
+  /* --- finalize: counts, rare redo (skew overflow / cap exceeded), column
+   * assembly, concat --- */
+  std::vector<std::unique_ptr<cudf::table>> batch_results;
+  for (int b = 0; b < over_decom_factor; b++) {
+    Batch& bt = batches[b];
+    struct {
+      int64_t count;
+      int error;
+      int any_overflow;
+    } meta;
+    DJ_HIP_CALL(hipMemcpy(&meta, bt.meta.p, 16, hipMemcpyDeviceToHost));
+    DJ_CHECK_ERROR(meta.error == 0,
+                   "join build: key equal to the empty sentinel (-1) is unsupported");
+    if (bt.ln == 0 || bt.rn == 0 || meta.count == 0) {
+      /* empty batch result with the output schema */
+      std::vector<std::unique_ptr<cudf::column>> cols;
+      auto empty_col2 = [&](cudf::column_view v) {
+        if (v.type().id() == cudf::type_id::STRING)
+          return std::make_unique<cudf::column>((cudf::size_type)0, (int64_t)0);
+        return std::make_unique<cudf::column>(v.type(), (cudf::size_type)0);
+      };
+      for (cudf::size_type c = 0; c < left.num_columns(); c++)
+        cols.push_back(empty_col2(left.column(c)));
+      for (cudf::size_type c = 0; c < right.num_columns(); c++)
+        cols.push_back(empty_col2(right.column(c)));
+      batch_results.push_back(std::make_unique<cudf::table>(std::move(cols)));
+      continue;
+    }
+    if (meta.any_overflow || meta.count > bt.cap) {
+      /* rare path: redo this batch synchronously (handles skewed buckets
+       * via the global-table fallback and exact capacity) */
+      batch_results.push_back(
+        local_inner_join(bt.lrecv->view(), bt.rrecv->view(), left_on[0], right_on[0]));
+      continue;
+    }
+    const int64_t nout = meta.count;
+    std::vector<std::unique_ptr<cudf::column>> cols;
+    if (fast2) {
+      auto adopt = [&](DBuf& d) {
+        auto col = std::make_unique<cudf::column>(cudf::data_type(cudf::type_id::INT64),
+                                                  (cudf::size_type)nout, d.p);
+        d.p = nullptr;
+        return col;
+      };
+      cols.push_back(adopt(bt.o0));
+      cols.push_back(adopt(bt.o1));
+      cols.push_back(adopt(bt.o2));
+      cols.push_back(adopt(bt.o3));
+    } else {
+      auto gather_col = [&](cudf::column_view src, DBuf& idx) {
+        if (src.type().id() == cudf::type_id::STRING)
+          return gather_string_column(src, idx.i64(), nout);
+        auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
+        if (src.type().id() == cudf::type_id::INT64)
+          hipLaunchKernelGGL(gather_i64_kernel, dim3(grid_for_n(nout)), dim3(kBlock), 0, st,
+                             src.head<int64_t>(), idx.i64(), nout, (int64_t*)col->head());
+        else
+          hipLaunchKernelGGL(gather_i32_kernel, dim3(grid_for_n(nout)), dim3(kBlock), 0, st,
+                             src.head<int32_t>(), idx.i64(), nout, (int32_t*)col->head());
+        return col;
+      };
+      for (cudf::size_type c = 0; c < left.num_columns(); c++)
+        cols.push_back(gather_col(bt.lrecv->view().column(c), bt.o1));
+      for (cudf::size_type c = 0; c < right.num_columns(); c++)
+        cols.push_back(gather_col(bt.rrecv->view().column(c), bt.o3));
+      DJ_HIP_CALL(hipStreamSynchronize(st));
+    }
+    batch_results.push_back(std::make_unique<cudf::table>(std::move(cols)));
   }
   return concat_tables(batch_results);
 }
