@@ -68,6 +68,7 @@ void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s)
 {
   if (n <= 0) return;
   hipLaunchKernelGGL(fill_i64_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, s, d_dst, value, n);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 /* ------------------------------------------------------------- generator */
@@ -105,6 +106,7 @@ void generate_build(int64_t* d_keys, int64_t* d_pay, int64_t n_global, int64_t r
   if (nrows <= 0) return;
   hipLaunchKernelGGL(gen_build_kernel, dim3(grid_for(nrows)), dim3(BLOCK), 0, s, d_keys, d_pay,
                      n_global, rand_max, seed, (int)uniq, row0, nrows);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void generate_probe(int64_t* d_keys, int64_t* d_pay, int64_t build_n_global, int64_t rand_max,
@@ -113,6 +115,7 @@ void generate_probe(int64_t* d_keys, int64_t* d_pay, int64_t build_n_global, int
   if (nrows <= 0) return;
   hipLaunchKernelGGL(gen_probe_kernel, dim3(grid_for(nrows)), dim3(BLOCK), 0, s, d_keys, d_pay,
                      build_n_global, rand_max, selectivity, seed, row0, nrows);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 /* ------------------------------------------------------- stable partition */
@@ -275,6 +278,7 @@ void partition_count(const int64_t* d_keys, int64_t n, int nparts, int hash_fn,
   PartGeom g = part_geom(n);
   hipLaunchKernelGGL(part_count_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, n, nparts,
                      hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void partition_scan(int64_t n, int nparts, void* d_scratch, int64_t* d_offsets, hipStream_t s)
@@ -288,7 +292,9 @@ void partition_scan(int64_t n, int nparts, void* d_scratch, int64_t* d_offsets, 
   int64_t* totals = wave_counts + g.nwaves * nparts;
   hipLaunchKernelGGL(part_scan_kernel, dim3(nparts), dim3(BLOCK), 0, s, wave_counts, g.nwaves,
                      nparts, totals);
+  DJ_HIP_CALL(hipGetLastError());
   hipLaunchKernelGGL(part_offsets_kernel, dim3(1), dim3(64), 0, s, totals, nparts, d_offsets);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void partition_scatter(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
@@ -300,6 +306,7 @@ void partition_scatter(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
   hipLaunchKernelGGL(part_scatter_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, d_pay, n,
                      nparts, hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch,
                      d_offsets, d_out_keys, d_out_pay);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
@@ -364,6 +371,7 @@ void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d
   hipLaunchKernelGGL(join_build_kernel<false>, dim3(grid_for(ln)), dim3(BLOCK), 0, s, d_lk,
                      d_lp, (const longlong2*)nullptr, ln, (longlong2*)d_table,
                      (uint64_t)(nslots - 1), d_error);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int64_t nslots,
@@ -373,6 +381,7 @@ void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int
   hipLaunchKernelGGL(join_build_kernel<true>, dim3(grid_for(ln)), dim3(BLOCK), 0, s,
                      (const int64_t*)nullptr, (const int64_t*)nullptr, d_rows, ln,
                      (longlong2*)d_table, (uint64_t)(nslots - 1), d_error);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 /* Probe with wave-aggregated output append: matches are emitted via one
@@ -452,6 +461,7 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
                      d_rp, (const longlong2*)nullptr, rn, (const longlong2*)d_table,
                      (uint64_t)(nslots - 1), d_out0, d_out1, d_out2, d_out3, cap,
                      (unsigned long long*)d_counter);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_table,
@@ -463,6 +473,7 @@ void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_tabl
                      (const int64_t*)nullptr, (const int64_t*)nullptr, d_rows, rn,
                      (const longlong2*)d_table, (uint64_t)(nslots - 1), d_out0, d_out1,
                      d_out2, d_out3, cap, (unsigned long long*)d_counter);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 /* ----------------------------------------------- bucketed LDS join ------ */
@@ -768,22 +779,29 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
   DJ_CHECK_ERROR(PA >= 1 && PA <= 1024, "bucket_partition: B out of range");
   if (PA == 1) {
     hipLaunchKernelGGL(set_segoff1_kernel, dim3(1), dim3(1), 0, s, d_segoff, n);
+    DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_subpart_kernel<true>, dim3(PA), dim3(BUCKET_THREADS), 0, s,
                        (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B, d_out_pairs,
                        d_offsets);
+    DJ_HIP_CALL(hipGetLastError());
   } else {
     size_t lds = (size_t)PA * sizeof(uint32_t);
     hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
                        d_keys, n, PA, d_counts);
+    DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_scanA_kernel, dim3(PA), dim3(BUCKET_BLOCKS), 0, s, d_counts, PA,
                        d_totals);
+    DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, PA,
                        d_segoff);
+    DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
                        d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmp_pairs);
+    DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_subpart_kernel<false>, dim3(PA), dim3(BUCKET_THREADS), 0, s,
                        d_tmp_pairs, (const int64_t*)nullptr, (const int64_t*)nullptr, d_segoff,
                        B, d_out_pairs, d_offsets);
+    DJ_HIP_CALL(hipGetLastError());
   }
 }
 
@@ -798,6 +816,7 @@ void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* 
                      d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
                      (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
                      d_error);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 }  // namespace dj

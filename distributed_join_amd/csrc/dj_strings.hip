@@ -53,6 +53,7 @@ void sizes_from_offsets(const int32_t* d_offsets, int64_t n, int32_t* d_sizes, h
   if (n <= 0) return;
   hipLaunchKernelGGL(sizes_from_offsets_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s, d_offsets,
                      n, d_sizes);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 /* ---- exclusive scan of int32 sizes -> int32 offsets[n+1], offsets[0]=0 ---- */
@@ -132,10 +133,13 @@ void offsets_from_sizes(const int32_t* d_sizes, int64_t n, int32_t* d_offsets, v
   int64_t* partials = (int64_t*)d_scratch;
   hipLaunchKernelGGL(scan_partials_kernel, dim3(sgrid(nchunks)), dim3(SBLOCK), 0, s, d_sizes, n,
                      nchunks, partials);
+  DJ_HIP_CALL(hipGetLastError());
   hipLaunchKernelGGL(scan_partials_exclusive_kernel, dim3(1), dim3(1024), 0, s, partials,
                      nchunks);
+  DJ_HIP_CALL(hipGetLastError());
   hipLaunchKernelGGL(scan_finalize_kernel, dim3(sgrid(nchunks)), dim3(SBLOCK), 0, s, d_sizes, n,
                      nchunks, partials, d_offsets);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 /* ---- permutation gathers ---- */
@@ -158,6 +162,7 @@ void gather_sizes(const int32_t* d_src_off, const int64_t* d_idx, int64_t n, int
   if (n <= 0) return;
   hipLaunchKernelGGL(gather_sizes_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s, d_src_off, d_idx,
                      n, d_sizes);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 __global__ void gather_chars_kernel(const int32_t* __restrict__ src_off,
@@ -182,6 +187,7 @@ void gather_chars(const int32_t* d_src_off, const uint8_t* d_src_chars, const in
   if (n <= 0) return;
   hipLaunchKernelGGL(gather_chars_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s, d_src_off,
                      d_src_chars, d_idx, n, d_dst_off, d_dst_chars);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 /* ---- deterministic test/bench string payload (string_payload.cu:50-94) ---- */
@@ -212,6 +218,7 @@ void make_test_string_sizes(const int64_t* d_keys, int64_t n, int32_t* d_sizes, 
   if (n <= 0) return;
   hipLaunchKernelGGL(test_string_sizes_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s, d_keys, n,
                      d_sizes);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void fill_test_strings(const int64_t* d_keys, int64_t n, const int32_t* d_offsets,
@@ -220,6 +227,7 @@ void fill_test_strings(const int64_t* d_keys, int64_t n, const int32_t* d_offset
   if (n <= 0) return;
   hipLaunchKernelGGL(fill_test_strings_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s, d_keys, n,
                      d_offsets, d_chars);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 }  // namespace dj
