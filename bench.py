@@ -73,7 +73,7 @@ def main():
     ap.add_argument("--rows", type=int, default=100_000_000,
                     help="rows per GPU per table (build and probe)")
     ap.add_argument("--over-decom", type=int, default=1)
-    ap.add_argument("--cpu-baseline-rows", type=int, default=20_000_000)
+    ap.add_argument("--cpu-baseline-rows", type=int, default=100_000_000)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
