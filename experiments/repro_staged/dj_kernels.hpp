@@ -1,0 +1,114 @@
+/*
+ * dj_kernels.hpp — host-side launchers for the gfx950 HIP kernels of the
+ * distributed repartitioned hash join hot path.
+ *
+ * These replace the third-party cuDF 0.19 kernels the reference calls on its
+ * hot path (SURVEY.md §2 third-party kernel table):
+ *   - hash_partition  <- cudf::hash_partition (distributed_join.cpp:213-225,
+ *                        shuffle_on.cpp:59-60)
+ *   - build/probe     <- cudf::inner_join (distributed_join.cpp:79)
+ *   - generate_*      <- generate_dataset.cuh:40-260 (restated deterministic)
+ *
+ * All launchers are stream-ordered on the given hipStream_t and operate on
+ * raw device pointers (columnar int64 arrays). No torch types.
+ */
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace dj {
+
+/* int64 value marking an empty hash-table slot; build rows with this key are
+ * rejected (loud error), see dj_kernels.hip. Set via 0xFF memset. */
+constexpr int64_t kEmptyKey = -1;
+
+constexpr int kMaxPartitions = 64;  // nparts = world_size x over_decom <= 64
+
+/* ----- synthetic inputs (deterministic, spec in dj_rng.h) ----- */
+void generate_build(int64_t* d_keys, int64_t* d_pay, int64_t n_global, int64_t rand_max,
+                    uint64_t seed, bool uniq, int64_t row0, int64_t nrows, hipStream_t s);
+void generate_probe(int64_t* d_keys, int64_t* d_pay, int64_t build_n_global, int64_t rand_max,
+                    double selectivity, uint64_t seed, int64_t row0, int64_t nrows, hipStream_t s);
+
+/* ----- stable hash partition (histogram + scan + wave-ballot scatter) ----- */
+size_t hash_partition_scratch_bytes(int64_t n, int nparts);
+/* sub-steps, so the C ABI can time each kernel separately */
+void partition_count(const int64_t* d_keys, int64_t n, int nparts, int hash_fn,
+                     uint32_t hash_seed, void* d_scratch, hipStream_t s);
+void partition_scan(int64_t n, int nparts, void* d_scratch, int64_t* d_offsets, hipStream_t s);
+void partition_scatter(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                       int hash_fn, uint32_t hash_seed, const int64_t* d_offsets,
+                       void* d_scratch, int64_t* d_out_keys, int64_t* d_out_pay,
+                       hipStream_t s);
+/* d_offsets: device array of nparts+1 int64 partition offsets (offsets[0]=0).
+ * Stable: rows keep input order inside each partition. nparts <= 64. */
+void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                    int hash_fn, uint32_t hash_seed, int64_t* d_out_keys, int64_t* d_out_pay,
+                    int64_t* d_offsets, void* d_scratch, hipStream_t s);
+
+/* ----- local inner join: open-addressing build + probe-append ----- */
+/* Number of table slots for ln build rows (power of two, <=50% fill).
+ * The table buffer holds nslots interleaved 16 B {key,val} pairs
+ * (2*nslots int64). */
+int64_t join_table_slots(int64_t ln);
+/* Initialize table pairs to kEmptyKey (async memset over 16*nslots bytes). */
+void join_table_init(int64_t* d_table, int64_t nslots, hipStream_t s);
+/* Insert build rows. d_error (device int32) set to 1 if any key==kEmptyKey. */
+void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d_table,
+                int64_t nslots, int* d_error, hipStream_t s);
+/* Probe rows; append matches (lkey, lpay, rkey, rpay) to the 4 output
+ * columns at positions drawn from d_counter (device int64, caller-zeroed),
+ * one wave-aggregated atomic per emit round. Rows beyond `cap` are counted
+ * but not written (caller re-runs bigger). */
+void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int64_t* d_table,
+                int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s);
+
+/* ----- bucketed LDS join (the product local-join path; see dj_kernels.hip
+ * "bucketed LDS join" comment block) ----- */
+constexpr int kBucketBlocks = 512;
+constexpr int kSubBuckets = 256;
+constexpr int kJoinBucketRowCap = 3072;  // 75% of the 4096-slot LDS table
+int bucket_count_for(int64_t ln, int64_t rn);
+/* Two-level non-stable partition into B buckets (B = PA*256, PA<=1024) of
+ * interleaved 16 B {key,payload} pairs. d_tmp_pairs: longlong2[n] pass-A
+ * staging; d_counts: u32[kBucketBlocks*PA]; d_totals: u32[PA]; d_segoff:
+ * int64[PA+1]; d_offsets: int64[B+1]. */
+void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
+                       longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
+                       int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
+                       hipStream_t s);
+/* Fused per-bucket LDS build+probe over bucketed pair tables. Buckets whose
+ * build side exceeds kJoinBucketRowCap set overflow_flags[b]/any_overflow
+ * and are skipped (host runs the global-table path on them). */
+void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
+              const int64_t* d_roff, int B, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+              int64_t* d_out3, int64_t cap, int64_t* d_counter, uint32_t* d_overflow_flags,
+              int* d_any_overflow, int* d_error, hipStream_t s);
+/* Global-table build/probe over interleaved pair inputs (skew fallback). */
+void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int64_t nslots,
+                      int* d_error, hipStream_t s);
+void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_table,
+                      int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                      int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s);
+
+/* ----- strings-column kernels (dj_strings.hip; see its header comment for
+ * the reference helpers each replaces) ----- */
+void sizes_from_offsets(const int32_t* d_offsets, int64_t n, int32_t* d_sizes, hipStream_t s);
+size_t offsets_from_sizes_scratch_bytes(int64_t n);
+void offsets_from_sizes(const int32_t* d_sizes, int64_t n, int32_t* d_offsets, void* d_scratch,
+                        hipStream_t s);
+void gather_sizes(const int32_t* d_src_off, const int64_t* d_idx, int64_t n, int32_t* d_sizes,
+                  hipStream_t s);
+void gather_chars(const int32_t* d_src_off, const uint8_t* d_src_chars, const int64_t* d_idx,
+                  int64_t n, const int32_t* d_dst_off, uint8_t* d_dst_chars, hipStream_t s);
+void make_test_string_sizes(const int64_t* d_keys, int64_t n, int32_t* d_sizes, hipStream_t s);
+void fill_test_strings(const int64_t* d_keys, int64_t n, const int32_t* d_offsets,
+                       uint8_t* d_chars, hipStream_t s);
+
+/* ----- small utilities ----- */
+void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s);
+
+}  // namespace dj

@@ -1,0 +1,877 @@
+/*
+ * dj_kernels.hip — hand-written CDNA4 (gfx950) kernels for the distributed
+ * repartitioned hash-join hot path. MI355X-first design, not a CUDA port:
+ * 64-wide wavefront ballot/prefix-sum scatter, coalesced int64 loads,
+ * grid-stride launches sized for 256 CUs / 8 XCDs.
+ *
+ * Replaces (SURVEY.md §2 third-party kernel table):
+ *  - cudf::hash_partition  (reference calls: distributed_join.cpp:213-225,
+ *    shuffle_on.cpp:59-60): stable partition = per-wave histogram pass +
+ *    device scan + wave-ballot rank scatter (no LDS atomics, no sort).
+ *  - cudf::inner_join      (reference call: distributed_join.cpp:79):
+ *    open-addressing (linear probing, <=50%% fill, power-of-two slots)
+ *    atomicCAS build + probe with single-pass wave-aggregated append
+ *    (replaces cuDF's count+gather two-pass; row order is unspecified by the
+ *    API — reference tests sort before comparing,
+ *    compare_against_single_gpu.cu:167-174).
+ *  - generate_dataset.cuh:40-260: deterministic counter-based restatement
+ *    (spec in dj_rng.h) — bit-identical to the CPU oracle on any device.
+ */
+#include "dj_error.hpp"
+#include "dj_hash.h"
+#include "dj_kernels.hpp"
+#include "dj_rng.h"
+
+#include <hip/hip_runtime.h>
+
+namespace dj {
+
+constexpr int BLOCK = 256;
+constexpr int WAVE = 64;
+constexpr int WPB = BLOCK / WAVE;  // waves per block
+
+/* Non-temporal loads for single-use streams: partition/scatter inputs are
+ * read once and must not evict the partially-written output lines from the
+ * per-XCD L2 (PMC measured 1.24-1.49x write amplification on the scatter
+ * passes from exactly that eviction). */
+template <typename T>
+__device__ __forceinline__ T nt_load(const T* p)
+{
+  return __builtin_nontemporal_load(p);
+}
+__device__ __forceinline__ longlong2 nt_load2(const longlong2* p)
+{
+  longlong2 v;
+  v.x = nt_load(&p->x);
+  v.y = nt_load(&p->y);
+  return v;
+}
+
+/* ------------------------------------------------------------------ misc */
+
+__global__ void fill_i64_kernel(int64_t* dst, int64_t value, int64_t n)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = value;
+}
+
+static int grid_for(int64_t n)
+{
+  int64_t blocks = (n + BLOCK - 1) / BLOCK;
+  if (blocks > 2048) blocks = 2048;  // 256 CU x 8 blocks, grid-stride the rest
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s)
+{
+  if (n <= 0) return;
+  hipLaunchKernelGGL(fill_i64_kernel, dim3(grid_for(n)), dim3(BLOCK), 0, s, d_dst, value, n);
+}
+
+/* ------------------------------------------------------------- generator */
+
+__global__ void gen_build_kernel(int64_t* keys, int64_t* pay, int64_t n_global,
+                                 int64_t rand_max, uint64_t seed, int uniq, int64_t row0,
+                                 int64_t nrows)
+{
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; t < nrows; t += stride) {
+    int64_t i = row0 + t;
+    keys[t] = uniq ? dj_build_key((uint64_t)i, (uint64_t)n_global, rand_max, seed)
+                   : dj_build_key_nonuniq((uint64_t)i, rand_max, seed);
+    if (pay) pay[t] = i;
+  }
+}
+
+__global__ void gen_probe_kernel(int64_t* keys, int64_t* pay, int64_t build_n_global,
+                                 int64_t rand_max, double selectivity, uint64_t seed,
+                                 int64_t row0, int64_t nrows)
+{
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; t < nrows; t += stride) {
+    int64_t j = row0 + t;
+    keys[t] = dj_probe_key((uint64_t)j, (uint64_t)build_n_global, rand_max, selectivity, seed);
+    if (pay) pay[t] = j;
+  }
+}
+
+void generate_build(int64_t* d_keys, int64_t* d_pay, int64_t n_global, int64_t rand_max,
+                    uint64_t seed, bool uniq, int64_t row0, int64_t nrows, hipStream_t s)
+{
+  if (nrows <= 0) return;
+  hipLaunchKernelGGL(gen_build_kernel, dim3(grid_for(nrows)), dim3(BLOCK), 0, s, d_keys, d_pay,
+                     n_global, rand_max, seed, (int)uniq, row0, nrows);
+}
+
+void generate_probe(int64_t* d_keys, int64_t* d_pay, int64_t build_n_global, int64_t rand_max,
+                    double selectivity, uint64_t seed, int64_t row0, int64_t nrows, hipStream_t s)
+{
+  if (nrows <= 0) return;
+  hipLaunchKernelGGL(gen_probe_kernel, dim3(grid_for(nrows)), dim3(BLOCK), 0, s, d_keys, d_pay,
+                     build_n_global, rand_max, selectivity, seed, row0, nrows);
+}
+
+/* ------------------------------------------------------- stable partition */
+/*
+ * Wave-contiguous assignment: wave w owns rows [w*rpw, (w+1)*rpw). Lanes read
+ * row base+it*64+lane, so within one iteration lane order == row order and
+ * global memory loads coalesce to 512 B per instruction.
+ *
+ * Pass 1 (count): per-wave histogram held in registers — lane q accumulates
+ * the count for partition q from __ballot(p == q) (64-bit wave64 ballots;
+ * nparts <= 64).
+ * Scan: per-partition block scan over waves (exclusive), then a tiny scan of
+ * partition totals into global partition offsets.
+ * Pass 2 (scatter): lane q holds the running cursor of partition q; each row
+ * gets dst = shfl(cursor, p) + rank, rank = popcount(ballot-mask of its own
+ * partition & lanes-below mask) — conflict-free, stable, no atomics.
+ */
+
+struct PartGeom {
+  int64_t nwaves;
+  int64_t rows_per_wave;
+  int blocks;
+};
+
+static PartGeom part_geom(int64_t n)
+{
+  PartGeom g;
+  int64_t target = (n + 255) / 256;  // >=256 rows per wave
+  g.nwaves = target < 1 ? 1 : (target > 8192 ? 8192 : target);
+  g.blocks = (int)((g.nwaves + WPB - 1) / WPB);
+  g.nwaves = (int64_t)g.blocks * WPB;
+  g.rows_per_wave = (n + g.nwaves - 1) / g.nwaves;
+  return g;
+}
+
+size_t hash_partition_scratch_bytes(int64_t n, int nparts)
+{
+  PartGeom g = part_geom(n);
+  /* wave_counts/prefix [nwaves][nparts] (int64) + totals [nparts] */
+  return (size_t)(g.nwaves * nparts + nparts) * sizeof(int64_t);
+}
+
+__device__ __forceinline__ uint32_t part_of(int64_t key, int hash_fn, uint32_t seed, int nparts)
+{
+  return dj_row_hash(key, hash_fn, seed) % (uint32_t)nparts;
+}
+
+__global__ void part_count_kernel(const int64_t* __restrict__ keys, int64_t n, int nparts,
+                                  int hash_fn, uint32_t seed, int64_t rows_per_wave,
+                                  int64_t* __restrict__ wave_counts)
+{
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t w = (int64_t)blockIdx.x * WPB + (threadIdx.x >> 6);
+  const int64_t start = w * rows_per_wave;
+  const int64_t end = min(start + rows_per_wave, n);
+  int64_t my_count = 0;  // lane q counts partition q
+  for (int64_t base = start; base < end; base += WAVE) {
+    int64_t i = base + lane;
+    uint32_t p = 0xFFFFFFFFu;
+    if (i < end) p = part_of(keys[i], hash_fn, seed, nparts);
+    for (int q = 0; q < nparts; q++) {
+      uint64_t m = __ballot(p == (uint32_t)q);
+      if (lane == q) my_count += __popcll(m);
+    }
+  }
+  if (lane < nparts) wave_counts[w * nparts + lane] = my_count;
+}
+
+/* grid = nparts blocks; block p scans column p over nwaves (exclusive),
+ * in-place, and writes the partition total to totals[p]. */
+__global__ void part_scan_kernel(int64_t* wave_counts, int64_t nwaves, int nparts,
+                                 int64_t* totals)
+{
+  const int p = blockIdx.x;
+  __shared__ int64_t sh[BLOCK];
+  int64_t running = 0;
+  for (int64_t base = 0; base < nwaves; base += BLOCK) {
+    int64_t w = base + threadIdx.x;
+    int64_t v = (w < nwaves) ? wave_counts[w * nparts + p] : 0;
+    /* Hillis-Steele inclusive scan in LDS */
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < BLOCK; off <<= 1) {
+      int64_t add = (threadIdx.x >= off) ? sh[threadIdx.x - off] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += add;
+      __syncthreads();
+    }
+    int64_t incl = sh[threadIdx.x];
+    if (w < nwaves) wave_counts[w * nparts + p] = running + incl - v;  // exclusive
+    int64_t chunk_total = sh[BLOCK - 1];
+    running += chunk_total;
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) totals[p] = running;
+}
+
+/* single block: exclusive scan of totals -> offsets[nparts+1] */
+__global__ void part_offsets_kernel(const int64_t* totals, int nparts, int64_t* offsets)
+{
+  if (threadIdx.x == 0) {
+    int64_t acc = 0;
+    for (int p = 0; p < nparts; p++) {
+      offsets[p] = acc;
+      acc += totals[p];
+    }
+    offsets[nparts] = acc;
+  }
+}
+
+__global__ void part_scatter_kernel(const int64_t* __restrict__ keys,
+                                    const int64_t* __restrict__ pay, int64_t n, int nparts,
+                                    int hash_fn, uint32_t seed, int64_t rows_per_wave,
+                                    const int64_t* __restrict__ wave_prefix,
+                                    const int64_t* __restrict__ offsets,
+                                    int64_t* __restrict__ out_keys,
+                                    int64_t* __restrict__ out_pay)
+{
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int64_t w = (int64_t)blockIdx.x * WPB + (threadIdx.x >> 6);
+  const int64_t start = w * rows_per_wave;
+  const int64_t end = min(start + rows_per_wave, n);
+  /* lane q holds the cursor for partition q */
+  int64_t cursor = 0;
+  if (lane < nparts) cursor = offsets[lane] + wave_prefix[w * nparts + lane];
+  const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+  for (int64_t base = start; base < end; base += WAVE) {
+    int64_t i = base + lane;
+    bool valid = i < end;
+    int64_t k = 0, v = 0;
+    uint32_t p = 0xFFFFFFFFu;
+    if (valid) {
+      k = keys[i];
+      v = pay ? pay[i] : 0;
+      p = part_of(k, hash_fn, seed, nparts);
+    }
+    uint64_t my_q_mask = 0;  // lane q's ballot mask for partition q
+    for (int q = 0; q < nparts; q++) {
+      uint64_t m = __ballot(p == (uint32_t)q);
+      if (lane == q) my_q_mask = m;
+    }
+    /* mask and cursor of MY partition, fetched from lane p */
+    uint64_t m_p = __shfl((unsigned long long)my_q_mask, (int)(valid ? p : 0));
+    int64_t base_dst = __shfl(cursor, (int)(valid ? p : 0));
+    if (valid) {
+      int rank = __popcll(m_p & lt_mask);
+      int64_t dst = base_dst + rank;
+      out_keys[dst] = k;
+      if (out_pay) out_pay[dst] = v;
+    }
+    cursor += __popcll(my_q_mask);
+  }
+}
+
+void partition_count(const int64_t* d_keys, int64_t n, int nparts, int hash_fn,
+                     uint32_t hash_seed, void* d_scratch, hipStream_t s)
+{
+  DJ_CHECK_ERROR(nparts >= 1 && nparts <= kMaxPartitions, "nparts must be in [1,64]");
+  if (n <= 0) return;
+  PartGeom g = part_geom(n);
+  hipLaunchKernelGGL(part_count_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, n, nparts,
+                     hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch);
+}
+
+void partition_scan(int64_t n, int nparts, void* d_scratch, int64_t* d_offsets, hipStream_t s)
+{
+  if (n <= 0) {
+    DJ_HIP_CALL(hipMemsetAsync(d_offsets, 0, (size_t)(nparts + 1) * sizeof(int64_t), s));
+    return;
+  }
+  PartGeom g = part_geom(n);
+  int64_t* wave_counts = (int64_t*)d_scratch;
+  int64_t* totals = wave_counts + g.nwaves * nparts;
+  hipLaunchKernelGGL(part_scan_kernel, dim3(nparts), dim3(BLOCK), 0, s, wave_counts, g.nwaves,
+                     nparts, totals);
+  hipLaunchKernelGGL(part_offsets_kernel, dim3(1), dim3(64), 0, s, totals, nparts, d_offsets);
+}
+
+void partition_scatter(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                       int hash_fn, uint32_t hash_seed, const int64_t* d_offsets,
+                       void* d_scratch, int64_t* d_out_keys, int64_t* d_out_pay, hipStream_t s)
+{
+  if (n <= 0) return;
+  PartGeom g = part_geom(n);
+  hipLaunchKernelGGL(part_scatter_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, d_pay, n,
+                     nparts, hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch,
+                     d_offsets, d_out_keys, d_out_pay);
+}
+
+void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
+                    int hash_fn, uint32_t hash_seed, int64_t* d_out_keys, int64_t* d_out_pay,
+                    int64_t* d_offsets, void* d_scratch, hipStream_t s)
+{
+  partition_count(d_keys, n, nparts, hash_fn, hash_seed, d_scratch, s);
+  partition_scan(n, nparts, d_scratch, d_offsets, s);
+  partition_scatter(d_keys, d_pay, n, nparts, hash_fn, hash_seed, d_offsets, d_scratch,
+                    d_out_keys, d_out_pay, s);
+}
+
+/* ------------------------------------------------------------ local join */
+
+int64_t join_table_slots(int64_t ln)
+{
+  int64_t p = 1;
+  while (p < 2 * ln + 1) p <<= 1;  // <=50% fill
+  return p;
+}
+
+void join_table_init(int64_t* d_table, int64_t nslots, hipStream_t s)
+{
+  /* kEmptyKey == -1 == all bytes 0xFF: one HBM-rate memset over the
+   * interleaved {key,val} pairs */
+  DJ_HIP_CALL(hipMemsetAsync(d_table, 0xFF, (size_t)nslots * 2 * sizeof(int64_t), s));
+}
+
+/* Table layout: nslots interleaved 16 B {key, val} pairs — one random
+ * cache-line fetch serves both the key compare and the payload read.
+ * PAIRS: input rows as interleaved longlong2 instead of two columns. */
+template <bool PAIRS>
+__global__ void join_build_kernel(const int64_t* __restrict__ lk, const int64_t* __restrict__ lp,
+                                  const longlong2* __restrict__ lrows,
+                                  int64_t ln, longlong2* __restrict__ table, uint64_t mask,
+                                  int* error)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < ln; i += stride) {
+    int64_t key = PAIRS ? lrows[i].x : lk[i];
+    if (key == kEmptyKey) {
+      *error = 1;  // reserved sentinel; loud failure, never silent wrong data
+      continue;
+    }
+    uint64_t slot = dj_mix64((uint64_t)key) & mask;
+    for (;;) {
+      unsigned long long old = atomicCAS((unsigned long long*)&table[slot].x,
+                                         (unsigned long long)kEmptyKey,
+                                         (unsigned long long)key);
+      if (old == (unsigned long long)kEmptyKey) break;
+      slot = (slot + 1) & mask;
+    }
+    table[slot].y = PAIRS ? lrows[i].y : (lp ? lp[i] : i);
+  }
+}
+
+void join_build(const int64_t* d_lk, const int64_t* d_lp, int64_t ln, int64_t* d_table,
+                int64_t nslots, int* d_error, hipStream_t s)
+{
+  if (ln <= 0) return;
+  hipLaunchKernelGGL(join_build_kernel<false>, dim3(grid_for(ln)), dim3(BLOCK), 0, s, d_lk,
+                     d_lp, (const longlong2*)nullptr, ln, (longlong2*)d_table,
+                     (uint64_t)(nslots - 1), d_error);
+}
+
+void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int64_t nslots,
+                      int* d_error, hipStream_t s)
+{
+  if (ln <= 0) return;
+  hipLaunchKernelGGL(join_build_kernel<true>, dim3(grid_for(ln)), dim3(BLOCK), 0, s,
+                     (const int64_t*)nullptr, (const int64_t*)nullptr, d_rows, ln,
+                     (longlong2*)d_table, (uint64_t)(nslots - 1), d_error);
+}
+
+/* Probe with wave-aggregated output append: matches are emitted via one
+ * atomicAdd per wave (ballot leader) instead of one per lane — the single
+ * global counter would otherwise serialize the whole kernel. All lanes of a
+ * wave iterate in lockstep so the ballots are well-defined. */
+template <bool PAIRS>
+__global__ void join_probe_kernel(const int64_t* __restrict__ rk, const int64_t* __restrict__ rp,
+                                  const longlong2* __restrict__ rrows,
+                                  int64_t rn, const longlong2* __restrict__ table,
+                                  uint64_t mask, int64_t* __restrict__ out0,
+                                  int64_t* __restrict__ out1, int64_t* __restrict__ out2,
+                                  int64_t* __restrict__ out3, int64_t cap,
+                                  unsigned long long* counter)
+{
+  const int lane = threadIdx.x & (WAVE - 1);
+  const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;; j += stride) {
+    const bool row_valid = j < rn;
+    if (__ballot(row_valid) == 0) break;  // whole wave done (uniform)
+    int64_t key = 0, payload = 0;
+    uint64_t slot = 0;
+    bool walking = row_valid;
+    if (row_valid) {
+      if (PAIRS) {
+        longlong2 r = rrows[j];
+        key = r.x;
+        payload = r.y;
+      } else {
+        key = rk[j];
+        payload = rp ? rp[j] : j;
+      }
+      slot = dj_mix64((uint64_t)key) & mask;
+    }
+    for (;;) {
+      int64_t mval = 0;
+      bool have = false;
+      while (walking) {
+        longlong2 e = table[slot];
+        if (e.x == kEmptyKey) {
+          walking = false;
+          break;
+        }
+        slot = (slot + 1) & mask;
+        if (e.x == key) {
+          mval = e.y;
+          have = true;
+          break;
+        }
+      }
+      uint64_t m = __ballot(have);
+      if (m == 0) break;
+      const int leader = (int)(__ffsll((unsigned long long)m) - 1);
+      unsigned long long base = 0;
+      if (lane == leader) base = atomicAdd(counter, (unsigned long long)__popcll(m));
+      base = __shfl(base, leader);
+      if (have) {
+        int64_t idx = (int64_t)base + __popcll(m & lt_mask);
+        if (idx < cap) {
+          out0[idx] = key;
+          out1[idx] = mval;
+          out2[idx] = key;
+          out3[idx] = payload;
+        }
+      }
+    }
+  }
+}
+
+void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int64_t* d_table,
+                int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s)
+{
+  if (rn <= 0) return;
+  hipLaunchKernelGGL(join_probe_kernel<false>, dim3(grid_for(rn)), dim3(BLOCK), 0, s, d_rk,
+                     d_rp, (const longlong2*)nullptr, rn, (const longlong2*)d_table,
+                     (uint64_t)(nslots - 1), d_out0, d_out1, d_out2, d_out3, cap,
+                     (unsigned long long*)d_counter);
+}
+
+void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_table,
+                      int64_t nslots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                      int64_t* d_out3, int64_t cap, int64_t* d_counter, hipStream_t s)
+{
+  if (rn <= 0) return;
+  hipLaunchKernelGGL(join_probe_kernel<true>, dim3(grid_for(rn)), dim3(BLOCK), 0, s,
+                     (const int64_t*)nullptr, (const int64_t*)nullptr, d_rows, rn,
+                     (const longlong2*)d_table, (uint64_t)(nslots - 1), d_out0, d_out1,
+                     d_out2, d_out3, cap, (unsigned long long*)d_counter);
+}
+
+/* ----------------------------------------------- bucketed LDS join ------ */
+/*
+ * MI355X-native local join. A single HBM-resident hash table is bound far
+ * below the HBM roofline by random 128 B line fetches, device-scope atomic
+ * throughput (~18 G scattered CAS/s measured) and, worst, the single output
+ * counter (~83 M serialized atomics/s measured — experiments/membench.hip).
+ * Instead: partition both tables into B buckets small enough that a
+ * bucket's hash table fits in LDS, then one fused kernel per bucket builds
+ * a 4096-slot LDS table (ds atomicCAS) and probes it twice (count, then
+ * write) — all random access is on-chip, HBM sees only streaming traffic,
+ * and the global output counter is touched ONCE per bucket.
+ *
+ * The bucket partition is two-level so every scatter pass has at most 1024
+ * write streams alive per block (partial-line writes then merge in the
+ * per-XCD L2 — a single-pass 32768-way scatter measurably thrashes it):
+ *   pass A: P_A = B/256 groups by bits [40, 40+log2(P_A)) of dj_mix64(key)
+ *           (block-chunked count + cross-block scan + LDS-cursor scatter);
+ *   pass B: one block per group: 256 sub-buckets by bits [32,40), count +
+ *           in-block scan + scatter, emitting the final bucket offsets.
+ * The decomposition is internal to the local join (result row order is
+ * unspecified by the API), so both passes are NON-stable — unlike the
+ * rank-level hash_partition above, which stays stable (reference pin:
+ * SURVEY.md appendix, batch offsets).
+ * LDS slot hash = low 32 bits of dj_mix64; the rank-level partition uses
+ * MurmurHash3%G, so bucket occupancy is independent of rank placement.
+ */
+
+constexpr int BUCKET_BLOCKS = kBucketBlocks;  // chunking blocks for pass A (2/CU); scratch sizing uses the same constant
+constexpr int BUCKET_THREADS = 1024;
+constexpr int JOIN_LDS_SLOTS = 4096;        // 64 KiB of longlong2 pairs -> 2 blocks/CU
+constexpr int SUB_BUCKETS = 256;            // pass-B fanout (fixed)
+
+int bucket_count_for(int64_t ln, int64_t rn)
+{
+  int64_t maxn = ln > rn ? ln : rn;
+  int64_t B = 256;
+  while (B < 262144 && maxn / B > 1600) B <<= 1;
+  return (int)B;
+}
+
+/* final bucket id = bits [32, 32+log2(B)) of dj_mix64(key) */
+__device__ __forceinline__ uint32_t bucket_of(int64_t key, int B)
+{
+  return (uint32_t)(dj_mix64((uint64_t)key) >> 32) & (uint32_t)(B - 1);
+}
+/* pass-A group = bucket id >> 8 ; pass-B sub-bucket = bucket id & 255 */
+__device__ __forceinline__ uint32_t groupA_of(int64_t key, int PA)
+{
+  return (uint32_t)(dj_mix64((uint64_t)key) >> 40) & (uint32_t)(PA - 1);
+}
+__device__ __forceinline__ uint32_t subB_of(int64_t key)
+{
+  return (uint32_t)(dj_mix64((uint64_t)key) >> 32) & (uint32_t)(SUB_BUCKETS - 1);
+}
+
+/* ---- pass A kernels (P groups, P <= 1024) ---- */
+
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_count_kernel(
+  const int64_t* __restrict__ keys, int64_t n, int P, uint32_t* __restrict__ counts)
+{
+  extern __shared__ uint32_t hist[];
+  for (int p = threadIdx.x; p < P; p += blockDim.x) hist[p] = 0;
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t start = (int64_t)blockIdx.x * chunk;
+  const int64_t end = min(start + chunk, n);
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    atomicAdd(&hist[groupA_of(nt_load(&keys[i]), P)], 1u);
+  __syncthreads();
+  for (int p = threadIdx.x; p < P; p += blockDim.x)
+    counts[(size_t)blockIdx.x * P + p] = hist[p];
+}
+
+/* grid = P blocks x BUCKET_BLOCKS threads: exclusive scan of counts[:, p]
+ * over blocks (in place), totals[p] = column sum */
+__global__ void bucket_scanA_kernel(uint32_t* counts, int P, uint32_t* totals)
+{
+  const int p = blockIdx.x;
+  __shared__ uint32_t sh[BUCKET_BLOCKS];
+  uint32_t v = counts[(size_t)threadIdx.x * P + p];
+  sh[threadIdx.x] = v;
+  __syncthreads();
+  for (int off = 1; off < BUCKET_BLOCKS; off <<= 1) {
+    uint32_t add = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
+    __syncthreads();
+    sh[threadIdx.x] += add;
+    __syncthreads();
+  }
+  counts[(size_t)threadIdx.x * P + p] = sh[threadIdx.x] - v;  // exclusive
+  if (threadIdx.x == BUCKET_BLOCKS - 1) totals[p] = sh[threadIdx.x];
+}
+
+/* single block: exclusive scan of totals[P] -> segoff[P+1] (int64) */
+__global__ void bucket_scanB_kernel(const uint32_t* totals, int P, int64_t* segoff)
+{
+  __shared__ int64_t sh[BUCKET_THREADS];
+  __shared__ int64_t running_sh;
+  if (threadIdx.x == 0) running_sh = 0;
+  __syncthreads();
+  for (int base = 0; base < P; base += BUCKET_THREADS) {
+    int p = base + threadIdx.x;
+    int64_t v = (p < P) ? (int64_t)totals[p] : 0;
+    sh[threadIdx.x] = v;
+    __syncthreads();
+    for (int off = 1; off < BUCKET_THREADS; off <<= 1) {
+      int64_t add = (threadIdx.x >= (unsigned)off) ? sh[threadIdx.x - off] : 0;
+      __syncthreads();
+      sh[threadIdx.x] += add;
+      __syncthreads();
+    }
+    int64_t rbase = running_sh;
+    __syncthreads();  // all reads of running_sh precede the update below
+    if (p < P) segoff[p] = rbase + sh[threadIdx.x] - v;
+    if (threadIdx.x == BUCKET_THREADS - 1) running_sh = rbase + sh[threadIdx.x];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) segoff[P] = running_sh;
+}
+
+/* Tile-staged scatter: rows of a 4096-row tile are ranked per group in LDS
+ * (hist atomics), staged grouped into an LDS tile buffer, then each group's
+ * run is flushed linearly — global writes coalesce into ~cnt[g]-row runs
+ * (full 128 B lines) instead of 64 scattered 16 B transactions per wave.
+ * The direct-scatter variant measured 59% SQ issue-stall on store-pipe
+ * pressure; staging trades cheap LDS traffic for it. Requires P <= 1024
+ * (= BUCKET_THREADS: one group per thread in the hist scan). */
+constexpr int SCATTER_TILE = 4096;  // 64 KiB of staged pairs
+
+template <bool SINGLE_LEVEL>
+__device__ __forceinline__ longlong2 load_row(const int64_t* keys, const int64_t* pay,
+                                              const longlong2* pairs, int64_t i)
+{
+  longlong2 r;
+  if (SINGLE_LEVEL) {
+    r.x = nt_load(&keys[i]);
+    r.y = pay ? nt_load(&pay[i]) : i;
+  } else {
+    r = nt_load2(&pairs[i]);
+  }
+  return r;
+}
+
+/* staged scatter over [start, end); cur[P] holds GLOBAL destination
+ * cursors; GROUP_FN: 0 = groupA_of(P), 1 = subB_of */
+template <int GROUP_FN, bool SINGLE_LEVEL>
+__device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
+                                    const longlong2* in_pairs, int64_t start, int64_t end,
+                                    int P, longlong2* tbuf, uint32_t* hist, uint32_t* base,
+                                    uint32_t* gcur, longlong2* out_pairs)
+{
+  constexpr int VPT = SCATTER_TILE / BUCKET_THREADS;  // 4
+  const int tid = threadIdx.x;
+  for (int64_t t0 = start; t0 < end; t0 += SCATTER_TILE) {
+    const int count = (int)min((int64_t)SCATTER_TILE, end - t0);
+    if (tid < P) hist[tid] = 0;
+    __syncthreads();
+    longlong2 r[VPT];
+    uint32_t g[VPT], rank[VPT];
+    int nv = 0;
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
+      if (i < end) {
+        r[v] = load_row<SINGLE_LEVEL>(keys, pay, in_pairs, i);
+        g[v] = GROUP_FN == 0 ? groupA_of(r[v].x, P) : subB_of(r[v].x);
+        rank[v] = atomicAdd(&hist[g[v]], 1u);
+        nv = v + 1;
+      }
+    }
+    __syncthreads();
+    /* exclusive scan of hist -> base (P <= blockDim: one group per thread) */
+    if (tid < P) base[tid] = hist[tid];
+    __syncthreads();
+    for (int off = 1; off < P; off <<= 1) {
+      uint32_t add = (tid < P && tid >= off) ? base[tid - off] : 0;
+      __syncthreads();
+      if (tid < P) base[tid] += add;
+      __syncthreads();
+    }
+    if (tid < P) base[tid] -= hist[tid];  // inclusive -> exclusive
+    __syncthreads();
+    for (int v = 0; v < nv; v++) tbuf[base[g[v]] + rank[v]] = r[v];
+    __syncthreads();
+    /* flush linearly: per-group runs coalesce into full lines */
+    for (int pos = tid; pos < count; pos += blockDim.x) {
+      longlong2 row = tbuf[pos];
+      uint32_t gg = GROUP_FN == 0 ? groupA_of(row.x, P) : subB_of(row.x);
+      out_pairs[gcur[gg] + (pos - base[gg])] = row;
+    }
+    __syncthreads();
+    if (tid < P) gcur[tid] += hist[tid];
+    __syncthreads();
+  }
+}
+
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_kernel(
+  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int P,
+  const uint32_t* __restrict__ counts, const int64_t* __restrict__ segoff,
+  longlong2* __restrict__ out_pairs)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbuf = (longlong2*)smem;
+  uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
+  uint32_t* base = hist + P;
+  uint32_t* gcur = base + P;
+  if (threadIdx.x < P)
+    gcur[threadIdx.x] = (uint32_t)segoff[threadIdx.x] +
+                        counts[(size_t)blockIdx.x * P + threadIdx.x];
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t start = (int64_t)blockIdx.x * chunk;
+  const int64_t end = min(start + chunk, n);
+  staged_scatter_span<0, true>(keys, pay, nullptr, start, end, P, tbuf, hist, base, gcur,
+                               out_pairs);
+}
+
+/* trivial segoff = {0, n} for the single-group (B == 256) case */
+__global__ void set_segoff1_kernel(int64_t* segoff, int64_t n)
+{
+  segoff[0] = 0;
+  segoff[1] = n;
+}
+
+/* ---- pass B: one block per pass-A group; 256 sub-buckets in-block ---- */
+/* pass B: one block per pass-A group; 256 sub-buckets, tile-staged scatter.
+ * single_level: input is the original two column arrays (B == 256 case). */
+template <bool SINGLE_LEVEL>
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
+  const longlong2* __restrict__ in_pairs, const int64_t* __restrict__ keys,
+  const int64_t* __restrict__ pay, const int64_t* __restrict__ segoff, int B,
+  longlong2* __restrict__ out_pairs, int64_t* __restrict__ bucket_offsets /* B+1 */)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbuf = (longlong2*)smem;
+  uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
+  uint32_t* base = hist + SUB_BUCKETS;
+  uint32_t* gcur = base + SUB_BUCKETS;
+  uint32_t* seghist = gcur + SUB_BUCKETS;
+  const int tid = threadIdx.x;
+  const int a = blockIdx.x;
+  const int64_t s0 = segoff[a], s1 = segoff[a + 1];
+  if (tid < SUB_BUCKETS) seghist[tid] = 0;
+  __syncthreads();
+  for (int64_t i = s0 + tid; i < s1; i += blockDim.x) {
+    int64_t k = SINGLE_LEVEL ? nt_load(&keys[i]) : nt_load(&in_pairs[i].x);
+    atomicAdd(&seghist[subB_of(k)], 1u);
+  }
+  __syncthreads();
+  if (tid == 0) {
+    uint32_t acc = 0;
+    for (int j = 0; j < SUB_BUCKETS; j++) {
+      uint32_t c = seghist[j];
+      gcur[j] = (uint32_t)s0 + acc;
+      bucket_offsets[(size_t)a * SUB_BUCKETS + j] = s0 + acc;
+      acc += c;
+    }
+    if (a == gridDim.x - 1) bucket_offsets[B] = s1;
+  }
+  __syncthreads();
+  staged_scatter_span<1, SINGLE_LEVEL>(keys, pay, in_pairs, s0, s1, SUB_BUCKETS, tbuf, hist,
+                                       base, gcur, out_pairs);
+}
+
+/* ---- fused per-bucket LDS build + two-phase probe (count, then write) ---- */
+__global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
+  const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
+  const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
+  int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
+  int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter,
+  uint32_t* __restrict__ overflow_flags, int* __restrict__ any_overflow,
+  int* __restrict__ error)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbl = (longlong2*)smem;
+  long long* base_sh = (long long*)(smem + JOIN_LDS_SLOTS * sizeof(longlong2));
+  uint32_t* total_sh = (uint32_t*)(base_sh + 1);
+  uint32_t* cur_sh = total_sh + 1;
+  const uint32_t smask = JOIN_LDS_SLOTS - 1;
+
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const int64_t l0 = loff[b], l1 = loff[b + 1];
+    const int64_t r0 = roff[b], r1 = roff[b + 1];
+    const int64_t lnb = l1 - l0;
+    if (lnb == 0 || r1 == r0) continue;
+    if (lnb > kJoinBucketRowCap) {  // skew overflow: host-side fallback joins it
+      if (threadIdx.x == 0) {
+        overflow_flags[b] = 1;
+        *any_overflow = 1;
+      }
+      continue;
+    }
+    for (int s = threadIdx.x; s < JOIN_LDS_SLOTS; s += blockDim.x) tbl[s].x = kEmptyKey;
+    if (threadIdx.x == 0) {
+      *total_sh = 0;
+      *cur_sh = 0;
+    }
+    __syncthreads();
+    /* build */
+    for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
+      longlong2 row = lrows[i];
+      if (row.x == kEmptyKey) {
+        *error = 1;
+        continue;
+      }
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)row.x) & smask;
+      for (;;) {
+        unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
+                                           (unsigned long long)kEmptyKey,
+                                           (unsigned long long)row.x);
+        if (old == (unsigned long long)kEmptyKey) break;
+        slot = (slot + 1) & smask;
+      }
+      tbl[slot].y = row.y;
+    }
+    __syncthreads();
+    /* probe phase 1: count my matches */
+    uint32_t my = 0;
+    for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
+      int64_t key = rrows[j].x;
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
+      for (;;) {
+        longlong2 e = tbl[slot];
+        if (e.x == kEmptyKey) break;
+        if (e.x == key) my++;
+        slot = (slot + 1) & smask;
+      }
+    }
+    if (my) atomicAdd(total_sh, my);
+    __syncthreads();
+    /* ONE global atomic per bucket reserves the output range */
+    if (threadIdx.x == 0 && *total_sh)
+      *base_sh = (long long)atomicAdd(counter, (unsigned long long)*total_sh);
+    __syncthreads();
+    if (*total_sh) {
+      const long long base = *base_sh;
+      uint32_t w = my ? atomicAdd(cur_sh, my) : 0;
+      /* probe phase 2: rewalk in the same order, write */
+      if (my) {
+        for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
+          longlong2 prow = rrows[j];
+          int64_t key = prow.x;
+          int64_t payload = prow.y;
+          uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
+          for (;;) {
+            longlong2 e = tbl[slot];
+            if (e.x == kEmptyKey) break;
+            if (e.x == key) {
+              long long idx = base + (long long)w;
+              if (idx < cap) {
+                out0[idx] = key;
+                out1[idx] = e.y;
+                out2[idx] = key;
+                out3[idx] = payload;
+              }
+              w++;
+            }
+            slot = (slot + 1) & smask;
+          }
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
+                       longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
+                       int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
+                       hipStream_t s)
+{
+  DJ_CHECK_ERROR(n < (int64_t)UINT32_MAX, "bucket_partition: n must be < 2^32");
+  const int PA = B / SUB_BUCKETS;
+  DJ_CHECK_ERROR(PA >= 1 && PA <= 1024, "bucket_partition: B out of range");
+  if (PA == 1) {
+    size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
+    hipLaunchKernelGGL(set_segoff1_kernel, dim3(1), dim3(1), 0, s, d_segoff, n);
+    hipLaunchKernelGGL(bucket_subpart_kernel<true>, dim3(PA), dim3(BUCKET_THREADS),
+                       subpart_lds, s, (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B,
+                       d_out_pairs, d_offsets);
+  } else {
+    size_t hist_lds = (size_t)PA * sizeof(uint32_t);
+    size_t scatter_lds = SCATTER_TILE * sizeof(longlong2) + 3 * hist_lds;
+    size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
+    hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), hist_lds,
+                       s, d_keys, n, PA, d_counts);
+    hipLaunchKernelGGL(bucket_scanA_kernel, dim3(PA), dim3(BUCKET_BLOCKS), 0, s, d_counts, PA,
+                       d_totals);
+    hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, PA,
+                       d_segoff);
+    hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
+                       scatter_lds, s, d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmp_pairs);
+    hipLaunchKernelGGL(bucket_subpart_kernel<false>, dim3(PA), dim3(BUCKET_THREADS),
+                       subpart_lds, s, d_tmp_pairs, (const int64_t*)nullptr,
+                       (const int64_t*)nullptr, d_segoff, B, d_out_pairs, d_offsets);
+  }
+}
+
+void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
+              const int64_t* d_roff, int B, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+              int64_t* d_out3, int64_t cap, int64_t* d_counter, uint32_t* d_overflow_flags,
+              int* d_any_overflow, int* d_error, hipStream_t s)
+{
+  int grid = B < 8192 ? B : 8192;
+  size_t lds = JOIN_LDS_SLOTS * sizeof(longlong2) + 16;
+  hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lrows,
+                     d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
+                     (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
+                     d_error);
+}
+
+}  // namespace dj
