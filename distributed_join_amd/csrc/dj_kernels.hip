@@ -595,25 +595,100 @@ __global__ void bucket_scanB_kernel(const uint32_t* totals, int P, int64_t* sego
   if (threadIdx.x == 0) segoff[P] = running_sh;
 }
 
+/* Tile-staged scatter: rows of a 4096-row tile are ranked per group in LDS
+ * (hist atomics), staged grouped into an LDS tile buffer, then each group's
+ * run is flushed linearly — global writes coalesce into ~cnt[g]-row runs
+ * (full 128 B lines) instead of 64 scattered 16 B transactions per wave.
+ * The direct-scatter variant measured 59% SQ issue-stall on store-pipe
+ * pressure; staging trades cheap LDS traffic for it. Requires P <= 1024
+ * (= BUCKET_THREADS: one group per thread in the hist scan). */
+constexpr int SCATTER_TILE = 4096;  // 64 KiB of staged pairs
+
+template <bool SINGLE_LEVEL>
+__device__ __forceinline__ longlong2 load_row(const int64_t* keys, const int64_t* pay,
+                                              const longlong2* pairs, int64_t i)
+{
+  longlong2 r;
+  if (SINGLE_LEVEL) {
+    r.x = nt_load(&keys[i]);
+    r.y = pay ? nt_load(&pay[i]) : i;
+  } else {
+    r = nt_load2(&pairs[i]);
+  }
+  return r;
+}
+
+/* staged scatter over [start, end); cur[P] holds GLOBAL destination
+ * cursors; GROUP_FN: 0 = groupA_of(P), 1 = subB_of */
+template <int GROUP_FN, bool SINGLE_LEVEL>
+__device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
+                                    const longlong2* in_pairs, int64_t start, int64_t end,
+                                    int P, longlong2* tbuf, uint32_t* hist, uint32_t* base,
+                                    uint32_t* gcur, longlong2* out_pairs)
+{
+  constexpr int VPT = SCATTER_TILE / BUCKET_THREADS;  // 4
+  const int tid = threadIdx.x;
+  for (int64_t t0 = start; t0 < end; t0 += SCATTER_TILE) {
+    const int count = (int)min((int64_t)SCATTER_TILE, end - t0);
+    if (tid < P) hist[tid] = 0;
+    __syncthreads();
+    longlong2 r[VPT];
+    uint32_t g[VPT], rank[VPT];
+    int nv = 0;
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
+      if (i < end) {
+        r[v] = load_row<SINGLE_LEVEL>(keys, pay, in_pairs, i);
+        g[v] = GROUP_FN == 0 ? groupA_of(r[v].x, P) : subB_of(r[v].x);
+        rank[v] = atomicAdd(&hist[g[v]], 1u);
+        nv = v + 1;
+      }
+    }
+    __syncthreads();
+    /* exclusive scan of hist -> base (P <= blockDim: one group per thread) */
+    if (tid < P) base[tid] = hist[tid];
+    __syncthreads();
+    for (int off = 1; off < P; off <<= 1) {
+      uint32_t add = (tid < P && tid >= off) ? base[tid - off] : 0;
+      __syncthreads();
+      if (tid < P) base[tid] += add;
+      __syncthreads();
+    }
+    if (tid < P) base[tid] -= hist[tid];  // inclusive -> exclusive
+    __syncthreads();
+    for (int v = 0; v < nv; v++) tbuf[base[g[v]] + rank[v]] = r[v];
+    __syncthreads();
+    /* flush linearly: per-group runs coalesce into full lines */
+    for (int pos = tid; pos < count; pos += blockDim.x) {
+      longlong2 row = tbuf[pos];
+      uint32_t gg = GROUP_FN == 0 ? groupA_of(row.x, P) : subB_of(row.x);
+      out_pairs[gcur[gg] + (pos - base[gg])] = row;
+    }
+    __syncthreads();
+    if (tid < P) gcur[tid] += hist[tid];
+    __syncthreads();
+  }
+}
+
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_kernel(
   const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int P,
   const uint32_t* __restrict__ counts, const int64_t* __restrict__ segoff,
   longlong2* __restrict__ out_pairs)
 {
-  extern __shared__ uint32_t cursor[];
-  for (int p = threadIdx.x; p < P; p += blockDim.x)
-    cursor[p] = (uint32_t)segoff[p] + counts[(size_t)blockIdx.x * P + p];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbuf = (longlong2*)smem;
+  uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
+  uint32_t* base = hist + P;
+  uint32_t* gcur = base + P;
+  if (threadIdx.x < P)
+    gcur[threadIdx.x] = (uint32_t)segoff[threadIdx.x] +
+                        counts[(size_t)blockIdx.x * P + threadIdx.x];
   __syncthreads();
   const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
   const int64_t start = (int64_t)blockIdx.x * chunk;
   const int64_t end = min(start + chunk, n);
-  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x) {
-    longlong2 row;
-    row.x = nt_load(&keys[i]);
-    row.y = pay ? nt_load(&pay[i]) : i;
-    uint32_t dst = atomicAdd(&cursor[groupA_of(row.x, P)], 1u);
-    out_pairs[dst] = row;  // one 16 B store: half the write streams of 2 columns
-  }
+  staged_scatter_span<0, true>(keys, pay, nullptr, start, end, P, tbuf, hist, base, gcur,
+                               out_pairs);
 }
 
 /* trivial segoff = {0, n} for the single-group (B == 256) case */
@@ -624,47 +699,43 @@ __global__ void set_segoff1_kernel(int64_t* segoff, int64_t n)
 }
 
 /* ---- pass B: one block per pass-A group; 256 sub-buckets in-block ---- */
-/* pass B over interleaved pairs. single_level: input is the original two
- * column arrays instead of pass-A pairs (B == 256 case). */
+/* pass B: one block per pass-A group; 256 sub-buckets, tile-staged scatter.
+ * single_level: input is the original two column arrays (B == 256 case). */
 template <bool SINGLE_LEVEL>
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
   const longlong2* __restrict__ in_pairs, const int64_t* __restrict__ keys,
   const int64_t* __restrict__ pay, const int64_t* __restrict__ segoff, int B,
   longlong2* __restrict__ out_pairs, int64_t* __restrict__ bucket_offsets /* B+1 */)
 {
-  __shared__ uint32_t hist[SUB_BUCKETS];
-  __shared__ uint32_t cur[SUB_BUCKETS];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbuf = (longlong2*)smem;
+  uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
+  uint32_t* base = hist + SUB_BUCKETS;
+  uint32_t* gcur = base + SUB_BUCKETS;
+  uint32_t* seghist = gcur + SUB_BUCKETS;
+  const int tid = threadIdx.x;
   const int a = blockIdx.x;
   const int64_t s0 = segoff[a], s1 = segoff[a + 1];
-  for (int p = threadIdx.x; p < SUB_BUCKETS; p += blockDim.x) hist[p] = 0;
+  if (tid < SUB_BUCKETS) seghist[tid] = 0;
   __syncthreads();
-  for (int64_t i = s0 + threadIdx.x; i < s1; i += blockDim.x) {
+  for (int64_t i = s0 + tid; i < s1; i += blockDim.x) {
     int64_t k = SINGLE_LEVEL ? nt_load(&keys[i]) : nt_load(&in_pairs[i].x);
-    atomicAdd(&hist[subB_of(k)], 1u);
+    atomicAdd(&seghist[subB_of(k)], 1u);
   }
   __syncthreads();
-  if (threadIdx.x == 0) {
+  if (tid == 0) {
     uint32_t acc = 0;
     for (int j = 0; j < SUB_BUCKETS; j++) {
-      uint32_t c = hist[j];
-      cur[j] = acc;
+      uint32_t c = seghist[j];
+      gcur[j] = (uint32_t)s0 + acc;
       bucket_offsets[(size_t)a * SUB_BUCKETS + j] = s0 + acc;
       acc += c;
     }
     if (a == gridDim.x - 1) bucket_offsets[B] = s1;
   }
   __syncthreads();
-  for (int64_t i = s0 + threadIdx.x; i < s1; i += blockDim.x) {
-    longlong2 row;
-    if (SINGLE_LEVEL) {
-      row.x = nt_load(&keys[i]);
-      row.y = pay ? nt_load(&pay[i]) : i;
-    } else {
-      row = nt_load2(&in_pairs[i]);
-    }
-    uint32_t dst = atomicAdd(&cur[subB_of(row.x)], 1u);
-    out_pairs[s0 + dst] = row;
-  }
+  staged_scatter_span<1, SINGLE_LEVEL>(keys, pay, in_pairs, s0, s1, SUB_BUCKETS, tbuf, hist,
+                                       base, gcur, out_pairs);
 }
 
 /* ---- fused per-bucket LDS build + two-phase probe (count, then write) ---- */
@@ -778,7 +849,10 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
   const int PA = B / SUB_BUCKETS;
   DJ_CHECK_ERROR(PA >= 1 && PA <= 1024, "bucket_partition: B out of range");
   if (PA == 1) {
+    size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
     hipLaunchKernelGGL(set_segoff1_kernel, dim3(1), dim3(1), 0, s, d_segoff, n);
+    DJ_HIP_CALL(hipGetLastError());
+<<<<<<< HEAD
     DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_subpart_kernel<true>, dim3(PA), dim3(BUCKET_THREADS), 0, s,
                        (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B, d_out_pairs,
@@ -789,11 +863,26 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
     hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
                        d_keys, n, PA, d_counts);
     DJ_HIP_CALL(hipGetLastError());
+=======
+    hipLaunchKernelGGL(bucket_subpart_kernel<true>, dim3(PA), dim3(BUCKET_THREADS),
+                       subpart_lds, s, (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B,
+                       d_out_pairs, d_offsets);
+    DJ_HIP_CALL(hipGetLastError());
+  } else {
+    size_t hist_lds = (size_t)PA * sizeof(uint32_t);
+    size_t scatter_lds = SCATTER_TILE * sizeof(longlong2) + 3 * hist_lds;
+    size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
+    hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), hist_lds,
+                       s, d_keys, n, PA, d_counts);
+    DJ_HIP_CALL(hipGetLastError());
+>>>>>>> parent of de22b5e (Revert "tile-staged scatter for both bucket levels")
     hipLaunchKernelGGL(bucket_scanA_kernel, dim3(PA), dim3(BUCKET_BLOCKS), 0, s, d_counts, PA,
                        d_totals);
     DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, PA,
                        d_segoff);
+    DJ_HIP_CALL(hipGetLastError());
+<<<<<<< HEAD
     DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
                        d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmp_pairs);
@@ -802,6 +891,15 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        d_tmp_pairs, (const int64_t*)nullptr, (const int64_t*)nullptr, d_segoff,
                        B, d_out_pairs, d_offsets);
     DJ_HIP_CALL(hipGetLastError());
+=======
+    hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
+                       scatter_lds, s, d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmp_pairs);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(bucket_subpart_kernel<false>, dim3(PA), dim3(BUCKET_THREADS),
+                       subpart_lds, s, d_tmp_pairs, (const int64_t*)nullptr,
+                       (const int64_t*)nullptr, d_segoff, B, d_out_pairs, d_offsets);
+    DJ_HIP_CALL(hipGetLastError());
+>>>>>>> parent of de22b5e (Revert "tile-staged scatter for both bucket levels")
   }
 }
 
