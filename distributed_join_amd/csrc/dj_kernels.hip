@@ -852,18 +852,6 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
     size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
     hipLaunchKernelGGL(set_segoff1_kernel, dim3(1), dim3(1), 0, s, d_segoff, n);
     DJ_HIP_CALL(hipGetLastError());
-<<<<<<< HEAD
-    DJ_HIP_CALL(hipGetLastError());
-    hipLaunchKernelGGL(bucket_subpart_kernel<true>, dim3(PA), dim3(BUCKET_THREADS), 0, s,
-                       (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B, d_out_pairs,
-                       d_offsets);
-    DJ_HIP_CALL(hipGetLastError());
-  } else {
-    size_t lds = (size_t)PA * sizeof(uint32_t);
-    hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
-                       d_keys, n, PA, d_counts);
-    DJ_HIP_CALL(hipGetLastError());
-=======
     hipLaunchKernelGGL(bucket_subpart_kernel<true>, dim3(PA), dim3(BUCKET_THREADS),
                        subpart_lds, s, (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B,
                        d_out_pairs, d_offsets);
@@ -875,23 +863,12 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
     hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), hist_lds,
                        s, d_keys, n, PA, d_counts);
     DJ_HIP_CALL(hipGetLastError());
->>>>>>> parent of de22b5e (Revert "tile-staged scatter for both bucket levels")
     hipLaunchKernelGGL(bucket_scanA_kernel, dim3(PA), dim3(BUCKET_BLOCKS), 0, s, d_counts, PA,
                        d_totals);
     DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, PA,
                        d_segoff);
     DJ_HIP_CALL(hipGetLastError());
-<<<<<<< HEAD
-    DJ_HIP_CALL(hipGetLastError());
-    hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), lds, s,
-                       d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmp_pairs);
-    DJ_HIP_CALL(hipGetLastError());
-    hipLaunchKernelGGL(bucket_subpart_kernel<false>, dim3(PA), dim3(BUCKET_THREADS), 0, s,
-                       d_tmp_pairs, (const int64_t*)nullptr, (const int64_t*)nullptr, d_segoff,
-                       B, d_out_pairs, d_offsets);
-    DJ_HIP_CALL(hipGetLastError());
-=======
     hipLaunchKernelGGL(bucket_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
                        scatter_lds, s, d_keys, d_pay, n, PA, d_counts, d_segoff, d_tmp_pairs);
     DJ_HIP_CALL(hipGetLastError());
@@ -899,7 +876,6 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        subpart_lds, s, d_tmp_pairs, (const int64_t*)nullptr,
                        (const int64_t*)nullptr, d_segoff, B, d_out_pairs, d_offsets);
     DJ_HIP_CALL(hipGetLastError());
->>>>>>> parent of de22b5e (Revert "tile-staged scatter for both bucket levels")
   }
 }
 
