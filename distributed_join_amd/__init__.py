@@ -91,6 +91,7 @@ def lib():
                                                       vp, vp, vp, i64, i64, i32, i32], vp),
             "dj_cpp_distributed_inner_join_cols": ([vp, vp, i32, i64, vp, i32, i64,
                                                     i32, i32, i32, i32], vp),
+            "dj_cpp_distribute_collect_roundtrip_i64": ([vp, vp, vp, i64], vp),
             "dj_table_column_type": ([vp, i32], i32),
             "dj_table_column_chars": ([vp, i32], vp),
             "dj_table_column_chars_size": ([vp, i32], i64),

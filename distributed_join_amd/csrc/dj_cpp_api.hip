@@ -1509,6 +1509,23 @@ void* dj_cpp_distributed_inner_join_cols(void* comm, const dj_col_desc* lcols, i
   return result.release();
 }
 
+/* behavior hook for the parity tests: distribute a 2-column int64 table
+ * from rank 0 and collect it back (distribute_table.hpp round trip);
+ * returns the collected table on rank 0, nullptr elsewhere */
+void* dj_cpp_distribute_collect_roundtrip_i64(void* comm, const int64_t* d_keys,
+                                              const int64_t* d_pay, int64_t n)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  cudf::table_view global(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)n, d_keys),
+     column_view(data_type(type_id::INT64), (cudf::size_type)n, d_pay)});
+  auto local = distribute_table(global, (Communicator*)comm);
+  auto merged = collect_tables(local->view(), (Communicator*)comm);
+  return merged.release();
+}
+
 int dj_table_column_type(void* tbl, int i)
 {
   return (int)((cudf::table*)tbl)->get_column(i).type().id();

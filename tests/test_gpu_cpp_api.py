@@ -95,3 +95,14 @@ def test_cpp_join_empty(dj, comm):
     empty = dj.DeviceArray(1)
     got = dj.cpp_distributed_inner_join(comm, dlk, dlp, n, empty, empty, 0)
     assert all(len(c) == 0 for c in got)
+
+
+def test_distribute_and_collect_roundtrip_single_rank(dj, comm):
+    # distribute_table / collect_tables round trip (reference
+    # distribute_table.hpp:36-49); world size 1: collected == original
+    n = 100_000
+    k, p = oracle.gen_probe(n, n)
+    dk, dp = dj.DeviceArray.from_numpy(k), dj.DeviceArray.from_numpy(p)
+    t = dj.lib().dj_cpp_distribute_collect_roundtrip_i64(comm.ptr, dk.ptr, dp.ptr, n)
+    c0, c1 = dj.table_to_numpy(t)
+    assert (c0 == k).all() and (c1 == p).all()
