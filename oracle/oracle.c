@@ -60,11 +60,7 @@ ORACLE_API void oracle_gen_build(int64_t *keys,
                                  int64_t row0,
                                  int64_t nrows)
 {
-<<<<<<< HEAD
-  #pragma omp parallel for schedule(static)
-=======
 #pragma omp parallel for schedule(static)
->>>>>>> parent of de22b5e (Revert "tile-staged scatter for both bucket levels")
   for (int64_t t = 0; t < nrows; t++) {
     int64_t i = row0 + t;
     keys[t] = uniq ? dj_build_key((uint64_t)i, (uint64_t)build_n_global, rand_max, seed)
@@ -82,11 +78,7 @@ ORACLE_API void oracle_gen_probe(int64_t *keys,
                                  int64_t row0,
                                  int64_t nrows)
 {
-<<<<<<< HEAD
-  #pragma omp parallel for schedule(static)
-=======
 #pragma omp parallel for schedule(static)
->>>>>>> parent of de22b5e (Revert "tile-staged scatter for both bucket levels")
   for (int64_t t = 0; t < nrows; t++) {
     int64_t j = row0 + t;
     keys[t]   = dj_probe_key((uint64_t)j, (uint64_t)build_n_global, rand_max, selectivity, seed);
