@@ -94,7 +94,7 @@ def main():
 
     dj.require_gpu()
     L = dj.lib()
-    L.dj_set_device(local_rank)
+    L.dj_set_device(local_rank % max(L.dj_device_count(), 1))
 
     # RCCL bootstrap: rank 0's unique id broadcast over gloo
     id_bytes = None

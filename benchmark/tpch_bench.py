@@ -91,7 +91,7 @@ def main():
 
     dj.require_gpu()
     L = dj.lib()
-    L.dj_set_device(local_rank)
+    L.dj_set_device(local_rank % max(L.dj_device_count(), 1))
     id_bytes = None
     if world > 1:
         import torch
