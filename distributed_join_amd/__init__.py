@@ -230,6 +230,11 @@ def local_inner_join_global(d_lk, d_lp, ln, d_rk, d_rp, rn, cap=None):
 
 
 TYPE_INT8, TYPE_INT32, TYPE_INT64, TYPE_STRING = 1, 2, 3, 4
+# chrono reps (4-byte DAYS, 8-byte the rest) — join on the integer rep
+TYPE_TIMESTAMP_DAYS, TYPE_TIMESTAMP_S, TYPE_TIMESTAMP_MS = 5, 6, 7
+TYPE_TIMESTAMP_US, TYPE_TIMESTAMP_NS = 8, 9
+TYPE_DURATION_DAYS, TYPE_DURATION_S, TYPE_DURATION_MS = 10, 11, 12
+TYPE_DURATION_US, TYPE_DURATION_NS = 13, 14
 
 
 def table_to_numpy(tbl_ptr):
@@ -251,7 +256,7 @@ def table_to_numpy(tbl_ptr):
             if nch:
                 L.dj_memcpy_d2h(ch.ctypes.data, L.dj_table_column_chars(tbl_ptr, c), nch)
             cols.append((off, ch[:nch]))
-        elif t == TYPE_INT32:
+        elif t in (TYPE_INT32, TYPE_TIMESTAMP_DAYS, TYPE_DURATION_DAYS):
             out = np.empty(n, dtype=np.int32)
             if n:
                 L.dj_memcpy_d2h(out.ctypes.data, data, n * 4)

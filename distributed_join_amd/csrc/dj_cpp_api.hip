@@ -175,9 +175,9 @@ void sync_streams()
  * int64 keys — INT32 keys are widened, identity placement k % G preserved) */
 const int64_t* key_as_i64(cudf::column_view col, DBuf& tmp)
 {
-  if (col.type().id() == cudf::type_id::INT64) return col.head<int64_t>();
-  DJ_CHECK_ERROR(col.type().id() == cudf::type_id::INT32,
-                 "join/shuffle key column must be INT32 or INT64");
+  if (cudf::is_rep_int64(col.type())) return col.head<int64_t>();
+  DJ_CHECK_ERROR(cudf::is_rep_int32(col.type()),
+                 "join/shuffle key column must have a 4- or 8-byte integer rep");
   tmp = DBuf((size_t)col.size() * 8);
   hipLaunchKernelGGL(widen_i32_kernel, dim3(grid_for_n(col.size())), dim3(kBlock), 0,
                      dj_rt_stream(), col.head<int32_t>(), (int64_t)col.size(), tmp.i64());
@@ -454,9 +454,8 @@ void append_to_all_to_all_comm_buffers(cudf::table_view input,
 {
   check_no_compression(compression_options);
   for (cudf::size_type c = 0; c < input.num_columns(); c++) {
-    DJ_CHECK_ERROR(input.column(c).type().id() == cudf::type_id::INT64 ||
-                     input.column(c).type().id() == cudf::type_id::INT32,
-                   "all-to-all: fixed-width INT32/INT64 columns only in this build");
+    DJ_CHECK_ERROR(cudf::size_of(input.column(c).type()) > 0,
+                   "all-to-all: fixed-width (4/8-byte rep) columns only in this build");
     std::vector<int64_t> soff(send_offsets.begin(), send_offsets.end());
     all_to_all_comm_buffers.emplace_back(
       input.column(c).head<int8_t>(), output.column(c).head<int8_t>(), soff, recv_offsets,
@@ -790,11 +789,11 @@ PartitionedTable partition_table(cudf::table_view in, cudf::size_type key_col, i
                             keys_out.i64(), perm.i64(), st);
     }
     for (cudf::size_type c = 0; c < in.num_columns(); c++) {
-      if (in.column(c).type().id() == cudf::type_id::INT64) {
+      if (cudf::is_rep_int64(in.column(c).type())) {
         hipLaunchKernelGGL(gather_i64_kernel, dim3(grid_for_n(n)), dim3(kBlock), 0, st,
                            in.column(c).head<int64_t>(), perm.i64(), n,
                            (int64_t*)out->get_column(c).head());
-      } else if (in.column(c).type().id() == cudf::type_id::INT32) {
+      } else if (cudf::is_rep_int32(in.column(c).type())) {
         hipLaunchKernelGGL(gather_i32_kernel, dim3(grid_for_n(n)), dim3(kBlock), 0, st,
                            in.column(c).head<int32_t>(), perm.i64(), n,
                            (int32_t*)out->get_column(c).head());
@@ -903,7 +902,7 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
           return gather_string_column(src, idx.i64(), nout);
         auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
         if (nout > 0) {
-          if (src.type().id() == cudf::type_id::INT64)
+          if (cudf::is_rep_int64(src.type()))
             hipLaunchKernelGGL(gather_i64_kernel, dim3(grid_for_n(nout)), dim3(kBlock), 0, st,
                                src.head<int64_t>(), idx.i64(), nout, (int64_t*)col->head());
           else
@@ -1093,7 +1092,7 @@ std::unique_ptr<cudf::table> distributed_inner_join(
       rp = (const int64_t*)bt.rrecv->get_column(1).head();
     } else {
       auto widen_or_use = [&](cudf::column_view col, DBuf& w) -> const int64_t* {
-        if (col.type().id() == cudf::type_id::INT64) return col.head<int64_t>();
+        if (cudf::is_rep_int64(col.type())) return col.head<int64_t>();
         hipLaunchKernelGGL(widen_i32_kernel, dim3(grid_for_n(col.size())), dim3(kBlock), 0, st,
                            col.head<int32_t>(), (int64_t)col.size(), w.i64());
         return w.i64();
@@ -1168,7 +1167,7 @@ std::unique_ptr<cudf::table> distributed_inner_join(
         if (src.type().id() == cudf::type_id::STRING)
           return gather_string_column(src, idx.i64(), nout);
         auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
-        if (src.type().id() == cudf::type_id::INT64)
+        if (cudf::is_rep_int64(src.type()))
           hipLaunchKernelGGL(gather_i64_kernel, dim3(grid_for_n(nout)), dim3(kBlock), 0, st,
                              src.head<int64_t>(), idx.i64(), nout, (int64_t*)col->head());
         else

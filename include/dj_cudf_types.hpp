@@ -25,7 +25,26 @@ namespace cudf {
 
 using size_type = int32_t;
 
-enum class type_id : int32_t { EMPTY = 0, INT8, INT32, INT64, STRING };
+enum class type_id : int32_t {
+  EMPTY = 0,
+  INT8,
+  INT32,
+  INT64,
+  STRING,
+  /* chrono types (reference dtype coverage, compare_against_single_gpu.cu:
+   * 237-268): joins operate on the integer representation — DAYS is 4-byte,
+   * the rest 8-byte, matching cuDF's reps */
+  TIMESTAMP_DAYS,
+  TIMESTAMP_SECONDS,
+  TIMESTAMP_MILLISECONDS,
+  TIMESTAMP_MICROSECONDS,
+  TIMESTAMP_NANOSECONDS,
+  DURATION_DAYS,
+  DURATION_SECONDS,
+  DURATION_MILLISECONDS,
+  DURATION_MICROSECONDS,
+  DURATION_NANOSECONDS,
+};
 
 struct data_type {
   data_type() = default;
@@ -42,11 +61,25 @@ inline constexpr size_type size_of(data_type t)
 {
   switch (t.id()) {
     case type_id::INT8: return 1;
-    case type_id::INT32: return 4;
-    case type_id::INT64: return 8;
+    case type_id::INT32:
+    case type_id::TIMESTAMP_DAYS:
+    case type_id::DURATION_DAYS: return 4;
+    case type_id::INT64:
+    case type_id::TIMESTAMP_SECONDS:
+    case type_id::TIMESTAMP_MILLISECONDS:
+    case type_id::TIMESTAMP_MICROSECONDS:
+    case type_id::TIMESTAMP_NANOSECONDS:
+    case type_id::DURATION_SECONDS:
+    case type_id::DURATION_MILLISECONDS:
+    case type_id::DURATION_MICROSECONDS:
+    case type_id::DURATION_NANOSECONDS: return 8;
     default: return 0;  // STRING handled via children
   }
 }
+
+/* fixed-width 4-/8-byte classification used by the join/partition engine */
+inline constexpr bool is_rep_int32(data_type t) { return size_of(t) == 4; }
+inline constexpr bool is_rep_int64(data_type t) { return size_of(t) == 8; }
 
 /* mirrors cudf::hash_id as used by shuffle_on.hpp:49 */
 enum class hash_id : int32_t { HASH_IDENTITY = 0, HASH_MURMUR3 = 1 };

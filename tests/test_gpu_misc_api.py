@@ -100,3 +100,44 @@ def test_key_not_first_column(dj, comm):
     assert len(g[0]) == len(w[0])
     for a, b in zip(g, w):
         assert (a == b).all()
+
+
+def test_timestamp_duration_dtypes(dj, comm):
+    # chrono-typed key/payload columns (reference dtype coverage,
+    # compare_against_single_gpu.cu:237-268): join on the integer rep
+    n = 100_000
+    bk, bp = oracle.gen_build(n)
+    pk, pp = oracle.gen_probe(n, n)
+    d = {name: dj.DeviceArray.from_numpy(a) for name, a in
+         [("bk", bk), ("bp", bp), ("pk", pk), ("pp", pp)]}
+    cols = dj.cpp_distributed_inner_join_cols(
+        comm,
+        [(dj.TYPE_TIMESTAMP_NS, d["bk"].ptr), (dj.TYPE_DURATION_MS, d["bp"].ptr)], n,
+        [(dj.TYPE_TIMESTAMP_NS, d["pk"].ptr), (dj.TYPE_INT64, d["pp"].ptr)], n)
+    want = oracle.inner_join(bk, bp, pk, pp)
+    g = oracle.sort_rows(*[c.astype(np.int64) for c in cols])
+    w = oracle.sort_rows(*want)
+    assert len(g[0]) == len(w[0])
+    for a, b in zip(g, w):
+        assert (a == b).all()
+
+
+def test_timestamp_days_int32_rep(dj, comm):
+    n = 50_000
+    bk64, bp64 = oracle.gen_build(n, rand_max=min(2 * n, 2**31 - 2))
+    pk64, pp64 = oracle.gen_probe(n, n, rand_max=min(2 * n, 2**31 - 2))
+    d_bk = _upload_i32(dj, bk64.astype(np.int32))
+    d_pk = _upload_i32(dj, pk64.astype(np.int32))
+    d_bp = _upload_i32(dj, bp64.astype(np.int32))
+    d_pp = _upload_i32(dj, pp64.astype(np.int32))
+    cols = dj.cpp_distributed_inner_join_cols(
+        comm, [(dj.TYPE_TIMESTAMP_DAYS, d_bk), (dj.TYPE_TIMESTAMP_DAYS, d_bp)], n,
+        [(dj.TYPE_TIMESTAMP_DAYS, d_pk), (dj.TYPE_TIMESTAMP_DAYS, d_pp)], n)
+    want = oracle.inner_join(bk64, bp64, pk64, pp64)
+    g = oracle.sort_rows(*[c.astype(np.int64) for c in cols])
+    w = oracle.sort_rows(*want)
+    assert len(g[0]) == len(w[0])
+    for a, b in zip(g, w):
+        assert (a == b).all()
+    for p in (d_bk, d_pk, d_bp, d_pp):
+        dj.lib().dj_dfree(p)
