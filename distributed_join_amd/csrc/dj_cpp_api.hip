@@ -23,7 +23,9 @@
 #include <rccl/rccl.h>
 
 #include <algorithm>
+#include <chrono>
 #include <cstring>
+#include <iostream>
 #include <map>
 #include <mutex>
 #include <unordered_map>
@@ -1006,11 +1008,22 @@ std::unique_ptr<cudf::table> distributed_inner_join(
   DJ_CHECK_ERROR(nparts <= dj::kMaxPartitions,
                  "world_size x over_decom_factor must be <= 64");
 
+  /* report_timing mirrors the reference's per-phase prints
+   * (distributed_join.cpp:120-130, 235-240) */
+  auto now = [] { return std::chrono::steady_clock::now(); };
+  auto ms = [](auto a, auto b) {
+    return std::chrono::duration<double, std::milli>(b - a).count();
+  };
+  auto t_part0 = now();
+
   /* rank-level stable partition, seed 12345678 (distributed_join.cpp:211-226) */
   PartitionedTable lpart =
     partition_table(left, left_on[0], nparts, DJ_HASH_MURMUR3, DJ_SEED_INTRA);
   PartitionedTable rpart =
     partition_table(right, right_on[0], nparts, DJ_HASH_MURMUR3, DJ_SEED_INTRA);
+  if (report_timing)
+    std::cout << "Rank " << communicator->mpi_rank << ": hash partition takes "
+              << ms(t_part0, now()) << "ms" << std::endl;
 
   /* --- pre-phase: size exchanges + every allocation (pool allocs sync the
    * compute stream, so none happen inside the pipeline) --- */
@@ -1071,12 +1084,17 @@ std::unique_ptr<cudf::table> distributed_inner_join(
    * launch_communication's stop; join kernels are enqueued without syncs) —
    * replaces the reference's join thread + atomic-flag busy wait
    * (distributed_join.cpp:283-329) with stream ordering --- */
+  auto t_pipe0 = now();
   for (int b = 0; b < over_decom_factor; b++) {
     Batch& bt = batches[b];
+    auto t_comm0 = now();
     bt.latoa->launch_communication(bt.lrecv->mutable_view(), report_timing,
                                    preallocated_pinned_buffer);
     bt.ratoa->launch_communication(bt.rrecv->mutable_view(), report_timing,
                                    preallocated_pinned_buffer);
+    if (report_timing)
+      std::cout << "Rank " << communicator->mpi_rank << ": all-to-all communication (batch "
+                << b << ") takes " << ms(t_comm0, now()) << "ms" << std::endl;
     if (bt.ln == 0 || bt.rn == 0) {
       DJ_HIP_CALL(hipMemsetAsync(bt.meta.p, 0, 16, st));
       continue;
@@ -1113,6 +1131,10 @@ std::unique_ptr<cudf::table> distributed_inner_join(
                                  (int*)((char*)bt.meta.p + 12), scratch.p);
   }
   DJ_HIP_CALL(hipStreamSynchronize(st));
+  if (report_timing)
+    std::cout << "Rank " << communicator->mpi_rank
+              << ": communication + local join pipeline takes " << ms(t_pipe0, now()) << "ms"
+              << std::endl;
 
   /* --- finalize: counts, rare redo (skew overflow / cap exceeded), column
    * assembly, concat --- */
@@ -1183,7 +1205,12 @@ std::unique_ptr<cudf::table> distributed_inner_join(
     }
     batch_results.push_back(std::make_unique<cudf::table>(std::move(cols)));
   }
-  return concat_tables(batch_results);
+  auto t_cat0 = now();
+  auto result = concat_tables(batch_results);
+  if (report_timing && over_decom_factor > 1)
+    std::cout << "Rank " << communicator->mpi_rank << ": concatenation takes "
+              << ms(t_cat0, now()) << "ms" << std::endl;
+  return result;
 }
 
 /* ------------------------------------------------------------- shuffle_on */
