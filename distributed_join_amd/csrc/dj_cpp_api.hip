@@ -1068,9 +1068,9 @@ std::unique_ptr<cudf::table> distributed_inner_join(
     bt.o3 = DBuf((size_t)bt.cap * 8);
     bt.meta = DBuf(16);
     if (!fast2) {
-      if (bt.ln && left.column(left_on[0]).type().id() == cudf::type_id::INT32)
+      if (bt.ln && cudf::is_rep_int32(left.column(left_on[0]).type()))
         bt.lkw = DBuf((size_t)bt.ln * 8);
-      if (bt.rn && right.column(right_on[0]).type().id() == cudf::type_id::INT32)
+      if (bt.rn && cudf::is_rep_int32(right.column(right_on[0]).type()))
         bt.rkw = DBuf((size_t)bt.rn * 8);
       bt.liota = DBuf((size_t)std::max<int64_t>(bt.ln, 1) * 8);
       bt.riota = DBuf((size_t)std::max<int64_t>(bt.rn, 1) * 8);
