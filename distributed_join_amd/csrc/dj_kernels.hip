@@ -505,14 +505,16 @@ void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_tabl
 
 constexpr int BUCKET_BLOCKS = kBucketBlocks;  // chunking blocks for pass A (2/CU); scratch sizing uses the same constant
 constexpr int BUCKET_THREADS = 1024;
-constexpr int JOIN_LDS_SLOTS = 4096;        // 64 KiB of longlong2 pairs -> 2 blocks/CU
+constexpr int JOIN_LDS_SLOTS = 2048;        // 32 KiB of longlong2 pairs
+constexpr int JOIN_STAGE_ROWS = 1024;       // 32 KiB staged output rows (4 x i64)
+/* table + stage = 64 KiB + 16 B -> 2 blocks/CU */
 constexpr int SUB_BUCKETS = 256;            // pass-B fanout (fixed)
 
 int bucket_count_for(int64_t ln, int64_t rn)
 {
   int64_t maxn = ln > rn ? ln : rn;
   int64_t B = 256;
-  while (B < 262144 && maxn / B > 1600) B <<= 1;
+  while (B < 262144 && maxn / B > 800) B <<= 1;
   return (int)B;
 }
 
@@ -738,7 +740,13 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
                                        base, gcur, out_pairs);
 }
 
-/* ---- fused per-bucket LDS build + two-phase probe (count, then write) ---- */
+/* ---- fused per-bucket LDS build + single-pass probe with staged output ----
+ * The probe emits matches into an LDS output stage (SoA, 4 x i64 per row) and
+ * flushes once per bucket with ONE global counter atomic + coalesced column
+ * writes. (The earlier two-phase count-then-rewalk design paid a full extra
+ * probe walk — ~0.8 ms at 100 M rows, profiles/r01_ablation.txt.) Buckets
+ * whose matches exceed the stage spill directly to the global counter per
+ * lane — correct for any duplication factor, slower only on such buckets. */
 __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
   const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
@@ -749,10 +757,11 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbl = (longlong2*)smem;
-  long long* base_sh = (long long*)(smem + JOIN_LDS_SLOTS * sizeof(longlong2));
-  uint32_t* total_sh = (uint32_t*)(base_sh + 1);
-  uint32_t* cur_sh = total_sh + 1;
+  int64_t* stage = (int64_t*)(tbl + JOIN_LDS_SLOTS);  // SoA: stage[c*JOIN_STAGE_ROWS + i]
+  long long* base_sh = (long long*)(stage + 4 * JOIN_STAGE_ROWS);
+  uint32_t* cur_sh = (uint32_t*)(base_sh + 1);
   const uint32_t smask = JOIN_LDS_SLOTS - 1;
+  constexpr int S = JOIN_STAGE_ROWS;
 
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
     const int64_t l0 = loff[b], l1 = loff[b + 1];
@@ -767,10 +776,7 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
       continue;
     }
     for (int s = threadIdx.x; s < JOIN_LDS_SLOTS; s += blockDim.x) tbl[s].x = kEmptyKey;
-    if (threadIdx.x == 0) {
-      *total_sh = 0;
-      *cur_sh = 0;
-    }
+    if (threadIdx.x == 0) *cur_sh = 0;
     __syncthreads();
     /* build */
     for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
@@ -790,49 +796,49 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
       tbl[slot].y = row.y;
     }
     __syncthreads();
-    /* probe phase 1: count my matches */
-    uint32_t my = 0;
+    /* single-pass probe: matches append to the LDS stage */
     for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
-      int64_t key = rrows[j].x;
-      uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
+      longlong2 prow = rrows[j];
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)prow.x) & smask;
       for (;;) {
         longlong2 e = tbl[slot];
         if (e.x == kEmptyKey) break;
-        if (e.x == key) my++;
+        if (e.x == prow.x) {
+          uint32_t pos = atomicAdd(cur_sh, 1u);
+          if (pos < (uint32_t)S) {
+            stage[0 * S + pos] = prow.x;
+            stage[1 * S + pos] = e.y;
+            stage[2 * S + pos] = prow.x;
+            stage[3 * S + pos] = prow.y;
+          } else {
+            /* stage overflow (heavy duplication): spill directly; the
+             * compiler wave-aggregates the counter atomic */
+            long long idx = (long long)atomicAdd(counter, 1ull);
+            if (idx < cap) {
+              out0[idx] = prow.x;
+              out1[idx] = e.y;
+              out2[idx] = prow.x;
+              out3[idx] = prow.y;
+            }
+          }
+        }
         slot = (slot + 1) & smask;
       }
     }
-    if (my) atomicAdd(total_sh, my);
     __syncthreads();
-    /* ONE global atomic per bucket reserves the output range */
-    if (threadIdx.x == 0 && *total_sh)
-      *base_sh = (long long)atomicAdd(counter, (unsigned long long)*total_sh);
+    const uint32_t total = min(*cur_sh, (uint32_t)S);
+    if (threadIdx.x == 0 && total)
+      *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
     __syncthreads();
-    if (*total_sh) {
+    if (total) {
       const long long base = *base_sh;
-      uint32_t w = my ? atomicAdd(cur_sh, my) : 0;
-      /* probe phase 2: rewalk in the same order, write */
-      if (my) {
-        for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
-          longlong2 prow = rrows[j];
-          int64_t key = prow.x;
-          int64_t payload = prow.y;
-          uint32_t slot = (uint32_t)dj_mix64((uint64_t)key) & smask;
-          for (;;) {
-            longlong2 e = tbl[slot];
-            if (e.x == kEmptyKey) break;
-            if (e.x == key) {
-              long long idx = base + (long long)w;
-              if (idx < cap) {
-                out0[idx] = key;
-                out1[idx] = e.y;
-                out2[idx] = key;
-                out3[idx] = payload;
-              }
-              w++;
-            }
-            slot = (slot + 1) & smask;
-          }
+      for (uint32_t i = threadIdx.x; i < total; i += blockDim.x) {
+        long long idx = base + (long long)i;
+        if (idx < cap) {
+          out0[idx] = stage[0 * S + i];
+          out1[idx] = stage[1 * S + i];
+          out2[idx] = stage[2 * S + i];
+          out3[idx] = stage[3 * S + i];
         }
       }
     }
@@ -885,7 +891,7 @@ void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* 
               int* d_any_overflow, int* d_error, hipStream_t s)
 {
   int grid = B < 8192 ? B : 8192;
-  size_t lds = JOIN_LDS_SLOTS * sizeof(longlong2) + 16;
+  size_t lds = JOIN_LDS_SLOTS * sizeof(longlong2) + 4 * JOIN_STAGE_ROWS * sizeof(int64_t) + 16;
   hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lrows,
                      d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
                      (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,

@@ -70,7 +70,7 @@ void join_probe(const int64_t* d_rk, const int64_t* d_rp, int64_t rn, const int6
  * "bucketed LDS join" comment block) ----- */
 constexpr int kBucketBlocks = 512;
 constexpr int kSubBuckets = 256;
-constexpr int kJoinBucketRowCap = 3072;  // 75% of the 4096-slot LDS table
+constexpr int kJoinBucketRowCap = 1536;  // 75% of the 2048-slot LDS table
 int bucket_count_for(int64_t ln, int64_t rn);
 /* Two-level non-stable partition into B buckets (B = PA*256, PA<=1024) of
  * interleaved 16 B {key,payload} pairs. d_tmp_pairs: longlong2[n] pass-A

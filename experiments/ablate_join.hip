@@ -232,6 +232,8 @@ static double bench_ms(F launch, int iters)
 
 int main()
 {
+  setbuf(stdout, NULL);
+  printf("start\n");
   const int64_t n = 100'000'000;
   const int PA = 256, B = 65536;
   longlong2 *in, *out;
@@ -241,6 +243,7 @@ int main()
   CHECK(hipMalloc(&out, n * 16));
   CHECK(hipMalloc(&segoff, (PA + 1) * 8));
   CHECK(hipMalloc(&sink, 4));
+  printf("allocs done\n");
   hipLaunchKernelGGL(gen_pairs, dim3(2048), dim3(256), 0, 0, in, n, 7);
   {
     /* equal segments for the ablation (uniform hash -> near-equal anyway) */
@@ -248,6 +251,7 @@ int main()
     for (int i = 0; i <= PA; i++) h[i] = n * i / PA;
     CHECK(hipMemcpy(segoff, h, (PA + 1) * 8, hipMemcpyHostToDevice));
   }
+  printf("segoff done\n");
   size_t lds = TILE * 16 + 4 * SUB * 4;
   double t;
 #define RUN(c, r, s, f, name)                                                            \
@@ -266,13 +270,17 @@ int main()
 
   /* join ablation over bucketed data: reuse out as "bucketed" input (the
    * real pipeline's distribution); offsets B+1 from uniform split */
+  printf("subpart done\n");
   int64_t* boff;
+  printf("A\n");
   CHECK(hipMalloc(&boff, (B + 1) * 8));
+  printf("B\n");
   {
     int64_t* h = new int64_t[B + 1];
     for (int i = 0; i <= B; i++) h[i] = n * (int64_t)i / B;
     CHECK(hipMemcpy(boff, h, (B + 1) * 8, hipMemcpyHostToDevice));
   }
+  printf("C\n");
   int64_t *o0, *o1, *o2, *o3;
   unsigned long long* ctr;
   int64_t cap = n + (n >> 3);
@@ -281,12 +289,13 @@ int main()
   CHECK(hipMalloc(&o2, cap * 8));
   CHECK(hipMalloc(&o3, cap * 8));
   CHECK(hipMalloc(&ctr, 8));
+  printf("D\n");
   size_t jlds = 4096 * 16 + 16;
 #define JRUN(i, bl, p1, p2, name)                                                      \
   CHECK(hipMemset(ctr, 0, 8));                                                         \
   t = bench_ms(                                                                        \
     [&] {                                                                              \
-      CHECK(hipMemsetAsync(ctr, 0, 8));                                                \
+      (void)hipMemsetAsync(ctr, 0, 8);                                                 \
       hipLaunchKernelGGL((join_ablate<i, bl, p1, p2>), dim3(8192), dim3(THREADS),      \
                          jlds, 0, out, boff, out, boff, B, o0, o1, o2, o3, cap, ctr);  \
     },                                                                                 \
