@@ -9,6 +9,7 @@ over-decomposition batching (distributed_join.cpp:244-329) which must not
 change the result.
 """
 import numpy as np
+import os
 import pytest
 
 import oracle
@@ -106,3 +107,24 @@ def test_distribute_and_collect_roundtrip_single_rank(dj, comm):
     t = dj.lib().dj_cpp_distribute_collect_roundtrip_i64(comm.ptr, dk.ptr, dp.ptr, n)
     c0, c1 = dj.table_to_numpy(t)
     assert (c0 == k).all() and (c1 == p).all()
+
+
+def test_multirank_loopback():
+    """Multi-rank C++ orchestration on one GPU: 2 and 4 ranks as threads with
+    a user-style loopback Communicator (tests/cpp/multirank_loopback.cpp).
+    Concatenated multi-rank result must equal the single-rank result."""
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exe = os.path.join(repo, "tests", "cpp", "multirank_loopback")
+    if not os.path.exists(exe):
+        subprocess.run(
+            ["hipcc", "--offload-arch=gfx950", "-O2", "-std=c++17",
+             "-I", os.path.join(repo, "include"),
+             os.path.join(repo, "tests", "cpp", "multirank_loopback.cpp"),
+             "-o", exe, "-L", os.path.join(repo, "distributed_join_amd"), "-ldistjoin",
+             "-Wl,-rpath," + os.path.join(repo, "distributed_join_amd")],
+            check=True, capture_output=True)
+    for args in (["2", "2"], ["4", "1"]):
+        r = subprocess.run([exe] + args, capture_output=True, text=True, timeout=240)
+        assert r.returncode == 0, r.stdout + r.stderr
+        assert "MULTIRANK OK" in r.stdout
