@@ -17,7 +17,9 @@
 #include "all_to_all_comm.hpp"
 #include "communicator.hpp"
 #include "compression.hpp"
+#include "distribute_table.hpp"
 #include "distributed_join.hpp"
+#include "shuffle_on.hpp"
 
 #include "../../distributed_join_amd/csrc/dj_rng.h"
 
@@ -208,6 +210,67 @@ int main(int argc, char** argv)
   for (int r = 0; r < G; r++) {
     got_sum += sums[r];
     got_rows += rows[r];
+  }
+
+  /* shuffle_on (identity hash): every received key must be ≡ rank (mod G) —
+   * the reference's placement pin (test_shuffle_on.cpp:78-83) — and
+   * distribute_table/collect_tables must round-trip the global table. */
+  {
+    Mailbox mb2;
+    std::vector<int> shuffle_ok(G, 0), roundtrip_ok(G, 0);
+    std::vector<std::thread> th2;
+    const int64_t n = 100000;
+    for (int r = 0; r < G; r++) {
+      th2.emplace_back([&, r] {
+        CHECK(hipSetDevice(0));
+        LoopbackCommunicator comm(r, G, &mb2);
+        /* per-rank input: arbitrary int64 keys */
+        auto t = make_rank_tables(n, n / G, r * (n / G), false);
+        auto opts = generate_compression_options_distributed(t->view(), false);
+        auto shuffled = shuffle_on(t->view(), {0}, &comm, opts,
+                                   cudf::hash_id::HASH_IDENTITY, 0);
+        int64_t m = shuffled->num_rows();
+        std::vector<int64_t> keys(m);
+        CHECK(hipMemcpy(keys.data(), shuffled->get_column(0).head(), m * 8,
+                        hipMemcpyDeviceToHost));
+        bool ok = true;
+        for (int64_t i = 0; i < m; i++)
+          ok &= ((uint32_t)((uint64_t)keys[i] & 0xFFFFFFFFu) % (uint32_t)G == (uint32_t)r);
+        shuffle_ok[r] = ok ? 1 : 0;
+
+        /* distribute/collect round trip (global table significant on root) */
+        std::unique_ptr<cudf::table> global;
+        cudf::table_view gview;
+        if (r == 0) {
+          global = make_rank_tables(n, n, 0, true);
+          gview = global->view();
+        }
+        auto local = distribute_table(gview, &comm);
+        auto merged = collect_tables(local->view(), &comm);
+        if (r == 0) {
+          std::vector<int64_t> a(n), b(n);
+          CHECK(hipMemcpy(a.data(), global->get_column(0).head(), n * 8,
+                          hipMemcpyDeviceToHost));
+          CHECK(hipMemcpy(b.data(), merged->get_column(0).head(), n * 8,
+                          hipMemcpyDeviceToHost));
+          roundtrip_ok[0] = (merged->num_rows() == n && a == b) ? 1 : 0;
+        } else {
+          roundtrip_ok[r] = (merged == nullptr) ? 1 : 0;
+        }
+      });
+    }
+    for (auto& t : th2) t.join();
+    for (int r = 0; r < G; r++) {
+      if (!shuffle_ok[r]) {
+        printf("SHUFFLE PLACEMENT MISMATCH rank %d\n", r);
+        return 1;
+      }
+      if (!roundtrip_ok[r]) {
+        printf("DISTRIBUTE/COLLECT MISMATCH rank %d\n", r);
+        return 1;
+      }
+    }
+    printf("shuffle identity placement + distribute/collect roundtrip OK\n");
   }
   printf("single-rank: rows=%lld sum=%llx\n", (long long)want_rows,
          (unsigned long long)want_sum);
