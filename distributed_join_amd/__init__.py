@@ -84,6 +84,9 @@ def lib():
             "dj_cpp_comm_destroy": ([vp], None),
             "dj_cpp_distributed_inner_join_i64": ([vp, vp, vp, i64, vp, vp, i64, i32, i32], vp),
             "dj_cpp_shuffle_on_i64": ([vp, vp, vp, i64, i32, u32], vp),
+            "dj_cpp_shuffle_on_i64_comp": ([vp, vp, vp, i64, i32, u32, i32], vp),
+            "dj_cpp_distributed_inner_join_i64_opts": ([vp, vp, vp, i64, vp, vp, i64,
+                                                        i32, i32, i32], vp),
             "dj_gen_test_strings": ([vp, i64, ctypes.POINTER(ctypes.c_void_p),
                                      ctypes.POINTER(ctypes.c_void_p),
                                      ctypes.POINTER(ctypes.c_int64)], None),
@@ -340,8 +343,13 @@ def cpp_distributed_inner_join_cols(comm, lcols, ln, rcols, rn, key_l=0, key_r=0
     return table_to_numpy(t)
 
 
-def cpp_shuffle_on(comm, d_keys, d_pay, n, hash_fn=HASH_MURMUR3, seed=0):
-    t = lib().dj_cpp_shuffle_on_i64(comm.ptr, d_keys.ptr, d_pay.ptr, n, hash_fn, seed)
+def cpp_shuffle_on(comm, d_keys, d_pay, n, hash_fn=HASH_MURMUR3, seed=0,
+                   compression=False):
+    if compression:
+        t = lib().dj_cpp_shuffle_on_i64_comp(comm.ptr, d_keys.ptr, d_pay.ptr, n, hash_fn,
+                                             seed, 1)
+    else:
+        t = lib().dj_cpp_shuffle_on_i64(comm.ptr, d_keys.ptr, d_pay.ptr, n, hash_fn, seed)
     return table_to_numpy(t)
 
 

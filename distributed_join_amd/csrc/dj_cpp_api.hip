@@ -186,13 +186,28 @@ const int64_t* key_as_i64(cudf::column_view col, DBuf& tmp)
   return tmp.i64();
 }
 
+void validate_compression(std::vector<ColumnCompressionOptions> const& opts)
+{
+  for (auto& o : opts) {
+    if (o.compression_method == CompressionMethod::none) continue;
+    if (o.compression_method == CompressionMethod::lz4)
+      throw std::runtime_error("lz4 compression is not implemented (cascaded or none)");
+    if (o.cascaded_format.num_RLEs > 0)
+      throw std::runtime_error(
+        "cascaded RLE passes are not implemented (delta/bitpack only; dj_compress.hip)");
+    if (o.cascaded_format.num_deltas < 0 || o.cascaded_format.num_deltas > 1)
+      throw std::runtime_error("cascaded num_deltas must be 0 or 1");
+  }
+}
+/* kept under the old name for call sites that still require none (public
+ * all_to_all_comm exec path — compression happens in launch_communication) */
 void check_no_compression(std::vector<ColumnCompressionOptions> const& opts)
 {
   for (auto& o : opts)
     if (o.compression_method != CompressionMethod::none)
       throw std::runtime_error(
-        "compression is not implemented in this build (CompressionMethod::none only; "
-        "SURVEY.md §8f rank 3)");
+        "this entry point executes uncompressed buffers only; use "
+        "AllToAllCommunicator::launch_communication for cascaded compression");
 }
 
 Communicator* g_default_comm = nullptr;
@@ -356,11 +371,21 @@ Communicator* default_communicator()
 std::vector<ColumnCompressionOptions> generate_compression_options_distributed(
   cudf::table_view input, bool compression)
 {
-  if (compression)
-    throw std::runtime_error(
-      "compression is not implemented in this build (CompressionMethod::none only)");
-  return std::vector<ColumnCompressionOptions>((size_t)input.num_columns(),
-                                               ColumnCompressionOptions(CompressionMethod::none));
+  if (!compression)
+    return std::vector<ColumnCompressionOptions>(
+      (size_t)input.num_columns(), ColumnCompressionOptions(CompressionMethod::none));
+  /* fixed policy standing in for the reference's nvcomp auto-selector
+   * (compression.hpp:253-292): bitpack-only cascaded for fixed-width columns
+   * and the row-size wire of STRING columns; chars are never compressed
+   * (reference policy, compression.cpp:44-60) */
+  std::vector<ColumnCompressionOptions> opts;
+  nvcompCascadedFormatOpts bp{};
+  bp.num_RLEs = 0;
+  bp.num_deltas = 0;
+  bp.use_bp = 1;
+  for (cudf::size_type c = 0; c < input.num_columns(); c++)
+    opts.emplace_back(CompressionMethod::cascaded, bp);
+  return opts;
 }
 
 /* ------------------------------------------------------ CommunicationGroup */
@@ -542,7 +567,7 @@ AllToAllCommunicator::AllToAllCommunicator(
     send_offsets(std::move(offsets)),
     compression_options(std::move(compression_options_))
 {
-  check_no_compression(compression_options);
+  validate_compression(compression_options);
   DJ_CHECK_ERROR((int)send_offsets.size() == comm_group.size() + 1,
                  "AllToAllCommunicator: offsets must have comm_group.size()+1 entries");
   communicate_sizes(send_offsets, recv_offsets, comm_group, communicator);
@@ -660,42 +685,197 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
                                                 bool report_timing,
                                                 void* preallocated_pinned_buffer)
 {
-  std::vector<AllToAllCommBuffer> bufs;
+  const int G = comm_group.size();
+  const int me = comm_group.get_local_idx();
+  const bool include_self = !explicit_copy_to_current_rank;
+  hipStream_t st = dj_rt_stream();
   std::vector<int64_t> soff(send_offsets.begin(), send_offsets.end());
+
+  /* wire plan: per column one or two buffers, each either plain or
+   * cascaded-compressed (the reference's compression branch,
+   * all_to_all_comm.cpp:358-478: compress -> exchange compressed sizes ->
+   * exchange compressed slices -> decompress receiver-side) */
+  struct Plain {
+    const int8_t* src;
+    int8_t* dst;
+    int esize;
+    const std::vector<int64_t>* soff;
+    const std::vector<int64_t>* roff;
+  };
+  struct Comp {
+    const uint8_t* src;       // uncompressed source buffer
+    uint8_t* dst;             // uncompressed destination
+    int esize;
+    int delta, bp;
+    const std::vector<int64_t>* soff;  // element offsets (source)
+    const std::vector<int64_t>* roff;  // element offsets (destination)
+    DBuf comp;                         // compressed send slices (bound-sized slots)
+    DBuf bits;                         // u32 per peer
+    std::vector<size_t> slot;          // slot byte offsets (G+1)
+    std::vector<int64_t> csize;        // compressed bytes per peer
+    std::vector<int64_t> csend_off;    // prefix of csize (G+1)
+    std::vector<int64_t> crecv_off;    // recv byte offsets (G+1)
+    DBuf comp_recv;
+  };
+  std::vector<Plain> plains;
+  std::vector<Comp> comps;
+  auto add_buffer = [&](const void* src, void* dst, int esize,
+                        const std::vector<int64_t>* so, const std::vector<int64_t>* ro,
+                        const ColumnCompressionOptions& opt) {
+    if (opt.compression_method == CompressionMethod::cascaded && (esize == 4 || esize == 8)) {
+      Comp c;
+      c.src = (const uint8_t*)src;
+      c.dst = (uint8_t*)dst;
+      c.esize = esize;
+      c.delta = opt.cascaded_format.num_deltas;
+      c.bp = opt.cascaded_format.use_bp;
+      c.soff = so;
+      c.roff = ro;
+      comps.push_back(std::move(c));
+    } else {
+      plains.push_back(Plain{(const int8_t*)src, (int8_t*)dst, esize, so, ro});
+    }
+  };
+
+  /* per-column element-offset vectors must outlive the exchange */
+  std::vector<std::vector<int64_t>> offsets_storage;
+  offsets_storage.reserve(input_table.num_columns() * 2 + 2);
+  offsets_storage.push_back(soff);
+  const std::vector<int64_t>* rows_soff = &offsets_storage.back();
   for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
     auto in = input_table.column(c);
     auto out = communicated_table.column(c);
     if (in.type().id() == cudf::type_id::STRING) {
-      /* two wire buffers per string column: row SIZES (row offsets) and the
-       * chars bytes (char offsets) — strings_column.cu strategy */
-      bufs.emplace_back(strings->sizes_to_send[c].p, strings->sizes_received[c].p, soff,
-                        recv_offsets, cudf::data_type(cudf::type_id::INT32),
-                        CompressionMethod::none, nvcompCascadedFormatOpts{});
-      bufs.emplace_back(in.chars(), out.chars(), strings->send_char_offsets[c],
-                        strings->recv_char_offsets[c], cudf::data_type(cudf::type_id::INT8),
-                        CompressionMethod::none, nvcompCascadedFormatOpts{});
+      /* sizes on the wire (may be compressed); chars never compressed
+       * (reference policy, compression.cpp:44-60) */
+      add_buffer(strings->sizes_to_send[c].p, strings->sizes_received[c].p, 4, rows_soff,
+                 &recv_offsets, compression_options[c]);
+      plains.push_back(Plain{(const int8_t*)in.chars(), (int8_t*)out.chars(), 1,
+                             &strings->send_char_offsets[c], &strings->recv_char_offsets[c]});
     } else {
-      bufs.emplace_back(in.head<int8_t>(), out.head<int8_t>(), soff, recv_offsets, in.type(),
-                        compression_options[c].compression_method,
-                        compression_options[c].cascaded_format);
+      add_buffer(in.head<int8_t>(), out.head<int8_t>(), cudf::size_of(in.type()), rows_soff,
+                 &recv_offsets, compression_options[c]);
     }
   }
+
+  /* compress send slices (compute stream), then read final sizes */
+  int64_t max_recv_count = 0;
+  if (!comps.empty()) {
+    for (auto& c : comps) {
+      c.slot.resize(G + 1);
+      size_t acc = 0;
+      for (int i = 0; i < G; i++) {
+        c.slot[i] = acc;
+        int64_t cnt = (*c.soff)[i + 1] - (*c.soff)[i];
+        acc += (dj::compress_bound(cnt, c.esize) + 255) & ~(size_t)255;
+      }
+      c.slot[G] = acc;
+      c.comp = DBuf(acc);
+      c.bits = DBuf((size_t)G * 4);
+      for (int i = 0; i < G; i++) {
+        if (i == me && !include_self) continue;
+        int64_t cnt = (*c.soff)[i + 1] - (*c.soff)[i];
+        dj::compress_slice_async(c.src + (*c.soff)[i] * c.esize, cnt, c.esize, c.delta, c.bp,
+                                 (uint8_t*)c.comp.p + c.slot[i], (uint32_t*)c.bits.p + i, st);
+      }
+    }
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    for (auto& c : comps) {
+      std::vector<uint32_t> bits(G, 0);
+      DJ_HIP_CALL(hipMemcpy(bits.data(), c.bits.p, (size_t)G * 4, hipMemcpyDeviceToHost));
+      c.csize.assign(G, 0);
+      c.csend_off.assign(G + 1, 0);
+      for (int i = 0; i < G; i++) {
+        int64_t cnt = (*c.soff)[i + 1] - (*c.soff)[i];
+        c.csize[i] = (i == me && !include_self)
+                       ? 0
+                       : (int64_t)dj::compressed_size_from_bits(cnt, c.esize, bits[i]);
+        c.csend_off[i + 1] = c.csend_off[i] + c.csize[i];
+      }
+      /* exchange compressed byte counts (communicate_sizes over bytes) */
+      communicate_sizes(c.csend_off, c.crecv_off, comm_group, communicator);
+      c.comp_recv = DBuf((size_t)std::max<int64_t>(c.crecv_off.back(), 1));
+      for (int i = 0; i < G; i++)
+        max_recv_count = std::max(max_recv_count, (*c.roff)[i + 1] - (*c.roff)[i]);
+    }
+  }
+
   {
     dj_timing::Scope t(DJ_PHASE_COMM, dj_rt_comm_stream());
     communicator->start();
-    all_to_all_comm(bufs, comm_group, communicator, !explicit_copy_to_current_rank,
-                    report_timing, preallocated_pinned_buffer);
+    for (auto& b : plains) {
+      for (int i = 0; i < G; i++) {
+        int64_t scount = (*b.soff)[i + 1] - (*b.soff)[i];
+        int64_t rcount = (*b.roff)[i + 1] - (*b.roff)[i];
+        const int8_t* src = b.src + (*b.soff)[i] * b.esize;
+        int8_t* dst = b.dst + (*b.roff)[i] * b.esize;
+        if (i == me) {
+          if (include_self && scount > 0)
+            DJ_HIP_CALL(hipMemcpyAsync(dst, src, (size_t)scount * b.esize,
+                                       hipMemcpyDeviceToDevice, dj_rt_comm_stream()));
+          continue;
+        }
+        int peer = comm_group.get_global_rank(i);
+        communicator->send(src, scount, b.esize, peer);
+        communicator->recv(dst, rcount, b.esize, peer);
+      }
+    }
+    for (auto& c : comps) {
+      for (int i = 0; i < G; i++) {
+        int64_t rbytes = c.crecv_off[i + 1] - c.crecv_off[i];
+        if (i == me) {
+          if (include_self && c.csize[i] > 0)
+            DJ_HIP_CALL(hipMemcpyAsync((uint8_t*)c.comp_recv.p + c.crecv_off[i],
+                                       (uint8_t*)c.comp.p + c.slot[i], (size_t)c.csize[i],
+                                       hipMemcpyDeviceToDevice, dj_rt_comm_stream()));
+          continue;
+        }
+        int peer = comm_group.get_global_rank(i);
+        communicator->send((uint8_t*)c.comp.p + c.slot[i], c.csize[i], 1, peer);
+        communicator->recv((uint8_t*)c.comp_recv.p + c.crecv_off[i], rbytes, 1, peer);
+      }
+    }
     communicator->stop();  // blocks the host (all_to_all_comm.hpp:331 contract)
   }
-  postprocess_all_to_all_comm(bufs, comm_group, communicator, !explicit_copy_to_current_rank,
-                              report_timing);
+
+  /* decompress received slices */
+  if (!comps.empty()) {
+    DBuf scratch(dj::compress_scratch_bytes(std::max<int64_t>(max_recv_count, 1)));
+    for (auto& c : comps) {
+      for (int i = 0; i < G; i++) {
+        int64_t rbytes = c.crecv_off[i + 1] - c.crecv_off[i];
+        int64_t rcount = (*c.roff)[i + 1] - (*c.roff)[i];
+        if (rbytes == 0 || rcount == 0) continue;
+        dj::CompSliceHeader h;
+        DJ_HIP_CALL(hipMemcpy(&h, (uint8_t*)c.comp_recv.p + c.crecv_off[i], 16,
+                              hipMemcpyDeviceToHost));
+        DJ_CHECK_ERROR((int64_t)h.count == rcount,
+                       "cascaded: received slice count mismatch");
+        dj::decompress_slice_async((uint8_t*)c.comp_recv.p + c.crecv_off[i], h, c.esize,
+                                   c.dst + (*c.roff)[i] * c.esize, scratch.p, st);
+      }
+    }
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    if (report_timing) {
+      int64_t raw = 0, wire = 0;
+      for (auto& c : comps) {
+        for (int i = 0; i < G; i++) {
+          if (i == me) continue;
+          raw += ((*c.soff)[i + 1] - (*c.soff)[i]) * c.esize;
+          wire += c.csize[i];
+        }
+      }
+      if (raw > 0)
+        std::cout << "Rank " << communicator->mpi_rank << ": cascaded wire "
+                  << wire / 1.0e6 << "MB from " << raw / 1.0e6 << "MB ("
+                  << (double)raw / (wire ? wire : 1) << "x)" << std::endl;
+    }
+  }
+  (void)preallocated_pinned_buffer;
+
   /* receiver-side: rebuild string offsets from the received sizes by scan
    * with offset[0]=0 (strings_column.cu:111-131) */
   if (strings) {
-    /* enqueued on the compute stream without a host sync so the offsets
-     * rebuild pipelines with the local join under over-decomposition;
-     * compute-stream consumers are ordered behind it */
-    hipStream_t st = dj_rt_stream();
     const int64_t n_recv = recv_offsets.back();
     for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
       if (input_table.column(c).type().id() != cudf::type_id::STRING) continue;
@@ -999,8 +1179,8 @@ std::unique_ptr<cudf::table> distributed_inner_join(
   (void)nvlink_domain_size;  // flat xGMI all-to-all; see distributed_join.hpp header
   DJ_CHECK_ERROR(left_on.size() == 1 && right_on.size() == 1,
                  "this build joins on a single key column (the hot-path shape)");
-  check_no_compression(left_compression_options);
-  check_no_compression(right_compression_options);
+  validate_compression(left_compression_options);
+  validate_compression(right_compression_options);
 
   const int G = communicator->mpi_size;
   CommunicationGroup group(G, 1, communicator->mpi_rank);
@@ -1427,6 +1607,46 @@ void* dj_cpp_distributed_inner_join_i64(void* comm, const int64_t* d_lk, const i
   auto opts = generate_compression_options_distributed(left, false);
   auto result = distributed_inner_join(left, right, {0}, {0}, (Communicator*)comm, opts, opts,
                                        over_decom, report_timing != 0, nullptr, 1);
+  return result.release();
+}
+
+/* join with compression options (cascaded bitpack when compression != 0) */
+void* dj_cpp_distributed_inner_join_i64_opts(void* comm, const int64_t* d_lk,
+                                             const int64_t* d_lp, int64_t ln,
+                                             const int64_t* d_rk, const int64_t* d_rp,
+                                             int64_t rn, int over_decom, int report_timing,
+                                             int compression)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  cudf::table_view left(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)ln, d_lk),
+     column_view(data_type(type_id::INT64), (cudf::size_type)ln, d_lp)});
+  cudf::table_view right(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)rn, d_rk),
+     column_view(data_type(type_id::INT64), (cudf::size_type)rn, d_rp)});
+  auto opts = generate_compression_options_distributed(left, compression != 0);
+  auto result = distributed_inner_join(left, right, {0}, {0}, (Communicator*)comm, opts, opts,
+                                       over_decom, report_timing != 0, nullptr, 1);
+  return result.release();
+}
+
+void* dj_cpp_shuffle_on_i64_comp(void* comm, const int64_t* d_keys, const int64_t* d_pay,
+                                 int64_t n, int hash_function, uint32_t seed, int compression)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  cudf::table_view input(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)n, d_keys),
+     column_view(data_type(type_id::INT64), (cudf::size_type)n, d_pay)});
+  auto opts = generate_compression_options_distributed(input, compression != 0);
+  auto result =
+    shuffle_on(input, {0}, (Communicator*)comm, opts,
+               hash_function == DJ_HASH_IDENTITY ? cudf::hash_id::HASH_IDENTITY
+                                                 : cudf::hash_id::HASH_MURMUR3,
+               seed, false, nullptr);
   return result.release();
 }
 

@@ -108,6 +108,20 @@ void make_test_string_sizes(const int64_t* d_keys, int64_t n, int32_t* d_sizes, 
 void fill_test_strings(const int64_t* d_keys, int64_t n, const int32_t* d_offsets,
                        uint8_t* d_chars, hipStream_t s);
 
+/* ----- cascaded codec (dj_compress.hip; wire format in its header) ----- */
+struct CompSliceHeader {
+  uint32_t bits;
+  uint32_t scheme;
+  uint64_t count;
+};
+size_t compress_bound(int64_t count, int elem_size);
+size_t compress_scratch_bytes(int64_t count);
+void compress_slice_async(const void* d_in, int64_t count, int elem_size, int num_deltas,
+                          int use_bp, uint8_t* d_out, uint32_t* d_bits_tmp, hipStream_t s);
+size_t compressed_size_from_bits(int64_t count, int elem_size, uint32_t bits);
+void decompress_slice_async(const uint8_t* d_comp, const CompSliceHeader& h, int elem_size,
+                            void* d_out, void* d_scratch, hipStream_t s);
+
 /* ----- small utilities ----- */
 void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s);
 
