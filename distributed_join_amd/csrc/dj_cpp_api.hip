@@ -24,6 +24,7 @@
 
 #include <algorithm>
 #include <chrono>
+#include <cmath>
 #include <cstring>
 #include <iostream>
 #include <map>
@@ -1163,6 +1164,16 @@ std::unique_ptr<cudf::table> concat_tables(std::vector<std::unique_ptr<cudf::tab
 
 /* --------------------------------------------------- distributed_inner_join */
 
+/* replicates the reference's divisor search exactly
+ * (distributed_join.cpp:55-69, including its sqrt starting point) */
+static int get_nvl_partition_size(int mpi_size, int nvlink_domain_size)
+{
+  if (nvlink_domain_size >= mpi_size) return mpi_size;
+  for (int size = (int)ceil(sqrt((double)mpi_size)); size > 0; size--)
+    if (mpi_size % size == 0 && size <= nvlink_domain_size) return size;
+  return 1;
+}
+
 std::unique_ptr<cudf::table> distributed_inner_join(
   cudf::table_view left,
   cudf::table_view right,
@@ -1176,17 +1187,38 @@ std::unique_ptr<cudf::table> distributed_inner_join(
   void* preallocated_pinned_buffer,
   int nvlink_domain_size)
 {
-  (void)nvlink_domain_size;  // flat xGMI all-to-all; see distributed_join.hpp header
   DJ_CHECK_ERROR(left_on.size() == 1 && right_on.size() == 1,
                  "this build joins on a single key column (the hot-path shape)");
   validate_compression(left_compression_options);
   validate_compression(right_compression_options);
 
-  const int G = communicator->mpi_size;
-  CommunicationGroup group(G, 1, communicator->mpi_rank);
+  const int world = communicator->mpi_size;
+  const int nvl = get_nvl_partition_size(world, nvlink_domain_size < 1 ? 1 : nvlink_domain_size);
+
+  /* 2-level hierarchy (distributed_join.cpp:152-199): when the join group is
+   * smaller than the world, first shuffle both tables across domains (the
+   * reference's InfiniBand stage, seed 87654321), then run the flat batched
+   * pipeline within each domain (seed 12345678). On one xGMI node
+   * nvlink_domain_size >= world collapses this to the flat path. */
+  std::unique_ptr<cudf::table> l_ib, r_ib;
+  if (nvl != world) {
+    CommunicationGroup inter(world, nvl, communicator->mpi_rank);
+    l_ib = shuffle_on(left, left_on, inter, communicator, left_compression_options,
+                      cudf::hash_id::HASH_MURMUR3, DJ_SEED_INTER, report_timing,
+                      preallocated_pinned_buffer);
+    r_ib = shuffle_on(right, right_on, inter, communicator, right_compression_options,
+                      cudf::hash_id::HASH_MURMUR3, DJ_SEED_INTER, report_timing,
+                      preallocated_pinned_buffer);
+    left = l_ib->view();
+    right = r_ib->view();
+  }
+  if (nvl == 1) return local_inner_join(left, right, left_on[0], right_on[0]);
+
+  const int G = nvl;
+  CommunicationGroup group(nvl, 1, communicator->mpi_rank);
   const int nparts = G * over_decom_factor;
   DJ_CHECK_ERROR(nparts <= dj::kMaxPartitions,
-                 "world_size x over_decom_factor must be <= 64");
+                 "join-group size x over_decom_factor must be <= 64");
 
   /* report_timing mirrors the reference's per-phase prints
    * (distributed_join.cpp:120-130, 235-240) */

@@ -197,9 +197,14 @@ int main(int argc, char** argv)
       auto left = make_rank_tables(n_global, per, r * per, true);
       auto right = make_rank_tables(n_global, per, r * per, false);
       LoopbackCommunicator comm(r, G, &mb);
-      auto opts = generate_compression_options_distributed(left->view(), false);
+      /* alternate per-rank-invariant variants across runs of this binary via
+       * argv: nvlink_domain_size and compression are exercised by the python
+       * wrapper invoking multiple configurations */
+      const int nvl = (argc > 4) ? atoi(argv[4]) : 1;
+      const bool compress = (argc > 5) && atoi(argv[5]) != 0;
+      auto opts = generate_compression_options_distributed(left->view(), compress);
       auto res = distributed_inner_join(left->view(), right->view(), {0}, {0}, &comm, opts,
-                                        opts, over_decom, false, nullptr, 1);
+                                        opts, over_decom, false, nullptr, nvl);
       checksum(*res, &sums[r], &rows[r]);
     });
   }

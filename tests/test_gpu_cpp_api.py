@@ -124,7 +124,11 @@ def test_multirank_loopback():
              "-o", exe, "-L", os.path.join(repo, "distributed_join_amd"), "-ldistjoin",
              "-Wl,-rpath," + os.path.join(repo, "distributed_join_amd")],
             check=True, capture_output=True)
-    for args in (["2", "2"], ["4", "1"]):
+    # (G, over_decom, n_global, nvlink_domain_size, compression)
+    for args in (["2", "2"], ["4", "1"],
+                 ["4", "1", "400000", "2", "0"],   # 2-level hierarchy
+                 ["4", "1", "400000", "1", "0"],   # nvl=1: full shuffle + local join
+                 ["2", "2", "400000", "2", "1"]):  # compressed wire
         r = subprocess.run([exe] + args, capture_output=True, text=True, timeout=240)
-        assert r.returncode == 0, r.stdout + r.stderr
+        assert r.returncode == 0, " ".join(args) + "\n" + r.stdout + r.stderr
         assert "MULTIRANK OK" in r.stdout
