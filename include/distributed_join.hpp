@@ -14,13 +14,14 @@
  *  - over_decom_factor batches the exchange+join (distributed_join.cpp:244-329)
  *  - partition seed 12345678 intra-node (distributed_join.cpp:211)
  *
- * Changed vs reference: nvlink_domain_size is accepted but the exchange is
- * always the flat single-level xGMI all-to-all — on one MI355X node every
- * GPU pair has a direct link, so the reference's 2-level IB x NVLink
- * hierarchy (distributed_join.cpp:152-199) collapses;
- * the join result is partition-invariant, so results are identical for any
- * nvlink_domain_size. The engine below the API is the bucketed-LDS join
- * (dj_kernels.hip), not cuDF.
+ * The 2-level hierarchy is implemented with the reference's semantics
+ * (distributed_join.cpp:55-69,152-199): nvlink_domain_size selects the
+ * join-group size via the same divisor search; when it is smaller than the
+ * world, both tables are first shuffled across domains (seed 87654321) and
+ * the batched pipeline runs within each domain. On one 8x MI355X node
+ * (full xGMI mesh) nvlink_domain_size >= world collapses to the flat
+ * single-level all-to-all, which is bandwidth-optimal there. The engine
+ * below the API is the bucketed-LDS join (dj_kernels.hip), not cuDF.
  */
 #pragma once
 
