@@ -1162,7 +1162,259 @@ std::unique_ptr<cudf::table> concat_tables(std::vector<std::unique_ptr<cudf::tab
 
 }  // namespace
 
+/* access hook for the fused path (friend declared in all_to_all_comm.hpp) */
+struct AllToAllCommunicatorAccess {
+  static const std::vector<int64_t>& recv_offsets(const AllToAllCommunicator& a)
+  {
+    return a.recv_offsets;
+  }
+};
+
+namespace {
+
+/* exchange per-peer int64 vectors (host) through the communicator */
+void exchange_vecs(CommunicationGroup group, Communicator* communicator,
+                   const std::vector<std::vector<int64_t>>& send,
+                   std::vector<std::vector<int64_t>>& recv)
+{
+  const int G = group.size();
+  const int me = group.get_local_idx();
+  const int V = (int)send[0].size();
+  recv.assign(G, std::vector<int64_t>(V, 0));
+  recv[me] = send[me];
+  if (G == 1) return;
+  DBuf ds((size_t)G * V * 8), dr((size_t)G * V * 8);
+  std::vector<int64_t> flat((size_t)G * V);
+  for (int i = 0; i < G; i++)
+    std::copy(send[i].begin(), send[i].end(), flat.begin() + (size_t)i * V);
+  DJ_HIP_CALL(hipMemcpyAsync(ds.p, flat.data(), flat.size() * 8, hipMemcpyHostToDevice,
+                             dj_rt_comm_stream()));
+  DJ_HIP_CALL(hipStreamSynchronize(dj_rt_comm_stream()));
+  communicator->start();
+  for (int i = 0; i < G; i++) {
+    if (i == me) continue;
+    int peer = group.get_global_rank(i);
+    communicator->send(ds.i64() + (size_t)i * V, V, 8, peer);
+    communicator->recv(dr.i64() + (size_t)i * V, V, 8, peer);
+  }
+  communicator->stop();
+  DJ_HIP_CALL(hipMemcpyAsync(flat.data(), dr.p, flat.size() * 8, hipMemcpyDeviceToHost,
+                             dj_rt_comm_stream()));
+  DJ_HIP_CALL(hipStreamSynchronize(dj_rt_comm_stream()));
+  for (int i = 0; i < G; i++)
+    if (i != me) recv[i].assign(flat.begin() + (size_t)i * V, flat.begin() + (size_t)(i + 1) * V);
+}
+
+/* the fused wire path (2 x INT64 columns, keys at column 0): ONE staged
+ * scatter produces the rank/batch slices pre-grouped by PA groups; the
+ * receiver runs pass B over per-peer segment lists straight into the LDS
+ * join — the separate stable rank partition and bucket pass A disappear
+ * (DESIGN.md §3 ablation: the partition passes dominated the step). */
+std::unique_ptr<cudf::table> distributed_inner_join_fused(
+  cudf::table_view left, cudf::table_view right, Communicator* communicator,
+  CommunicationGroup group, std::vector<ColumnCompressionOptions> const& lopts,
+  std::vector<ColumnCompressionOptions> const& ropts, int od, int PA, bool report_timing,
+  void* pinned)
+{
+  hipStream_t st = dj_rt_stream();
+  const int G = group.size();
+  const int P = G * od * PA;
+  const int64_t ln = left.num_rows(), rn = right.num_rows();
+
+  /* fused partition of both tables (columnar outputs for the wire) */
+  DBuf counts((size_t)dj::kBucketBlocks * P * 4), totals((size_t)P * 4);
+  DBuf d_poff((size_t)(P + 1) * 8);
+  DBuf lkp((size_t)std::max<int64_t>(ln, 1) * 8), lpp((size_t)std::max<int64_t>(ln, 1) * 8);
+  DBuf rkp((size_t)std::max<int64_t>(rn, 1) * 8), rpp((size_t)std::max<int64_t>(rn, 1) * 8);
+  std::vector<int64_t> lposf(P + 1, 0), rposf(P + 1, 0);
+  {
+    dj_timing::Scope t(DJ_PHASE_PART_SCATTER, st);
+    dj::fused_partition(left.column(0).head<int64_t>(), left.column(1).head<int64_t>(), ln,
+                        G * od, DJ_SEED_INTRA, PA, (uint32_t*)counts.p, (uint32_t*)totals.p,
+                        d_poff.i64(), lkp.i64(), lpp.i64(), st);
+    DJ_HIP_CALL(hipMemcpyAsync(lposf.data(), d_poff.p, (size_t)(P + 1) * 8,
+                               hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    dj::fused_partition(right.column(0).head<int64_t>(), right.column(1).head<int64_t>(), rn,
+                        G * od, DJ_SEED_INTRA, PA, (uint32_t*)counts.p, (uint32_t*)totals.p,
+                        d_poff.i64(), rkp.i64(), rpp.i64(), st);
+    DJ_HIP_CALL(hipMemcpyAsync(rposf.data(), d_poff.p, (size_t)(P + 1) * 8,
+                               hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+  }
+
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  cudf::table_view lpart_view({column_view(data_type(type_id::INT64), (cudf::size_type)ln,
+                                           lkp.p),
+                               column_view(data_type(type_id::INT64), (cudf::size_type)ln,
+                                           lpp.p)});
+  cudf::table_view rpart_view({column_view(data_type(type_id::INT64), (cudf::size_type)rn,
+                                           rkp.p),
+                               column_view(data_type(type_id::INT64), (cudf::size_type)rn,
+                                           rpp.p)});
+
+  struct Batch {
+    std::unique_ptr<AllToAllCommunicator> latoa, ratoa;
+    std::unique_ptr<cudf::table> lrecv, rrecv;
+    std::vector<std::vector<int64_t>> lsub_recv, rsub_recv;  // [G][PA]
+    DBuf lseg, rseg;       // device [G][PA+1]
+    DBuf lgb, rgb;         // device [PA+1] group bases
+    DBuf lpairs, rpairs;   // bucketed pairs
+    DBuf loff, roff;       // int64[B+1]
+    DBuf flags;            // u32[B]
+    DBuf o0, o1, o2, o3, meta;
+    int F{0};
+    int64_t cap{0}, lrows{0}, rrows{0};
+  };
+  std::vector<Batch> batches(od);
+
+  auto slice_of = [&](const std::vector<int64_t>& pofs, int b) {
+    std::vector<cudf::size_type> sl(G + 1);
+    for (int r = 0; r <= G; r++) sl[r] = (cudf::size_type)pofs[(size_t)(b * G + r) * PA];
+    return sl;
+  };
+  auto subcounts_of = [&](const std::vector<int64_t>& pofs, int b) {
+    std::vector<std::vector<int64_t>> sc(G, std::vector<int64_t>(PA));
+    for (int r = 0; r < G; r++)
+      for (int g = 0; g < PA; g++) {
+        size_t base = (size_t)(b * G + r) * PA + g;
+        sc[r][g] = pofs[base + 1] - pofs[base];
+      }
+    return sc;
+  };
+
+  /* pre-phase: exchanges + allocations */
+  for (int b = 0; b < od; b++) {
+    Batch& bt = batches[b];
+    bt.latoa = std::make_unique<AllToAllCommunicator>(lpart_view, slice_of(lposf, b), group,
+                                                      communicator, lopts, true);
+    bt.ratoa = std::make_unique<AllToAllCommunicator>(rpart_view, slice_of(rposf, b), group,
+                                                      communicator, ropts, true);
+    exchange_vecs(group, communicator, subcounts_of(lposf, b), bt.lsub_recv);
+    exchange_vecs(group, communicator, subcounts_of(rposf, b), bt.rsub_recv);
+    bt.lrecv = bt.latoa->allocate_communicated_table();
+    bt.rrecv = bt.ratoa->allocate_communicated_table();
+    bt.lrows = bt.lrecv->num_rows();
+    bt.rrows = bt.rrecv->num_rows();
+    /* seg bounds + group bases (host -> device) */
+    auto build_segs = [&](const std::vector<std::vector<int64_t>>& sub,
+                          const AllToAllCommunicator& atoa, DBuf& dseg, DBuf& dgb) {
+      const auto& roff = AllToAllCommunicatorAccess::recv_offsets(atoa);
+      std::vector<int64_t> seg((size_t)G * (PA + 1));
+      std::vector<int64_t> gtot(PA + 1, 0);
+      for (int r2 = 0; r2 < G; r2++) {
+        int64_t acc = roff[r2];
+        for (int g = 0; g <= PA; g++) {
+          seg[(size_t)r2 * (PA + 1) + g] = acc;
+          if (g < PA) acc += sub[r2][g];
+        }
+        for (int g = 0; g < PA; g++) gtot[g + 1] += sub[r2][g];
+      }
+      for (int g = 0; g < PA; g++) gtot[g + 1] += gtot[g];
+      dseg = DBuf(seg.size() * 8);
+      dgb = DBuf((size_t)(PA + 1) * 8);
+      DJ_HIP_CALL(hipMemcpyAsync(dseg.p, seg.data(), seg.size() * 8, hipMemcpyHostToDevice,
+                                 st));
+      DJ_HIP_CALL(hipMemcpyAsync(dgb.p, gtot.data(), (size_t)(PA + 1) * 8,
+                                 hipMemcpyHostToDevice, st));
+    };
+    build_segs(bt.lsub_recv, *bt.latoa, bt.lseg, bt.lgb);
+    build_segs(bt.rsub_recv, *bt.ratoa, bt.rseg, bt.rgb);
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    /* sub-bucket fanout: target ~760 rows per final bucket */
+    int64_t maxn = std::max(bt.lrows, bt.rrows);
+    int F = 64;
+    while (F < 1024 && (int64_t)PA * F * 760 < maxn) F <<= 1;
+    bt.F = F;
+    int64_t B = (int64_t)PA * F;
+    bt.lpairs = DBuf((size_t)std::max<int64_t>(bt.lrows, 1) * 16);
+    bt.rpairs = DBuf((size_t)std::max<int64_t>(bt.rrows, 1) * 16);
+    bt.loff = DBuf((size_t)(B + 1) * 8);
+    bt.roff = DBuf((size_t)(B + 1) * 8);
+    bt.flags = DBuf((size_t)B * 4);
+    bt.cap = std::max<int64_t>(bt.rrows + (bt.rrows >> 3), 1024);
+    bt.o0 = DBuf((size_t)bt.cap * 8);
+    bt.o1 = DBuf((size_t)bt.cap * 8);
+    bt.o2 = DBuf((size_t)bt.cap * 8);
+    bt.o3 = DBuf((size_t)bt.cap * 8);
+    bt.meta = DBuf(16);
+  }
+
+  /* pipeline: comm(b) then enqueue passB+join(b); comm(b+1) overlaps */
+  for (int b = 0; b < od; b++) {
+    Batch& bt = batches[b];
+    bt.latoa->launch_communication(bt.lrecv->mutable_view(), report_timing, pinned);
+    bt.ratoa->launch_communication(bt.rrecv->mutable_view(), report_timing, pinned);
+    DJ_HIP_CALL(hipMemsetAsync(bt.meta.p, 0, 16, st));
+    if (bt.lrows == 0 || bt.rrows == 0) continue;
+    int64_t B = (int64_t)PA * bt.F;
+    DJ_HIP_CALL(hipMemsetAsync(bt.flags.p, 0, (size_t)B * 4, st));
+    {
+      dj_timing::Scope t(DJ_PHASE_BUCKET_SCATTER, st);
+      dj::subpart_lists((const int64_t*)bt.lrecv->get_column(0).head(),
+                        (const int64_t*)bt.lrecv->get_column(1).head(), bt.lseg.i64(), G, PA,
+                        bt.F, bt.lgb.i64(), (longlong2*)bt.lpairs.p, bt.loff.i64(), st);
+      dj::subpart_lists((const int64_t*)bt.rrecv->get_column(0).head(),
+                        (const int64_t*)bt.rrecv->get_column(1).head(), bt.rseg.i64(), G, PA,
+                        bt.F, bt.rgb.i64(), (longlong2*)bt.rpairs.p, bt.roff.i64(), st);
+    }
+    {
+      dj_timing::Scope t(DJ_PHASE_JOIN_FUSED, st);
+      dj::lds_join((const longlong2*)bt.lpairs.p, bt.loff.i64(),
+                   (const longlong2*)bt.rpairs.p, bt.roff.i64(), (int)B, bt.o0.i64(),
+                   bt.o1.i64(), bt.o2.i64(), bt.o3.i64(), bt.cap, bt.meta.i64(),
+                   (uint32_t*)bt.flags.p, (int*)((char*)bt.meta.p + 12),
+                   (int*)((char*)bt.meta.p + 8), st);
+    }
+  }
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+
+  /* finalize */
+  std::vector<std::unique_ptr<cudf::table>> results;
+  for (int b = 0; b < od; b++) {
+    Batch& bt = batches[b];
+    struct {
+      int64_t count;
+      int error;
+      int any_overflow;
+    } meta;
+    DJ_HIP_CALL(hipMemcpy(&meta, bt.meta.p, 16, hipMemcpyDeviceToHost));
+    DJ_CHECK_ERROR(meta.error == 0,
+                   "join build: key equal to the empty sentinel (-1) is unsupported");
+    if (bt.lrows == 0 || bt.rrows == 0 || meta.count == 0) {
+      std::vector<std::unique_ptr<cudf::column>> cols;
+      for (int c = 0; c < 4; c++)
+        cols.push_back(std::make_unique<cudf::column>(data_type(type_id::INT64),
+                                                      (cudf::size_type)0));
+      results.push_back(std::make_unique<cudf::table>(std::move(cols)));
+      continue;
+    }
+    if (meta.any_overflow || meta.count > bt.cap) {
+      results.push_back(local_inner_join(bt.lrecv->view(), bt.rrecv->view(), 0, 0));
+      continue;
+    }
+    std::vector<std::unique_ptr<cudf::column>> cols;
+    auto adopt = [&](DBuf& d) {
+      auto col = std::make_unique<cudf::column>(data_type(type_id::INT64),
+                                                (cudf::size_type)meta.count, d.p);
+      d.p = nullptr;
+      return col;
+    };
+    cols.push_back(adopt(bt.o0));
+    cols.push_back(adopt(bt.o1));
+    cols.push_back(adopt(bt.o2));
+    cols.push_back(adopt(bt.o3));
+    results.push_back(std::make_unique<cudf::table>(std::move(cols)));
+  }
+  return concat_tables(results);
+}
+
+}  // namespace
+
 /* --------------------------------------------------- distributed_inner_join */
+
 
 /* replicates the reference's divisor search exactly
  * (distributed_join.cpp:55-69, including its sqrt starting point) */
@@ -1219,6 +1471,27 @@ std::unique_ptr<cudf::table> distributed_inner_join(
   const int nparts = G * over_decom_factor;
   DJ_CHECK_ERROR(nparts <= dj::kMaxPartitions,
                  "join-group size x over_decom_factor must be <= 64");
+
+  /* fused wire path for the hot shape (2 x INT64 columns, key at 0):
+   * partitions carry the bucket grouping on the wire, removing the separate
+   * stable rank partition and bucket pass A */
+  {
+    const bool fast2f = left.num_columns() == 2 && right.num_columns() == 2 &&
+                        left_on[0] == 0 && right_on[0] == 0 &&
+                        left.column(0).type().id() == cudf::type_id::INT64 &&
+                        left.column(1).type().id() == cudf::type_id::INT64 &&
+                        right.column(0).type().id() == cudf::type_id::INT64 &&
+                        right.column(1).type().id() == cudf::type_id::INT64;
+    if (fast2f) {
+      int pa = 1;
+      while (pa * 2 * nparts <= 1024) pa *= 2;
+      if (pa >= 4)
+        return distributed_inner_join_fused(left, right, communicator, group,
+                                            left_compression_options,
+                                            right_compression_options, over_decom_factor, pa,
+                                            report_timing, preallocated_pinned_buffer);
+    }
+  }
 
   /* report_timing mirrors the reference's per-phase prints
    * (distributed_join.cpp:120-130, 235-240) */

@@ -627,6 +627,8 @@ __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
                                     const longlong2* in_pairs, int64_t start, int64_t end,
                                     int P, longlong2* tbuf, uint32_t* hist, uint32_t* base,
                                     uint32_t* gcur, longlong2* out_pairs)
+/* GROUP_FN 0: groupA_of(P); GROUP_FN 1: sub-bucket = (mix64>>32) & (P-1)
+ * (P is the runtime fanout) */
 {
   constexpr int VPT = SCATTER_TILE / BUCKET_THREADS;  // 4
   const int tid = threadIdx.x;
@@ -641,7 +643,9 @@ __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
       int64_t i = t0 + (int64_t)v * blockDim.x + tid;
       if (i < end) {
         r[v] = load_row<SINGLE_LEVEL>(keys, pay, in_pairs, i);
-        g[v] = GROUP_FN == 0 ? groupA_of(r[v].x, P) : subB_of(r[v].x);
+        g[v] = GROUP_FN == 0 ? groupA_of(r[v].x, P)
+                             : ((uint32_t)(dj_mix64((uint64_t)r[v].x) >> 32) &
+                                (uint32_t)(P - 1));
         rank[v] = atomicAdd(&hist[g[v]], 1u);
         nv = v + 1;
       }
@@ -663,7 +667,9 @@ __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
     /* flush linearly: per-group runs coalesce into full lines */
     for (int pos = tid; pos < count; pos += blockDim.x) {
       longlong2 row = tbuf[pos];
-      uint32_t gg = GROUP_FN == 0 ? groupA_of(row.x, P) : subB_of(row.x);
+      uint32_t gg = GROUP_FN == 0 ? groupA_of(row.x, P)
+                                  : ((uint32_t)(dj_mix64((uint64_t)row.x) >> 32) &
+                                     (uint32_t)(P - 1));
       out_pairs[gcur[gg] + (pos - base[gg])] = row;
     }
     __syncthreads();
@@ -738,6 +744,67 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
   __syncthreads();
   staged_scatter_span<1, SINGLE_LEVEL>(keys, pay, in_pairs, s0, s1, SUB_BUCKETS, tbuf, hist,
                                        base, gcur, out_pairs);
+}
+
+/* pass B over per-peer segment lists (fused wire path): block = one pass-A
+ * group g, whose rows arrived pre-grouped inside each peer slice; the
+ * block's sub-buckets fan out by F (runtime power of two). */
+__global__ __launch_bounds__(BUCKET_THREADS) void subpart_lists_kernel(
+  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay,
+  const int64_t* __restrict__ seg_bounds /* [nseg][PA+1] absolute row offsets */, int nseg,
+  int PA, int F, const int64_t* __restrict__ group_base /* [PA+1] output bases */,
+  longlong2* __restrict__ out_pairs, int64_t* __restrict__ bucket_offsets /* PA*F+1 */)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbuf = (longlong2*)smem;
+  uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
+  uint32_t* base = hist + F;
+  uint32_t* gcur = base + F;
+  uint32_t* seghist = gcur + F;
+  const int tid = threadIdx.x;
+  const int g = blockIdx.x;
+  for (int j = tid; j < F; j += blockDim.x) seghist[j] = 0;
+  __syncthreads();
+  for (int sgi = 0; sgi < nseg; sgi++) {
+    const int64_t s0 = seg_bounds[(size_t)sgi * (PA + 1) + g];
+    const int64_t s1 = seg_bounds[(size_t)sgi * (PA + 1) + g + 1];
+    for (int64_t i = s0 + tid; i < s1; i += blockDim.x)
+      atomicAdd(&seghist[(uint32_t)(dj_mix64((uint64_t)nt_load(&keys[i])) >> 32) &
+                         (uint32_t)(F - 1)],
+                1u);
+  }
+  __syncthreads();
+  if (tid == 0) {
+    uint32_t acc = 0;
+    const int64_t gb = group_base[g];
+    for (int j = 0; j < F; j++) {
+      uint32_t c = seghist[j];
+      gcur[j] = (uint32_t)(gb + acc);
+      bucket_offsets[(size_t)g * F + j] = gb + acc;
+      acc += c;
+    }
+    if (g == gridDim.x - 1) bucket_offsets[(size_t)PA * F] = group_base[PA];
+  }
+  __syncthreads();
+  for (int sgi = 0; sgi < nseg; sgi++) {
+    const int64_t s0 = seg_bounds[(size_t)sgi * (PA + 1) + g];
+    const int64_t s1 = seg_bounds[(size_t)sgi * (PA + 1) + g + 1];
+    staged_scatter_span<1, true>(keys, pay, nullptr, s0, s1, F, tbuf, hist, base, gcur,
+                                 out_pairs);
+  }
+}
+
+void subpart_lists(const int64_t* d_keys, const int64_t* d_pay, const int64_t* d_seg_bounds,
+                   int nseg, int PA, int F, const int64_t* d_group_base,
+                   longlong2* d_out_pairs, int64_t* d_bucket_offsets, hipStream_t s)
+{
+  DJ_CHECK_ERROR(F >= 64 && F <= 1024 && (F & (F - 1)) == 0,
+                 "subpart_lists: F must be a power of two in [64,1024]");
+  size_t lds = SCATTER_TILE * sizeof(longlong2) + 4 * (size_t)F * 4;
+  hipLaunchKernelGGL(subpart_lists_kernel, dim3(PA), dim3(BUCKET_THREADS), lds, s, d_keys,
+                     d_pay, d_seg_bounds, nseg, PA, F, d_group_base, d_out_pairs,
+                     d_bucket_offsets);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 /* ---- fused per-bucket LDS build + single-pass probe with staged output ----
@@ -844,6 +911,133 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
     }
     __syncthreads();
   }
+}
+
+/* ----------------- fused rank+group partition (fast2 wire path) -----------
+ * One staged scatter replaces the stable rank partition AND bucket pass A:
+ * partition id = p * PA + groupA, where p = murmur(key) %% (G*od) is the
+ * reference's rank/batch partition (distributed_join.cpp:211-226 semantics:
+ * batch = p / G, rank = p %% G) and groupA = high mix64 bits. Rank/batch
+ * slices stay contiguous for the exchange; each peer slice arrives grouped
+ * by groupA, so the local join needs only pass B (over per-peer segment
+ * lists) + the fused LDS join. Row order within a slice is NOT stable —
+ * internal wire layout only (the drop-in hash_partition C ABI stays
+ * stable). Outputs are columnar (key col + payload col) for the existing
+ * column-wise exchange and compression paths. */
+
+__device__ __forceinline__ uint32_t fused_pid(int64_t key, int nparts_rank, uint32_t seed,
+                                              int PA)
+{
+  uint32_t p = dj_murmur3_int64(key, seed) % (uint32_t)nparts_rank;
+  uint32_t g = (uint32_t)(dj_mix64((uint64_t)key) >> 40) & (uint32_t)(PA - 1);
+  return p * (uint32_t)PA + g;
+}
+
+__global__ __launch_bounds__(BUCKET_THREADS) void fused_count_kernel(
+  const int64_t* __restrict__ keys, int64_t n, int nparts_rank, uint32_t seed, int PA,
+  uint32_t* __restrict__ counts)
+{
+  extern __shared__ uint32_t hist[];
+  const int P = nparts_rank * PA;
+  for (int p = threadIdx.x; p < P; p += blockDim.x) hist[p] = 0;
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t start = (int64_t)blockIdx.x * chunk;
+  const int64_t end = min(start + chunk, n);
+  for (int64_t i = start + threadIdx.x; i < end; i += blockDim.x)
+    atomicAdd(&hist[fused_pid(nt_load(&keys[i]), nparts_rank, seed, PA)], 1u);
+  __syncthreads();
+  for (int p = threadIdx.x; p < P; p += blockDim.x)
+    counts[(size_t)blockIdx.x * P + p] = hist[p];
+}
+
+/* staged scatter with columnar outputs (stage pairs, flush per column) */
+__global__ __launch_bounds__(BUCKET_THREADS) void fused_scatter_kernel(
+  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n,
+  int nparts_rank, uint32_t seed, int PA, const uint32_t* __restrict__ counts,
+  const int64_t* __restrict__ offsets, int64_t* __restrict__ out_keys,
+  int64_t* __restrict__ out_pay)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbuf = (longlong2*)smem;
+  const int P = nparts_rank * PA;
+  uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
+  uint32_t* base = hist + P;
+  uint32_t* gcur = base + P;
+  const int tid = threadIdx.x;
+  if (tid < P)
+    gcur[tid] = (uint32_t)offsets[tid] + counts[(size_t)blockIdx.x * P + tid];
+  __syncthreads();
+  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
+  const int64_t start = (int64_t)blockIdx.x * chunk;
+  const int64_t end = min(start + chunk, n);
+  constexpr int VPT = SCATTER_TILE / BUCKET_THREADS;
+  for (int64_t t0 = start; t0 < end; t0 += SCATTER_TILE) {
+    const int count = (int)min((int64_t)SCATTER_TILE, end - t0);
+    if (tid < P) hist[tid] = 0;
+    __syncthreads();
+    longlong2 r[VPT];
+    uint32_t g[VPT], rank[VPT];
+    int nv = 0;
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
+      if (i < end) {
+        r[v].x = nt_load(&keys[i]);
+        r[v].y = pay ? nt_load(&pay[i]) : i;
+        g[v] = fused_pid(r[v].x, nparts_rank, seed, PA);
+        rank[v] = atomicAdd(&hist[g[v]], 1u);
+        nv = v + 1;
+      }
+    }
+    __syncthreads();
+    if (tid < P) base[tid] = hist[tid];
+    __syncthreads();
+    for (int off = 1; off < P; off <<= 1) {
+      uint32_t add = (tid < P && tid >= off) ? base[tid - off] : 0;
+      __syncthreads();
+      if (tid < P) base[tid] += add;
+      __syncthreads();
+    }
+    if (tid < P) base[tid] -= hist[tid];
+    __syncthreads();
+    for (int v = 0; v < nv; v++) tbuf[base[g[v]] + rank[v]] = r[v];
+    __syncthreads();
+    for (int pos = tid; pos < count; pos += blockDim.x) {
+      longlong2 row = tbuf[pos];
+      uint32_t gg = fused_pid(row.x, nparts_rank, seed, PA);
+      uint32_t dst = gcur[gg] + (uint32_t)(pos - base[gg]);
+      out_keys[dst] = row.x;
+      out_pay[dst] = row.y;
+    }
+    __syncthreads();
+    if (tid < P) gcur[tid] += hist[tid];
+    __syncthreads();
+  }
+}
+
+void fused_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts_rank,
+                     uint32_t seed, int PA, uint32_t* d_counts, uint32_t* d_totals,
+                     int64_t* d_offsets, int64_t* d_out_keys, int64_t* d_out_pay,
+                     hipStream_t s)
+{
+  const int P = nparts_rank * PA;
+  DJ_CHECK_ERROR(P >= 1 && P <= 1024, "fused_partition: nparts_rank*PA must be <= 1024");
+  DJ_CHECK_ERROR(n < (int64_t)UINT32_MAX, "fused_partition: n must be < 2^32");
+  size_t hist_lds = (size_t)P * 4;
+  size_t scatter_lds = SCATTER_TILE * sizeof(longlong2) + 3 * hist_lds;
+  hipLaunchKernelGGL(fused_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), hist_lds,
+                     s, d_keys, n, nparts_rank, seed, PA, d_counts);
+  DJ_HIP_CALL(hipGetLastError());
+  hipLaunchKernelGGL(bucket_scanA_kernel, dim3(P), dim3(BUCKET_BLOCKS), 0, s, d_counts, P,
+                     d_totals);
+  DJ_HIP_CALL(hipGetLastError());
+  hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, P,
+                     d_offsets);
+  DJ_HIP_CALL(hipGetLastError());
+  hipLaunchKernelGGL(fused_scatter_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
+                     scatter_lds, s, d_keys, d_pay, n, nparts_rank, seed, PA, d_counts,
+                     d_offsets, d_out_keys, d_out_pay);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,

@@ -108,6 +108,16 @@ void make_test_string_sizes(const int64_t* d_keys, int64_t n, int32_t* d_sizes, 
 void fill_test_strings(const int64_t* d_keys, int64_t n, const int32_t* d_offsets,
                        uint8_t* d_chars, hipStream_t s);
 
+/* ----- fused rank+group partition + segment-list pass B (fast2 wire path;
+ * see dj_kernels.hip "fused rank+group partition" block) ----- */
+void fused_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts_rank,
+                     uint32_t seed, int PA, uint32_t* d_counts, uint32_t* d_totals,
+                     int64_t* d_offsets, int64_t* d_out_keys, int64_t* d_out_pay,
+                     hipStream_t s);
+void subpart_lists(const int64_t* d_keys, const int64_t* d_pay, const int64_t* d_seg_bounds,
+                   int nseg, int PA, int F, const int64_t* d_group_base,
+                   longlong2* d_out_pairs, int64_t* d_bucket_offsets, hipStream_t s);
+
 /* ----- cascaded codec (dj_compress.hip; wire format in its header) ----- */
 struct CompSliceHeader {
   uint32_t bits;
