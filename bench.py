@@ -166,21 +166,31 @@ def main():
               for name, pid in dj.PHASES.items()}
 
     # roofline for the dominant join kernel (algorithmic bytes per step for
-    # this rank; derivation in DESIGN.md §Measurement):
-    #   part_scatter   — rank-level stable scatter, both tables:
-    #                    16 B read + 16 B write per row
-    #   bucket_scatter — full two-level bucket partition of both local
-    #                    tables: 8 B count-read + 16 B read + 16 B write per
-    #                    row per level (pass A skipped when B == 256)
-    #   join_fused     — streaming read of both bucketed tables (16 B/row) +
+    # this rank; derivation in DESIGN.md §Measurement).
+    # N==1 (reference single-rank semantics: local join only, no
+    # partition/shuffle stage — distributed_join.cpp:200-214):
+    #   bucket_scatter = the two-level local bucket partition
+    #     (8 B count-read + 16 B read + 16 B write per row per level)
+    # N>1 (fused wire path):
+    #   part_scatter   = the fused rank+group partition (8 B count-read +
+    #                    16 B read + 16 B write per row, both tables)
+    #   bucket_scatter = pass B over per-peer segment lists (8 B count +
+    #                    16 B read + 16 B pair write per received row)
+    #   join_fused     = streaming read of both bucketed tables (16 B/row) +
     #                    32 B per output row; LDS traffic is on-chip
     lrows = rrows = float(rows)  # hash-uniform => received ~= sent rows
-    nlevels = 2 if rows / 256 > 1600 else 1
-    alg = {
-        "part_scatter": 32.0 * rows * 2,
-        "bucket_scatter": 40.0 * nlevels * (lrows + rrows),
-        "join_fused": 16.0 * (lrows + rrows) + 32.0 * matches,
-    }
+    if N == 1:
+        alg = {
+            "part_scatter": 1.0,  # not run at N==1
+            "bucket_scatter": 40.0 * 2 * (lrows + rrows),
+            "join_fused": 16.0 * (lrows + rrows) + 32.0 * matches,
+        }
+    else:
+        alg = {
+            "part_scatter": 40.0 * rows * 2,
+            "bucket_scatter": 40.0 * (lrows + rrows),
+            "join_fused": 16.0 * (lrows + rrows) + 32.0 * matches,
+        }
     dom = max(alg.keys(), key=lambda k: phases[k]["ms"])
     dom_ms = phases[dom]["ms"]
     achieved = (alg[dom] / 1e9) / (dom_ms / 1e3) if dom_ms > 0 else None
@@ -240,7 +250,9 @@ def main():
                 "selectivity": SELECTIVITY,
                 "output_rows_per_gpu": int(matches),
                 "output_rows_per_sec": out_rows_s,
-                "engine": "C++ distributed_inner_join (drop-in path)",
+                "engine": "C++ distributed_inner_join (drop-in path); at N=1 the "
+                          "reference's single-rank semantics apply (local join only, "
+                          "no partition/shuffle stage)",
                 "phases_ms": {k: round(v["ms"], 4) for k, v in phases.items()},
                 "all_to_all_GBs_per_gpu": all_to_all_GBs,
             },
