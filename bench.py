@@ -198,14 +198,13 @@ def main():
     # profiles/r01_pmc_traffic.json (rocprofv3 --pmc FETCH_SIZE / WRITE_SIZE
     # passes; FETCH x2 gfx950 correction — see that file)
     traffic = None
-    try:
-        with open(os.path.join(REPO, "profiles", "r01_pmc_traffic.json")) as fh:
-            coeff = json.load(fh)["phase_bytes_per_input_row_pair"]
-        traffic = coeff[dom] * (lrows + rrows)
-        if dom == "join_fused":
-            traffic += 0.0  # output writes already included in the calibration
-    except Exception:
-        pass
+    if N == 1:  # coefficients were calibrated on the N=1 kernel set only
+        try:
+            with open(os.path.join(REPO, "profiles", "r01_pmc_traffic.json")) as fh:
+                coeff = json.load(fh)["phase_bytes_per_input_row_pair"]
+            traffic = coeff[dom] * (lrows + rrows)
+        except Exception:
+            pass
     roofline = {
         "bound": "hbm",
         "kernel": dom,

@@ -18,6 +18,7 @@
 #include "../../include/shuffle_on.hpp"
 
 #include "dj_hash.h"
+#include "dj_rng.h"
 
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
@@ -1892,6 +1893,34 @@ void dj_cpp_comm_destroy(void* comm)
   c->finalize();
   if (g_default_comm == c) g_default_comm = nullptr;
   delete c;
+}
+
+/* RCCL self-test: world-1 ncclCommInitRank + grouped self send/recv through
+ * the real RCCLCommunicator start/send/recv/stop path (the same calls the
+ * N>1 exchange makes per peer slice). Validates RCCL init + group + stream
+ * semantics on a single GPU before any multi-GPU run. Returns 0 on success,
+ * 1 on payload mismatch (RCCL/HIP errors abort via DJ_*_CALL). */
+int dj_rccl_selftest(int64_t n)
+{
+  std::vector<uint8_t> idb((size_t)rccl_unique_id_size());
+  rccl_unique_id(idb.data());
+  RCCLCommunicator comm(0, 1, idb.data());
+  std::vector<uint8_t> h_src((size_t)n), h_dst((size_t)n, 0);
+  for (int64_t i = 0; i < n; i++) h_src[(size_t)i] = (uint8_t)(dj_mix64((uint64_t)i) & 0xFF);
+  void *d_src = nullptr, *d_dst = nullptr;
+  DJ_HIP_CALL(hipMalloc(&d_src, (size_t)n));
+  DJ_HIP_CALL(hipMalloc(&d_dst, (size_t)n));
+  DJ_HIP_CALL(hipMemcpy(d_src, h_src.data(), (size_t)n, hipMemcpyHostToDevice));
+  DJ_HIP_CALL(hipMemset(d_dst, 0, (size_t)n));
+  comm.start();
+  comm.send(d_src, n, 1, 0);
+  comm.recv(d_dst, n, 1, 0);
+  comm.stop();
+  DJ_HIP_CALL(hipMemcpy(h_dst.data(), d_dst, (size_t)n, hipMemcpyDeviceToHost));
+  DJ_HIP_CALL(hipFree(d_src));
+  DJ_HIP_CALL(hipFree(d_dst));
+  comm.finalize();
+  return h_src == h_dst ? 0 : 1;
 }
 
 /* full distributed_inner_join over int64 key/payload columns; returns an

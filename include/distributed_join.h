@@ -150,6 +150,10 @@ void dj_all_to_all_i64(const int64_t* d_send, const int64_t* h_send_offsets,
 /* exchange per-peer row counts (host arrays of size comm_size):
  * replaces communicate_sizes (all_to_all_comm.cpp:54-100) */
 void dj_exchange_sizes(const int64_t* h_send_counts, int64_t* h_recv_counts);
+/* world-1 RCCL init + grouped self send/recv of n bytes through the
+ * RCCLCommunicator path (the same calls the N>1 exchange makes per peer);
+ * returns 0 on success, 1 on payload mismatch */
+int dj_rccl_selftest(int64_t n);
 
 #ifdef __cplusplus
 }

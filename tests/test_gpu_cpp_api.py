@@ -132,3 +132,11 @@ def test_multirank_loopback():
         r = subprocess.run([exe] + args, capture_output=True, text=True, timeout=240)
         assert r.returncode == 0, " ".join(args) + "\n" + r.stdout + r.stderr
         assert "MULTIRANK OK" in r.stdout
+
+
+def test_rccl_selftest():
+    """RCCL proper (not the loopback transport): world-1 ncclCommInitRank +
+    grouped self send/recv through RCCLCommunicator start/send/recv/stop —
+    the exact calls the N>1 peer-slice exchange makes. De-risks the driver's
+    multi-GPU runs on a single-GPU box."""
+    assert dj.lib().dj_rccl_selftest(1 << 20) == 0
