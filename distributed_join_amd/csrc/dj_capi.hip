@@ -269,7 +269,7 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
   BucketScratch s;
   s.lpairs = (longlong2*)take((size_t)ln * 16);
   s.rpairs = (longlong2*)take((size_t)rn * 16);
-  s.tmp_pairs = (longlong2*)take((size_t)maxn * 16);
+  s.tmp_pairs = (longlong2*)take((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);
   s.loff = (int64_t*)take((size_t)(B + 1) * 8);
   s.roff = (int64_t*)take((size_t)(B + 1) * 8);
   s.segoff = (int64_t*)take((size_t)(PA + 1) * 8);
@@ -291,7 +291,7 @@ int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
   auto add = [&](size_t b) { bytes += (b + 255) & ~(size_t)255; };
   add((size_t)ln * 16);
   add((size_t)rn * 16);
-  add((size_t)maxn * 16);
+  add((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);  // pass-A slack staging (>= maxn)
   add((size_t)(B + 1) * 8);
   add((size_t)(B + 1) * 8);
   add((size_t)(PA + 1) * 8);
@@ -321,15 +321,15 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
   const int B = dj::bucket_count_for(ln, rn);
   BucketScratch s = carve_bucket_scratch(d_scratch, ln, rn, B);
   hipStream_t st = stream();
+  DJ_HIP_CALL(hipMemsetAsync(d_any_overflow, 0, sizeof(int), st));
+  DJ_HIP_CALL(hipMemsetAsync(s.flags, 0, (size_t)B * 4, st));
   {
     PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
     dj::bucket_partition2(d_lk, d_lp, ln, B, s.tmp_pairs, s.counts, s.totals, s.segoff,
-                          s.loff, s.lpairs, st);
+                          s.loff, s.lpairs, d_any_overflow, st);
     dj::bucket_partition2(d_rk, d_rp, rn, B, s.tmp_pairs, s.counts, s.totals, s.segoff,
-                          s.roff, s.rpairs, st);
+                          s.roff, s.rpairs, d_any_overflow, st);
   }
-  DJ_HIP_CALL(hipMemsetAsync(d_any_overflow, 0, sizeof(int), st));
-  DJ_HIP_CALL(hipMemsetAsync(s.flags, 0, (size_t)B * 4, st));
   {
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
     dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, d_out0, d_out1, d_out2,
@@ -353,6 +353,22 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
   int any = 0;
   DJ_HIP_CALL(hipMemcpyAsync(&any, s.any_overflow, sizeof(int), hipMemcpyDeviceToHost, st));
   DJ_HIP_CALL(hipStreamSynchronize(st));
+  if (any & 2) {
+    /* slack-partition overflow (skew beyond the pass-A slack): the bucketed
+     * data is incomplete -- redo the WHOLE join via the global-table path
+     * on the original inputs */
+    DJ_HIP_CALL(hipMemsetAsync(d_counter, 0, sizeof(int64_t), st));
+    DJ_HIP_CALL(hipMemsetAsync(d_error, 0, sizeof(int), st));
+    int64_t nslots = dj::join_table_slots(ln);
+    int64_t* d_table = (int64_t*)dj_dmalloc(nslots * 2 * sizeof(int64_t));
+    dj_join_table_init(d_table, nslots);
+    dj_join_build(d_lk, d_lp, ln, d_table, nslots, d_error);
+    dj_join_probe(d_rk, d_rp, rn, d_table, nslots, d_out0, d_out1, d_out2, d_out3, cap,
+                  d_counter);
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    dj_dfree(d_table);
+    return;
+  }
   if (any) {
     std::vector<uint32_t> flags((size_t)B);
     std::vector<int64_t> loff((size_t)B + 1), roff((size_t)B + 1);

@@ -72,14 +72,26 @@ constexpr int kBucketBlocks = 512;
 constexpr int kSubBuckets = 256;
 constexpr int kJoinBucketRowCap = 1536;  // 75% of the 2048-slot LDS table
 int bucket_count_for(int64_t ln, int64_t rn);
+/* pass-A slack segment capacity (rows per group): mean + ~6% + 1024 covers
+ * hash-uniform inputs w.h.p.; skewed inputs overflow -> bit 2 of
+ * any_overflow -> the caller redoes the join exactly */
+inline int64_t slack_capA(int64_t n, int PA)
+{
+  const int64_t m = n / PA;
+  return m + m / 16 + 1024;
+}
 /* Two-level non-stable partition into B buckets (B = PA*256, PA<=1024) of
- * interleaved 16 B {key,payload} pairs. d_tmp_pairs: longlong2[n] pass-A
- * staging; d_counts: u32[kBucketBlocks*PA]; d_totals: u32[PA]; d_segoff:
- * int64[PA+1]; d_offsets: int64[B+1]. */
+ * interleaved 16 B {key,payload} pairs. d_tmp_pairs: pass-A staging —
+ * longlong2[PA * slack_capA(n, PA)] when d_any_overflow is non-null (the
+ * slack path: no count pass; group skew beyond the slack sets BIT 2 of
+ * *d_any_overflow and the partition output is INCOMPLETE — caller must redo
+ * the whole join), else longlong2[n] (exact two-pass path). d_counts:
+ * u32[kBucketBlocks*PA]; d_totals: u32[PA]; d_segoff: int64[PA+1];
+ * d_offsets: int64[B+1]. */
 void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
                        longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
                        int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
-                       hipStream_t s);
+                       int* d_any_overflow, hipStream_t s);
 /* Fused per-bucket LDS build+probe over bucketed pair tables. Buckets whose
  * build side exceeds kJoinBucketRowCap set overflow_flags[b]/any_overflow
  * and are skipped (host runs the global-table path on them). */

@@ -134,7 +134,7 @@ def test_multirank_loopback():
         assert "MULTIRANK OK" in r.stdout
 
 
-def test_rccl_selftest():
+def test_rccl_selftest(dj):
     """RCCL proper (not the loopback transport): world-1 ncclCommInitRank +
     grouped self send/recv through RCCLCommunicator start/send/recv/stop —
     the exact calls the N>1 peer-slice exchange makes. De-risks the driver's
