@@ -1061,6 +1061,9 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
   uint32_t* hist = (uint32_t*)(tbuf + SLACK_TILE);
   uint32_t* base = hist + P;
   uint32_t* gcur = base + P;
+  uint32_t* glim = gcur + P;  // per-group segment end (precomputed: the
+                              // per-row bound check must not pay a 64-bit
+                              // multiply in the flush loop)
   __shared__ int s_ovf;
   const int tid = threadIdx.x;
   if (tid == 0) s_ovf = 0;
@@ -1097,8 +1100,11 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
     if (tid < P) base[tid] -= hist[tid];
     __syncthreads();
     /* claim this tile's run in each group's slack segment */
-    if (tid < P)
-      gcur[tid] = (uint32_t)((int64_t)tid * capA) + atomicAdd(&gcursor[tid], hist[tid]);
+    if (tid < P) {
+      const int64_t seg0 = (int64_t)tid * capA;
+      gcur[tid] = (uint32_t)seg0 + atomicAdd(&gcursor[tid], hist[tid]);
+      glim[tid] = (uint32_t)(seg0 + capA);
+    }
     __syncthreads();
     for (int v = 0; v < nv; v++) tbuf[base[g[v]] + rank[v]] = r[v];
     __syncthreads();
@@ -1106,7 +1112,7 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
       longlong2 row = tbuf[pos];
       uint32_t gg = groupA_of(row.x, P);
       uint32_t dst = gcur[gg] + (uint32_t)(pos - base[gg]);
-      if (dst < (uint32_t)((int64_t)(gg + 1) * capA))
+      if (dst < glim[gg])
         out_pairs[dst] = row;
       else
         s_ovf = 1;  // benign LDS race: any 1 wins
@@ -1186,7 +1192,7 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
              (int64_t)PA * slack_capA(n, PA) + n < (int64_t)UINT32_MAX) {
     /* slack path: no count pass (see bucket_scatter_slack_kernel header) */
     const int64_t capA = slack_capA(n, PA);
-    size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 3 * (size_t)PA * sizeof(uint32_t);
+    size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 4 * (size_t)PA * sizeof(uint32_t);
     size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
     DJ_HIP_CALL(hipMemsetAsync(d_totals, 0, (size_t)PA * 4, s));
     hipLaunchKernelGGL(bucket_scatter_slack_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
