@@ -636,9 +636,12 @@ __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
     const int count = (int)min((int64_t)SCATTER_TILE, end - t0);
     if (tid < P) hist[tid] = 0;
     __syncthreads();
+    /* compile-time-bounded, predicated loops: a runtime trip count would
+     * force r/g/rank into scratch (the arrays must stay in VGPRs — spilled
+     * staging re-reads cost more HBM traffic than the payload itself) */
     longlong2 r[VPT];
     uint32_t g[VPT], rank[VPT];
-    int nv = 0;
+#pragma unroll
     for (int v = 0; v < VPT; v++) {
       int64_t i = t0 + (int64_t)v * blockDim.x + tid;
       if (i < end) {
@@ -647,7 +650,6 @@ __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
                              : ((uint32_t)(dj_mix64((uint64_t)r[v].x) >> 32) &
                                 (uint32_t)(P - 1));
         rank[v] = atomicAdd(&hist[g[v]], 1u);
-        nv = v + 1;
       }
     }
     __syncthreads();
@@ -662,7 +664,11 @@ __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
     }
     if (tid < P) base[tid] -= hist[tid];  // inclusive -> exclusive
     __syncthreads();
-    for (int v = 0; v < nv; v++) tbuf[base[g[v]] + rank[v]] = r[v];
+#pragma unroll
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
+      if (i < end) tbuf[base[g[v]] + rank[v]] = r[v];
+    }
     __syncthreads();
     /* flush linearly: per-group runs coalesce into full lines */
     for (int pos = tid; pos < count; pos += blockDim.x) {
@@ -978,7 +984,7 @@ __global__ __launch_bounds__(BUCKET_THREADS) void fused_scatter_kernel(
     __syncthreads();
     longlong2 r[VPT];
     uint32_t g[VPT], rank[VPT];
-    int nv = 0;
+#pragma unroll
     for (int v = 0; v < VPT; v++) {
       int64_t i = t0 + (int64_t)v * blockDim.x + tid;
       if (i < end) {
@@ -986,7 +992,6 @@ __global__ __launch_bounds__(BUCKET_THREADS) void fused_scatter_kernel(
         r[v].y = pay ? nt_load(&pay[i]) : i;
         g[v] = fused_pid(r[v].x, nparts_rank, seed, PA);
         rank[v] = atomicAdd(&hist[g[v]], 1u);
-        nv = v + 1;
       }
     }
     __syncthreads();
@@ -1000,7 +1005,11 @@ __global__ __launch_bounds__(BUCKET_THREADS) void fused_scatter_kernel(
     }
     if (tid < P) base[tid] -= hist[tid];
     __syncthreads();
-    for (int v = 0; v < nv; v++) tbuf[base[g[v]] + rank[v]] = r[v];
+#pragma unroll
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
+      if (i < end) tbuf[base[g[v]] + rank[v]] = r[v];
+    }
     __syncthreads();
     for (int pos = tid; pos < count; pos += blockDim.x) {
       longlong2 row = tbuf[pos];
@@ -1077,7 +1086,7 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
     __syncthreads();
     longlong2 r[VPT];
     uint32_t g[VPT], rank[VPT];
-    int nv = 0;
+#pragma unroll
     for (int v = 0; v < VPT; v++) {
       int64_t i = t0 + (int64_t)v * blockDim.x + tid;
       if (i < end) {
@@ -1085,7 +1094,6 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
         r[v].y = pay ? nt_load(&pay[i]) : i;
         g[v] = groupA_of(r[v].x, P);
         rank[v] = atomicAdd(&hist[g[v]], 1u);
-        nv = v + 1;
       }
     }
     __syncthreads();
@@ -1106,7 +1114,11 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
       glim[tid] = (uint32_t)(seg0 + capA);
     }
     __syncthreads();
-    for (int v = 0; v < nv; v++) tbuf[base[g[v]] + rank[v]] = r[v];
+#pragma unroll
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
+      if (i < end) tbuf[base[g[v]] + rank[v]] = r[v];
+    }
     __syncthreads();
     for (int pos = tid; pos < count; pos += blockDim.x) {
       longlong2 row = tbuf[pos];
