@@ -1,0 +1,213 @@
+/*
+ * join_layout.hip — isolates the lds_join slowdown seen when bucket input
+ * moved from compact offsets to slack regions (b*capB + sizes[b]).
+ * Same kernel copy, same bucket contents, five layouts:
+ *   compact (offsets), slack capB=1080 (product), 1024 (pow2 stride),
+ *   1160 (odd line count), 2048 (extreme gaps).
+ * Diagnostic only.
+ */
+#include "../distributed_join_amd/csrc/dj_rng.h"
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+
+#define CHECK(c)                                                      \
+  do {                                                                \
+    hipError_t e = (c);                                               \
+    if (e != hipSuccess) {                                            \
+      printf("HIP error %s at %d\n", hipGetErrorString(e), __LINE__); \
+      exit(1);                                                        \
+    }                                                                 \
+  } while (0)
+
+constexpr int THREADS = 1024;
+constexpr int SLOTS = 2048;
+constexpr int STAGE = 1024;
+constexpr int64_t EMPTY = -1;
+constexpr int ROWCAP = 1536;
+
+__global__ __launch_bounds__(THREADS) void join_kernel(
+  const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
+  const uint32_t* __restrict__ lsizes, int64_t capBl, const longlong2* __restrict__ rrows,
+  const int64_t* __restrict__ roff, const uint32_t* __restrict__ rsizes, int64_t capBr, int B,
+  int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
+  int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbl = (longlong2*)smem;
+  int64_t* stage = (int64_t*)(tbl + SLOTS);
+  long long* base_sh = (long long*)(stage + 4 * STAGE);
+  uint32_t* cur_sh = (uint32_t*)(base_sh + 1);
+  const uint32_t smask = SLOTS - 1;
+  constexpr int S = STAGE;
+
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const int64_t l0 = lsizes ? (int64_t)b * capBl : loff[b];
+    const int64_t l1 = lsizes ? l0 + lsizes[b] : loff[b + 1];
+    const int64_t r0 = rsizes ? (int64_t)b * capBr : roff[b];
+    const int64_t r1 = rsizes ? r0 + rsizes[b] : roff[b + 1];
+    const int64_t lnb = l1 - l0;
+    if (lnb == 0 || r1 == r0 || lnb > ROWCAP) continue;
+    for (int s = threadIdx.x; s < SLOTS; s += blockDim.x) tbl[s].x = EMPTY;
+    if (threadIdx.x == 0) *cur_sh = 0;
+    __syncthreads();
+    for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
+      longlong2 row = lrows[i];
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)row.x) & smask;
+      for (;;) {
+        unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
+                                           (unsigned long long)EMPTY,
+                                           (unsigned long long)row.x);
+        if (old == (unsigned long long)EMPTY) break;
+        slot = (slot + 1) & smask;
+      }
+      tbl[slot].y = row.y;
+    }
+    __syncthreads();
+    for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
+      longlong2 prow = rrows[j];
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)prow.x) & smask;
+      for (;;) {
+        longlong2 e = tbl[slot];
+        if (e.x == EMPTY) break;
+        if (e.x == prow.x) {
+          uint32_t pos = atomicAdd(cur_sh, 1u);
+          if (pos < (uint32_t)S) {
+            stage[0 * S + pos] = prow.x;
+            stage[1 * S + pos] = e.y;
+            stage[2 * S + pos] = prow.x;
+            stage[3 * S + pos] = prow.y;
+          } else {
+            long long idx = (long long)atomicAdd(counter, 1ull);
+            if (idx < cap) {
+              out0[idx] = prow.x;
+              out1[idx] = e.y;
+              out2[idx] = prow.x;
+              out3[idx] = prow.y;
+            }
+          }
+        }
+        slot = (slot + 1) & smask;
+      }
+    }
+    __syncthreads();
+    const uint32_t total = min(*cur_sh, (uint32_t)S);
+    if (threadIdx.x == 0 && total)
+      *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
+    __syncthreads();
+    if (total) {
+      const long long base = *base_sh;
+      for (uint32_t i = threadIdx.x; i < total; i += blockDim.x) {
+        long long idx = base + (long long)i;
+        if (idx < cap) {
+          out0[idx] = stage[0 * S + i];
+          out1[idx] = stage[1 * S + i];
+          out2[idx] = stage[2 * S + i];
+          out3[idx] = stage[3 * S + i];
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
+/* fill bucket b with nrows rows at 'where': build keys b*4096+i (i<nrows),
+ * probe keys b*4096+i for i<match else b*4096+2048+i (no match) */
+__global__ void fill_kernel(longlong2* rows, const int64_t* starts, int B, int nrows,
+                            int match, int probe)
+{
+  int b = blockIdx.x;
+  for (int i = threadIdx.x; i < nrows; i += blockDim.x) {
+    int64_t k = (int64_t)b * 4096 + (probe && i >= match ? 2048 + i : i);
+    rows[starts[b] + i] = {k, (int64_t)i};
+  }
+}
+
+static double run_case(const char* name, int B, int nrows, int match, int64_t capB, int reps)
+{
+  /* layout: capB==0 => compact */
+  int64_t stride = capB ? capB : nrows;
+  int64_t total = (int64_t)B * stride;
+  longlong2 *lrows, *rrows;
+  int64_t *loff, *starts;
+  uint32_t* sizes;
+  int64_t *o0, *o1, *o2, *o3;
+  unsigned long long* counter;
+  int64_t cap = (int64_t)B * match + 1024;
+  CHECK(hipMalloc(&lrows, total * 16));
+  CHECK(hipMalloc(&rrows, total * 16));
+  CHECK(hipMalloc(&loff, (B + 1) * 8));
+  CHECK(hipMalloc(&starts, B * 8));
+  CHECK(hipMalloc(&sizes, B * 4));
+  CHECK(hipMalloc(&o0, cap * 8));
+  CHECK(hipMalloc(&o1, cap * 8));
+  CHECK(hipMalloc(&o2, cap * 8));
+  CHECK(hipMalloc(&o3, cap * 8));
+  CHECK(hipMalloc(&counter, 8));
+  std::vector<int64_t> h_off(B + 1), h_starts(B);
+  std::vector<uint32_t> h_sizes(B, (uint32_t)nrows);
+  for (int b = 0; b <= B; b++) h_off[b] = (int64_t)b * stride;
+  for (int b = 0; b < B; b++) h_starts[b] = (int64_t)b * stride;
+  CHECK(hipMemcpy(loff, h_off.data(), (B + 1) * 8, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(starts, h_starts.data(), B * 8, hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(sizes, h_sizes.data(), B * 4, hipMemcpyHostToDevice));
+  fill_kernel<<<B, 256>>>(lrows, starts, B, nrows, match, 0);
+  fill_kernel<<<B, 256>>>(rrows, starts, B, nrows, match, 1);
+  CHECK(hipDeviceSynchronize());
+
+  size_t lds = SLOTS * 16 + 4 * STAGE * 8 + 16;
+  int grid = B < 8192 ? B : 8192;
+  hipEvent_t e0, e1;
+  CHECK(hipEventCreate(&e0));
+  CHECK(hipEventCreate(&e1));
+  double ms_sum = 0;
+  unsigned long long nout = 0;
+  for (int rep = 0; rep < reps; rep++) {
+    CHECK(hipMemset(counter, 0, 8));
+    CHECK(hipEventRecord(e0));
+    if (capB)
+      join_kernel<<<grid, THREADS, lds>>>(lrows, nullptr, sizes, capB, rrows, nullptr, sizes,
+                                          capB, B, o0, o1, o2, o3, cap, counter);
+    else
+      join_kernel<<<grid, THREADS, lds>>>(lrows, loff, nullptr, 0, rrows, loff, nullptr, 0, B,
+                                          o0, o1, o2, o3, cap, counter);
+    CHECK(hipEventRecord(e1));
+    CHECK(hipEventSynchronize(e1));
+    float ms;
+    CHECK(hipEventElapsedTime(&ms, e0, e1));
+    if (rep) ms_sum += ms;
+    CHECK(hipMemcpy(&nout, counter, 8, hipMemcpyDeviceToHost));
+  }
+  CHECK(hipGetLastError());
+  double avg = ms_sum / (reps - 1);
+  printf("%-28s capB=%5lld  %8.3f ms  (out=%llu, want=%lld)\n", name, (long long)capB, avg,
+         nout, (int64_t)B * match);
+  CHECK(hipFree(lrows)); CHECK(hipFree(rrows)); CHECK(hipFree(loff));
+  CHECK(hipFree(starts)); CHECK(hipFree(sizes)); CHECK(hipFree(o0)); CHECK(hipFree(o1));
+  CHECK(hipFree(o2)); CHECK(hipFree(o3)); CHECK(hipFree(counter));
+  CHECK(hipEventDestroy(e0)); CHECK(hipEventDestroy(e1));
+  return avg;
+}
+
+int main(int argc, char** argv)
+{
+  int B = argc > 1 ? atoi(argv[1]) : 131072;
+  int nrows = argc > 2 ? atoi(argv[2]) : 762;
+  int match = (int)(nrows * 0.3);
+  int reps = 4;
+  run_case("compact+offsets", B, nrows, match, 0, reps);
+  run_case("sizes, slack 1080", B, nrows, match, 1080, reps);
+  run_case("sizes, slack 1024", B, nrows, match, 1024, reps);
+  run_case("sizes, slack 1160", B, nrows, match, 1160, reps);
+  run_case("sizes, slack 2048", B, nrows, match, 2048, reps);
+  /* sizes code path but zero gaps: distinguishes code path from layout */
+  {
+    int64_t capB = nrows;
+    run_case("sizes, stride=nrows", B, nrows, match, capB, reps);
+  }
+  return 0;
+}
