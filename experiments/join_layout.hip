@@ -115,21 +115,24 @@ __global__ __launch_bounds__(THREADS) void join_kernel(
   }
 }
 
-/* fill bucket b with nrows rows at 'where': build keys b*4096+i (i<nrows),
- * probe keys b*4096+i for i<match else b*4096+2048+i (no match) */
-__global__ void fill_kernel(longlong2* rows, const int64_t* starts, int B, int nrows,
-                            int match, int probe)
+/* fill bucket b with sizes[b] rows: build keys b*4096+i, probe keys
+ * b*4096+i for i<30% else non-matching */
+__global__ void fill_kernel(longlong2* rows, const int64_t* starts,
+                            const uint32_t* sizes, int B, int probe)
 {
   int b = blockIdx.x;
+  int nrows = sizes[b];
+  int match = (int)(nrows * 0.3);
   for (int i = threadIdx.x; i < nrows; i += blockDim.x) {
     int64_t k = (int64_t)b * 4096 + (probe && i >= match ? 2048 + i : i);
     rows[starts[b] + i] = {k, (int64_t)i};
   }
 }
 
-static double run_case(const char* name, int B, int nrows, int match, int64_t capB, int reps)
+static double run_case(const char* name, int B, int nrows, int match, int64_t capB, int reps,
+                       bool arena_carve, bool dirty, bool poisson)
 {
-  /* layout: capB==0 => compact */
+  /* layout: capB==0 => compact (only valid with uniform sizes) */
   int64_t stride = capB ? capB : nrows;
   int64_t total = (int64_t)B * stride;
   longlong2 *lrows, *rrows;
@@ -138,25 +141,47 @@ static double run_case(const char* name, int B, int nrows, int match, int64_t ca
   int64_t *o0, *o1, *o2, *o3;
   unsigned long long* counter;
   int64_t cap = (int64_t)B * match + 1024;
-  CHECK(hipMalloc(&lrows, total * 16));
-  CHECK(hipMalloc(&rrows, total * 16));
-  CHECK(hipMalloc(&loff, (B + 1) * 8));
-  CHECK(hipMalloc(&starts, B * 8));
-  CHECK(hipMalloc(&sizes, B * 4));
-  CHECK(hipMalloc(&o0, cap * 8));
-  CHECK(hipMalloc(&o1, cap * 8));
-  CHECK(hipMalloc(&o2, cap * 8));
-  CHECK(hipMalloc(&o3, cap * 8));
-  CHECK(hipMalloc(&counter, 8));
+  char* arena = nullptr;
+  if (arena_carve) {
+    size_t need = 2 * (size_t)total * 16 + (size_t)(B + 1) * 8 + (size_t)B * 12 +
+                  4 * (size_t)cap * 8 + 4096;
+    CHECK(hipMalloc(&arena, need));
+    char* pp = arena;
+    auto take = [&](size_t b) { void* r = pp; pp += (b + 255) & ~(size_t)255; return r; };
+    lrows = (longlong2*)take((size_t)total * 16);
+    rrows = (longlong2*)take((size_t)total * 16);
+    loff = (int64_t*)take((size_t)(B + 1) * 8);
+    starts = (int64_t*)take((size_t)B * 8);
+    sizes = (uint32_t*)take((size_t)B * 4);
+    o0 = (int64_t*)take((size_t)cap * 8);
+    o1 = (int64_t*)take((size_t)cap * 8);
+    o2 = (int64_t*)take((size_t)cap * 8);
+    o3 = (int64_t*)take((size_t)cap * 8);
+    counter = (unsigned long long*)take(8);
+  } else {
+    CHECK(hipMalloc(&lrows, total * 16));
+    CHECK(hipMalloc(&rrows, total * 16));
+    CHECK(hipMalloc(&loff, (B + 1) * 8));
+    CHECK(hipMalloc(&starts, B * 8));
+    CHECK(hipMalloc(&sizes, B * 4));
+    CHECK(hipMalloc(&o0, cap * 8));
+    CHECK(hipMalloc(&o1, cap * 8));
+    CHECK(hipMalloc(&o2, cap * 8));
+    CHECK(hipMalloc(&o3, cap * 8));
+    CHECK(hipMalloc(&counter, 8));
+  }
   std::vector<int64_t> h_off(B + 1), h_starts(B);
-  std::vector<uint32_t> h_sizes(B, (uint32_t)nrows);
+  std::vector<uint32_t> h_sizes(B);
+  for (int b = 0; b < B; b++)
+    h_sizes[b] = poisson ? (uint32_t)(nrows - 84 + (int)(dj_mix64((uint64_t)b) % 169))
+                         : (uint32_t)nrows;  // ~uniform +-84 (~3 sigma-ish spread)
   for (int b = 0; b <= B; b++) h_off[b] = (int64_t)b * stride;
   for (int b = 0; b < B; b++) h_starts[b] = (int64_t)b * stride;
   CHECK(hipMemcpy(loff, h_off.data(), (B + 1) * 8, hipMemcpyHostToDevice));
   CHECK(hipMemcpy(starts, h_starts.data(), B * 8, hipMemcpyHostToDevice));
   CHECK(hipMemcpy(sizes, h_sizes.data(), B * 4, hipMemcpyHostToDevice));
-  fill_kernel<<<B, 256>>>(lrows, starts, B, nrows, match, 0);
-  fill_kernel<<<B, 256>>>(rrows, starts, B, nrows, match, 1);
+  fill_kernel<<<B, 256>>>(lrows, starts, sizes, B, 0);
+  fill_kernel<<<B, 256>>>(rrows, starts, sizes, B, 1);
   CHECK(hipDeviceSynchronize());
 
   size_t lds = SLOTS * 16 + 4 * STAGE * 8 + 16;
@@ -168,6 +193,10 @@ static double run_case(const char* name, int B, int nrows, int match, int64_t ca
   unsigned long long nout = 0;
   for (int rep = 0; rep < reps; rep++) {
     CHECK(hipMemset(counter, 0, 8));
+    if (dirty) {  // re-write inputs so the join starts with a dirty L2
+      fill_kernel<<<B, 256>>>(lrows, starts, sizes, B, 0);
+      fill_kernel<<<B, 256>>>(rrows, starts, sizes, B, 1);
+    }
     CHECK(hipEventRecord(e0));
     if (capB)
       join_kernel<<<grid, THREADS, lds>>>(lrows, nullptr, sizes, capB, rrows, nullptr, sizes,
@@ -184,11 +213,15 @@ static double run_case(const char* name, int B, int nrows, int match, int64_t ca
   }
   CHECK(hipGetLastError());
   double avg = ms_sum / (reps - 1);
-  printf("%-28s capB=%5lld  %8.3f ms  (out=%llu, want=%lld)\n", name, (long long)capB, avg,
-         nout, (int64_t)B * match);
-  CHECK(hipFree(lrows)); CHECK(hipFree(rrows)); CHECK(hipFree(loff));
-  CHECK(hipFree(starts)); CHECK(hipFree(sizes)); CHECK(hipFree(o0)); CHECK(hipFree(o1));
-  CHECK(hipFree(o2)); CHECK(hipFree(o3)); CHECK(hipFree(counter));
+  printf("%-34s capB=%5lld  %8.3f ms  (out=%llu)\n", name, (long long)capB, avg,
+         (unsigned long long)nout);
+  if (arena_carve) {
+    CHECK(hipFree(arena));
+  } else {
+    CHECK(hipFree(lrows)); CHECK(hipFree(rrows)); CHECK(hipFree(loff));
+    CHECK(hipFree(starts)); CHECK(hipFree(sizes)); CHECK(hipFree(o0)); CHECK(hipFree(o1));
+    CHECK(hipFree(o2)); CHECK(hipFree(o3)); CHECK(hipFree(counter));
+  }
   CHECK(hipEventDestroy(e0)); CHECK(hipEventDestroy(e1));
   return avg;
 }
@@ -199,15 +232,11 @@ int main(int argc, char** argv)
   int nrows = argc > 2 ? atoi(argv[2]) : 762;
   int match = (int)(nrows * 0.3);
   int reps = 4;
-  run_case("compact+offsets", B, nrows, match, 0, reps);
-  run_case("sizes, slack 1080", B, nrows, match, 1080, reps);
-  run_case("sizes, slack 1024", B, nrows, match, 1024, reps);
-  run_case("sizes, slack 1160", B, nrows, match, 1160, reps);
-  run_case("sizes, slack 2048", B, nrows, match, 2048, reps);
-  /* sizes code path but zero gaps: distinguishes code path from layout */
-  {
-    int64_t capB = nrows;
-    run_case("sizes, stride=nrows", B, nrows, match, capB, reps);
-  }
+  run_case("base: sizes slack1080", B, nrows, match, 1080, reps, false, false, false);
+  run_case("+arena", B, nrows, match, 1080, reps, true, false, false);
+  run_case("+dirty-producer", B, nrows, match, 1080, reps, false, true, false);
+  run_case("+varying-sizes", B, nrows, match, 1080, reps, false, false, true);
+  run_case("+all", B, nrows, match, 1080, reps, true, true, true);
+  run_case("compact+offsets +dirty", B, nrows, match, 0, reps, false, true, false);
   return 0;
 }
