@@ -253,18 +253,8 @@ struct BucketScratch {
   uint32_t* counts;            // u32[kBucketBlocks*PA]
   uint32_t* totals;            // u32[PA]
   uint32_t* flags;             // u32[B]
-  uint32_t *lsizes, *rsizes;   // u32[B] slack-layout bucket sizes
   int* any_overflow;           // int[1]
 };
-
-/* the bucketed pair region must hold either layout (slack b*capB regions or
- * exact-compact), and stay monotone in n so batch-wise carving under a
- * max-sized allocation is safe */
-static size_t pairs_region_rows(int64_t n, int B)
-{
-  const int64_t slack = (int64_t)B * dj::slack_capB(n, B);
-  return (size_t)(slack > n ? slack : n);
-}
 
 BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
 {
@@ -277,8 +267,8 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
     return r;
   };
   BucketScratch s;
-  s.lpairs = (longlong2*)take(pairs_region_rows(ln, B) * 16);
-  s.rpairs = (longlong2*)take(pairs_region_rows(rn, B) * 16);
+  s.lpairs = (longlong2*)take((size_t)ln * 16);
+  s.rpairs = (longlong2*)take((size_t)rn * 16);
   s.tmp_pairs = (longlong2*)take((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);
   s.loff = (int64_t*)take((size_t)(B + 1) * 8);
   s.roff = (int64_t*)take((size_t)(B + 1) * 8);
@@ -286,8 +276,6 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
   s.counts = (uint32_t*)take((size_t)dj::kBucketBlocks * PA * 4);
   s.totals = (uint32_t*)take((size_t)PA * 4);
   s.flags = (uint32_t*)take((size_t)B * 4);
-  s.lsizes = (uint32_t*)take((size_t)B * 4);
-  s.rsizes = (uint32_t*)take((size_t)B * 4);
   s.any_overflow = (int*)take(16);
   return s;
 }
@@ -301,16 +289,14 @@ int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
   const int64_t maxn = ln > rn ? ln : rn;
   size_t bytes = 0;
   auto add = [&](size_t b) { bytes += (b + 255) & ~(size_t)255; };
-  add(pairs_region_rows(ln, B) * 16);
-  add(pairs_region_rows(rn, B) * 16);
+  add((size_t)ln * 16);
+  add((size_t)rn * 16);
   add((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);  // pass-A slack staging (>= maxn)
   add((size_t)(B + 1) * 8);
   add((size_t)(B + 1) * 8);
   add((size_t)(PA + 1) * 8);
   add((size_t)dj::kBucketBlocks * PA * 4);
   add((size_t)PA * 4);
-  add((size_t)B * 4);
-  add((size_t)B * 4);
   add((size_t)B * 4);
   add(16);
   return (int64_t)bytes;
@@ -337,21 +323,17 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
   hipStream_t st = stream();
   DJ_HIP_CALL(hipMemsetAsync(d_any_overflow, 0, sizeof(int), st));
   DJ_HIP_CALL(hipMemsetAsync(s.flags, 0, (size_t)B * 4, st));
-  const bool slack_l = dj::slack_partition_ok(ln, B);
-  const bool slack_r = dj::slack_partition_ok(rn, B);
   {
     PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
     dj::bucket_partition2(d_lk, d_lp, ln, B, s.tmp_pairs, s.counts, s.totals, s.segoff,
-                          s.loff, s.lpairs, d_any_overflow, s.lsizes, st);
+                          s.loff, s.lpairs, d_any_overflow, st);
     dj::bucket_partition2(d_rk, d_rp, rn, B, s.tmp_pairs, s.counts, s.totals, s.segoff,
-                          s.roff, s.rpairs, d_any_overflow, s.rsizes, st);
+                          s.roff, s.rpairs, d_any_overflow, st);
   }
   {
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
-    dj::lds_join(s.lpairs, s.loff, slack_l ? s.lsizes : nullptr, dj::slack_capB(ln, B),
-                 s.rpairs, s.roff, slack_r ? s.rsizes : nullptr, dj::slack_capB(rn, B), B,
-                 d_out0, d_out1, d_out2, d_out3, cap, d_counter, s.flags, d_any_overflow,
-                 d_error, st);
+    dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, d_out0, d_out1, d_out2,
+                 d_out3, cap, d_counter, s.flags, d_any_overflow, d_error, st);
   }
 }
 
@@ -388,38 +370,21 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
     return;
   }
   if (any) {
-    const bool slack_l = dj::slack_partition_ok(ln, B);
-    const bool slack_r = dj::slack_partition_ok(rn, B);
-    const int64_t capBl = dj::slack_capB(ln, B), capBr = dj::slack_capB(rn, B);
-    std::vector<uint32_t> flags((size_t)B), lsz, rsz;
-    std::vector<int64_t> loff, roff;
+    std::vector<uint32_t> flags((size_t)B);
+    std::vector<int64_t> loff((size_t)B + 1), roff((size_t)B + 1);
     DJ_HIP_CALL(hipMemcpy(flags.data(), s.flags, (size_t)B * 4, hipMemcpyDeviceToHost));
-    if (slack_l) {
-      lsz.resize((size_t)B);
-      DJ_HIP_CALL(hipMemcpy(lsz.data(), s.lsizes, (size_t)B * 4, hipMemcpyDeviceToHost));
-    } else {
-      loff.resize((size_t)B + 1);
-      DJ_HIP_CALL(hipMemcpy(loff.data(), s.loff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
-    }
-    if (slack_r) {
-      rsz.resize((size_t)B);
-      DJ_HIP_CALL(hipMemcpy(rsz.data(), s.rsizes, (size_t)B * 4, hipMemcpyDeviceToHost));
-    } else {
-      roff.resize((size_t)B + 1);
-      DJ_HIP_CALL(hipMemcpy(roff.data(), s.roff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
-    }
+    DJ_HIP_CALL(hipMemcpy(loff.data(), s.loff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
+    DJ_HIP_CALL(hipMemcpy(roff.data(), s.roff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
     for (int b = 0; b < B; b++) {
       if (!flags[b]) continue;
-      int64_t l0 = slack_l ? (int64_t)b * capBl : loff[b];
-      int64_t lnb = slack_l ? (int64_t)lsz[b] : loff[b + 1] - loff[b];
-      int64_t r0 = slack_r ? (int64_t)b * capBr : roff[b];
-      int64_t rnb = slack_r ? (int64_t)rsz[b] : roff[b + 1] - roff[b];
+      int64_t lnb = loff[b + 1] - loff[b];
+      int64_t rnb = roff[b + 1] - roff[b];
       if (lnb == 0 || rnb == 0) continue;
       int64_t nslots = dj::join_table_slots(lnb);
       int64_t* d_table = (int64_t*)dj_dmalloc(nslots * 2 * sizeof(int64_t));
       dj_join_table_init(d_table, nslots);
-      dj::join_build_pairs(s.lpairs + l0, lnb, d_table, nslots, d_error, st);
-      dj::join_probe_pairs(s.rpairs + r0, rnb, d_table, nslots, d_out0, d_out1, d_out2,
+      dj::join_build_pairs(s.lpairs + loff[b], lnb, d_table, nslots, d_error, st);
+      dj::join_probe_pairs(s.rpairs + roff[b], rnb, d_table, nslots, d_out0, d_out1, d_out2,
                            d_out3, cap, d_counter, st);
       DJ_HIP_CALL(hipStreamSynchronize(st));
       dj_dfree(d_table);

@@ -1363,9 +1363,9 @@ std::unique_ptr<cudf::table> distributed_inner_join_fused(
     }
     {
       dj_timing::Scope t(DJ_PHASE_JOIN_FUSED, st);
-      dj::lds_join((const longlong2*)bt.lpairs.p, bt.loff.i64(), nullptr, 0,
-                   (const longlong2*)bt.rpairs.p, bt.roff.i64(), nullptr, 0, (int)B,
-                   bt.o0.i64(), bt.o1.i64(), bt.o2.i64(), bt.o3.i64(), bt.cap, bt.meta.i64(),
+      dj::lds_join((const longlong2*)bt.lpairs.p, bt.loff.i64(),
+                   (const longlong2*)bt.rpairs.p, bt.roff.i64(), (int)B, bt.o0.i64(),
+                   bt.o1.i64(), bt.o2.i64(), bt.o3.i64(), bt.cap, bt.meta.i64(),
                    (uint32_t*)bt.flags.p, (int*)((char*)bt.meta.p + 12),
                    (int*)((char*)bt.meta.p + 8), st);
     }

@@ -822,8 +822,7 @@ void subpart_lists(const int64_t* d_keys, const int64_t* d_pay, const int64_t* d
  * lane — correct for any duplication factor, slower only on such buckets. */
 __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
-  const uint32_t* __restrict__ lsizes, int64_t capBl, const longlong2* __restrict__ rrows,
-  const int64_t* __restrict__ roff, const uint32_t* __restrict__ rsizes, int64_t capBr, int B,
+  const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
   int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
   int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter,
   uint32_t* __restrict__ overflow_flags, int* __restrict__ any_overflow,
@@ -838,12 +837,8 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   constexpr int S = JOIN_STAGE_ROWS;
 
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
-    /* sizes-based layout (slack bucket regions at b*capB) or contiguous
-     * offsets, independently per side */
-    const int64_t l0 = lsizes ? (int64_t)b * capBl : loff[b];
-    const int64_t l1 = lsizes ? l0 + lsizes[b] : loff[b + 1];
-    const int64_t r0 = rsizes ? (int64_t)b * capBr : roff[b];
-    const int64_t r1 = rsizes ? r0 + rsizes[b] : roff[b + 1];
+    const int64_t l0 = loff[b], l1 = loff[b + 1];
+    const int64_t r0 = roff[b], r1 = roff[b + 1];
     const int64_t lnb = l1 - l0;
     if (lnb == 0 || r1 == r0) continue;
     if (lnb > kJoinBucketRowCap) {  // skew overflow: host-side fallback joins it
@@ -1149,95 +1144,50 @@ __global__ void clamp_seglen_kernel(uint32_t* seg_len, int P, uint32_t capA)
   if (i < P && seg_len[i] > capA) seg_len[i] = capA;
 }
 
-/* pass B over slack pass-A segments: block a owns group a's 256 sub-buckets
- * EXCLUSIVELY, so bucket cursors are private to the block (LDS registers
- * across tiles) — no count sweep over the segment, no scan kernel, no
- * global atomics. Rows go straight into slack bucket regions (bucket
- * b = a*256+j starts at b*capB); final bucket sizes (clamped to capB) feed
- * lds_join's sizes-based layout. Bucket overflow — skew beyond the slack —
- * skips rows and sets bit 2 of any_overflow (caller redoes the join). */
+/* pass B over slack pass-A segments: block a reads [a*capA, a*capA+len[a])
+ * and writes its sub-buckets compactly at segout[a] (exclusive scan of the
+ * lengths), so bucket_offsets keep the contiguous B+1 convention that
+ * lds_join consumes. */
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack_kernel(
   const longlong2* __restrict__ in_pairs, const uint32_t* __restrict__ seg_len, int64_t capA,
-  int64_t capB, longlong2* __restrict__ out_pairs, uint32_t* __restrict__ bsizes,
-  int* __restrict__ any_overflow)
+  const int64_t* __restrict__ segout /* PA+1 compact output bases */, int B,
+  longlong2* __restrict__ out_pairs, int64_t* __restrict__ bucket_offsets /* B+1 */)
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbuf = (longlong2*)smem;
   uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
   uint32_t* base = hist + SUB_BUCKETS;
   uint32_t* gcur = base + SUB_BUCKETS;
-  uint32_t* glim = gcur + SUB_BUCKETS;
-  __shared__ int s_ovf;
+  uint32_t* seghist = gcur + SUB_BUCKETS;
   const int tid = threadIdx.x;
   const int a = blockIdx.x;
   const int64_t s0 = (int64_t)a * capA;
   const int64_t s1 = s0 + seg_len[a];
-  if (tid == 0) s_ovf = 0;
-  if (tid < SUB_BUCKETS) {
-    const int64_t b0 = ((int64_t)a * SUB_BUCKETS + tid) * capB;
-    gcur[tid] = (uint32_t)b0;
-    glim[tid] = (uint32_t)(b0 + capB);
+  const int64_t ob = segout[a];
+  if (tid < SUB_BUCKETS) seghist[tid] = 0;
+  __syncthreads();
+  for (int64_t i = s0 + tid; i < s1; i += blockDim.x)
+    atomicAdd(&seghist[subB_of(nt_load(&in_pairs[i].x))], 1u);
+  __syncthreads();
+  if (tid == 0) {
+    uint32_t acc = 0;
+    for (int j = 0; j < SUB_BUCKETS; j++) {
+      uint32_t c = seghist[j];
+      gcur[j] = (uint32_t)(ob + acc);
+      bucket_offsets[(size_t)a * SUB_BUCKETS + j] = ob + acc;
+      acc += c;
+    }
+    if (a == gridDim.x - 1) bucket_offsets[B] = segout[gridDim.x];
   }
   __syncthreads();
-  constexpr int VPT = SCATTER_TILE / BUCKET_THREADS;
-  for (int64_t t0 = s0; t0 < s1; t0 += SCATTER_TILE) {
-    const int count = (int)min((int64_t)SCATTER_TILE, s1 - t0);
-    if (tid < SUB_BUCKETS) hist[tid] = 0;
-    __syncthreads();
-    longlong2 r[VPT];
-    uint32_t g[VPT], rank[VPT];
-#pragma unroll
-    for (int v = 0; v < VPT; v++) {
-      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
-      if (i < s1) {
-        r[v] = nt_load2(&in_pairs[i]);
-        g[v] = subB_of(r[v].x);
-        rank[v] = atomicAdd(&hist[g[v]], 1u);
-      }
-    }
-    __syncthreads();
-    if (tid < SUB_BUCKETS) base[tid] = hist[tid];
-    __syncthreads();
-    for (int off = 1; off < SUB_BUCKETS; off <<= 1) {
-      uint32_t add = (tid < SUB_BUCKETS && tid >= off) ? base[tid - off] : 0;
-      __syncthreads();
-      if (tid < SUB_BUCKETS) base[tid] += add;
-      __syncthreads();
-    }
-    if (tid < SUB_BUCKETS) base[tid] -= hist[tid];
-    __syncthreads();
-#pragma unroll
-    for (int v = 0; v < VPT; v++) {
-      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
-      if (i < s1) tbuf[base[g[v]] + rank[v]] = r[v];
-    }
-    __syncthreads();
-    for (int pos = tid; pos < count; pos += blockDim.x) {
-      longlong2 row = tbuf[pos];
-      uint32_t gg = subB_of(row.x);
-      uint32_t dst = gcur[gg] + (uint32_t)(pos - base[gg]);
-      if (dst < glim[gg])
-        out_pairs[dst] = row;
-      else
-        s_ovf = 1;  // benign LDS race: any 1 wins
-    }
-    __syncthreads();
-    if (tid < SUB_BUCKETS) gcur[tid] += hist[tid];
-    __syncthreads();
-  }
-  __syncthreads();
-  if (tid < SUB_BUCKETS) {
-    const int64_t b0 = ((int64_t)a * SUB_BUCKETS + tid) * capB;
-    uint32_t len = gcur[tid] - (uint32_t)b0;
-    bsizes[(size_t)a * SUB_BUCKETS + tid] = min(len, (uint32_t)capB);
-  }
-  if (tid == 0 && s_ovf) atomicOr(any_overflow, 2);
+  staged_scatter_span<1, false>(nullptr, nullptr, in_pairs, s0, s1, SUB_BUCKETS, tbuf, hist,
+                                base, gcur, out_pairs);
 }
 
 void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
                        longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
                        int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
-                       int* d_any_overflow, uint32_t* d_bsizes, hipStream_t s)
+                       int* d_any_overflow, hipStream_t s)
 {
   DJ_CHECK_ERROR(n < (int64_t)UINT32_MAX, "bucket_partition: n must be < 2^32");
   const int PA = B / SUB_BUCKETS;
@@ -1250,10 +1200,10 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        subpart_lds, s, (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B,
                        d_out_pairs, d_offsets);
     DJ_HIP_CALL(hipGetLastError());
-  } else if (d_any_overflow != nullptr && d_bsizes != nullptr && slack_partition_ok(n, B)) {
+  } else if (d_any_overflow != nullptr &&
+             (int64_t)PA * slack_capA(n, PA) + n < (int64_t)UINT32_MAX) {
     /* slack path: no count pass (see bucket_scatter_slack_kernel header) */
     const int64_t capA = slack_capA(n, PA);
-    const int64_t capB = slack_capB(n, B);
     size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 4 * (size_t)PA * sizeof(uint32_t);
     size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
     DJ_HIP_CALL(hipMemsetAsync(d_totals, 0, (size_t)PA * 4, s));
@@ -1264,9 +1214,11 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
     hipLaunchKernelGGL(clamp_seglen_kernel, dim3((PA + 255) / 256), dim3(256), 0, s, d_totals,
                        PA, (uint32_t)capA);
     DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(bucket_scanB_kernel, dim3(1), dim3(BUCKET_THREADS), 0, s, d_totals, PA,
+                       d_segoff);
+    DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_subpart_slack_kernel, dim3(PA), dim3(BUCKET_THREADS), subpart_lds,
-                       s, d_tmp_pairs, d_totals, capA, capB, d_out_pairs, d_bsizes,
-                       d_any_overflow);
+                       s, d_tmp_pairs, d_totals, capA, d_segoff, B, d_out_pairs, d_offsets);
     DJ_HIP_CALL(hipGetLastError());
   } else {
     size_t hist_lds = (size_t)PA * sizeof(uint32_t);
@@ -1291,18 +1243,17 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
   }
 }
 
-void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const uint32_t* d_lsizes,
-              int64_t capBl, const longlong2* d_rrows, const int64_t* d_roff,
-              const uint32_t* d_rsizes, int64_t capBr, int B, int64_t* d_out0, int64_t* d_out1,
-              int64_t* d_out2, int64_t* d_out3, int64_t cap, int64_t* d_counter,
-              uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error, hipStream_t s)
+void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
+              const int64_t* d_roff, int B, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+              int64_t* d_out3, int64_t cap, int64_t* d_counter, uint32_t* d_overflow_flags,
+              int* d_any_overflow, int* d_error, hipStream_t s)
 {
   int grid = B < 8192 ? B : 8192;
   size_t lds = JOIN_LDS_SLOTS * sizeof(longlong2) + 4 * JOIN_STAGE_ROWS * sizeof(int64_t) + 16;
   hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lrows,
-                     d_loff, d_lsizes, capBl, d_rrows, d_roff, d_rsizes, capBr, B, d_out0,
-                     d_out1, d_out2, d_out3, cap, (unsigned long long*)d_counter,
-                     d_overflow_flags, d_any_overflow, d_error);
+                     d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
+                     (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
+                     d_error);
   DJ_HIP_CALL(hipGetLastError());
 }
 

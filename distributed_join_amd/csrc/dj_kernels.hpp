@@ -80,45 +80,25 @@ inline int64_t slack_capA(int64_t n, int PA)
   const int64_t m = n / PA;
   return m + m / 16 + 1024;
 }
-/* slack bucket-region capacity for pass B (rows per bucket): mean + 25% +
- * 128 covers hash-uniform inputs beyond 9 sigma */
-inline int64_t slack_capB(int64_t n, int B)
-{
-  const int64_t m = n / B;
-  return m + m / 4 + 128;
-}
-/* both slack layouts must index with u32 cursors */
-inline bool slack_partition_ok(int64_t n, int B)
-{
-  const int PA = B / kSubBuckets;
-  return PA > 1 && (int64_t)PA * slack_capA(n, PA) + n < (int64_t)UINT32_MAX &&
-         (int64_t)B * slack_capB(n, B) < (int64_t)UINT32_MAX;
-}
 /* Two-level non-stable partition into B buckets (B = PA*256, PA<=1024) of
- * interleaved 16 B {key,payload} pairs. When d_any_overflow AND d_bsizes
- * are non-null and slack_partition_ok(n, B): the SLACK path — no count or
- * scan kernels at all; d_tmp_pairs is longlong2[PA*slack_capA(n,PA)],
- * d_out_pairs is longlong2[B*slack_capB(n,B)] with bucket b at
- * [b*capB, b*capB + d_bsizes[b]) (d_offsets unused); skew beyond the slack
- * sets BIT 2 of *d_any_overflow and the output is INCOMPLETE — the caller
- * must redo the whole join. Otherwise the exact two-pass path: d_tmp_pairs
- * longlong2[n], contiguous d_offsets[B+1]. d_counts: u32[kBucketBlocks*PA];
- * d_totals: u32[PA]; d_segoff: int64[PA+1]. */
+ * interleaved 16 B {key,payload} pairs. d_tmp_pairs: pass-A staging —
+ * longlong2[PA * slack_capA(n, PA)] when d_any_overflow is non-null (the
+ * slack path: no count pass; group skew beyond the slack sets BIT 2 of
+ * *d_any_overflow and the partition output is INCOMPLETE — caller must redo
+ * the whole join), else longlong2[n] (exact two-pass path). d_counts:
+ * u32[kBucketBlocks*PA]; d_totals: u32[PA]; d_segoff: int64[PA+1];
+ * d_offsets: int64[B+1]. */
 void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
                        longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
                        int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
-                       int* d_any_overflow, uint32_t* d_bsizes, hipStream_t s);
+                       int* d_any_overflow, hipStream_t s);
 /* Fused per-bucket LDS build+probe over bucketed pair tables. Buckets whose
  * build side exceeds kJoinBucketRowCap set overflow_flags[b]/any_overflow
  * and are skipped (host runs the global-table path on them). */
-/* d_lsizes/d_rsizes non-null => that side's bucket b occupies
- * [b*capB, b*capB + sizes[b]) (the slack-partition layout); null => the
- * contiguous offsets convention [off[b], off[b+1]). */
-void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const uint32_t* d_lsizes,
-              int64_t capBl, const longlong2* d_rrows, const int64_t* d_roff,
-              const uint32_t* d_rsizes, int64_t capBr, int B, int64_t* d_out0, int64_t* d_out1,
-              int64_t* d_out2, int64_t* d_out3, int64_t cap, int64_t* d_counter,
-              uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error, hipStream_t s);
+void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
+              const int64_t* d_roff, int B, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+              int64_t* d_out3, int64_t cap, int64_t* d_counter, uint32_t* d_overflow_flags,
+              int* d_any_overflow, int* d_error, hipStream_t s);
 /* Global-table build/probe over interleaved pair inputs (skew fallback). */
 void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int64_t nslots,
                       int* d_error, hipStream_t s);
