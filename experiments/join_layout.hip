@@ -226,6 +226,95 @@ __global__ __launch_bounds__(THREADS) void join_kernel_v2(
   }
 }
 
+/* K consecutive buckets share one LDS table: a probe key can only equal a
+ * build key of its own bucket (equal keys hash to the same bucket), so the
+ * merged table is correct by construction; rows of K buckets are contiguous
+ * (coalesced full-wave loads) and the per-bucket barrier overhead drops Kx. */
+template <int K, int SLOTS2, int STAGE2>
+__global__ __launch_bounds__(THREADS) void join_kernel_multi(
+  const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
+  const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
+  int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
+  int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbl = (longlong2*)smem;
+  int64_t* stage = (int64_t*)(tbl + SLOTS2);
+  long long* base_sh = (long long*)(stage + 4 * STAGE2);
+  uint32_t* cur_sh = (uint32_t*)(base_sh + 1);
+  const uint32_t smask = SLOTS2 - 1;
+  const int tid = threadIdx.x;
+  const int nsuper = (B + K - 1) / K;
+
+  for (int sb = blockIdx.x; sb < nsuper; sb += gridDim.x) {
+    const int b0 = sb * K;
+    const int bK = min(b0 + K, B);
+    const int64_t l0 = loff[b0], l1 = loff[bK];
+    const int64_t r0 = roff[b0], r1 = roff[bK];
+    if (l1 - l0 == 0 || r1 == r0 || l1 - l0 > (int64_t)(SLOTS2 * 3 / 4)) continue;
+    for (int s2 = tid; s2 < SLOTS2; s2 += blockDim.x) tbl[s2].x = EMPTY;
+    if (tid == 0) *cur_sh = 0;
+    __syncthreads();
+    for (int64_t i = l0 + tid; i < l1; i += blockDim.x) {
+      longlong2 row = lrows[i];
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)row.x) & smask;
+      for (;;) {
+        unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
+                                           (unsigned long long)EMPTY,
+                                           (unsigned long long)row.x);
+        if (old == (unsigned long long)EMPTY) break;
+        slot = (slot + 1) & smask;
+      }
+      tbl[slot].y = row.y;
+    }
+    __syncthreads();
+    for (int64_t j = r0 + tid; j < r1; j += blockDim.x) {
+      longlong2 prow = rrows[j];
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)prow.x) & smask;
+      for (;;) {
+        longlong2 e = tbl[slot];
+        if (e.x == EMPTY) break;
+        if (e.x == prow.x) {
+          uint32_t pos = atomicAdd(cur_sh, 1u);
+          if (pos < (uint32_t)STAGE2) {
+            stage[0 * STAGE2 + pos] = prow.x;
+            stage[1 * STAGE2 + pos] = e.y;
+            stage[2 * STAGE2 + pos] = prow.x;
+            stage[3 * STAGE2 + pos] = prow.y;
+          } else {
+            long long idx = (long long)atomicAdd(counter, 1ull);
+            if (idx < cap) {
+              out0[idx] = prow.x;
+              out1[idx] = e.y;
+              out2[idx] = prow.x;
+              out3[idx] = prow.y;
+            }
+          }
+        }
+        slot = (slot + 1) & smask;
+      }
+    }
+    __syncthreads();
+    const uint32_t total = min(*cur_sh, (uint32_t)STAGE2);
+    if (tid == 0 && total)
+      *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
+    __syncthreads();
+    if (total) {
+      const long long base = *base_sh;
+      for (uint32_t i = tid; i < total; i += blockDim.x) {
+        long long idx = base + (long long)i;
+        if (idx < cap) {
+          out0[idx] = stage[0 * STAGE2 + i];
+          out1[idx] = stage[1 * STAGE2 + i];
+          out2[idx] = stage[2 * STAGE2 + i];
+          out3[idx] = stage[3 * STAGE2 + i];
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
 /* fill bucket b with sizes[b] rows: build keys b*4096+i, probe keys
  * b*4096+i for i<30% else non-matching */
 __global__ void fill_kernel(longlong2* rows, const int64_t* starts,
@@ -312,6 +401,15 @@ static double run_case(const char* name, int B, int nrows, int match, int64_t ca
     if (capB == -1)
       join_kernel_v2<<<grid, THREADS, lds>>>(lrows, loff, rrows, loff, B, o0, o1, o2, o3, cap,
                                              counter);
+    else if (capB == -2) {
+      size_t lds2 = 4096 * 16 + 4 * 384 * 8 + 16;
+      join_kernel_multi<2, 4096, 384><<<grid, THREADS, lds2>>>(lrows, loff, rrows, loff, B, o0,
+                                                               o1, o2, o3, cap, counter);
+    } else if (capB == -4) {
+      size_t lds4 = 8192 * 16 + 4 * 768 * 8 + 16;
+      join_kernel_multi<4, 8192, 768><<<grid, THREADS, lds4>>>(lrows, loff, rrows, loff, B, o0,
+                                                               o1, o2, o3, cap, counter);
+    }
     else if (capB)
       join_kernel<<<grid, THREADS, lds>>>(lrows, nullptr, sizes, capB, rrows, nullptr, sizes,
                                           capB, B, o0, o1, o2, o3, cap, counter);
@@ -348,6 +446,7 @@ int main(int argc, char** argv)
   int reps = 4;
   run_case("compact+offsets (base)", B, nrows, match, 0, reps, false, true, false);
   run_case("pipelined v2", B, nrows, match, -1, reps, false, true, false);
-  run_case("pipelined v2 varying", B, nrows, match, -1, reps, false, true, true);
+  run_case("multi K=2 (2 blocks/CU)", B, nrows, match, -2, reps, false, true, false);
+  run_case("multi K=4 (1 block/CU)", B, nrows, match, -4, reps, false, true, false);
   return 0;
 }
