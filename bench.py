@@ -169,8 +169,9 @@ def main():
     # this rank; derivation in DESIGN.md §Measurement).
     # N==1 (reference single-rank semantics: local join only, no
     # partition/shuffle stage — distributed_join.cpp:200-214):
-    #   bucket_scatter = the two-level local bucket partition
-    #     (8 B count-read + 16 B read + 16 B write per row per level)
+    #   bucket_scatter = the two-level local bucket partition: slack pass A
+    #     (16 B read + 16 B write, no count pass) + pass B (8 B count read +
+    #     16 B read + 16 B write) = 72 B per row per table
     # N>1 (fused wire path):
     #   part_scatter   = the fused rank+group partition (8 B count-read +
     #                    16 B read + 16 B write per row, both tables)
@@ -182,7 +183,7 @@ def main():
     if N == 1:
         alg = {
             "part_scatter": 1.0,  # not run at N==1
-            "bucket_scatter": 40.0 * 2 * (lrows + rrows),
+            "bucket_scatter": 72.0 * (lrows + rrows),
             "join_fused": 16.0 * (lrows + rrows) + 32.0 * matches,
         }
     else:
