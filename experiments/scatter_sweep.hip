@@ -43,6 +43,33 @@ __device__ __forceinline__ uint32_t subB_of(int64_t k)
   return (uint32_t)(dj_mix64((uint64_t)k) >> 32) & (uint32_t)(SUB - 1);
 }
 
+/* exclusive scan of P per-thread values (tid<P holds base[tid]) via wave
+ * shfl scans + one cross-wave round: 2 barriers instead of 2*log2(P) */
+__device__ __forceinline__ void wave_excl_scan(uint32_t* base, uint32_t* wsum, int P, int tid)
+{
+  const int lane = tid & 63, wid = tid >> 6;
+  const int nw = P >> 6;
+  uint32_t v = (tid < P) ? base[tid] : 0;
+  uint32_t x = v;
+  for (int d = 1; d < 64; d <<= 1) {
+    uint32_t y = __shfl_up(x, d);
+    if (lane >= d) x += y;
+  }
+  if (tid < P && lane == 63) wsum[wid] = x;
+  __syncthreads();
+  if (wid == 0) {
+    uint32_t w = (lane < nw) ? wsum[lane] : 0;
+    uint32_t xx = w;
+    for (int d = 1; d < 64; d <<= 1) {
+      uint32_t y = __shfl_up(xx, d);
+      if (lane >= d) xx += y;
+    }
+    if (lane < nw) wsum[lane] = xx - w;  // exclusive wave bases
+  }
+  __syncthreads();
+  if (tid < P) base[tid] = x - v + wsum[wid];
+}
+
 __global__ void gen_kernel(int64_t* keys, int64_t* pay, int64_t n)
 {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -68,7 +95,7 @@ __global__ __launch_bounds__(THREADS) void countA_kernel(const int64_t* __restri
 
 /* pass A scatter, templated staging tile.  ATOMIC_CUR: per-tile global
  * atomicAdd cursor per group into slack segments (g*cap), no counts/segoff. */
-template <int TILE, bool ATOMIC_CUR>
+template <int TILE, bool ATOMIC_CUR, bool WAVESCAN = false>
 __global__ __launch_bounds__(THREADS) void scatterA_kernel(
   const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n,
   const uint32_t* __restrict__ counts, const int64_t* __restrict__ segoff,
@@ -79,6 +106,7 @@ __global__ __launch_bounds__(THREADS) void scatterA_kernel(
   uint32_t* hist = (uint32_t*)(tbuf + TILE);
   uint32_t* base = hist + PA;
   uint32_t* gcur = base + PA;
+  __shared__ uint32_t wsum[16];
   const int tid = threadIdx.x;
   if (!ATOMIC_CUR && tid < PA)
     gcur[tid] = (uint32_t)segoff[tid] + counts[(size_t)blockIdx.x * PA + tid];
@@ -106,13 +134,17 @@ __global__ __launch_bounds__(THREADS) void scatterA_kernel(
     __syncthreads();
     if (tid < PA) base[tid] = hist[tid];
     __syncthreads();
-    for (int off = 1; off < PA; off <<= 1) {
-      uint32_t add = (tid < PA && tid >= off) ? base[tid - off] : 0;
-      __syncthreads();
-      if (tid < PA) base[tid] += add;
-      __syncthreads();
+    if (WAVESCAN) {
+      wave_excl_scan(base, wsum, PA, tid);
+    } else {
+      for (int off = 1; off < PA; off <<= 1) {
+        uint32_t add = (tid < PA && tid >= off) ? base[tid - off] : 0;
+        __syncthreads();
+        if (tid < PA) base[tid] += add;
+        __syncthreads();
+      }
+      if (tid < PA) base[tid] -= hist[tid];
     }
-    if (tid < PA) base[tid] -= hist[tid];
     __syncthreads();
     if (ATOMIC_CUR && tid < PA)
       gcur[tid] = (uint32_t)((int64_t)tid * cap) +
@@ -132,7 +164,7 @@ __global__ __launch_bounds__(THREADS) void scatterA_kernel(
 }
 
 /* pass B: one block per pass-A group segment, SUB-way sub-split */
-template <int TILE>
+template <int TILE, bool WAVESCAN = false>
 __global__ __launch_bounds__(THREADS) void scatterB_kernel(const longlong2* __restrict__ in,
                                                            const int64_t* __restrict__ segoff,
                                                            longlong2* __restrict__ out)
@@ -143,6 +175,7 @@ __global__ __launch_bounds__(THREADS) void scatterB_kernel(const longlong2* __re
   uint32_t* base = hist + SUB;
   uint32_t* gcur = base + SUB;
   uint32_t* seghist = gcur + SUB;
+  __shared__ uint32_t wsum[16];
   const int tid = threadIdx.x;
   const int64_t s0 = segoff[blockIdx.x], s1 = segoff[blockIdx.x + 1];
   if (tid < SUB) seghist[tid] = 0;
@@ -179,13 +212,17 @@ __global__ __launch_bounds__(THREADS) void scatterB_kernel(const longlong2* __re
     __syncthreads();
     if (tid < SUB) base[tid] = hist[tid];
     __syncthreads();
-    for (int off = 1; off < SUB; off <<= 1) {
-      uint32_t add = (tid < SUB && tid >= off) ? base[tid - off] : 0;
-      __syncthreads();
-      if (tid < SUB) base[tid] += add;
-      __syncthreads();
+    if (WAVESCAN) {
+      wave_excl_scan(base, wsum, SUB, tid);
+    } else {
+      for (int off = 1; off < SUB; off <<= 1) {
+        uint32_t add = (tid < SUB && tid >= off) ? base[tid - off] : 0;
+        __syncthreads();
+        if (tid < SUB) base[tid] += add;
+        __syncthreads();
+      }
+      if (tid < SUB) base[tid] -= hist[tid];
     }
-    if (tid < SUB) base[tid] -= hist[tid];
     __syncthreads();
     for (int v = 0; v < nv; v++) tbuf[base[g[v]] + rank[v]] = r[v];
     __syncthreads();
@@ -230,15 +267,26 @@ void run_variant(const int64_t* keys, const int64_t* pay, int64_t n, uint32_t* c
   CHECK(hipEventCreate(&e1));
   float msA = 0, msB = 0, msAat = 0;
   const int64_t cap = n / PA + n / PA / 16 + 1024;  // ~6% slack
+  float msAw = 0, msBw = 0;
   for (int rep = 0; rep < reps; rep++) {
+    float ms;
+    CHECK(hipMemset(gcursor, 0, PA * 8));
     CHECK(hipEventRecord(e0));
-    scatterA_kernel<TILE, false>
+    scatterA_kernel<TILE, true>
       <<<BLOCKS, THREADS, ldsA>>>(keys, pay, n, counts, segoff_a, gcursor, cap, mid);
     CHECK(hipEventRecord(e1));
     CHECK(hipEventSynchronize(e1));
-    float ms;
     CHECK(hipEventElapsedTime(&ms, e0, e1));
-    if (rep) msA += ms;
+    if (rep) msAat += ms;
+
+    CHECK(hipMemset(gcursor, 0, PA * 8));
+    CHECK(hipEventRecord(e0));
+    scatterA_kernel<TILE, true, true>
+      <<<BLOCKS, THREADS, ldsA>>>(keys, pay, n, counts, segoff_a, gcursor, cap, mid);
+    CHECK(hipEventRecord(e1));
+    CHECK(hipEventSynchronize(e1));
+    CHECK(hipEventElapsedTime(&ms, e0, e1));
+    if (rep) msAw += ms;
 
     CHECK(hipEventRecord(e0));
     scatterB_kernel<TILE><<<PA, THREADS, ldsB>>>(mid, segoff_a, out);
@@ -247,21 +295,17 @@ void run_variant(const int64_t* keys, const int64_t* pay, int64_t n, uint32_t* c
     CHECK(hipEventElapsedTime(&ms, e0, e1));
     if (rep) msB += ms;
 
-    CHECK(hipMemset(gcursor, 0, PA * 8));
     CHECK(hipEventRecord(e0));
-    scatterA_kernel<TILE, true>
-      <<<BLOCKS, THREADS, ldsA>>>(keys, pay, n, counts, segoff_a, gcursor, cap, out);
+    scatterB_kernel<TILE, true><<<PA, THREADS, ldsB>>>(mid, segoff_a, out);
     CHECK(hipEventRecord(e1));
     CHECK(hipEventSynchronize(e1));
     CHECK(hipEventElapsedTime(&ms, e0, e1));
-    if (rep) msAat += ms;
+    if (rep) msBw += ms;
   }
   CHECK(hipGetLastError());
   int r = reps - 1;
-  double gb = 32.0 * n / 1e9;  // payload bytes moved per pass (16 r + 16 w)
-  printf("TILE %5d | passA %7.3f ms (%6.1f GB/s) | passB %7.3f ms (%6.1f GB/s) | "
-         "passA-atomiccur %7.3f ms\n",
-         TILE, msA / r, gb / (msA / r / 1e3), msB / r, gb / (msB / r / 1e3), msAat / r);
+  printf("TILE %5d | A-atomic %7.3f -> wavescan %7.3f ms | B %7.3f -> wavescan %7.3f ms\n",
+         TILE, msAat / r, msAw / r, msB / r, msBw / r);
   CHECK(hipEventDestroy(e0));
   CHECK(hipEventDestroy(e1));
 }
