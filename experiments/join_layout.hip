@@ -30,7 +30,8 @@ constexpr int STAGE = 1024;
 constexpr int64_t EMPTY = -1;
 constexpr int ROWCAP = 1536;
 
-__global__ __launch_bounds__(THREADS) void join_kernel(
+template <int SL = SLOTS, int ST = STAGE, int TH = THREADS>
+__global__ __launch_bounds__(TH) void join_kernel(
   const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
   const uint32_t* __restrict__ lsizes, int64_t capBl, const longlong2* __restrict__ rrows,
   const int64_t* __restrict__ roff, const uint32_t* __restrict__ rsizes, int64_t capBr, int B,
@@ -39,11 +40,11 @@ __global__ __launch_bounds__(THREADS) void join_kernel(
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbl = (longlong2*)smem;
-  int64_t* stage = (int64_t*)(tbl + SLOTS);
-  long long* base_sh = (long long*)(stage + 4 * STAGE);
+  int64_t* stage = (int64_t*)(tbl + SL);
+  long long* base_sh = (long long*)(stage + 4 * ST);
   uint32_t* cur_sh = (uint32_t*)(base_sh + 1);
-  const uint32_t smask = SLOTS - 1;
-  constexpr int S = STAGE;
+  const uint32_t smask = SL - 1;
+  constexpr int S = ST;
 
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
     const int64_t l0 = lsizes ? (int64_t)b * capBl : loff[b];
@@ -51,8 +52,8 @@ __global__ __launch_bounds__(THREADS) void join_kernel(
     const int64_t r0 = rsizes ? (int64_t)b * capBr : roff[b];
     const int64_t r1 = rsizes ? r0 + rsizes[b] : roff[b + 1];
     const int64_t lnb = l1 - l0;
-    if (lnb == 0 || r1 == r0 || lnb > ROWCAP) continue;
-    for (int s = threadIdx.x; s < SLOTS; s += blockDim.x) tbl[s].x = EMPTY;
+    if (lnb == 0 || r1 == r0 || lnb > SL * 3 / 4) continue;
+    for (int s = threadIdx.x; s < SL; s += blockDim.x) tbl[s].x = EMPTY;
     if (threadIdx.x == 0) *cur_sh = 0;
     __syncthreads();
     for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
@@ -398,7 +399,17 @@ static double run_case(const char* name, int B, int nrows, int match, int64_t ca
       fill_kernel<<<B, 256>>>(rrows, starts, sizes, B, 1);
     }
     CHECK(hipEventRecord(e0));
-    if (capB == -1)
+    if (capB == -5) {  /* bigger buckets: 4096-slot table, 1 block/CU */
+      size_t lds5 = 4096 * 16 + 4 * 1024 * 8 + 16;
+      join_kernel<4096, 1024, 1024><<<grid, 1024, lds5>>>(lrows, loff, nullptr, 0, rrows, loff,
+                                                          nullptr, 0, B, o0, o1, o2, o3, cap,
+                                                          counter);
+    } else if (capB == -6) {  /* smaller buckets: 1024-slot, 512 thr, 4 blocks/CU */
+      size_t lds6 = 1024 * 16 + 4 * 512 * 8 + 16;
+      join_kernel<1024, 512, 512><<<grid, 512, lds6>>>(lrows, loff, nullptr, 0, rrows, loff,
+                                                       nullptr, 0, B, o0, o1, o2, o3, cap,
+                                                       counter);
+    } else if (capB == -1)
       join_kernel_v2<<<grid, THREADS, lds>>>(lrows, loff, rrows, loff, B, o0, o1, o2, o3, cap,
                                              counter);
     else if (capB == -2) {
@@ -411,11 +422,11 @@ static double run_case(const char* name, int B, int nrows, int match, int64_t ca
                                                                o1, o2, o3, cap, counter);
     }
     else if (capB)
-      join_kernel<<<grid, THREADS, lds>>>(lrows, nullptr, sizes, capB, rrows, nullptr, sizes,
-                                          capB, B, o0, o1, o2, o3, cap, counter);
+      join_kernel<><<<grid, THREADS, lds>>>(lrows, nullptr, sizes, capB, rrows, nullptr, sizes,
+                                            capB, B, o0, o1, o2, o3, cap, counter);
     else
-      join_kernel<<<grid, THREADS, lds>>>(lrows, loff, nullptr, 0, rrows, loff, nullptr, 0, B,
-                                          o0, o1, o2, o3, cap, counter);
+      join_kernel<><<<grid, THREADS, lds>>>(lrows, loff, nullptr, 0, rrows, loff, nullptr, 0,
+                                            B, o0, o1, o2, o3, cap, counter);
     CHECK(hipEventRecord(e1));
     CHECK(hipEventSynchronize(e1));
     float ms;
@@ -444,9 +455,10 @@ int main(int argc, char** argv)
   int nrows = argc > 2 ? atoi(argv[2]) : 762;
   int match = (int)(nrows * 0.3);
   int reps = 4;
-  run_case("compact+offsets (base)", B, nrows, match, 0, reps, false, true, false);
-  run_case("pipelined v2", B, nrows, match, -1, reps, false, true, false);
-  run_case("multi K=2 (2 blocks/CU)", B, nrows, match, -2, reps, false, true, false);
-  run_case("multi K=4 (1 block/CU)", B, nrows, match, -4, reps, false, true, false);
+  run_case("base B=131072 r=762", B, nrows, match, 0, reps, false, true, false);
+  run_case("big B/2 r=1526 1blk/CU", B / 2, nrows * 2, (int)(nrows * 2 * 0.3), -5, reps,
+           false, true, false);
+  run_case("small 2B r=381 512thr", B * 2, nrows / 2, (int)(nrows / 2 * 0.3), -6, reps,
+           false, true, false);
   return 0;
 }
