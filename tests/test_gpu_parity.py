@@ -180,3 +180,22 @@ def test_negative_keys(dj):
     rk = np.array([-3, 7, -5, 123, -2**62], dtype=np.int64)
     rp = np.arange(5, dtype=np.int64)
     _join_parity(dj, lk, lp, rk, rp)
+
+
+def test_slack_partition_overflow_redo(dj):
+    # at this size the local partition takes the slack pass-A path (PA > 1,
+    # capA ~ n/PA + 6%); a 50k-copy hot key pushes its pass-A group past the
+    # slack (bit 2 of any_overflow) and the whole join must redo via the
+    # global-table path with identical results (dj_capi.hip
+    # dj_bucket_local_join). 50k duplicates keep the fallback's collision
+    # chain quadratic-cost bounded (~1.2e9 slot probes).
+    n = 1_000_000
+    rng = np.random.default_rng(7)
+    lk = rng.integers(1 << 40, size=n).astype(np.int64)
+    lk[:50_000] = 42  # skew one pass-A group past its slack capacity
+    lp = np.arange(n, dtype=np.int64)
+    rk = np.concatenate(
+        [np.array([42], dtype=np.int64), rng.integers(1 << 40, size=5000).astype(np.int64)]
+    )
+    rp = np.arange(rk.size, dtype=np.int64)
+    _join_parity(dj, lk, lp, rk, rp)
