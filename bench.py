@@ -73,6 +73,12 @@ def main():
     ap.add_argument("--rows", type=int, default=100_000_000,
                     help="rows per GPU per table (build and probe)")
     ap.add_argument("--over-decom", type=int, default=1)
+    ap.add_argument("--nvlink-domain-size", type=int, default=0,
+                    help="0 = world size (one xGMI node is ONE domain -> batched "
+                         "all-to-all pipeline / fused wire path). The reference's "
+                         "README benchmark ran its default of 1 (IB shuffle + local "
+                         "join, distributed_join.cpp:152-214); pass 1 to reproduce "
+                         "that configuration.")
     ap.add_argument("--cpu-baseline-rows", type=int, default=100_000_000)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
@@ -124,9 +130,12 @@ def main():
 
     state = {"matches": 0}
 
+    nvl = args.nvlink_domain_size if args.nvlink_domain_size > 0 else N
+
     def step():
-        t = L.dj_cpp_distributed_inner_join_i64(comm.ptr, bk.ptr, bp.ptr, rows,
-                                                pk.ptr, pp.ptr, rows, args.over_decom, 0)
+        t = L.dj_cpp_distributed_inner_join_i64_full(comm.ptr, bk.ptr, bp.ptr, rows,
+                                                     pk.ptr, pp.ptr, rows, args.over_decom,
+                                                     0, 0, nvl)
         state["matches"] = L.dj_table_num_rows(t)
         L.dj_table_free(t)
 
@@ -248,6 +257,7 @@ def main():
                             f"{rows//1_000_000}M rows/GPU/table, over_decom {args.over_decom}",
                 "rows_per_gpu": rows,
                 "selectivity": SELECTIVITY,
+                "nvlink_domain_size": nvl,
                 "output_rows_per_gpu": int(matches),
                 "output_rows_per_sec": out_rows_s,
                 "engine": "C++ distributed_inner_join (drop-in path); at N=1 the "

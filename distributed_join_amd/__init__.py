@@ -88,6 +88,8 @@ def lib():
             "dj_cpp_shuffle_on_i64_comp": ([vp, vp, vp, i64, i32, u32, i32], vp),
             "dj_cpp_distributed_inner_join_i64_opts": ([vp, vp, vp, i64, vp, vp, i64,
                                                         i32, i32, i32], vp),
+            "dj_cpp_distributed_inner_join_i64_full": ([vp, vp, vp, i64, vp, vp, i64,
+                                                        i32, i32, i32, i32], vp),
             "dj_gen_test_strings": ([vp, i64, ctypes.POINTER(ctypes.c_void_p),
                                      ctypes.POINTER(ctypes.c_void_p),
                                      ctypes.POINTER(ctypes.c_int64)], None),

@@ -1966,6 +1966,34 @@ void* dj_cpp_distributed_inner_join_i64_opts(void* comm, const int64_t* d_lk,
   return result.release();
 }
 
+/* full-featured variant: adds nvlink_domain_size. The reference's README
+ * benchmark runs with its default nvlink_domain_size=1 (IB-domain shuffle of
+ * both tables + local join, distributed_join.cpp:152-199 + README.md:73-86);
+ * on one 8x MI355X node the whole node is ONE xGMI domain, so
+ * nvlink_domain_size = world engages the batched all-to-all pipeline (fused
+ * wire path + comm/compute overlap) — the MI355X-correct configuration. */
+void* dj_cpp_distributed_inner_join_i64_full(void* comm, const int64_t* d_lk,
+                                             const int64_t* d_lp, int64_t ln,
+                                             const int64_t* d_rk, const int64_t* d_rp,
+                                             int64_t rn, int over_decom, int report_timing,
+                                             int compression, int nvlink_domain_size)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  cudf::table_view left(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)ln, d_lk),
+     column_view(data_type(type_id::INT64), (cudf::size_type)ln, d_lp)});
+  cudf::table_view right(
+    {column_view(data_type(type_id::INT64), (cudf::size_type)rn, d_rk),
+     column_view(data_type(type_id::INT64), (cudf::size_type)rn, d_rp)});
+  auto opts = generate_compression_options_distributed(left, compression != 0);
+  auto result = distributed_inner_join(left, right, {0}, {0}, (Communicator*)comm, opts, opts,
+                                       over_decom, report_timing != 0, nullptr,
+                                       nvlink_domain_size);
+  return result.release();
+}
+
 void* dj_cpp_shuffle_on_i64_comp(void* comm, const int64_t* d_keys, const int64_t* d_pay,
                                  int64_t n, int hash_function, uint32_t seed, int compression)
 {

@@ -140,3 +140,22 @@ def test_rccl_selftest(dj):
     the exact calls the N>1 peer-slice exchange makes. De-risks the driver's
     multi-GPU runs on a single-GPU box."""
     assert dj.lib().dj_rccl_selftest(1 << 20) == 0
+
+
+def test_full_variant_nvl(dj, comm):
+    """The full-featured C ABI wrapper (adds nvlink_domain_size) matches the
+    oracle at world=1 for any domain size (get_nvl_partition_size collapses
+    to the local join, distributed_join.cpp:200-214)."""
+    n = 200_000
+    lk, lp = oracle.gen_build(n)
+    rk, rp = oracle.gen_probe(n, n)
+    dlk, dlp = dj.DeviceArray.from_numpy(lk), dj.DeviceArray.from_numpy(lp)
+    drk, drp = dj.DeviceArray.from_numpy(rk), dj.DeviceArray.from_numpy(rp)
+    want = oracle.sort_rows(*oracle.inner_join(lk, lp, rk, rp))
+    for nvl in (1, 8):
+        t = dj.lib().dj_cpp_distributed_inner_join_i64_full(
+            comm.ptr, dlk.ptr, dlp.ptr, n, drk.ptr, drp.ptr, n, 1, 0, 0, nvl)
+        got = oracle.sort_rows(*dj.table_to_numpy(t))
+        dj.lib().dj_table_free(t)
+        for a, b in zip(got, want):
+            assert (a == b).all()
