@@ -332,7 +332,7 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
   }
   {
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
-    dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, d_out0, d_out1, d_out2,
+    dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, 2048, d_out0, d_out1, d_out2,
                  d_out3, cap, d_counter, s.flags, d_any_overflow, d_error, st);
   }
 }

@@ -1267,6 +1267,7 @@ std::unique_ptr<cudf::table> distributed_inner_join_fused(
     DBuf flags;            // u32[B]
     DBuf o0, o1, o2, o3, meta;
     int F{0};
+    int join_slots{2048};
     int64_t cap{0}, lrows{0}, rrows{0};
   };
   std::vector<Batch> batches(od);
@@ -1324,11 +1325,22 @@ std::unique_ptr<cudf::table> distributed_inner_join_fused(
     build_segs(bt.lsub_recv, *bt.latoa, bt.lseg, bt.lgb);
     build_segs(bt.rsub_recv, *bt.ratoa, bt.rseg, bt.rgb);
     DJ_HIP_CALL(hipStreamSynchronize(st));
-    /* sub-bucket fanout: target ~760 rows per final bucket */
+    /* sub-bucket fanout: target ~760 rows per final bucket. F caps at 1024
+     * (the staged span's one-group-per-thread scan); when buckets then run
+     * big (e.g. G=8 od=1: 100M over PA*F=65536 => ~1526 rows) the join
+     * switches to its 4096-slot table instead of overflowing every bucket. */
     int64_t maxn = std::max(bt.lrows, bt.rrows);
     int F = 64;
     while (F < 1024 && (int64_t)PA * F * 760 < maxn) F <<= 1;
     bt.F = F;
+    if (const char* ff = getenv("DJ_FORCE_FUSED_F")) {
+      /* TEST HOOK (tests/test_gpu_cpp_api.py::test_fused_big_buckets): force
+       * a small fan-out so the 4096-slot join path is exercisable on one
+       * GPU without an 8-rank 100M-row workload */
+      int v = atoi(ff);
+      if (v >= 64 && v <= 1024 && (v & (v - 1)) == 0) F = bt.F = v;
+    }
+    bt.join_slots = (maxn / ((int64_t)PA * F) > 1300) ? 4096 : 2048;
     int64_t B = (int64_t)PA * F;
     bt.lpairs = DBuf((size_t)std::max<int64_t>(bt.lrows, 1) * 16);
     bt.rpairs = DBuf((size_t)std::max<int64_t>(bt.rrows, 1) * 16);
@@ -1364,8 +1376,8 @@ std::unique_ptr<cudf::table> distributed_inner_join_fused(
     {
       dj_timing::Scope t(DJ_PHASE_JOIN_FUSED, st);
       dj::lds_join((const longlong2*)bt.lpairs.p, bt.loff.i64(),
-                   (const longlong2*)bt.rpairs.p, bt.roff.i64(), (int)B, bt.o0.i64(),
-                   bt.o1.i64(), bt.o2.i64(), bt.o3.i64(), bt.cap, bt.meta.i64(),
+                   (const longlong2*)bt.rpairs.p, bt.roff.i64(), (int)B, bt.join_slots,
+                   bt.o0.i64(), bt.o1.i64(), bt.o2.i64(), bt.o3.i64(), bt.cap, bt.meta.i64(),
                    (uint32_t*)bt.flags.p, (int*)((char*)bt.meta.p + 12),
                    (int*)((char*)bt.meta.p + 8), st);
     }

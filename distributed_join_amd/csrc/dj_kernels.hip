@@ -820,6 +820,12 @@ void subpart_lists(const int64_t* d_keys, const int64_t* d_pay, const int64_t* d
  * probe walk — ~0.8 ms at 100 M rows, profiles/r01_ablation.txt.) Buckets
  * whose matches exceed the stage spill directly to the global counter per
  * lane — correct for any duplication factor, slower only on such buckets. */
+/* SLOTS2 = 2048 (2 blocks/CU, buckets <= 1536 build rows — the local-join
+ * shape) or 4096 (1 block/CU, buckets <= 3072 — used by the G>1 fused wire
+ * path when its PA*F fan-out cap leaves ~1.5-3k rows/bucket, e.g. G=8 od=1;
+ * measured 1.98 vs 1.67 ms at the same total rows,
+ * experiments/join_layout.hip — far cheaper than whole-batch redos). */
+template <int SLOTS2>
 __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
   const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
@@ -830,10 +836,10 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbl = (longlong2*)smem;
-  int64_t* stage = (int64_t*)(tbl + JOIN_LDS_SLOTS);  // SoA: stage[c*JOIN_STAGE_ROWS + i]
+  int64_t* stage = (int64_t*)(tbl + SLOTS2);  // SoA: stage[c*JOIN_STAGE_ROWS + i]
   long long* base_sh = (long long*)(stage + 4 * JOIN_STAGE_ROWS);
   uint32_t* cur_sh = (uint32_t*)(base_sh + 1);
-  const uint32_t smask = JOIN_LDS_SLOTS - 1;
+  const uint32_t smask = SLOTS2 - 1;
   constexpr int S = JOIN_STAGE_ROWS;
 
   for (int b = blockIdx.x; b < B; b += gridDim.x) {
@@ -841,14 +847,14 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
     const int64_t r0 = roff[b], r1 = roff[b + 1];
     const int64_t lnb = l1 - l0;
     if (lnb == 0 || r1 == r0) continue;
-    if (lnb > kJoinBucketRowCap) {  // skew overflow: host-side fallback joins it
+    if (lnb > SLOTS2 * 3 / 4) {  // skew overflow: host-side fallback joins it
       if (threadIdx.x == 0) {
         overflow_flags[b] = 1;
         atomicOr(any_overflow, 1);  // bit 2 is the slack-partition overflow
       }
       continue;
     }
-    for (int s = threadIdx.x; s < JOIN_LDS_SLOTS; s += blockDim.x) tbl[s].x = kEmptyKey;
+    for (int s = threadIdx.x; s < SLOTS2; s += blockDim.x) tbl[s].x = kEmptyKey;
     if (threadIdx.x == 0) *cur_sh = 0;
     __syncthreads();
     /* build */
@@ -1244,16 +1250,25 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
 }
 
 void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
-              const int64_t* d_roff, int B, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
-              int64_t* d_out3, int64_t cap, int64_t* d_counter, uint32_t* d_overflow_flags,
-              int* d_any_overflow, int* d_error, hipStream_t s)
+              const int64_t* d_roff, int B, int table_slots, int64_t* d_out0, int64_t* d_out1,
+              int64_t* d_out2, int64_t* d_out3, int64_t cap, int64_t* d_counter,
+              uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error, hipStream_t s)
 {
+  DJ_CHECK_ERROR(table_slots == 2048 || table_slots == 4096,
+                 "lds_join: table_slots must be 2048 or 4096");
   int grid = B < 8192 ? B : 8192;
-  size_t lds = JOIN_LDS_SLOTS * sizeof(longlong2) + 4 * JOIN_STAGE_ROWS * sizeof(int64_t) + 16;
-  hipLaunchKernelGGL(lds_join_kernel, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lrows,
-                     d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3, cap,
-                     (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
-                     d_error);
+  size_t lds =
+    (size_t)table_slots * sizeof(longlong2) + 4 * JOIN_STAGE_ROWS * sizeof(int64_t) + 16;
+  if (table_slots == 4096)
+    hipLaunchKernelGGL(lds_join_kernel<4096>, dim3(grid), dim3(BUCKET_THREADS), lds, s,
+                       d_lrows, d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3,
+                       cap, (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
+                       d_error);
+  else
+    hipLaunchKernelGGL(lds_join_kernel<2048>, dim3(grid), dim3(BUCKET_THREADS), lds, s,
+                       d_lrows, d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3,
+                       cap, (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
+                       d_error);
   DJ_HIP_CALL(hipGetLastError());
 }
 

@@ -92,13 +92,16 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
                        int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
                        int* d_any_overflow, hipStream_t s);
-/* Fused per-bucket LDS build+probe over bucketed pair tables. Buckets whose
- * build side exceeds kJoinBucketRowCap set overflow_flags[b]/any_overflow
- * and are skipped (host runs the global-table path on them). */
+/* Fused per-bucket LDS build+probe over bucketed pair tables. table_slots:
+ * 2048 (2 blocks/CU, bucket cap 1536 build rows = kJoinBucketRowCap) or
+ * 4096 (1 block/CU, cap 3072 — for the fused wire path when the PA*F
+ * fan-out cap leaves big buckets). Buckets whose build side exceeds the cap
+ * set overflow_flags[b]/any_overflow and are skipped (host runs the
+ * global-table path on them / redoes the batch). */
 void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
-              const int64_t* d_roff, int B, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
-              int64_t* d_out3, int64_t cap, int64_t* d_counter, uint32_t* d_overflow_flags,
-              int* d_any_overflow, int* d_error, hipStream_t s);
+              const int64_t* d_roff, int B, int table_slots, int64_t* d_out0, int64_t* d_out1,
+              int64_t* d_out2, int64_t* d_out3, int64_t cap, int64_t* d_counter,
+              uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error, hipStream_t s);
 /* Global-table build/probe over interleaved pair inputs (skew fallback). */
 void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int64_t nslots,
                       int* d_error, hipStream_t s);

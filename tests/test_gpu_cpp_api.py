@@ -158,3 +158,20 @@ def test_full_variant_nvl(dj, comm):
         got = oracle.sort_rows(*dj.table_to_numpy(t))  # frees t
         for a, b in zip(got, want):
             assert (a == b).all()
+
+
+def test_fused_big_buckets():
+    """G=8 od=1 shape on one GPU: when the fused wire path's PA*F fan-out cap
+    leaves >1300 rows per bucket, the join must take its 4096-slot LDS table
+    (not overflow every bucket into whole-batch redos). DJ_FORCE_FUSED_F=64
+    shrinks the fan-out so 2 loopback ranks at 80M global rows reproduce the
+    big-bucket regime (~2400 rows/bucket)."""
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exe = os.path.join(repo, "tests", "cpp", "multirank_loopback")
+    assert os.path.exists(exe), "built by test_multirank_loopback"
+    env = dict(os.environ, DJ_FORCE_FUSED_F="64")
+    r = subprocess.run([exe, "2", "1", "80000000", "2", "0"], capture_output=True,
+                       text=True, timeout=420, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "MULTIRANK OK" in r.stdout
