@@ -390,6 +390,160 @@ std::vector<ColumnCompressionOptions> generate_compression_options_distributed(
   return opts;
 }
 
+std::vector<ColumnCompressionOptions> generate_none_compression_options(cudf::table_view input)
+{
+  std::vector<ColumnCompressionOptions> opts;
+  for (cudf::size_type c = 0; c < input.num_columns(); c++) {
+    if (input.column(c).type().id() == cudf::type_id::STRING) {
+      std::vector<ColumnCompressionOptions> kids(
+        2, ColumnCompressionOptions(CompressionMethod::none));
+      opts.emplace_back(CompressionMethod::none, nvcompCascadedFormatOpts{}, kids);
+    } else {
+      opts.emplace_back(CompressionMethod::none);
+    }
+  }
+  return opts;
+}
+
+namespace {
+
+/* sampling selector over our executable cascaded schemes ({0|1 deltas} +
+ * bitpack; RLE passes are not implemented — dj_compress.hip header): picks
+ * the scheme with the smaller packed-bit total on a host-side sample, in
+ * the role of nvcomp's CascadedSelector (reference compression.hpp:253-292,
+ * compression.cpp:36-69). Selection is data-dependent, not parity-pinned
+ * (nvcomp is absent). */
+nvcompCascadedFormatOpts select_cascaded_on_sample(const void* d_data, int64_t n, int esize)
+{
+  nvcompCascadedFormatOpts o{};
+  o.use_bp = 1;
+  const int64_t sample = std::min<int64_t>(n, 65536);
+  if (sample < 2) return o;
+  std::vector<int64_t> h((size_t)sample);
+  if (esize == 8) {
+    DJ_HIP_CALL(hipMemcpy(h.data(), d_data, (size_t)sample * 8, hipMemcpyDeviceToHost));
+  } else {
+    std::vector<int32_t> h32((size_t)sample);
+    DJ_HIP_CALL(hipMemcpy(h32.data(), d_data, (size_t)sample * 4, hipMemcpyDeviceToHost));
+    for (int64_t i = 0; i < sample; i++) h[(size_t)i] = h32[(size_t)i];
+  }
+  auto zigzag = [](int64_t v) { return ((uint64_t)v << 1) ^ (uint64_t)(v >> 63); };
+  auto packed_bits = [&](bool delta) {
+    uint64_t total = 0;
+    for (int64_t g = 0; g < sample; g += 32) {
+      int w = 0;
+      const int64_t e = std::min<int64_t>(g + 32, sample);
+      for (int64_t i = g; i < e; i++) {
+        int64_t v = delta ? (i ? h[(size_t)i] - h[(size_t)i - 1] : h[0]) : h[(size_t)i];
+        uint64_t z = zigzag(v) | 1;
+        w = std::max(w, 64 - __builtin_clzll(z));
+      }
+      total += (uint64_t)w * 32;
+    }
+    return total;
+  };
+  o.num_deltas = packed_bits(true) < packed_bits(false) ? 1 : 0;
+  return o;
+}
+
+}  // namespace
+
+std::vector<ColumnCompressionOptions> generate_auto_select_compression_options(
+  cudf::table_view input_table)
+{
+  /* reference compression.cpp:36-69: per-column sampling selection; STRING
+   * columns: select on the offsets child, never compress chars */
+  std::vector<ColumnCompressionOptions> opts;
+  for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
+    cudf::column_view col = input_table.column(c);
+    if (col.type().id() == cudf::type_id::STRING) {
+      std::vector<ColumnCompressionOptions> kids;
+      kids.emplace_back(
+        CompressionMethod::cascaded,
+        select_cascaded_on_sample(col.child(0).head<int32_t>(), col.size() + 1, 4));
+      kids.emplace_back(CompressionMethod::none);
+      opts.emplace_back(CompressionMethod::none, nvcompCascadedFormatOpts{}, kids);
+    } else {
+      opts.emplace_back(
+        CompressionMethod::cascaded,
+        select_cascaded_on_sample(col.head<char>(), col.size(),
+                                  cudf::is_rep_int64(col.type()) ? 8 : 4));
+    }
+  }
+  return opts;
+}
+
+ColumnCompressionOptions broadcast_compression_options(cudf::column_view input_column,
+                                                       ColumnCompressionOptions input_options,
+                                                       Communicator* comm)
+{
+  /* reference compression.cpp:95-130 used MPI_Bcast on MPI_COMM_WORLD; here
+   * rank 0's choices travel over the given Communicator (grouped send/recv
+   * of a POD through device staging — RCCL transports device buffers
+   * only) */
+  struct Pod {
+    int method, rles, deltas, bp;
+  };
+  Pod pod{(int)input_options.compression_method, input_options.cascaded_format.num_RLEs,
+          input_options.cascaded_format.num_deltas, input_options.cascaded_format.use_bp};
+  if (comm->mpi_size > 1) {
+    DBuf stage(sizeof(Pod));
+    if (comm->mpi_rank == 0)
+      DJ_HIP_CALL(hipMemcpy(stage.p, &pod, sizeof(Pod), hipMemcpyHostToDevice));
+    comm->start();
+    if (comm->mpi_rank == 0) {
+      for (int r = 1; r < comm->mpi_size; r++) comm->send(stage.p, sizeof(Pod), 1, r);
+    } else {
+      comm->recv(stage.p, sizeof(Pod), 1, 0);
+    }
+    comm->stop();
+    DJ_HIP_CALL(hipMemcpy(&pod, stage.p, sizeof(Pod), hipMemcpyDeviceToHost));
+  }
+  nvcompCascadedFormatOpts fmt{};
+  fmt.num_RLEs = pod.rles;
+  fmt.num_deltas = pod.deltas;
+  fmt.use_bp = pod.bp;
+  std::vector<ColumnCompressionOptions> kids;
+  if (input_column.type().id() == cudf::type_id::STRING) {
+    for (size_t k = 0; k < 2; k++) {
+      ColumnCompressionOptions child;
+      if (comm->mpi_rank == 0) {
+        DJ_CHECK_ERROR(input_options.children_compression_options.size() == 2,
+                       "STRING compression options need 2 children");
+        child = input_options.children_compression_options[k];
+      }
+      kids.push_back(broadcast_compression_options(input_column.child((cudf::size_type)k),
+                                                   child, comm));
+    }
+  }
+  return ColumnCompressionOptions((CompressionMethod)pod.method, fmt, kids);
+}
+
+std::vector<ColumnCompressionOptions> broadcast_compression_options(
+  cudf::table_view input_table, std::vector<ColumnCompressionOptions> input_options,
+  Communicator* comm)
+{
+  std::vector<ColumnCompressionOptions> out;
+  for (cudf::size_type c = 0; c < input_table.num_columns(); c++) {
+    ColumnCompressionOptions col_opts;
+    if (comm->mpi_rank == 0) col_opts = input_options[(size_t)c];
+    out.push_back(broadcast_compression_options(input_table.column(c), col_opts, comm));
+  }
+  return out;
+}
+
+ColumnCompressionOptions broadcast_compression_options(cudf::column_view input_column,
+                                                       ColumnCompressionOptions input_options)
+{
+  return broadcast_compression_options(input_column, input_options, default_communicator());
+}
+
+std::vector<ColumnCompressionOptions> broadcast_compression_options(
+  cudf::table_view input_table, std::vector<ColumnCompressionOptions> input_options)
+{
+  return broadcast_compression_options(input_table, input_options, default_communicator());
+}
+
 /* ------------------------------------------------------ CommunicationGroup */
 
 CommunicationGroup::CommunicationGroup(int grid_size, int stride)

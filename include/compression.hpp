@@ -1,14 +1,14 @@
 /*
- * compression.hpp — signature-level mirror of the reference's compression
- * options (reference: src/compression.hpp:42-58). Per the hot-path scope
- * (nvcomp layer replaced by plain RCCL over xGMI;
- * CompressionMethod::none kept in signatures for drop-in), only
- * CompressionMethod::none is executable in this build; requesting cascaded
- * or lz4 raises std::runtime_error at use. A HIP cascaded codec is the
- * ranked-next item.
+ * compression.hpp — mirror of the reference's compression-options surface
+ * (reference: src/compression.hpp:42-58, 253-361). Executable here:
+ * CompressionMethod::none and cascaded with {0|1 delta passes} + bitpack
+ * (our own wire format, dj_compress.hip — the nvcomp format is
+ * parity-unpinned); RLE passes and lz4 raise
+ * std::runtime_error at use with a clear message.
  */
 #pragma once
 
+#include "communicator.hpp"
 #include "dj_cudf_types.hpp"
 
 #include <stdexcept>
@@ -40,7 +40,38 @@ struct ColumnCompressionOptions {
 };
 
 /* mirrors generate_compression_options_distributed (compression.cpp:97-150):
- * per-column options for a table. With compression=false returns `none` for
- * every column; compression=true is not implemented in this build. */
+ * per-column options for a table. compression=false -> `none` for every
+ * column; compression=true -> bitpack cascaded (fixed policy; use
+ * generate_auto_select_compression_options for sampling-based choice). */
 std::vector<ColumnCompressionOptions> generate_compression_options_distributed(
   cudf::table_view input, bool compression);
+
+/* mirrors generate_none_compression_options (compression.hpp:313): `none`
+ * for every column (STRING columns get two `none` children) */
+std::vector<ColumnCompressionOptions> generate_none_compression_options(
+  cudf::table_view input_table);
+
+/* mirrors generate_auto_select_compression_options (compression.hpp:302,
+ * compression.cpp:36-69): per-column sampling selection over the executable
+ * cascaded schemes ({0|1 deltas} + bitpack); STRING columns select on the
+ * offsets child and never compress chars. Data-dependent, not parity-pinned
+ * (the reference sampled through nvcomp's CascadedSelector). */
+std::vector<ColumnCompressionOptions> generate_auto_select_compression_options(
+  cudf::table_view input_table);
+
+/* mirror broadcast_compression_options (compression.hpp:328,345): rank 0's
+ * choices distributed to every rank — over the registered Communicator
+ * (the reference used MPI_Bcast on MPI_COMM_WORLD). Non-root ranks may pass
+ * empty/default options. */
+ColumnCompressionOptions broadcast_compression_options(cudf::column_view input_column,
+                                                       ColumnCompressionOptions input_options);
+std::vector<ColumnCompressionOptions> broadcast_compression_options(
+  cudf::table_view input_table, std::vector<ColumnCompressionOptions> input_options);
+/* additive overloads taking the Communicator explicitly (for user-supplied
+ * transports; the 2-argument forms use the registered default) */
+ColumnCompressionOptions broadcast_compression_options(cudf::column_view input_column,
+                                                       ColumnCompressionOptions input_options,
+                                                       Communicator* comm);
+std::vector<ColumnCompressionOptions> broadcast_compression_options(
+  cudf::table_view input_table, std::vector<ColumnCompressionOptions> input_options,
+  Communicator* comm);

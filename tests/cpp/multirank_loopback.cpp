@@ -202,7 +202,21 @@ int main(int argc, char** argv)
        * wrapper invoking multiple configurations */
       const int nvl = (argc > 4) ? atoi(argv[4]) : 1;
       const bool compress = (argc > 5) && atoi(argv[5]) != 0;
-      auto opts = generate_compression_options_distributed(left->view(), compress);
+      /* sampling selector on rank 0 + broadcast over the loopback transport
+       * (reference compression.cpp:36-130 flow); every rank must end up
+       * with rank 0's choice */
+      auto sel = (r == 0) ? generate_auto_select_compression_options(left->view())
+                          : std::vector<ColumnCompressionOptions>{};
+      auto bopts = broadcast_compression_options(left->view(), sel, &comm);
+      if (bopts.size() != 2 ||
+          bopts[0].compression_method != CompressionMethod::cascaded ||
+          bopts[0].cascaded_format.use_bp != 1) {
+        printf("BCAST OPTS WRONG rank %d
+", r);
+        exit(1);
+      }
+      auto opts = compress ? bopts
+                           : generate_compression_options_distributed(left->view(), compress);
       auto res = distributed_inner_join(left->view(), right->view(), {0}, {0}, &comm, opts,
                                         opts, over_decom, false, nullptr, nvl);
       checksum(*res, &sums[r], &rows[r]);

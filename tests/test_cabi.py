@@ -76,6 +76,13 @@ std::unique_ptr<cudf::table> user_distribute(cudf::table_view global, Communicat
   return collect_tables(local->view(), comm);
 }
 
+std::vector<ColumnCompressionOptions> user_compression_options(cudf::table_view t, int rank)
+{
+  auto none = generate_none_compression_options(t);
+  auto chosen = rank == 0 ? generate_auto_select_compression_options(t) : none;
+  return broadcast_compression_options(t, chosen);
+}
+
 void user_all_to_all(cudf::table_view t, Communicator* comm,
                      std::vector<cudf::size_type> offsets)
 {

@@ -68,3 +68,4 @@ def test_join_with_compression_option(dj, comm):
     assert len(g[0]) == len(w[0])
     for a, b in zip(g, w):
         assert (a == b).all()
+
