@@ -211,8 +211,7 @@ int main(int argc, char** argv)
       if (bopts.size() != 2 ||
           bopts[0].compression_method != CompressionMethod::cascaded ||
           bopts[0].cascaded_format.use_bp != 1) {
-        printf("BCAST OPTS WRONG rank %d
-", r);
+        printf("BCAST OPTS WRONG rank %d\n", r);
         exit(1);
       }
       auto opts = compress ? bopts
