@@ -79,6 +79,9 @@ def main():
                          "README benchmark ran its default of 1 (IB shuffle + local "
                          "join, distributed_join.cpp:152-214); pass 1 to reproduce "
                          "that configuration.")
+    ap.add_argument("--compression", action="store_true",
+                    help="cascaded (delta+bitpack) compression on the wire — config 4's "
+                         "knob; rarely pays on intra-node xGMI")
     ap.add_argument("--cpu-baseline-rows", type=int, default=100_000_000)
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
@@ -135,7 +138,7 @@ def main():
     def step():
         t = L.dj_cpp_distributed_inner_join_i64_full(comm.ptr, bk.ptr, bp.ptr, rows,
                                                      pk.ptr, pp.ptr, rows, args.over_decom,
-                                                     0, 0, nvl)
+                                                     0, int(args.compression), nvl)
         state["matches"] = L.dj_table_num_rows(t)
         L.dj_table_free(t)
 
@@ -258,6 +261,7 @@ def main():
                 "rows_per_gpu": rows,
                 "selectivity": SELECTIVITY,
                 "nvlink_domain_size": nvl,
+                "compression": bool(args.compression),
                 "output_rows_per_gpu": int(matches),
                 "output_rows_per_sec": out_rows_s,
                 "engine": "C++ distributed_inner_join (drop-in path); at N=1 the "
