@@ -332,7 +332,13 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
   }
   {
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
-    dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, 2048, d_out0, d_out1, d_out2,
+    /* B caps at 262144 (PA<=1024 x 256); beyond ~400M rows/table the
+     * buckets exceed the 2048-slot cap — switch to the 4096-slot table
+     * (cap 3072, covers the 800M single-GPU shape) instead of overflowing
+     * every bucket into the per-bucket fallback */
+    const int64_t maxn = ln > rn ? ln : rn;
+    const int slots = (maxn / B > 1300) ? 4096 : 2048;
+    dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, slots, d_out0, d_out1, d_out2,
                  d_out3, cap, d_counter, s.flags, d_any_overflow, d_error, st);
   }
 }
