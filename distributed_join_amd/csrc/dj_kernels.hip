@@ -532,6 +532,20 @@ __device__ __forceinline__ uint32_t subB_of(int64_t key)
 {
   return (uint32_t)(dj_mix64((uint64_t)key) >> 32) & (uint32_t)(SUB_BUCKETS - 1);
 }
+/* runtime-F sub-bucket (fused wire path): low 8 bits from [32,40) as subB_of
+ * (identical placement for F <= 256); F > 256 borrows its extra bits from
+ * [50,60) — DISJOINT from the pass-A group bits [40,50) (overlapping fields
+ * would collapse the effective fan-out: with group bits fixed inside a
+ * pass-A group, any shared bit halves the distinct sub-buckets) and from
+ * the LDS slot hash bits [0,11). */
+__device__ __forceinline__ uint32_t subF_of(int64_t key, int F)
+{
+  uint64_t m = dj_mix64((uint64_t)key);
+  if (F <= 256) return (uint32_t)(m >> 32) & (uint32_t)(F - 1);
+  uint32_t lo = (uint32_t)(m >> 32) & 255u;
+  uint32_t hi = (uint32_t)(m >> 50) & (uint32_t)((F >> 8) - 1);
+  return lo | (hi << 8);
+}
 
 /* ---- pass A kernels (P groups, P <= 1024) ---- */
 
@@ -621,14 +635,14 @@ __device__ __forceinline__ longlong2 load_row(const int64_t* keys, const int64_t
 }
 
 /* staged scatter over [start, end); cur[P] holds GLOBAL destination
- * cursors; GROUP_FN: 0 = groupA_of(P), 1 = subB_of */
+ * cursors; GROUP_FN: 0 = groupA_of(P), 1 = subF_of(P) */
 template <int GROUP_FN, bool SINGLE_LEVEL>
 __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
                                     const longlong2* in_pairs, int64_t start, int64_t end,
                                     int P, longlong2* tbuf, uint32_t* hist, uint32_t* base,
                                     uint32_t* gcur, longlong2* out_pairs)
-/* GROUP_FN 0: groupA_of(P); GROUP_FN 1: sub-bucket = (mix64>>32) & (P-1)
- * (P is the runtime fanout) */
+/* GROUP_FN 0: groupA_of(P); GROUP_FN 1: subF_of(P) (P is the runtime
+ * fanout; bit fields disjoint from groupA's, see subF_of) */
 {
   constexpr int VPT = SCATTER_TILE / BUCKET_THREADS;  // 4
   const int tid = threadIdx.x;
@@ -646,9 +660,7 @@ __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
       int64_t i = t0 + (int64_t)v * blockDim.x + tid;
       if (i < end) {
         r[v] = load_row<SINGLE_LEVEL>(keys, pay, in_pairs, i);
-        g[v] = GROUP_FN == 0 ? groupA_of(r[v].x, P)
-                             : ((uint32_t)(dj_mix64((uint64_t)r[v].x) >> 32) &
-                                (uint32_t)(P - 1));
+        g[v] = GROUP_FN == 0 ? groupA_of(r[v].x, P) : subF_of(r[v].x, P);
         rank[v] = atomicAdd(&hist[g[v]], 1u);
       }
     }
@@ -673,9 +685,7 @@ __device__ void staged_scatter_span(const int64_t* keys, const int64_t* pay,
     /* flush linearly: per-group runs coalesce into full lines */
     for (int pos = tid; pos < count; pos += blockDim.x) {
       longlong2 row = tbuf[pos];
-      uint32_t gg = GROUP_FN == 0 ? groupA_of(row.x, P)
-                                  : ((uint32_t)(dj_mix64((uint64_t)row.x) >> 32) &
-                                     (uint32_t)(P - 1));
+      uint32_t gg = GROUP_FN == 0 ? groupA_of(row.x, P) : subF_of(row.x, P);
       out_pairs[gcur[gg] + (pos - base[gg])] = row;
     }
     __syncthreads();
@@ -775,9 +785,7 @@ __global__ __launch_bounds__(BUCKET_THREADS) void subpart_lists_kernel(
     const int64_t s0 = seg_bounds[(size_t)sgi * (PA + 1) + g];
     const int64_t s1 = seg_bounds[(size_t)sgi * (PA + 1) + g + 1];
     for (int64_t i = s0 + tid; i < s1; i += blockDim.x)
-      atomicAdd(&seghist[(uint32_t)(dj_mix64((uint64_t)nt_load(&keys[i])) >> 32) &
-                         (uint32_t)(F - 1)],
-                1u);
+      atomicAdd(&seghist[subF_of(nt_load(&keys[i]), F)], 1u);
   }
   __syncthreads();
   if (tid == 0) {

@@ -170,8 +170,12 @@ def test_fused_big_buckets():
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     exe = os.path.join(repo, "tests", "cpp", "multirank_loopback")
     assert os.path.exists(exe), "built by test_multirank_loopback"
-    env = dict(os.environ, DJ_FORCE_FUSED_F="64")
-    r = subprocess.run([exe, "2", "1", "80000000", "2", "0"], capture_output=True,
-                       text=True, timeout=420, env=env)
-    assert r.returncode == 0, r.stdout + r.stderr
-    assert "MULTIRANK OK" in r.stdout
+    for force_f in ("64", "1024"):
+        # 64: ~2400 rows/bucket (4096-slot join); 1024: full fan-out, whose
+        # sub-bucket bits must stay disjoint from the pass-A group bits
+        # (subF_of) or the effective fan-out collapses 4x
+        env = dict(os.environ, DJ_FORCE_FUSED_F=force_f)
+        r = subprocess.run([exe, "2", "1", "80000000", "2", "0"], capture_output=True,
+                           text=True, timeout=420, env=env)
+        assert r.returncode == 0, force_f + "\n" + r.stdout + r.stderr
+        assert "MULTIRANK OK" in r.stdout
