@@ -258,7 +258,7 @@ struct BucketScratch {
 
 BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
 {
-  const int PA = B / dj::kSubBuckets;
+  const int PA = dj::bucket_groups_for(B);
   const int64_t maxn = ln > rn ? ln : rn;
   char* p = (char*)base;
   auto take = [&](size_t bytes) {
@@ -285,7 +285,7 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
 int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
 {
   int B = dj::bucket_count_for(ln, rn);
-  const int PA = B / dj::kSubBuckets;
+  const int PA = dj::bucket_groups_for(B);
   const int64_t maxn = ln > rn ? ln : rn;
   size_t bytes = 0;
   auto add = [&](size_t b) { bytes += (b + 255) & ~(size_t)255; };

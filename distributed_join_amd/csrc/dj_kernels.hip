@@ -514,7 +514,7 @@ int bucket_count_for(int64_t ln, int64_t rn)
 {
   int64_t maxn = ln > rn ? ln : rn;
   int64_t B = 256;
-  while (B < 262144 && maxn / B > 800) B <<= 1;
+  while (B < 1048576 && maxn / B > 800) B <<= 1;
   return (int)B;
 }
 
@@ -722,44 +722,46 @@ __global__ void set_segoff1_kernel(int64_t* segoff, int64_t n)
   segoff[1] = n;
 }
 
-/* ---- pass B: one block per pass-A group; 256 sub-buckets in-block ---- */
-/* pass B: one block per pass-A group; 256 sub-buckets, tile-staged scatter.
- * single_level: input is the original two column arrays (B == 256 case). */
+/* ---- pass B: one block per pass-A group; F sub-buckets in-block ---- */
+/* pass B: one block per pass-A group; F sub-buckets (runtime 256/512/1024,
+ * subF_of bit fields disjoint from the pass-A group bits), tile-staged
+ * scatter. single_level: input is the original two column arrays (the
+ * B == 256 single-pass case). */
 template <bool SINGLE_LEVEL>
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_kernel(
   const longlong2* __restrict__ in_pairs, const int64_t* __restrict__ keys,
-  const int64_t* __restrict__ pay, const int64_t* __restrict__ segoff, int B,
+  const int64_t* __restrict__ pay, const int64_t* __restrict__ segoff, int B, int F,
   longlong2* __restrict__ out_pairs, int64_t* __restrict__ bucket_offsets /* B+1 */)
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbuf = (longlong2*)smem;
   uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
-  uint32_t* base = hist + SUB_BUCKETS;
-  uint32_t* gcur = base + SUB_BUCKETS;
-  uint32_t* seghist = gcur + SUB_BUCKETS;
+  uint32_t* base = hist + F;
+  uint32_t* gcur = base + F;
+  uint32_t* seghist = gcur + F;
   const int tid = threadIdx.x;
   const int a = blockIdx.x;
   const int64_t s0 = segoff[a], s1 = segoff[a + 1];
-  if (tid < SUB_BUCKETS) seghist[tid] = 0;
+  if (tid < F) seghist[tid] = 0;
   __syncthreads();
   for (int64_t i = s0 + tid; i < s1; i += blockDim.x) {
     int64_t k = SINGLE_LEVEL ? nt_load(&keys[i]) : nt_load(&in_pairs[i].x);
-    atomicAdd(&seghist[subB_of(k)], 1u);
+    atomicAdd(&seghist[subF_of(k, F)], 1u);
   }
   __syncthreads();
   if (tid == 0) {
     uint32_t acc = 0;
-    for (int j = 0; j < SUB_BUCKETS; j++) {
+    for (int j = 0; j < F; j++) {
       uint32_t c = seghist[j];
       gcur[j] = (uint32_t)s0 + acc;
-      bucket_offsets[(size_t)a * SUB_BUCKETS + j] = s0 + acc;
+      bucket_offsets[(size_t)a * F + j] = s0 + acc;
       acc += c;
     }
     if (a == gridDim.x - 1) bucket_offsets[B] = s1;
   }
   __syncthreads();
-  staged_scatter_span<1, SINGLE_LEVEL>(keys, pay, in_pairs, s0, s1, SUB_BUCKETS, tbuf, hist,
-                                       base, gcur, out_pairs);
+  staged_scatter_span<1, SINGLE_LEVEL>(keys, pay, in_pairs, s0, s1, F, tbuf, hist, base, gcur,
+                                       out_pairs);
 }
 
 /* pass B over per-peer segment lists (fused wire path): block = one pass-A
@@ -1164,38 +1166,38 @@ __global__ void clamp_seglen_kernel(uint32_t* seg_len, int P, uint32_t capA)
  * lds_join consumes. */
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack_kernel(
   const longlong2* __restrict__ in_pairs, const uint32_t* __restrict__ seg_len, int64_t capA,
-  const int64_t* __restrict__ segout /* PA+1 compact output bases */, int B,
+  const int64_t* __restrict__ segout /* PA+1 compact output bases */, int B, int F,
   longlong2* __restrict__ out_pairs, int64_t* __restrict__ bucket_offsets /* B+1 */)
 {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbuf = (longlong2*)smem;
   uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
-  uint32_t* base = hist + SUB_BUCKETS;
-  uint32_t* gcur = base + SUB_BUCKETS;
-  uint32_t* seghist = gcur + SUB_BUCKETS;
+  uint32_t* base = hist + F;
+  uint32_t* gcur = base + F;
+  uint32_t* seghist = gcur + F;
   const int tid = threadIdx.x;
   const int a = blockIdx.x;
   const int64_t s0 = (int64_t)a * capA;
   const int64_t s1 = s0 + seg_len[a];
   const int64_t ob = segout[a];
-  if (tid < SUB_BUCKETS) seghist[tid] = 0;
+  if (tid < F) seghist[tid] = 0;
   __syncthreads();
   for (int64_t i = s0 + tid; i < s1; i += blockDim.x)
-    atomicAdd(&seghist[subB_of(nt_load(&in_pairs[i].x))], 1u);
+    atomicAdd(&seghist[subF_of(nt_load(&in_pairs[i].x), F)], 1u);
   __syncthreads();
   if (tid == 0) {
     uint32_t acc = 0;
-    for (int j = 0; j < SUB_BUCKETS; j++) {
+    for (int j = 0; j < F; j++) {
       uint32_t c = seghist[j];
       gcur[j] = (uint32_t)(ob + acc);
-      bucket_offsets[(size_t)a * SUB_BUCKETS + j] = ob + acc;
+      bucket_offsets[(size_t)a * F + j] = ob + acc;
       acc += c;
     }
     if (a == gridDim.x - 1) bucket_offsets[B] = segout[gridDim.x];
   }
   __syncthreads();
-  staged_scatter_span<1, false>(nullptr, nullptr, in_pairs, s0, s1, SUB_BUCKETS, tbuf, hist,
-                                base, gcur, out_pairs);
+  staged_scatter_span<1, false>(nullptr, nullptr, in_pairs, s0, s1, F, tbuf, hist, base, gcur,
+                                out_pairs);
 }
 
 void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
@@ -1204,22 +1206,24 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        int* d_any_overflow, hipStream_t s)
 {
   DJ_CHECK_ERROR(n < (int64_t)UINT32_MAX, "bucket_partition: n must be < 2^32");
-  const int PA = B / SUB_BUCKETS;
-  DJ_CHECK_ERROR(PA >= 1 && PA <= 1024, "bucket_partition: B out of range");
+  const int PA = bucket_groups_for(B);
+  const int F = B / PA;
+  DJ_CHECK_ERROR(PA >= 1 && PA <= 1024 && F >= 1 && F <= 1024,
+                 "bucket_partition: B out of range");
+  const size_t subpart_lds =
+    SCATTER_TILE * sizeof(longlong2) + 4 * (size_t)F * sizeof(uint32_t);
   if (PA == 1) {
-    size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
     hipLaunchKernelGGL(set_segoff1_kernel, dim3(1), dim3(1), 0, s, d_segoff, n);
     DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_subpart_kernel<true>, dim3(PA), dim3(BUCKET_THREADS),
                        subpart_lds, s, (const longlong2*)nullptr, d_keys, d_pay, d_segoff, B,
-                       d_out_pairs, d_offsets);
+                       F, d_out_pairs, d_offsets);
     DJ_HIP_CALL(hipGetLastError());
   } else if (d_any_overflow != nullptr &&
              (int64_t)PA * slack_capA(n, PA) + n < (int64_t)UINT32_MAX) {
     /* slack path: no count pass (see bucket_scatter_slack_kernel header) */
     const int64_t capA = slack_capA(n, PA);
     size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 4 * (size_t)PA * sizeof(uint32_t);
-    size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
     DJ_HIP_CALL(hipMemsetAsync(d_totals, 0, (size_t)PA * 4, s));
     hipLaunchKernelGGL(bucket_scatter_slack_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
                        scatter_lds, s, d_keys, d_pay, n, PA, capA, d_totals, d_any_overflow,
@@ -1232,12 +1236,11 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        d_segoff);
     DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_subpart_slack_kernel, dim3(PA), dim3(BUCKET_THREADS), subpart_lds,
-                       s, d_tmp_pairs, d_totals, capA, d_segoff, B, d_out_pairs, d_offsets);
+                       s, d_tmp_pairs, d_totals, capA, d_segoff, B, F, d_out_pairs, d_offsets);
     DJ_HIP_CALL(hipGetLastError());
   } else {
     size_t hist_lds = (size_t)PA * sizeof(uint32_t);
     size_t scatter_lds = SCATTER_TILE * sizeof(longlong2) + 3 * hist_lds;
-    size_t subpart_lds = SCATTER_TILE * sizeof(longlong2) + 4 * SUB_BUCKETS * sizeof(uint32_t);
     hipLaunchKernelGGL(bucket_count_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS), hist_lds,
                        s, d_keys, n, PA, d_counts);
     DJ_HIP_CALL(hipGetLastError());
@@ -1252,7 +1255,7 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
     DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(bucket_subpart_kernel<false>, dim3(PA), dim3(BUCKET_THREADS),
                        subpart_lds, s, d_tmp_pairs, (const int64_t*)nullptr,
-                       (const int64_t*)nullptr, d_segoff, B, d_out_pairs, d_offsets);
+                       (const int64_t*)nullptr, d_segoff, B, F, d_out_pairs, d_offsets);
     DJ_HIP_CALL(hipGetLastError());
   }
 }

@@ -72,6 +72,16 @@ constexpr int kBucketBlocks = 512;
 constexpr int kSubBuckets = 256;
 constexpr int kJoinBucketRowCap = 1536;  // 75% of the 2048-slot LDS table
 int bucket_count_for(int64_t ln, int64_t rn);
+/* decomposition B = PA x F: PA pass-A groups (<=1024, one per thread in the
+ * scans), F pass-B sub-buckets (256 up to B=262144 — the historical shape —
+ * then 512/1024, bit fields disjoint per subF_of) */
+inline int bucket_groups_for(int B)
+{
+  int PA = B / 256;
+  if (PA < 1) PA = 1;
+  if (PA > 1024) PA = 1024;
+  return PA;
+}
 /* pass-A slack segment capacity (rows per group): mean + ~6% + 1024 covers
  * hash-uniform inputs w.h.p.; skewed inputs overflow -> bit 2 of
  * any_overflow -> the caller redoes the join exactly */
@@ -80,7 +90,8 @@ inline int64_t slack_capA(int64_t n, int PA)
   const int64_t m = n / PA;
   return m + m / 16 + 1024;
 }
-/* Two-level non-stable partition into B buckets (B = PA*256, PA<=1024) of
+/* Two-level non-stable partition into B buckets (B = PA*F per
+ * bucket_groups_for; up to 1024x1024 = 1M buckets for ~800M-row tables) of
  * interleaved 16 B {key,payload} pairs. d_tmp_pairs: pass-A staging —
  * longlong2[PA * slack_capA(n, PA)] when d_any_overflow is non-null (the
  * slack path: no count pass; group skew beyond the slack sets BIT 2 of

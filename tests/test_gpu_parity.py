@@ -199,3 +199,14 @@ def test_slack_partition_overflow_redo(dj):
     )
     rp = np.arange(rk.size, dtype=np.int64)
     _join_parity(dj, lk, lp, rk, rp)
+
+
+def test_large_table_f512_parity(dj):
+    # >209M build rows push B past 262144 into the PA=1024 x F=512
+    # decomposition (runtime-F pass B, subF_of extended bit field); parity
+    # against the oracle with a small probe side keeps the test fast
+    n = 250_000_000
+    lk, lp = oracle.gen_build(n)
+    m = 1_000_000
+    rk, rp = oracle.gen_probe(m, n)
+    _join_parity(dj, lk, lp, rk, rp)
