@@ -518,12 +518,8 @@ int bucket_count_for(int64_t ln, int64_t rn)
   return (int)B;
 }
 
-/* final bucket id = bits [32, 32+log2(B)) of dj_mix64(key) */
-__device__ __forceinline__ uint32_t bucket_of(int64_t key, int B)
-{
-  return (uint32_t)(dj_mix64((uint64_t)key) >> 32) & (uint32_t)(B - 1);
-}
-/* pass-A group = bucket id >> 8 ; pass-B sub-bucket = bucket id & 255 */
+/* bucket id = groupA * F + subF: pass-A group from mix64 bits [40,50),
+ * pass-B sub-bucket from bits [32,40) (+ [50,60) when F > 256, subF_of) */
 __device__ __forceinline__ uint32_t groupA_of(int64_t key, int PA)
 {
   return (uint32_t)(dj_mix64((uint64_t)key) >> 40) & (uint32_t)(PA - 1);
