@@ -125,6 +125,10 @@ class LoopbackCommunicator : public Communicator {
 
 /* ---------------- helpers ---------------- */
 
+extern "C" void dj_gen_test_strings(const int64_t* d_keys, int64_t n, void** out_offsets,
+                                    void** out_chars, int64_t* out_chars_bytes);
+extern "C" void dj_dfree(void* p);
+
 static std::unique_ptr<cudf::table> make_rank_tables(int64_t n_global, int64_t rows,
                                                      int64_t row0, bool build)
 {
@@ -160,6 +164,84 @@ static void checksum(const cudf::table& t, uint64_t* out_sum, int64_t* out_rows)
   for (int64_t i = 0; i < n; i++)
     s += dj_mix64((uint64_t)c0[i] ^ dj_mix64((uint64_t)c1[i] ^
                                              dj_mix64((uint64_t)c2[i] ^ (uint64_t)c3[i])));
+  *out_sum = s;
+  *out_rows = n;
+}
+
+/* (key int64, payload STRING) table for a rank slice — the config-4 shape
+ * (string_payload.cu:50-74 semantics via the library's test-string
+ * generator: len = k%%7+1, char = 'a'+k%%26) */
+struct StrTable {
+  std::unique_ptr<cudf::column> kcol;
+  void* offsets{nullptr};
+  void* chars{nullptr};
+  int64_t chars_bytes{0};
+  cudf::table_view view() const
+  {
+    return cudf::table_view(
+      {kcol->view(), cudf::column_view(cudf::data_type(cudf::type_id::STRING), kcol->size(),
+                                       offsets, chars, chars_bytes)});
+  }
+  StrTable() = default;
+  StrTable(StrTable&& o) noexcept
+    : kcol(std::move(o.kcol)), offsets(o.offsets), chars(o.chars), chars_bytes(o.chars_bytes)
+  {
+    o.offsets = o.chars = nullptr;
+  }
+  StrTable(const StrTable&) = delete;
+  ~StrTable()
+  {
+    if (offsets) dj_dfree(offsets);
+    if (chars) dj_dfree(chars);
+  }
+};
+
+static StrTable make_rank_strtable(int64_t n_global, int64_t rows, int64_t row0, bool build)
+{
+  std::vector<int64_t> keys(rows);
+  for (int64_t t = 0; t < rows; t++) {
+    int64_t i = row0 + t;
+    keys[t] = build ? dj_build_key((uint64_t)i, (uint64_t)n_global, 2 * n_global, 1234)
+                    : dj_probe_key((uint64_t)i, (uint64_t)n_global, 2 * n_global, 0.3, 1234);
+  }
+  StrTable st;
+  st.kcol = std::make_unique<cudf::column>(cudf::data_type(cudf::type_id::INT64),
+                                           (cudf::size_type)rows);
+  CHECK(hipMemcpy(st.kcol->head(), keys.data(), rows * 8, hipMemcpyHostToDevice));
+  dj_gen_test_strings((const int64_t*)st.kcol->head(), rows, &st.offsets, &st.chars,
+                      &st.chars_bytes);
+  return st;
+}
+
+/* order-insensitive checksum over (int64, STRING, int64, STRING) rows */
+static void checksum_str(const cudf::table& t, uint64_t* out_sum, int64_t* out_rows)
+{
+  int64_t n = t.num_rows();
+  std::vector<int64_t> c0(n), c2(n);
+  CHECK(hipMemcpy(c0.data(), t.get_column(0).head(), n * 8, hipMemcpyDeviceToHost));
+  CHECK(hipMemcpy(c2.data(), t.get_column(2).head(), n * 8, hipMemcpyDeviceToHost));
+  auto str_col = [&](int c, std::vector<int32_t>& off, std::vector<uint8_t>& ch) {
+    const cudf::column& col = t.get_column(c);
+    off.resize((size_t)n + 1);
+    CHECK(hipMemcpy(off.data(), col.head(), (n + 1) * 4, hipMemcpyDeviceToHost));
+    ch.resize((size_t)col.chars_size());
+    if (col.chars_size())
+      CHECK(hipMemcpy(ch.data(), col.chars(), col.chars_size(), hipMemcpyDeviceToHost));
+  };
+  std::vector<int32_t> off1, off3;
+  std::vector<uint8_t> ch1, ch3;
+  str_col(1, off1, ch1);
+  str_col(3, off3, ch3);
+  auto fold = [](const std::vector<int32_t>& off, const std::vector<uint8_t>& ch, int64_t i) {
+    uint64_t h = 1469598103934665603ull;
+    for (int32_t p = off[(size_t)i]; p < off[(size_t)i + 1]; p++)
+      h = (h ^ ch[(size_t)p]) * 1099511628211ull;
+    return h;
+  };
+  uint64_t s = 0;
+  for (int64_t i = 0; i < n; i++)
+    s += dj_mix64((uint64_t)c0[i] ^ dj_mix64(fold(off1, ch1, i)) ^
+                  dj_mix64((uint64_t)c2[i] ^ dj_mix64(fold(off3, ch3, i) + 1)));
   *out_sum = s;
   *out_rows = n;
 }
@@ -290,6 +372,57 @@ int main(int argc, char** argv)
     }
     printf("shuffle identity placement + distribute/collect roundtrip OK\n");
   }
+  /* strings payload multirank (config 4 shape): sizes on the wire,
+   * receiver-side offset rebuild (strings_column.cu:39-145 semantics),
+   * compressed sizes wire when argv[5] asks for compression */
+  {
+    const int64_t ns = 200000;
+    const int nvl = (argc > 4) ? atoi(argv[4]) : 1;
+    const bool compress = (argc > 5) && atoi(argv[5]) != 0;
+    uint64_t want_s;
+    int64_t want_r;
+    {
+      Mailbox mb1;
+      LoopbackCommunicator c1(0, 1, &mb1);
+      auto l = make_rank_strtable(ns, ns, 0, true);
+      auto r = make_rank_strtable(ns, ns, 0, false);
+      auto o = generate_compression_options_distributed(l.view(), compress);
+      auto res = distributed_inner_join(l.view(), r.view(), {0}, {0}, &c1, o, o, 1, false,
+                                        nullptr, 1);
+      checksum_str(*res, &want_s, &want_r);
+    }
+    Mailbox mb2;
+    std::vector<uint64_t> ssums(G);
+    std::vector<int64_t> srows(G);
+    std::vector<std::thread> th;
+    for (int r = 0; r < G; r++) {
+      th.emplace_back([&, r] {
+        CHECK(hipSetDevice(0));
+        int64_t per = ns / G;
+        auto l = make_rank_strtable(ns, per, r * per, true);
+        auto rt = make_rank_strtable(ns, per, r * per, false);
+        LoopbackCommunicator comm(r, G, &mb2);
+        auto o = generate_compression_options_distributed(l.view(), compress);
+        auto res = distributed_inner_join(l.view(), rt.view(), {0}, {0}, &comm, o, o,
+                                          over_decom, false, nullptr, nvl);
+        checksum_str(*res, &ssums[r], &srows[r]);
+      });
+    }
+    for (auto& t : th) t.join();
+    uint64_t gs = 0;
+    int64_t gr = 0;
+    for (int r = 0; r < G; r++) {
+      gs += ssums[r];
+      gr += srows[r];
+    }
+    if (gs != want_s || gr != want_r) {
+      printf("STRINGS MULTIRANK MISMATCH (%lld vs %lld rows)\n", (long long)gr,
+             (long long)want_r);
+      return 1;
+    }
+    printf("strings multirank OK (%lld rows)\n", (long long)gr);
+  }
+
   printf("single-rank: rows=%lld sum=%llx\n", (long long)want_rows,
          (unsigned long long)want_sum);
   printf("%d-rank(od=%d): rows=%lld sum=%llx\n", G, over_decom, (long long)got_rows,
