@@ -210,3 +210,14 @@ def test_large_table_f512_parity(dj):
     m = 1_000_000
     rk, rp = oracle.gen_probe(m, n)
     _join_parity(dj, lk, lp, rk, rp)
+
+
+def test_benchmark_size_parity(dj):
+    """Bit-exact (order-insensitive) parity against the CPU oracle AT the
+    headline config-2 size: 100M x 100M int64/int64, selectivity 0.3, unique
+    build keys — the exact workload bench.py times. The slowest parity test
+    (~2 min: oracle join + 30M-row lexsort) but the strongest claim."""
+    n = 100_000_000
+    lk, lp = oracle.gen_build(n)
+    rk, rp = oracle.gen_probe(n, n, selectivity=0.3)
+    _join_parity(dj, lk, lp, rk, rp)
