@@ -316,6 +316,98 @@ __global__ __launch_bounds__(THREADS) void join_kernel_multi(
   }
 }
 
+/* cumulative phase ablation of the CURRENT single-pass staged join:
+ * 1=init, 2=+build, 3=+probe walk (matches counted, discarded),
+ * 4=+stage append, 5=full (reserve + flush). Successive differences give
+ * the marginal cost of each phase. */
+template <int LEVEL>
+__global__ __launch_bounds__(THREADS) void join_ablate(
+  const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
+  const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
+  int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
+  int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter, int* __restrict__ sink)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbl = (longlong2*)smem;
+  int64_t* stage = (int64_t*)(tbl + SLOTS);
+  long long* base_sh = (long long*)(stage + 4 * STAGE);
+  uint32_t* cur_sh = (uint32_t*)(base_sh + 1);
+  const uint32_t smask = SLOTS - 1;
+  constexpr int S = STAGE;
+  uint32_t acc = 0;
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const int64_t l0 = loff[b], l1 = loff[b + 1];
+    const int64_t r0 = roff[b], r1 = roff[b + 1];
+    const int64_t lnb = l1 - l0;
+    if (lnb == 0 || r1 == r0 || lnb > ROWCAP) continue;
+    for (int s = threadIdx.x; s < SLOTS; s += blockDim.x) tbl[s].x = EMPTY;
+    if (threadIdx.x == 0) *cur_sh = 0;
+    __syncthreads();
+    if (LEVEL >= 2) {
+      for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
+        longlong2 row = lrows[i];
+        uint32_t slot = (uint32_t)dj_mix64((uint64_t)row.x) & smask;
+        for (;;) {
+          unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
+                                             (unsigned long long)EMPTY,
+                                             (unsigned long long)row.x);
+          if (old == (unsigned long long)EMPTY) break;
+          slot = (slot + 1) & smask;
+        }
+        tbl[slot].y = row.y;
+      }
+    }
+    __syncthreads();
+    if (LEVEL >= 3) {
+      for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
+        longlong2 prow = rrows[j];
+        uint32_t slot = (uint32_t)dj_mix64((uint64_t)prow.x) & smask;
+        for (;;) {
+          longlong2 e = tbl[slot];
+          if (e.x == EMPTY) break;
+          if (e.x == prow.x) {
+            if (LEVEL >= 4) {
+              uint32_t pos = atomicAdd(cur_sh, 1u);
+              if (pos < (uint32_t)S) {
+                stage[0 * S + pos] = prow.x;
+                stage[1 * S + pos] = e.y;
+                stage[2 * S + pos] = prow.x;
+                stage[3 * S + pos] = prow.y;
+              }
+            } else {
+              acc += (uint32_t)e.y;
+            }
+          }
+          slot = (slot + 1) & smask;
+        }
+      }
+    }
+    __syncthreads();
+    if (LEVEL >= 5) {
+      const uint32_t total = min(*cur_sh, (uint32_t)S);
+      if (threadIdx.x == 0 && total)
+        *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
+      __syncthreads();
+      if (total) {
+        const long long base = *base_sh;
+        for (uint32_t i = threadIdx.x; i < total; i += blockDim.x) {
+          long long idx = base + (long long)i;
+          if (idx < cap) {
+            out0[idx] = stage[0 * S + i];
+            out1[idx] = stage[1 * S + i];
+            out2[idx] = stage[2 * S + i];
+            out3[idx] = stage[3 * S + i];
+          }
+        }
+      }
+    } else if (LEVEL >= 4 && threadIdx.x == 0) {
+      acc += *cur_sh;
+    }
+    __syncthreads();
+  }
+  if (acc == 0xFFFFFFFFu) *sink = 1;  // keep acc live
+}
+
 /* fill bucket b with sizes[b] rows: build keys b*4096+i, probe keys
  * b*4096+i for i<30% else non-matching */
 __global__ void fill_kernel(longlong2* rows, const int64_t* starts,
@@ -455,6 +547,51 @@ int main(int argc, char** argv)
   int nrows = argc > 2 ? atoi(argv[2]) : 762;
   int match = (int)(nrows * 0.3);
   int reps = 4;
+  if (argc > 3 && atoi(argv[3]) == 1) {
+    /* cumulative phase ablation on bench-shaped buckets */
+    int64_t stride = nrows, total = (int64_t)B * stride;
+    longlong2 *lrows, *rrows; int64_t *loff, *starts; uint32_t* szs;
+    int64_t *o0, *o1, *o2, *o3; unsigned long long* counter; int* sink;
+    int64_t cap = (int64_t)B * match + 1024;
+    CHECK(hipMalloc(&lrows, total * 16)); CHECK(hipMalloc(&rrows, total * 16));
+    CHECK(hipMalloc(&loff, (B + 1) * 8)); CHECK(hipMalloc(&starts, B * 8));
+    CHECK(hipMalloc(&szs, B * 4));
+    CHECK(hipMalloc(&o0, cap * 8)); CHECK(hipMalloc(&o1, cap * 8));
+    CHECK(hipMalloc(&o2, cap * 8)); CHECK(hipMalloc(&o3, cap * 8));
+    CHECK(hipMalloc(&counter, 8)); CHECK(hipMalloc(&sink, 4));
+    std::vector<int64_t> h_off(B + 1), h_st(B); std::vector<uint32_t> h_sz(B, (uint32_t)nrows);
+    for (int b = 0; b <= B; b++) h_off[b] = (int64_t)b * stride;
+    for (int b = 0; b < B; b++) h_st[b] = (int64_t)b * stride;
+    CHECK(hipMemcpy(loff, h_off.data(), (B + 1) * 8, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(starts, h_st.data(), B * 8, hipMemcpyHostToDevice));
+    CHECK(hipMemcpy(szs, h_sz.data(), B * 4, hipMemcpyHostToDevice));
+    fill_kernel<<<B, 256>>>(lrows, starts, szs, B, 0);
+    fill_kernel<<<B, 256>>>(rrows, starts, szs, B, 1);
+    CHECK(hipDeviceSynchronize());
+    size_t lds = SLOTS * 16 + 4 * STAGE * 8 + 16;
+    int grid = B < 8192 ? B : 8192;
+    hipEvent_t e0, e1; CHECK(hipEventCreate(&e0)); CHECK(hipEventCreate(&e1));
+    auto run = [&](auto kern, const char* name) {
+      double ms_sum = 0;
+      for (int rep = 0; rep < reps; rep++) {
+        CHECK(hipMemset(counter, 0, 8));
+        CHECK(hipEventRecord(e0));
+        kern<<<grid, THREADS, lds>>>(lrows, loff, rrows, loff, B, o0, o1, o2, o3, cap,
+                                     counter, sink);
+        CHECK(hipEventRecord(e1)); CHECK(hipEventSynchronize(e1));
+        float ms; CHECK(hipEventElapsedTime(&ms, e0, e1));
+        if (rep) ms_sum += ms;
+      }
+      CHECK(hipGetLastError());
+      printf("%-28s %7.3f ms\n", name, ms_sum / (reps - 1));
+    };
+    run(join_ablate<1>, "init only");
+    run(join_ablate<2>, "+build");
+    run(join_ablate<3>, "+probe walk (no stage)");
+    run(join_ablate<4>, "+stage append (no flush)");
+    run(join_ablate<5>, "full (reserve+flush)");
+    return 0;
+  }
   run_case("base B=131072 r=762", B, nrows, match, 0, reps, false, true, false);
   run_case("big B/2 r=1526 1blk/CU", B / 2, nrows * 2, (int)(nrows * 2 * 0.3), -5, reps,
            false, true, false);
