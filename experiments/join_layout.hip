@@ -408,6 +408,88 @@ __global__ __launch_bounds__(THREADS) void join_ablate(
   if (acc == 0xFFFFFFFFu) *sink = 1;  // keep acc live
 }
 
+/* full join with a FLAT flush: all threads cover total*4 column writes
+ * (the per-column loop leaves 3/4 of lanes idle at ~229 staged rows) */
+__global__ __launch_bounds__(THREADS) void join_kernel_flat(
+  const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
+  const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
+  int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
+  int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter)
+{
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbl = (longlong2*)smem;
+  int64_t* stage = (int64_t*)(tbl + SLOTS);
+  long long* base_sh = (long long*)(stage + 4 * STAGE);
+  uint32_t* cur_sh = (uint32_t*)(base_sh + 1);
+  const uint32_t smask = SLOTS - 1;
+  constexpr int S = STAGE;
+  for (int b = blockIdx.x; b < B; b += gridDim.x) {
+    const int64_t l0 = loff[b], l1 = loff[b + 1];
+    const int64_t r0 = roff[b], r1 = roff[b + 1];
+    const int64_t lnb = l1 - l0;
+    if (lnb == 0 || r1 == r0 || lnb > ROWCAP) continue;
+    for (int s = threadIdx.x; s < SLOTS; s += blockDim.x) tbl[s].x = EMPTY;
+    if (threadIdx.x == 0) *cur_sh = 0;
+    __syncthreads();
+    for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
+      longlong2 row = lrows[i];
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)row.x) & smask;
+      for (;;) {
+        unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
+                                           (unsigned long long)EMPTY,
+                                           (unsigned long long)row.x);
+        if (old == (unsigned long long)EMPTY) break;
+        slot = (slot + 1) & smask;
+      }
+      tbl[slot].y = row.y;
+    }
+    __syncthreads();
+    for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
+      longlong2 prow = rrows[j];
+      uint32_t slot = (uint32_t)dj_mix64((uint64_t)prow.x) & smask;
+      for (;;) {
+        longlong2 e = tbl[slot];
+        if (e.x == EMPTY) break;
+        if (e.x == prow.x) {
+          uint32_t pos = atomicAdd(cur_sh, 1u);
+          if (pos < (uint32_t)S) {
+            stage[0 * S + pos] = prow.x;
+            stage[1 * S + pos] = e.y;
+            stage[2 * S + pos] = prow.x;
+            stage[3 * S + pos] = prow.y;
+          } else {
+            long long idx = (long long)atomicAdd(counter, 1ull);
+            if (idx < cap) {
+              out0[idx] = prow.x;
+              out1[idx] = e.y;
+              out2[idx] = prow.x;
+              out3[idx] = prow.y;
+            }
+          }
+        }
+        slot = (slot + 1) & smask;
+      }
+    }
+    __syncthreads();
+    const uint32_t total = min(*cur_sh, (uint32_t)S);
+    if (threadIdx.x == 0 && total)
+      *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
+    __syncthreads();
+    if (total) {
+      const long long base = *base_sh;
+      for (uint32_t i = threadIdx.x; i < 4u * total; i += blockDim.x) {
+        uint32_t c = i / total, r = i - c * total;
+        long long idx = base + (long long)r;
+        if (idx < cap) {
+          int64_t v = stage[(size_t)c * S + r];
+          (c == 0 ? out0 : c == 1 ? out1 : c == 2 ? out2 : out3)[idx] = v;
+        }
+      }
+    }
+    __syncthreads();
+  }
+}
+
 /* fill bucket b with sizes[b] rows: build keys b*4096+i, probe keys
  * b*4096+i for i<30% else non-matching */
 __global__ void fill_kernel(longlong2* rows, const int64_t* starts,
@@ -501,6 +583,9 @@ static double run_case(const char* name, int B, int nrows, int match, int64_t ca
       join_kernel<1024, 512, 512><<<grid, 512, lds6>>>(lrows, loff, nullptr, 0, rrows, loff,
                                                        nullptr, 0, B, o0, o1, o2, o3, cap,
                                                        counter);
+    } else if (capB == -7) {
+      join_kernel_flat<<<grid, THREADS, lds>>>(lrows, loff, rrows, loff, B, o0, o1, o2, o3,
+                                               cap, counter);
     } else if (capB == -1)
       join_kernel_v2<<<grid, THREADS, lds>>>(lrows, loff, rrows, loff, B, o0, o1, o2, o3, cap,
                                              counter);
@@ -593,6 +678,7 @@ int main(int argc, char** argv)
     return 0;
   }
   run_case("base B=131072 r=762", B, nrows, match, 0, reps, false, true, false);
+  run_case("flat flush", B, nrows, match, -7, reps, false, true, false);
   run_case("big B/2 r=1526 1blk/CU", B / 2, nrows * 2, (int)(nrows * 2 * 0.3), -5, reps,
            false, true, false);
   run_case("small 2B r=381 512thr", B * 2, nrows / 2, (int)(nrows / 2 * 0.3), -6, reps,
