@@ -179,3 +179,37 @@ def test_fused_big_buckets():
                            text=True, timeout=420, env=env)
         assert r.returncode == 0, force_f + "\n" + r.stdout + r.stderr
         assert "MULTIRANK OK" in r.stdout
+
+
+def test_kernel_path_transport_world1(dj):
+    """The raw C-ABI transport (dj_comm_init + dj_all_to_all_i64 +
+    dj_exchange_sizes) at world 1 over real RCCL — our analogue of the
+    reference's transport-level test (test/buffer_communicator.cu: set a
+    pattern, exchange, assert). Self-exchange goes through the same grouped
+    ncclSend/ncclRecv the N>1 peer slices use."""
+    import ctypes
+    L = dj.lib()
+    nbytes = L.dj_rccl_unique_id_bytes()
+    idb = np.zeros(nbytes, dtype=np.uint8)
+    L.dj_rccl_get_unique_id(idb.ctypes.data)
+    L.dj_comm_init(0, 1, idb.ctypes.data)
+    try:
+        assert L.dj_comm_rank() == 0 and L.dj_comm_size() == 1
+        # sizes exchange: world of 1 -> recv == send
+        send_counts = np.array([123456], dtype=np.int64)
+        recv_counts = np.zeros(1, dtype=np.int64)
+        L.dj_exchange_sizes(send_counts.ctypes.data, recv_counts.ctypes.data)
+        assert recv_counts[0] == 123456
+        # buffer all-to-all: pattern round-trips through the self slice
+        n = 1 << 20
+        pattern = (np.arange(n, dtype=np.int64) * 2654435761) ^ 0x5DEECE66D
+        d_send = dj.DeviceArray.from_numpy(pattern)
+        d_recv = dj.DeviceArray(n)
+        send_off = np.array([0, n], dtype=np.int64)
+        recv_off = np.array([0, n], dtype=np.int64)
+        L.dj_all_to_all_i64(d_send.ptr, send_off.ctypes.data, d_recv.ptr,
+                            recv_off.ctypes.data)
+        L.dj_sync()
+        assert (d_recv.to_numpy() == pattern).all()
+    finally:
+        L.dj_comm_finalize()
