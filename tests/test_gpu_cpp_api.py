@@ -128,7 +128,8 @@ def test_multirank_loopback():
     for args in (["2", "2"], ["4", "1"],
                  ["4", "1", "400000", "2", "0"],   # 2-level hierarchy
                  ["4", "1", "400000", "1", "0"],   # nvl=1: full shuffle + local join
-                 ["2", "2", "400000", "2", "1"]):  # compressed wire
+                 ["2", "2", "400000", "2", "1"],   # compressed wire
+                 ["4", "2", "37", "4", "1"]):      # tiny: empty slices/buckets
         r = subprocess.run([exe] + args, capture_output=True, text=True, timeout=240)
         assert r.returncode == 0, " ".join(args) + "\n" + r.stdout + r.stderr
         assert "MULTIRANK OK" in r.stdout
