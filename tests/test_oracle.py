@@ -158,34 +158,73 @@ def test_golden_join():
     assert (got == want).all()
 
 
-def test_murmur3_known_vector():
-    # Standard MurmurHash3_x86_32 test vector: bytes 00..07 little-endian of
-    # key 0x0706050403020100 with seed 0 must hash like the canonical
-    # algorithm. Value computed from the public-domain reference algorithm.
-    key = 0x0706050403020100
-    got = oracle.row_hash(key, oracle.HASH_MURMUR3, 0)
-
-    # independent pure-python MurmurHash3_x86_32 over the 8 bytes
+def _murmur3_x86_32(data: bytes, seed: int) -> int:
+    """Independent pure-python restatement of MurmurHash3_x86_32 (Appleby's
+    public-domain algorithm), used only by the tests below. Validated against
+    the PUBLIC vectors in tests/golden/murmur3_public_vectors.json before it
+    is allowed to pin anything in this repo."""
     def rotl(x, r):
         return ((x << r) | (x >> (32 - r))) & 0xFFFFFFFF
 
-    data = key.to_bytes(8, "little")
-    h = 0
-    for blk in range(2):
-        k1 = int.from_bytes(data[4 * blk:4 * blk + 4], "little")
-        k1 = (k1 * 0xCC9E2D51) & 0xFFFFFFFF
+    c1, c2 = 0xCC9E2D51, 0x1B873593
+    h1 = seed & 0xFFFFFFFF
+    nblocks = len(data) // 4
+    for i in range(nblocks):
+        k1 = int.from_bytes(data[4 * i:4 * i + 4], "little")
+        k1 = (k1 * c1) & 0xFFFFFFFF
         k1 = rotl(k1, 15)
-        k1 = (k1 * 0x1B873593) & 0xFFFFFFFF
-        h ^= k1
-        h = rotl(h, 13)
-        h = (h * 5 + 0xE6546B64) & 0xFFFFFFFF
-    h ^= 8
-    h ^= h >> 16
-    h = (h * 0x85EBCA6B) & 0xFFFFFFFF
-    h ^= h >> 13
-    h = (h * 0xC2B2AE35) & 0xFFFFFFFF
-    h ^= h >> 16
-    assert got == h
+        k1 = (k1 * c2) & 0xFFFFFFFF
+        h1 ^= k1
+        h1 = rotl(h1, 13)
+        h1 = (h1 * 5 + 0xE6546B64) & 0xFFFFFFFF
+    tail = data[nblocks * 4:]
+    k1 = 0
+    if len(tail) >= 3:
+        k1 ^= tail[2] << 16
+    if len(tail) >= 2:
+        k1 ^= tail[1] << 8
+    if len(tail) >= 1:
+        k1 ^= tail[0]
+        k1 = (k1 * c1) & 0xFFFFFFFF
+        k1 = rotl(k1, 15)
+        k1 = (k1 * c2) & 0xFFFFFFFF
+        h1 ^= k1
+    h1 ^= len(data)
+    h1 ^= h1 >> 16
+    h1 = (h1 * 0x85EBCA6B) & 0xFFFFFFFF
+    h1 ^= h1 >> 13
+    h1 = (h1 * 0xC2B2AE35) & 0xFFFFFFFF
+    h1 ^= h1 >> 16
+    return h1
+
+
+def test_murmur3_public_vectors():
+    """Pin MurmurHash3_x86_32 to its PUBLISHED test vectors (fixture not
+    generated by this repo's code), then pin dj_murmur3_int64 (csrc/dj_hash.h,
+    the partition-placement hash replacing cudf's MurmurHash3 in
+    distributed_join.cpp:211-226 / shuffle_on.cpp:59-60) against the
+    so-validated restatement for 8-byte little-endian int64 inputs."""
+    import json
+    with open(os.path.join(GOLDEN, "murmur3_public_vectors.json")) as fh:
+        fixture = json.load(fh)
+    assert len(fixture["vectors"]) >= 13
+    for vec in fixture["vectors"]:
+        data = (bytes.fromhex(vec["data_hex"]) if "data_hex" in vec
+                else vec["data_utf8"].encode())
+        seed = int(vec["seed"], 16)
+        want = int(vec["hash"], 16)
+        assert _murmur3_x86_32(data, seed) == want, vec
+    # the restatement now carries the public pin down to our int64 form:
+    # cudf::hash_partition on ONE fundamental INT64 column hashes the 8
+    # little-endian bytes of each key (no multi-column seed-combine involved)
+    keys = [0, 1, -1, 3, 0x0706050403020100, 2**63 - 1, -2**63, 123456789,
+            -987654321, 2**32, 2**32 - 1]
+    rng = np.random.default_rng(7)
+    keys += [int(x) for x in rng.integers(-2**63, 2**63 - 1, size=64, dtype=np.int64)]
+    for seed in (0, 12345678, 87654321):  # cudf default + the reference's two
+        for k in keys:
+            want = _murmur3_x86_32((k & 0xFFFFFFFFFFFFFFFF).to_bytes(8, "little"), seed)
+            assert oracle.row_hash(k, oracle.HASH_MURMUR3, seed) == want, (k, seed)
 
 
 # -------------------------------------------------------------- partition
