@@ -41,6 +41,25 @@ def log(rank, *a):
         print(*a, file=sys.stderr, flush=True)
 
 
+def rccl_bootstrap(dist, rank, world, L):
+    """Exchange the RCCL unique id (rank 0 -> all ranks) over the already-
+    initialized gloo process group — the reference's MPI_Bcast of
+    ncclGetUniqueId (communicator.cpp:799-817). Returns the id bytes every
+    rank passes to CppCommunicator. Covered CPU-side (2-rank gloo, stubbed
+    ncclGetUniqueId) by tests/test_bootstrap_dryrun.py; the driver's N>1
+    bench executes exactly this function."""
+    import torch
+    nbytes = L.dj_rccl_unique_id_bytes()
+    if rank == 0:
+        buf = np.zeros(nbytes, dtype=np.uint8)
+        L.dj_rccl_get_unique_id(buf.ctypes.data)
+        t = torch.from_numpy(buf)
+    else:
+        t = torch.zeros(nbytes, dtype=torch.uint8)
+    dist.broadcast(t, src=0)
+    return np.ascontiguousarray(t.numpy())
+
+
 def cpu_baseline_leg(sample_rows):
     """Oracle OpenMP radix-partition join (the 'port' CPU baseline) on a
     bounded sample of the same workload, timed on this box's host cores.
@@ -106,18 +125,7 @@ def main():
     L.dj_set_device(local_rank % max(L.dj_device_count(), 1))
 
     # RCCL bootstrap: rank 0's unique id broadcast over gloo
-    id_bytes = None
-    if world > 1:
-        import torch
-        nbytes = L.dj_rccl_unique_id_bytes()
-        if rank == 0:
-            buf = np.zeros(nbytes, dtype=np.uint8)
-            L.dj_rccl_get_unique_id(buf.ctypes.data)
-            t = torch.from_numpy(buf)
-        else:
-            t = torch.zeros(nbytes, dtype=torch.uint8)
-        dist.broadcast(t, src=0)
-        id_bytes = np.ascontiguousarray(t.numpy())
+    id_bytes = rccl_bootstrap(dist, rank, world, L) if world > 1 else None
     comm = dj.CppCommunicator(rank, world, id_bytes)
 
     rows = args.rows
@@ -182,8 +190,9 @@ def main():
     # N==1 (reference single-rank semantics: local join only, no
     # partition/shuffle stage — distributed_join.cpp:200-214):
     #   bucket_scatter = the two-level local bucket partition: slack pass A
-    #     (16 B read + 16 B write, no count pass) + pass B (8 B count read +
-    #     16 B read + 16 B write) = 72 B per row per table
+    #     (16 B read + 16 B write, no count pass) + count-free slack pass B
+    #     (16 B read + 16 B write at analytic per-bucket starts) = 64 B per
+    #     row per table
     # N>1 (fused wire path):
     #   part_scatter   = the fused rank+group partition (8 B count-read +
     #                    16 B read + 16 B write per row, both tables)
@@ -195,7 +204,7 @@ def main():
     if N == 1:
         alg = {
             "part_scatter": 1.0,  # not run at N==1
-            "bucket_scatter": 72.0 * (lrows + rrows),
+            "bucket_scatter": 64.0 * (lrows + rrows),
             "join_fused": 16.0 * (lrows + rrows) + 32.0 * matches,
         }
     else:

@@ -245,13 +245,30 @@ int64_t dj_local_inner_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln
 
 namespace {
 
+/* round-2 local join uses the count-free slack bucket layout (pass B writes
+ * bucket b at b*capB with length lens[b] — bucket_partition2_slack) whenever
+ * its u32 row indices fit; otherwise the exact compact path. */
+bool bucket_slack_mode(int64_t ln, int64_t rn, int B)
+{
+  const int PA = dj::bucket_groups_for(B);
+  if (PA < 2) return false;
+  const int64_t maxn = ln > rn ? ln : rn;
+  if (maxn >= (int64_t)UINT32_MAX) return false;
+  if ((int64_t)PA * dj::slack_capA(maxn, PA) + maxn >= (int64_t)UINT32_MAX) return false;
+  if ((int64_t)B * dj::slack_capB(ln, B) >= (int64_t)UINT32_MAX) return false;
+  if ((int64_t)B * dj::slack_capB(rn, B) >= (int64_t)UINT32_MAX) return false;
+  return true;
+}
+
 struct BucketScratch {
-  longlong2 *lpairs, *rpairs;  // bucketed {key,payload} pairs (final)
+  longlong2 *lpairs, *rpairs;  // bucketed {key,payload} pairs (final; slack
+                               // layout B*capB when bucket_slack_mode)
   longlong2* tmp_pairs;        // pass-A staging (size max(ln,rn))
-  int64_t *loff, *roff;        // int64[B+1]
+  int64_t *loff, *roff;        // int64[B+1] (compact mode)
+  uint32_t *llen, *rlen;       // u32[B] (slack mode)
   int64_t* segoff;             // int64[PA+1]
   uint32_t* counts;            // u32[kBucketBlocks*PA]
-  uint32_t* totals;            // u32[PA]
+  uint32_t* totals;            // u32[PA] (also the slack pass-A cursors)
   uint32_t* flags;             // u32[B]
   int* any_overflow;           // int[1]
 };
@@ -260,6 +277,9 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
 {
   const int PA = dj::bucket_groups_for(B);
   const int64_t maxn = ln > rn ? ln : rn;
+  const bool slack = bucket_slack_mode(ln, rn, B);
+  const size_t lrows = slack ? (size_t)B * dj::slack_capB(ln, B) : (size_t)ln;
+  const size_t rrows = slack ? (size_t)B * dj::slack_capB(rn, B) : (size_t)rn;
   char* p = (char*)base;
   auto take = [&](size_t bytes) {
     void* r = p;
@@ -267,11 +287,13 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
     return r;
   };
   BucketScratch s;
-  s.lpairs = (longlong2*)take((size_t)ln * 16);
-  s.rpairs = (longlong2*)take((size_t)rn * 16);
+  s.lpairs = (longlong2*)take(lrows * 16);
+  s.rpairs = (longlong2*)take(rrows * 16);
   s.tmp_pairs = (longlong2*)take((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);
   s.loff = (int64_t*)take((size_t)(B + 1) * 8);
   s.roff = (int64_t*)take((size_t)(B + 1) * 8);
+  s.llen = (uint32_t*)take((size_t)B * 4);
+  s.rlen = (uint32_t*)take((size_t)B * 4);
   s.segoff = (int64_t*)take((size_t)(PA + 1) * 8);
   s.counts = (uint32_t*)take((size_t)dj::kBucketBlocks * PA * 4);
   s.totals = (uint32_t*)take((size_t)PA * 4);
@@ -287,13 +309,18 @@ int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
   int B = dj::bucket_count_for(ln, rn);
   const int PA = dj::bucket_groups_for(B);
   const int64_t maxn = ln > rn ? ln : rn;
+  const bool slack = bucket_slack_mode(ln, rn, B);
+  const size_t lrows = slack ? (size_t)B * dj::slack_capB(ln, B) : (size_t)ln;
+  const size_t rrows = slack ? (size_t)B * dj::slack_capB(rn, B) : (size_t)rn;
   size_t bytes = 0;
   auto add = [&](size_t b) { bytes += (b + 255) & ~(size_t)255; };
-  add((size_t)ln * 16);
-  add((size_t)rn * 16);
+  add(lrows * 16);
+  add(rrows * 16);
   add((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);  // pass-A slack staging (>= maxn)
   add((size_t)(B + 1) * 8);
   add((size_t)(B + 1) * 8);
+  add((size_t)B * 4);
+  add((size_t)B * 4);
   add((size_t)(PA + 1) * 8);
   add((size_t)dj::kBucketBlocks * PA * 4);
   add((size_t)PA * 4);
@@ -323,6 +350,26 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
   hipStream_t st = stream();
   DJ_HIP_CALL(hipMemsetAsync(d_any_overflow, 0, sizeof(int), st));
   DJ_HIP_CALL(hipMemsetAsync(s.flags, 0, (size_t)B * 4, st));
+  const int64_t maxn = ln > rn ? ln : rn;
+  /* beyond ~400M rows/table the buckets exceed the 2048-slot cap — switch
+   * to the 4096-slot table (cap 3072, covers the 800M single-GPU shape)
+   * instead of overflowing every bucket into the per-bucket fallback */
+  const int slots = (maxn / B > 1300) ? 4096 : 2048;
+  if (bucket_slack_mode(ln, rn, B)) {
+    const int64_t capBl = dj::slack_capB(ln, B), capBr = dj::slack_capB(rn, B);
+    {
+      PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
+      dj::bucket_partition2_slack(d_lk, d_lp, ln, B, s.tmp_pairs, s.totals, capBl, s.lpairs,
+                                  s.llen, d_any_overflow, st);
+      dj::bucket_partition2_slack(d_rk, d_rp, rn, B, s.tmp_pairs, s.totals, capBr, s.rpairs,
+                                  s.rlen, d_any_overflow, st);
+    }
+    PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
+    dj::lds_join_slack(s.lpairs, s.llen, capBl, s.rpairs, s.rlen, capBr, B, slots, d_out0,
+                       d_out1, d_out2, d_out3, cap, d_counter, s.flags, d_any_overflow,
+                       d_error, st);
+    return;
+  }
   {
     PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
     dj::bucket_partition2(d_lk, d_lp, ln, B, s.tmp_pairs, s.counts, s.totals, s.segoff,
@@ -332,12 +379,6 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
   }
   {
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
-    /* B caps at 262144 (PA<=1024 x 256); beyond ~400M rows/table the
-     * buckets exceed the 2048-slot cap — switch to the 4096-slot table
-     * (cap 3072, covers the 800M single-GPU shape) instead of overflowing
-     * every bucket into the per-bucket fallback */
-    const int64_t maxn = ln > rn ? ln : rn;
-    const int slots = (maxn / B > 1300) ? 4096 : 2048;
     dj::lds_join(s.lpairs, s.loff, s.rpairs, s.roff, B, slots, d_out0, d_out1, d_out2,
                  d_out3, cap, d_counter, s.flags, d_any_overflow, d_error, st);
   }
@@ -376,21 +417,34 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
     return;
   }
   if (any) {
+    const bool slack = bucket_slack_mode(ln, rn, B);
+    const int64_t capBl = slack ? dj::slack_capB(ln, B) : 0;
+    const int64_t capBr = slack ? dj::slack_capB(rn, B) : 0;
     std::vector<uint32_t> flags((size_t)B);
     std::vector<int64_t> loff((size_t)B + 1), roff((size_t)B + 1);
+    std::vector<uint32_t> llen, rlen;
     DJ_HIP_CALL(hipMemcpy(flags.data(), s.flags, (size_t)B * 4, hipMemcpyDeviceToHost));
-    DJ_HIP_CALL(hipMemcpy(loff.data(), s.loff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
-    DJ_HIP_CALL(hipMemcpy(roff.data(), s.roff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
+    if (slack) {
+      llen.resize((size_t)B);
+      rlen.resize((size_t)B);
+      DJ_HIP_CALL(hipMemcpy(llen.data(), s.llen, (size_t)B * 4, hipMemcpyDeviceToHost));
+      DJ_HIP_CALL(hipMemcpy(rlen.data(), s.rlen, (size_t)B * 4, hipMemcpyDeviceToHost));
+    } else {
+      DJ_HIP_CALL(hipMemcpy(loff.data(), s.loff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
+      DJ_HIP_CALL(hipMemcpy(roff.data(), s.roff, ((size_t)B + 1) * 8, hipMemcpyDeviceToHost));
+    }
     for (int b = 0; b < B; b++) {
       if (!flags[b]) continue;
-      int64_t lnb = loff[b + 1] - loff[b];
-      int64_t rnb = roff[b + 1] - roff[b];
+      int64_t l0 = slack ? (int64_t)b * capBl : loff[b];
+      int64_t r0 = slack ? (int64_t)b * capBr : roff[b];
+      int64_t lnb = slack ? (int64_t)llen[b] : loff[b + 1] - loff[b];
+      int64_t rnb = slack ? (int64_t)rlen[b] : roff[b + 1] - roff[b];
       if (lnb == 0 || rnb == 0) continue;
       int64_t nslots = dj::join_table_slots(lnb);
       int64_t* d_table = (int64_t*)dj_dmalloc(nslots * 2 * sizeof(int64_t));
       dj_join_table_init(d_table, nslots);
-      dj::join_build_pairs(s.lpairs + loff[b], lnb, d_table, nslots, d_error, st);
-      dj::join_probe_pairs(s.rpairs + roff[b], rnb, d_table, nslots, d_out0, d_out1, d_out2,
+      dj::join_build_pairs(s.lpairs + l0, lnb, d_table, nslots, d_error, st);
+      dj::join_probe_pairs(s.rpairs + r0, rnb, d_table, nslots, d_out0, d_out1, d_out2,
                            d_out3, cap, d_counter, st);
       DJ_HIP_CALL(hipStreamSynchronize(st));
       dj_dfree(d_table);

@@ -830,104 +830,163 @@ void subpart_lists(const int64_t* d_keys, const int64_t* d_pay, const int64_t* d
  * shape) or 4096 (1 block/CU, buckets <= 3072 — used by the G>1 fused wire
  * path when its PA*F fan-out cap leaves ~1.5-3k rows/bucket, e.g. G=8 od=1;
  * measured 1.98 vs 1.67 ms at the same total rows,
- * experiments/join_layout.hip — far cheaper than whole-batch redos). */
-template <int SLOTS2>
+ * experiments/join_layout.hip — far cheaper than whole-batch redos).
+ *
+ * Round 2: each block processes KBUK=4 buckets per flush group with a
+ * watermark — the per-bucket global reserve atomic was the largest marginal
+ * cost of the r1 kernel (profiles/r01_ablation.txt: reserve+flush 0.55 ms at
+ * 131072 buckets; a single hot counter absorbs one wave-op per ~12 ns);
+ * accumulating the stage across buckets and flushing at the 3/4 watermark
+ * (or group end) cuts the atomics ~4x: 1.69 -> 1.35 ms measured
+ * (experiments/join_v2/v4). Stage spills (high-duplication buckets, or a
+ * bucket landing on a nearly-full stage at high selectivity) go straight to
+ * the global counter (plain per-match atomic: the compiler wave-aggregates
+ * it; a hand-ballot version of this spill path — even though it never
+ * executes at sel 0.3 — cost 1.0 ms of probe-walk codegen, 2.42 vs 1.41 ms,
+ * experiments/join_v6).
+ *
+ * Bucket input layout: SLACK=false — compact offsets (loff/roff, B+1);
+ * SLACK=true — the slack layout of bucket_partition2_slack
+ * (b*capL + llen[b]). Compile-time split for the same reason. */
+template <int SLOTS2, bool SLACK>
 __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
-  const longlong2* __restrict__ rrows, const int64_t* __restrict__ roff, int B,
+  const uint32_t* __restrict__ llen, int64_t capL, const longlong2* __restrict__ rrows,
+  const int64_t* __restrict__ roff, const uint32_t* __restrict__ rlen, int64_t capR, int B,
   int64_t* __restrict__ out0, int64_t* __restrict__ out1, int64_t* __restrict__ out2,
   int64_t* __restrict__ out3, int64_t cap, unsigned long long* counter,
   uint32_t* __restrict__ overflow_flags, int* __restrict__ any_overflow,
   int* __restrict__ error)
 {
+  constexpr int S = JOIN_STAGE_ROWS;
+  constexpr int WATER = S - S / 4;
+  constexpr int KBUK = 4;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbl = (longlong2*)smem;
   int64_t* stage = (int64_t*)(tbl + SLOTS2);  // SoA: stage[c*JOIN_STAGE_ROWS + i]
-  long long* base_sh = (long long*)(stage + 4 * JOIN_STAGE_ROWS);
+  long long* base_sh = (long long*)(stage + 4 * S);
   uint32_t* cur_sh = (uint32_t*)(base_sh + 1);
   const uint32_t smask = SLOTS2 - 1;
-  constexpr int S = JOIN_STAGE_ROWS;
+  if (threadIdx.x == 0) *cur_sh = 0;
+  __syncthreads();
 
-  for (int b = blockIdx.x; b < B; b += gridDim.x) {
-    const int64_t l0 = loff[b], l1 = loff[b + 1];
-    const int64_t r0 = roff[b], r1 = roff[b + 1];
-    const int64_t lnb = l1 - l0;
-    if (lnb == 0 || r1 == r0) continue;
-    if (lnb > SLOTS2 * 3 / 4) {  // skew overflow: host-side fallback joins it
-      if (threadIdx.x == 0) {
-        overflow_flags[b] = 1;
-        atomicOr(any_overflow, 1);  // bit 2 is the slack-partition overflow
+  for (int bb = blockIdx.x * KBUK; bb < B; bb += gridDim.x * KBUK) {
+    for (int k = 0; k < KBUK; k++) {
+      const int b = bb + k;
+      if (b >= B) break;
+      int64_t l0, l1, r0, r1;
+      if (SLACK) {
+        l0 = (int64_t)b * capL;
+        l1 = l0 + llen[b];
+        r0 = (int64_t)b * capR;
+        r1 = r0 + rlen[b];
+      } else {
+        l0 = loff[b];
+        l1 = loff[b + 1];
+        r0 = roff[b];
+        r1 = roff[b + 1];
       }
-      continue;
-    }
-    for (int s = threadIdx.x; s < SLOTS2; s += blockDim.x) tbl[s].x = kEmptyKey;
-    if (threadIdx.x == 0) *cur_sh = 0;
-    __syncthreads();
-    /* build */
-    for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
-      longlong2 row = lrows[i];
-      if (row.x == kEmptyKey) {
-        *error = 1;
+      const int64_t lnb = l1 - l0;
+      if (lnb == 0 || r1 == r0) continue;
+      if (lnb > SLOTS2 * 3 / 4) {  // skew: host-side fallback joins it
+        if (threadIdx.x == 0) {
+          overflow_flags[b] = 1;
+          atomicOr(any_overflow, 1);  // bit 2 is the slack-partition overflow
+        }
         continue;
       }
-      uint32_t slot = (uint32_t)dj_mix64((uint64_t)row.x) & smask;
-      for (;;) {
-        unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
-                                           (unsigned long long)kEmptyKey,
-                                           (unsigned long long)row.x);
-        if (old == (unsigned long long)kEmptyKey) break;
-        slot = (slot + 1) & smask;
+      for (int s = threadIdx.x; s < SLOTS2; s += blockDim.x) tbl[s].x = kEmptyKey;
+      __syncthreads();
+      /* build */
+      for (int64_t i = l0 + threadIdx.x; i < l1; i += blockDim.x) {
+        longlong2 row = lrows[i];
+        if (row.x == kEmptyKey) {
+          *error = 1;
+          continue;
+        }
+        uint32_t slot = (uint32_t)dj_mix64((uint64_t)row.x) & smask;
+        for (;;) {
+          unsigned long long old = atomicCAS((unsigned long long*)&tbl[slot].x,
+                                             (unsigned long long)kEmptyKey,
+                                             (unsigned long long)row.x);
+          if (old == (unsigned long long)kEmptyKey) break;
+          slot = (slot + 1) & smask;
+        }
+        tbl[slot].y = row.y;
       }
-      tbl[slot].y = row.y;
-    }
-    __syncthreads();
-    /* single-pass probe: matches append to the LDS stage */
-    for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
-      longlong2 prow = rrows[j];
-      uint32_t slot = (uint32_t)dj_mix64((uint64_t)prow.x) & smask;
-      for (;;) {
-        longlong2 e = tbl[slot];
-        if (e.x == kEmptyKey) break;
-        if (e.x == prow.x) {
-          uint32_t pos = atomicAdd(cur_sh, 1u);
-          if (pos < (uint32_t)S) {
-            stage[0 * S + pos] = prow.x;
-            stage[1 * S + pos] = e.y;
-            stage[2 * S + pos] = prow.x;
-            stage[3 * S + pos] = prow.y;
-          } else {
-            /* stage overflow (heavy duplication): spill directly; the
-             * compiler wave-aggregates the counter atomic */
-            long long idx = (long long)atomicAdd(counter, 1ull);
-            if (idx < cap) {
-              out0[idx] = prow.x;
-              out1[idx] = e.y;
-              out2[idx] = prow.x;
-              out3[idx] = prow.y;
+      __syncthreads();
+      /* single-pass probe: matches append to the LDS stage */
+      for (int64_t j = r0 + threadIdx.x; j < r1; j += blockDim.x) {
+        longlong2 prow = rrows[j];
+        uint32_t slot = (uint32_t)dj_mix64((uint64_t)prow.x) & smask;
+        for (;;) {
+          longlong2 e = tbl[slot];
+          if (e.x == kEmptyKey) break;
+          if (e.x == prow.x) {
+            uint32_t pos = atomicAdd(cur_sh, 1u);
+            if (pos < (uint32_t)S) {
+              stage[0 * S + pos] = prow.x;
+              stage[1 * S + pos] = e.y;
+              stage[2 * S + pos] = prow.x;
+              stage[3 * S + pos] = prow.y;
+            } else {
+              /* stage overflow (heavy duplication): spill directly; the
+               * compiler wave-aggregates the counter atomic */
+              long long idx = (long long)atomicAdd(counter, 1ull);
+              if (idx < cap) {
+                out0[idx] = prow.x;
+                out1[idx] = e.y;
+                out2[idx] = prow.x;
+                out3[idx] = prow.y;
+              }
             }
           }
-        }
-        slot = (slot + 1) & smask;
-      }
-    }
-    __syncthreads();
-    const uint32_t total = min(*cur_sh, (uint32_t)S);
-    if (threadIdx.x == 0 && total)
-      *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
-    __syncthreads();
-    if (total) {
-      const long long base = *base_sh;
-      for (uint32_t i = threadIdx.x; i < total; i += blockDim.x) {
-        long long idx = base + (long long)i;
-        if (idx < cap) {
-          out0[idx] = stage[0 * S + i];
-          out1[idx] = stage[1 * S + i];
-          out2[idx] = stage[2 * S + i];
-          out3[idx] = stage[3 * S + i];
+          slot = (slot + 1) & smask;
         }
       }
+      __syncthreads();
+      /* watermark: flush early so the stage never spills structurally */
+      if (k < KBUK - 1 && *cur_sh < (uint32_t)WATER) continue;
+      const uint32_t total = min(*cur_sh, (uint32_t)S);
+      if (threadIdx.x == 0 && total)
+        *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
+      __syncthreads();
+      if (total) {
+        const long long base = *base_sh;
+        for (uint32_t i = threadIdx.x; i < total; i += blockDim.x) {
+          long long idx = base + (long long)i;
+          if (idx < cap) {
+            out0[idx] = stage[0 * S + i];
+            out1[idx] = stage[1 * S + i];
+            out2[idx] = stage[2 * S + i];
+            out3[idx] = stage[3 * S + i];
+          }
+        }
+      }
+      __syncthreads();
+      if (threadIdx.x == 0) *cur_sh = 0;
+      __syncthreads();
     }
-    __syncthreads();
+  }
+  /* epilogue: a group whose last bucket was empty, skew-flagged, or past B
+   * (B % KBUK != 0) skips the k==KBUK-1 flush above and carries its staged
+   * rows forward — flush whatever remains before the block exits */
+  __syncthreads();
+  const uint32_t total = min(*cur_sh, (uint32_t)S);
+  if (threadIdx.x == 0 && total)
+    *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
+  __syncthreads();
+  if (total) {
+    const long long base = *base_sh;
+    for (uint32_t i = threadIdx.x; i < total; i += blockDim.x) {
+      long long idx = base + (long long)i;
+      if (idx < cap) {
+        out0[idx] = stage[0 * S + i];
+        out1[idx] = stage[1 * S + i];
+        out2[idx] = stage[2 * S + i];
+        out3[idx] = stage[3 * S + i];
+      }
+    }
   }
 }
 
@@ -1072,11 +1131,50 @@ void fused_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int
  * redo the whole join (the partition output is incomplete). */
 constexpr int SLACK_TILE = 8192;
 
+/* block-level exclusive scan of hist[0..P) into base[0..P) via per-wave
+ * shfl scans + one cross-wave partial pass (2 barriers; the Hillis-Steele
+ * LDS scan costs 2*log2(P) barriers). partials: u32[16] LDS. */
+__device__ __forceinline__ void wave_excl_scan(const uint32_t* hist, uint32_t* base,
+                                               uint32_t* partials, int P)
+{
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid >> 6;
+  uint32_t v = (tid < P) ? hist[tid] : 0;
+  uint32_t incl = v;
+#pragma unroll
+  for (int off = 1; off < WAVE; off <<= 1) {
+    uint32_t up = __shfl_up(incl, off);
+    if (lane >= off) incl += up;
+  }
+  if (lane == WAVE - 1 && wid < (P + WAVE - 1) / WAVE) partials[wid] = incl;
+  __syncthreads();
+  if (wid == 0) {
+    const int nw = (P + WAVE - 1) / WAVE;
+    uint32_t pv = (lane < nw) ? partials[lane] : 0;
+    uint32_t pincl = pv;
+#pragma unroll
+    for (int off = 1; off < WAVE; off <<= 1) {
+      uint32_t up = __shfl_up(pincl, off);
+      if (lane >= off) pincl += up;
+    }
+    if (lane < nw) partials[lane] = pincl - pv;
+  }
+  __syncthreads();
+  if (tid < P) base[tid] = incl - v + partials[wid];
+}
+
+/* Round 2: software-pipelined — the next tile's 8 nontemporal loads per
+ * lane issue BEFORE the current tile's flush, so the HBM load latency hides
+ * under the store burst (this kernel runs 1 block/CU, so barriered phases
+ * cannot overlap across blocks; measured 0.90 -> ~0.86 ms/table with the
+ * wave scans, experiments/join_v4). */
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
   const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int P,
   int64_t capA, uint32_t* __restrict__ gcursor, int* __restrict__ any_overflow,
   longlong2* __restrict__ out_pairs)
 {
+  constexpr int VPT = SLACK_TILE / BUCKET_THREADS;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbuf = (longlong2*)smem;
   uint32_t* hist = (uint32_t*)(tbuf + SLACK_TILE);
@@ -1085,40 +1183,35 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
   uint32_t* glim = gcur + P;  // per-group segment end (precomputed: the
                               // per-row bound check must not pay a 64-bit
                               // multiply in the flush loop)
+  uint32_t* partials = glim + P;  // u32[16]
   __shared__ int s_ovf;
   const int tid = threadIdx.x;
   if (tid == 0) s_ovf = 0;
+  if (tid < P) hist[tid] = 0;
   const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
   const int64_t start = (int64_t)blockIdx.x * chunk;
   const int64_t end = min(start + chunk, n);
-  constexpr int VPT = SLACK_TILE / BUCKET_THREADS;
+  if (start >= end) return;
+  __syncthreads();
+
+  longlong2 r[VPT];
+  uint32_t g[VPT], rank[VPT];
+#pragma unroll
+  for (int v = 0; v < VPT; v++) {
+    int64_t i = start + (int64_t)v * BUCKET_THREADS + tid;
+    if (i < end) {
+      r[v].x = nt_load(&keys[i]);
+      r[v].y = pay ? nt_load(&pay[i]) : i;
+      g[v] = groupA_of(r[v].x, P);
+      rank[v] = atomicAdd(&hist[g[v]], 1u);
+    }
+  }
+  __syncthreads();
+
   for (int64_t t0 = start; t0 < end; t0 += SLACK_TILE) {
     const int count = (int)min((int64_t)SLACK_TILE, end - t0);
-    if (tid < P) hist[tid] = 0;
-    __syncthreads();
-    longlong2 r[VPT];
-    uint32_t g[VPT], rank[VPT];
-#pragma unroll
-    for (int v = 0; v < VPT; v++) {
-      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
-      if (i < end) {
-        r[v].x = nt_load(&keys[i]);
-        r[v].y = pay ? nt_load(&pay[i]) : i;
-        g[v] = groupA_of(r[v].x, P);
-        rank[v] = atomicAdd(&hist[g[v]], 1u);
-      }
-    }
-    __syncthreads();
-    if (tid < P) base[tid] = hist[tid];
-    __syncthreads();
-    for (int off = 1; off < P; off <<= 1) {
-      uint32_t add = (tid < P && tid >= off) ? base[tid - off] : 0;
-      __syncthreads();
-      if (tid < P) base[tid] += add;
-      __syncthreads();
-    }
-    if (tid < P) base[tid] -= hist[tid];
-    __syncthreads();
+    const int64_t t1 = t0 + SLACK_TILE;
+    wave_excl_scan(hist, base, partials, P);
     /* claim this tile's run in each group's slack segment */
     if (tid < P) {
       const int64_t seg0 = (int64_t)tid * capA;
@@ -1128,10 +1221,21 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
     __syncthreads();
 #pragma unroll
     for (int v = 0; v < VPT; v++) {
-      int64_t i = t0 + (int64_t)v * blockDim.x + tid;
+      int64_t i = t0 + (int64_t)v * BUCKET_THREADS + tid;
       if (i < end) tbuf[base[g[v]] + rank[v]] = r[v];
     }
     __syncthreads();
+    if (tid < P) hist[tid] = 0; /* free after the claim; re-ranked after flush */
+    /* issue the next tile's loads BEFORE the flush */
+    longlong2 r2[VPT];
+#pragma unroll
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t1 + (int64_t)v * BUCKET_THREADS + tid;
+      if (i < end) {
+        r2[v].x = nt_load(&keys[i]);
+        r2[v].y = pay ? nt_load(&pay[i]) : i;
+      }
+    }
     for (int pos = tid; pos < count; pos += blockDim.x) {
       longlong2 row = tbuf[pos];
       uint32_t gg = groupA_of(row.x, P);
@@ -1140,6 +1244,16 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
         out_pairs[dst] = row;
       else
         s_ovf = 1;  // benign LDS race: any 1 wins
+    }
+    __syncthreads();
+#pragma unroll
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t1 + (int64_t)v * BUCKET_THREADS + tid;
+      if (i < end) {
+        r[v] = r2[v];
+        g[v] = groupA_of(r[v].x, P);
+        rank[v] = atomicAdd(&hist[g[v]], 1u);
+      }
     }
     __syncthreads();
   }
@@ -1196,6 +1310,135 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack_kernel(
                                 out_pairs);
 }
 
+/* pass B without the count sweep (round 2): block = pass-A group a reads its
+ * slack segment [a*capA, a*capA+seg_len[a]) and scatters into F per-bucket
+ * SLACK segments at analytic starts b*capB with LDS cursors (only this
+ * block writes group a's buckets, so no global cursor atomics), pipelined:
+ * the next tile's loads issue before this tile's flush so the HBM load
+ * latency hides under the store burst. Bucket lengths out in d_lens. */
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_kernel(
+  const longlong2* __restrict__ in_pairs, const uint32_t* __restrict__ seg_len, int64_t capA,
+  int F, int64_t capB, longlong2* __restrict__ out_pairs, uint32_t* __restrict__ lens,
+  int* __restrict__ any_overflow)
+{
+  constexpr int VPT = SCATTER_TILE / BUCKET_THREADS; /* 4 */
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  longlong2* tbuf = (longlong2*)smem;
+  uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
+  uint32_t* base = hist + F;
+  uint32_t* gcur = base + F;
+  uint32_t* partials = gcur + F; /* 16 */
+  __shared__ int s_ovf;
+  const int tid = threadIdx.x;
+  const int a = blockIdx.x;
+  const int64_t s0 = (int64_t)a * capA;
+  const int64_t s1 = s0 + seg_len[a];
+  if (tid == 0) s_ovf = 0;
+  for (int j = tid; j < F; j += blockDim.x)
+    gcur[j] = (uint32_t)(((int64_t)a * F + j) * capB);
+  if (tid < F) hist[tid] = 0;
+  __syncthreads();
+  if (s0 >= s1) {
+    for (int j = tid; j < F; j += blockDim.x) lens[(size_t)a * F + j] = 0;
+    return;
+  }
+  longlong2 r[VPT];
+  uint32_t g[VPT], rank[VPT];
+#pragma unroll
+  for (int v = 0; v < VPT; v++) {
+    int64_t i = s0 + (int64_t)v * BUCKET_THREADS + tid;
+    if (i < s1) {
+      r[v].x = nt_load(&in_pairs[i].x);
+      r[v].y = nt_load(&in_pairs[i].y);
+      g[v] = subF_of(r[v].x, F);
+      rank[v] = atomicAdd(&hist[g[v]], 1u);
+    }
+  }
+  __syncthreads();
+  for (int64_t t0 = s0; t0 < s1; t0 += SCATTER_TILE) {
+    const int count = (int)min((int64_t)SCATTER_TILE, s1 - t0);
+    const int64_t t1 = t0 + SCATTER_TILE;
+    wave_excl_scan(hist, base, partials, F);
+    __syncthreads();
+#pragma unroll
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t0 + (int64_t)v * BUCKET_THREADS + tid;
+      if (i < s1) tbuf[base[g[v]] + rank[v]] = r[v];
+    }
+    __syncthreads();
+    longlong2 r2[VPT];
+#pragma unroll
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t1 + (int64_t)v * BUCKET_THREADS + tid;
+      if (i < s1) {
+        r2[v].x = nt_load(&in_pairs[i].x);
+        r2[v].y = nt_load(&in_pairs[i].y);
+      }
+    }
+    for (int pos = tid; pos < count; pos += blockDim.x) {
+      longlong2 row = tbuf[pos];
+      uint32_t gg = subF_of(row.x, F);
+      uint32_t dst = gcur[gg] + (uint32_t)(pos - base[gg]);
+      uint32_t limit = (uint32_t)(((int64_t)a * F + gg) * capB + capB);
+      if (dst < limit)
+        out_pairs[dst] = row; /* plain store: partial lines must merge in L2 */
+      else
+        s_ovf = 1;
+    }
+    __syncthreads();
+    if (tid < F) gcur[tid] += hist[tid];
+    if (tid < F) hist[tid] = 0;
+    __syncthreads();
+#pragma unroll
+    for (int v = 0; v < VPT; v++) {
+      int64_t i = t1 + (int64_t)v * BUCKET_THREADS + tid;
+      if (i < s1) {
+        r[v] = r2[v];
+        g[v] = subF_of(r[v].x, F);
+        rank[v] = atomicAdd(&hist[g[v]], 1u);
+      }
+    }
+    __syncthreads();
+  }
+  __syncthreads();
+  for (int j = tid; j < F; j += blockDim.x) {
+    int64_t b = (int64_t)a * F + j;
+    uint32_t len = gcur[j] - (uint32_t)(b * capB);
+    lens[b] = len > (uint32_t)capB ? (uint32_t)capB : len;
+  }
+  if (tid == 0 && s_ovf) atomicOr(any_overflow, 2);
+}
+
+void bucket_partition2_slack(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
+                             longlong2* d_tmp_pairs, uint32_t* d_cursors, int64_t capB,
+                             longlong2* d_out_pairs, uint32_t* d_lens, int* d_any_overflow,
+                             hipStream_t s)
+{
+  DJ_CHECK_ERROR(n < (int64_t)UINT32_MAX, "bucket_partition_slack: n must be < 2^32");
+  const int PA = bucket_groups_for(B);
+  const int F = B / PA;
+  const int64_t capA = slack_capA(n, PA);
+  DJ_CHECK_ERROR(PA >= 2 && PA <= 1024 && F >= 1 && F <= 1024,
+                 "bucket_partition_slack: B out of range (PA must be >= 2)");
+  DJ_CHECK_ERROR((int64_t)PA * capA + n < (int64_t)UINT32_MAX &&
+                   (int64_t)B * capB < (int64_t)UINT32_MAX,
+                 "bucket_partition_slack: slack layout exceeds u32 row index");
+  size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 4 * (size_t)PA * sizeof(uint32_t) + 64;
+  size_t sub_lds = SCATTER_TILE * sizeof(longlong2) + 3 * (size_t)F * sizeof(uint32_t) + 64;
+  DJ_HIP_CALL(hipMemsetAsync(d_cursors, 0, (size_t)PA * 4, s));
+  hipLaunchKernelGGL(bucket_scatter_slack_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
+                     scatter_lds, s, d_keys, d_pay, n, PA, capA, d_cursors, d_any_overflow,
+                     d_tmp_pairs);
+  DJ_HIP_CALL(hipGetLastError());
+  hipLaunchKernelGGL(clamp_seglen_kernel, dim3((PA + 255) / 256), dim3(256), 0, s, d_cursors,
+                     PA, (uint32_t)capA);
+  DJ_HIP_CALL(hipGetLastError());
+  hipLaunchKernelGGL(bucket_subpart_slack2_kernel, dim3(PA), dim3(BUCKET_THREADS), sub_lds, s,
+                     d_tmp_pairs, d_cursors, capA, F, capB, d_out_pairs, d_lens,
+                     d_any_overflow);
+  DJ_HIP_CALL(hipGetLastError());
+}
+
 void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
                        longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
                        int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
@@ -1219,7 +1462,7 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
              (int64_t)PA * slack_capA(n, PA) + n < (int64_t)UINT32_MAX) {
     /* slack path: no count pass (see bucket_scatter_slack_kernel header) */
     const int64_t capA = slack_capA(n, PA);
-    size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 4 * (size_t)PA * sizeof(uint32_t);
+    size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 4 * (size_t)PA * sizeof(uint32_t) + 64;
     DJ_HIP_CALL(hipMemsetAsync(d_totals, 0, (size_t)PA * 4, s));
     hipLaunchKernelGGL(bucket_scatter_slack_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
                        scatter_lds, s, d_keys, d_pay, n, PA, capA, d_totals, d_any_overflow,
@@ -1256,27 +1499,54 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
   }
 }
 
+static void lds_join_launch(const longlong2* d_lrows, const int64_t* d_loff,
+                            const uint32_t* d_llen, int64_t capL, const longlong2* d_rrows,
+                            const int64_t* d_roff, const uint32_t* d_rlen, int64_t capR,
+                            int B, int table_slots, int64_t* d_out0, int64_t* d_out1,
+                            int64_t* d_out2, int64_t* d_out3, int64_t cap, int64_t* d_counter,
+                            uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error,
+                            hipStream_t s)
+{
+  DJ_CHECK_ERROR(table_slots == 2048 || table_slots == 4096,
+                 "lds_join: table_slots must be 2048 or 4096");
+  int64_t groups = ((int64_t)B + 3) / 4; /* KBUK = 4 */
+  int grid = (int)(groups < 8192 ? groups : 8192);
+  size_t lds =
+    (size_t)table_slots * sizeof(longlong2) + 4 * JOIN_STAGE_ROWS * sizeof(int64_t) + 16;
+  const bool slack = d_llen != nullptr;
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lrows, d_loff,
+                       d_llen, capL, d_rrows, d_roff, d_rlen, capR, B, d_out0, d_out1,
+                       d_out2, d_out3, cap, (unsigned long long*)d_counter, d_overflow_flags,
+                       d_any_overflow, d_error);
+  };
+  if (table_slots == 4096)
+    slack ? launch(lds_join_kernel<4096, true>) : launch(lds_join_kernel<4096, false>);
+  else
+    slack ? launch(lds_join_kernel<2048, true>) : launch(lds_join_kernel<2048, false>);
+  DJ_HIP_CALL(hipGetLastError());
+}
+
 void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* d_rrows,
               const int64_t* d_roff, int B, int table_slots, int64_t* d_out0, int64_t* d_out1,
               int64_t* d_out2, int64_t* d_out3, int64_t cap, int64_t* d_counter,
               uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error, hipStream_t s)
 {
-  DJ_CHECK_ERROR(table_slots == 2048 || table_slots == 4096,
-                 "lds_join: table_slots must be 2048 or 4096");
-  int grid = B < 8192 ? B : 8192;
-  size_t lds =
-    (size_t)table_slots * sizeof(longlong2) + 4 * JOIN_STAGE_ROWS * sizeof(int64_t) + 16;
-  if (table_slots == 4096)
-    hipLaunchKernelGGL(lds_join_kernel<4096>, dim3(grid), dim3(BUCKET_THREADS), lds, s,
-                       d_lrows, d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3,
-                       cap, (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
-                       d_error);
-  else
-    hipLaunchKernelGGL(lds_join_kernel<2048>, dim3(grid), dim3(BUCKET_THREADS), lds, s,
-                       d_lrows, d_loff, d_rrows, d_roff, B, d_out0, d_out1, d_out2, d_out3,
-                       cap, (unsigned long long*)d_counter, d_overflow_flags, d_any_overflow,
-                       d_error);
-  DJ_HIP_CALL(hipGetLastError());
+  lds_join_launch(d_lrows, d_loff, nullptr, 0, d_rrows, d_roff, nullptr, 0, B, table_slots,
+                  d_out0, d_out1, d_out2, d_out3, cap, d_counter, d_overflow_flags,
+                  d_any_overflow, d_error, s);
+}
+
+void lds_join_slack(const longlong2* d_lrows, const uint32_t* d_llen, int64_t capL,
+                    const longlong2* d_rrows, const uint32_t* d_rlen, int64_t capR, int B,
+                    int table_slots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                    int64_t* d_out3, int64_t cap, int64_t* d_counter,
+                    uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error,
+                    hipStream_t s)
+{
+  lds_join_launch(d_lrows, nullptr, d_llen, capL, d_rrows, nullptr, d_rlen, capR, B,
+                  table_slots, d_out0, d_out1, d_out2, d_out3, cap, d_counter,
+                  d_overflow_flags, d_any_overflow, d_error, s);
 }
 
 }  // namespace dj

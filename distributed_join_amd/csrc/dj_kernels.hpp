@@ -90,6 +90,18 @@ inline int64_t slack_capA(int64_t n, int PA)
   const int64_t m = n / PA;
   return m + m / 16 + 1024;
 }
+/* per-bucket slack capacity for the count-free pass B (mean + ~7 sigma,
+ * rounded to 8): Poisson bucket occupancy overflows this with probability
+ * ~1e-12 per bucket; on overflow bit 2 of any_overflow is set and the caller
+ * redoes the join exactly (same contract as slack_capA) */
+inline int64_t slack_capB(int64_t n, int B)
+{
+  const int64_t lam = n / B < 1 ? 1 : n / B;
+  int64_t s = 1;
+  while (s * s < lam) s++;  // ceil(sqrt(lam))
+  int64_t cap = lam + 7 * s + 8;
+  return (cap + 7) & ~(int64_t)7;
+}
 /* Two-level non-stable partition into B buckets (B = PA*F per
  * bucket_groups_for; up to 1024x1024 = 1M buckets for ~800M-row tables) of
  * interleaved 16 B {key,payload} pairs. d_tmp_pairs: pass-A staging —
@@ -103,6 +115,21 @@ void bucket_partition2(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        longlong2* d_tmp_pairs, uint32_t* d_counts, uint32_t* d_totals,
                        int64_t* d_segoff, int64_t* d_offsets, longlong2* d_out_pairs,
                        int* d_any_overflow, hipStream_t s);
+/* Count-free two-level slack partition (round 2, the product local-join
+ * path): pass A as bucket_partition2's slack path, then pass B scatters each
+ * group's rows into per-bucket SLACK segments at analytic starts
+ * b*capB (capB = slack_capB(n, B)) with LDS cursors — no seghist sweep (the
+ * r1 pass B fetched full 16 B lines just to histogram 8 B keys: 3.28 GB PMC
+ * fetch vs 2.4 GB algorithmic). Bucket b holds d_lens[b] rows at
+ * d_out_pairs[b*capB]. Overflow of any slack segment sets bit 2 of
+ * *d_any_overflow (partition output incomplete -> caller redoes exactly).
+ * Requires PA >= 2, n < 2^32, PA*slack_capA(n,PA)+... and B*capB < 2^32
+ * (checked; errors loudly otherwise). d_tmp_pairs: longlong2[PA*capA];
+ * d_cursors: u32[PA]. */
+void bucket_partition2_slack(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,
+                             longlong2* d_tmp_pairs, uint32_t* d_cursors, int64_t capB,
+                             longlong2* d_out_pairs, uint32_t* d_lens, int* d_any_overflow,
+                             hipStream_t s);
 /* Fused per-bucket LDS build+probe over bucketed pair tables. table_slots:
  * 2048 (2 blocks/CU, bucket cap 1536 build rows = kJoinBucketRowCap) or
  * 4096 (1 block/CU, cap 3072 — for the fused wire path when the PA*F
@@ -113,6 +140,14 @@ void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* 
               const int64_t* d_roff, int B, int table_slots, int64_t* d_out0, int64_t* d_out1,
               int64_t* d_out2, int64_t* d_out3, int64_t cap, int64_t* d_counter,
               uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error, hipStream_t s);
+/* Same fused join over the slack bucket layout of bucket_partition2_slack:
+ * bucket b = d_lrows[b*capL .. +d_llen[b]) x d_rrows[b*capR .. +d_rlen[b]). */
+void lds_join_slack(const longlong2* d_lrows, const uint32_t* d_llen, int64_t capL,
+                    const longlong2* d_rrows, const uint32_t* d_rlen, int64_t capR, int B,
+                    int table_slots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                    int64_t* d_out3, int64_t cap, int64_t* d_counter,
+                    uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error,
+                    hipStream_t s);
 /* Global-table build/probe over interleaved pair inputs (skew fallback). */
 void join_build_pairs(const longlong2* d_rows, int64_t ln, int64_t* d_table, int64_t nslots,
                       int* d_error, hipStream_t s);
