@@ -292,8 +292,9 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
   s.tmp_pairs = (longlong2*)take((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);
   s.loff = (int64_t*)take((size_t)(B + 1) * 8);
   s.roff = (int64_t*)take((size_t)(B + 1) * 8);
-  s.llen = (uint32_t*)take((size_t)B * 4);
-  s.rlen = (uint32_t*)take((size_t)B * 4);
+  const size_t Bpad = ((size_t)B + 3) & ~(size_t)3;  // lds_join_slack KBUK pad
+  s.llen = (uint32_t*)take(Bpad * 4);
+  s.rlen = (uint32_t*)take(Bpad * 4);
   s.segoff = (int64_t*)take((size_t)(PA + 1) * 8);
   s.counts = (uint32_t*)take((size_t)dj::kBucketBlocks * PA * 4);
   s.totals = (uint32_t*)take((size_t)PA * 4);
@@ -319,8 +320,9 @@ int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
   add((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);  // pass-A slack staging (>= maxn)
   add((size_t)(B + 1) * 8);
   add((size_t)(B + 1) * 8);
-  add((size_t)B * 4);
-  add((size_t)B * 4);
+  const size_t Bpad = ((size_t)B + 3) & ~(size_t)3;
+  add(Bpad * 4);
+  add(Bpad * 4);
   add((size_t)(PA + 1) * 8);
   add((size_t)dj::kBucketBlocks * PA * 4);
   add((size_t)PA * 4);
@@ -357,6 +359,13 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
   const int slots = (maxn / B > 1300) ? 4096 : 2048;
   if (bucket_slack_mode(ln, rn, B)) {
     const int64_t capBl = dj::slack_capB(ln, B), capBr = dj::slack_capB(rn, B);
+    /* pad B to a multiple of 4 with zero-length buckets (the lds_join_slack
+     * KBUK contract; carve_bucket_scratch sized llen/rlen for it) */
+    const int Bpad = (B + 3) & ~3;
+    if (Bpad != B) {
+      DJ_HIP_CALL(hipMemsetAsync(s.llen + B, 0, (size_t)(Bpad - B) * 4, st));
+      DJ_HIP_CALL(hipMemsetAsync(s.rlen + B, 0, (size_t)(Bpad - B) * 4, st));
+    }
     {
       PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
       dj::bucket_partition2_slack(d_lk, d_lp, ln, B, s.tmp_pairs, s.totals, capBl, s.lpairs,
@@ -365,7 +374,7 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
                                   s.rlen, d_any_overflow, st);
     }
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
-    dj::lds_join_slack(s.lpairs, s.llen, capBl, s.rpairs, s.rlen, capBr, B, slots, d_out0,
+    dj::lds_join_slack(s.lpairs, s.llen, capBl, s.rpairs, s.rlen, capBr, Bpad, slots, d_out0,
                        d_out1, d_out2, d_out3, cap, d_counter, s.flags, d_any_overflow,
                        d_error, st);
     return;

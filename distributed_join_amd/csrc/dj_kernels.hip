@@ -847,8 +847,18 @@ void subpart_lists(const int64_t* d_keys, const int64_t* d_pay, const int64_t* d
  *
  * Bucket input layout: SLACK=false — compact offsets (loff/roff, B+1);
  * SLACK=true — the slack layout of bucket_partition2_slack
- * (b*capL + llen[b]). Compile-time split for the same reason. */
-template <int SLOTS2, bool SLACK>
+ * (b*capL + llen[b]). Compile-time split for the same reason.
+ *
+ * KBUK > 1 contract: B MUST be a multiple of KBUK with llen/rlen (or
+ * loff/roff) valid over all of it — callers pad with zero-length buckets
+ * (lds_join_slack checks). Padding keeps every flush-slot comparison
+ * against the constant KBUK-1: the b >= B tail break, a dynamic flush-slot
+ * register, and an in-range guard each put the kernel back at ~2.35 ms
+ * (experiments/join_v7 factorial; the goto-drain structure below measured
+ * 1.40 ms). KBUK == 1 (used for the compact wire path, whose B is not
+ * padded) flushes every bucket — the r1 behavior, ~0.3 ms slower per 100 M
+ * rows but contract-free. */
+template <int SLOTS2, bool SLACK, int KBUK>
 __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
   const uint32_t* __restrict__ llen, int64_t capL, const longlong2* __restrict__ rrows,
@@ -860,7 +870,6 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
 {
   constexpr int S = JOIN_STAGE_ROWS;
   constexpr int WATER = S - S / 4;
-  constexpr int KBUK = 4;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbl = (longlong2*)smem;
   int64_t* stage = (int64_t*)(tbl + SLOTS2);  // SoA: stage[c*JOIN_STAGE_ROWS + i]
@@ -871,9 +880,14 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   __syncthreads();
 
   for (int bb = blockIdx.x * KBUK; bb < B; bb += gridDim.x * KBUK) {
+    /* Exit-clean guarantee: every group's k == KBUK-1 slot flushes — valid
+     * buckets flush unconditionally there, and invalid slots (empty side,
+     * skew-routed, or caller-padded zero-length buckets) jump to the flush
+     * block instead of skipping it, so a block never exits carrying staged
+     * rows. */
     for (int k = 0; k < KBUK; k++) {
       const int b = bb + k;
-      if (b >= B) break;
+      if (b >= B) break;  // dead when KBUK > 1 (padded B); real for KBUK == 1
       int64_t l0, l1, r0, r1;
       if (SLACK) {
         l0 = (int64_t)b * capL;
@@ -887,13 +901,19 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
         r1 = roff[b + 1];
       }
       const int64_t lnb = l1 - l0;
-      if (lnb == 0 || r1 == r0) continue;
+      if (lnb == 0 || r1 == r0) {  // empty side (incl. padding buckets)
+        /* KBUK == 1 flushes every valid bucket, so there is never a carry
+         * to drain — plain skip (constant-folded) */
+        if (KBUK == 1 || k != KBUK - 1) continue;
+        goto flush;  // drain any carried stage at the group's flush slot
+      }
       if (lnb > SLOTS2 * 3 / 4) {  // skew: host-side fallback joins it
         if (threadIdx.x == 0) {
           overflow_flags[b] = 1;
           atomicOr(any_overflow, 1);  // bit 2 is the slack-partition overflow
         }
-        continue;
+        if (KBUK == 1 || k != KBUK - 1) continue;
+        goto flush;
       }
       for (int s = threadIdx.x; s < SLOTS2; s += blockDim.x) tbl[s].x = kEmptyKey;
       __syncthreads();
@@ -944,8 +964,10 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
           slot = (slot + 1) & smask;
         }
       }
+    flush:
       __syncthreads();
-      /* watermark: flush early so the stage never spills structurally */
+      /* watermark: flush early so the stage never spills structurally;
+       * unconditional at k == KBUK-1 (the block's exit-clean guarantee) */
       if (k < KBUK - 1 && *cur_sh < (uint32_t)WATER) continue;
       const uint32_t total = min(*cur_sh, (uint32_t)S);
       if (threadIdx.x == 0 && total)
@@ -966,26 +988,6 @@ __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
       __syncthreads();
       if (threadIdx.x == 0) *cur_sh = 0;
       __syncthreads();
-    }
-  }
-  /* epilogue: a group whose last bucket was empty, skew-flagged, or past B
-   * (B % KBUK != 0) skips the k==KBUK-1 flush above and carries its staged
-   * rows forward — flush whatever remains before the block exits */
-  __syncthreads();
-  const uint32_t total = min(*cur_sh, (uint32_t)S);
-  if (threadIdx.x == 0 && total)
-    *base_sh = (long long)atomicAdd(counter, (unsigned long long)total);
-  __syncthreads();
-  if (total) {
-    const long long base = *base_sh;
-    for (uint32_t i = threadIdx.x; i < total; i += blockDim.x) {
-      long long idx = base + (long long)i;
-      if (idx < cap) {
-        out0[idx] = stage[0 * S + i];
-        out1[idx] = stage[1 * S + i];
-        out2[idx] = stage[2 * S + i];
-        out3[idx] = stage[3 * S + i];
-      }
     }
   }
 }
@@ -1315,27 +1317,35 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack_kernel(
  * SLACK segments at analytic starts b*capB with LDS cursors (only this
  * block writes group a's buckets, so no global cursor atomics), pipelined:
  * the next tile's loads issue before this tile's flush so the HBM load
- * latency hides under the store burst. Bucket lengths out in d_lens. */
+ * latency hides under the store burst. Bucket lengths out in d_lens.
+ * BTILE 8192 (1 block/CU) measured 0.88 vs 1.01 ms/table over the 4096
+ * 2-block tile, and per-bucket limits are precomputed in LDS (glim) so the
+ * flush bound check pays no per-row 64-bit multiply (experiments/join_v7). */
+constexpr int BTILE = 8192;  // pass-B staging tile (128 KiB, 1 block/CU)
 __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_kernel(
   const longlong2* __restrict__ in_pairs, const uint32_t* __restrict__ seg_len, int64_t capA,
   int F, int64_t capB, longlong2* __restrict__ out_pairs, uint32_t* __restrict__ lens,
   int* __restrict__ any_overflow)
 {
-  constexpr int VPT = SCATTER_TILE / BUCKET_THREADS; /* 4 */
+  constexpr int VPT = BTILE / BUCKET_THREADS; /* 8 */
   extern __shared__ __attribute__((aligned(16))) char smem[];
   longlong2* tbuf = (longlong2*)smem;
-  uint32_t* hist = (uint32_t*)(tbuf + SCATTER_TILE);
+  uint32_t* hist = (uint32_t*)(tbuf + BTILE);
   uint32_t* base = hist + F;
   uint32_t* gcur = base + F;
-  uint32_t* partials = gcur + F; /* 16 */
+  uint32_t* glim = gcur + F;
+  uint32_t* partials = glim + F; /* 16 */
   __shared__ int s_ovf;
   const int tid = threadIdx.x;
   const int a = blockIdx.x;
   const int64_t s0 = (int64_t)a * capA;
   const int64_t s1 = s0 + seg_len[a];
   if (tid == 0) s_ovf = 0;
-  for (int j = tid; j < F; j += blockDim.x)
-    gcur[j] = (uint32_t)(((int64_t)a * F + j) * capB);
+  for (int j = tid; j < F; j += blockDim.x) {
+    const int64_t b0 = ((int64_t)a * F + j) * capB;
+    gcur[j] = (uint32_t)b0;
+    glim[j] = (uint32_t)(b0 + capB);
+  }
   if (tid < F) hist[tid] = 0;
   __syncthreads();
   if (s0 >= s1) {
@@ -1355,9 +1365,9 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_kernel(
     }
   }
   __syncthreads();
-  for (int64_t t0 = s0; t0 < s1; t0 += SCATTER_TILE) {
-    const int count = (int)min((int64_t)SCATTER_TILE, s1 - t0);
-    const int64_t t1 = t0 + SCATTER_TILE;
+  for (int64_t t0 = s0; t0 < s1; t0 += BTILE) {
+    const int count = (int)min((int64_t)BTILE, s1 - t0);
+    const int64_t t1 = t0 + BTILE;
     wave_excl_scan(hist, base, partials, F);
     __syncthreads();
 #pragma unroll
@@ -1379,8 +1389,7 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_kernel(
       longlong2 row = tbuf[pos];
       uint32_t gg = subF_of(row.x, F);
       uint32_t dst = gcur[gg] + (uint32_t)(pos - base[gg]);
-      uint32_t limit = (uint32_t)(((int64_t)a * F + gg) * capB + capB);
-      if (dst < limit)
+      if (dst < glim[gg])
         out_pairs[dst] = row; /* plain store: partial lines must merge in L2 */
       else
         s_ovf = 1;
@@ -1424,7 +1433,7 @@ void bucket_partition2_slack(const int64_t* d_keys, const int64_t* d_pay, int64_
                    (int64_t)B * capB < (int64_t)UINT32_MAX,
                  "bucket_partition_slack: slack layout exceeds u32 row index");
   size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 4 * (size_t)PA * sizeof(uint32_t) + 64;
-  size_t sub_lds = SCATTER_TILE * sizeof(longlong2) + 3 * (size_t)F * sizeof(uint32_t) + 64;
+  size_t sub_lds = BTILE * sizeof(longlong2) + 4 * (size_t)F * sizeof(uint32_t) + 64;
   DJ_HIP_CALL(hipMemsetAsync(d_cursors, 0, (size_t)PA * 4, s));
   hipLaunchKernelGGL(bucket_scatter_slack_kernel, dim3(BUCKET_BLOCKS), dim3(BUCKET_THREADS),
                      scatter_lds, s, d_keys, d_pay, n, PA, capA, d_cursors, d_any_overflow,
@@ -1509,11 +1518,14 @@ static void lds_join_launch(const longlong2* d_lrows, const int64_t* d_loff,
 {
   DJ_CHECK_ERROR(table_slots == 2048 || table_slots == 4096,
                  "lds_join: table_slots must be 2048 or 4096");
-  int64_t groups = ((int64_t)B + 3) / 4; /* KBUK = 4 */
+  /* slack path: KBUK = 4 (B padded to a multiple of 4 by the caller);
+   * compact path: KBUK = 1 (no padding contract, flush per bucket) */
+  const bool slack = d_llen != nullptr;
+  const int kbuk = slack ? 4 : 1;
+  int64_t groups = ((int64_t)B + kbuk - 1) / kbuk;
   int grid = (int)(groups < 8192 ? groups : 8192);
   size_t lds =
     (size_t)table_slots * sizeof(longlong2) + 4 * JOIN_STAGE_ROWS * sizeof(int64_t) + 16;
-  const bool slack = d_llen != nullptr;
   auto launch = [&](auto kern) {
     hipLaunchKernelGGL(kern, dim3(grid), dim3(BUCKET_THREADS), lds, s, d_lrows, d_loff,
                        d_llen, capL, d_rrows, d_roff, d_rlen, capR, B, d_out0, d_out1,
@@ -1521,9 +1533,9 @@ static void lds_join_launch(const longlong2* d_lrows, const int64_t* d_loff,
                        d_any_overflow, d_error);
   };
   if (table_slots == 4096)
-    slack ? launch(lds_join_kernel<4096, true>) : launch(lds_join_kernel<4096, false>);
+    slack ? launch(lds_join_kernel<4096, true, 4>) : launch(lds_join_kernel<4096, false, 1>);
   else
-    slack ? launch(lds_join_kernel<2048, true>) : launch(lds_join_kernel<2048, false>);
+    slack ? launch(lds_join_kernel<2048, true, 4>) : launch(lds_join_kernel<2048, false, 1>);
   DJ_HIP_CALL(hipGetLastError());
 }
 
@@ -1544,6 +1556,9 @@ void lds_join_slack(const longlong2* d_lrows, const uint32_t* d_llen, int64_t ca
                     uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error,
                     hipStream_t s)
 {
+  DJ_CHECK_ERROR(B % 4 == 0,
+                 "lds_join_slack: B must be padded to a multiple of 4 (zero-length "
+                 "buckets) — see lds_join_kernel's KBUK contract");
   lds_join_launch(d_lrows, nullptr, d_llen, capL, d_rrows, nullptr, d_rlen, capR, B,
                   table_slots, d_out0, d_out1, d_out2, d_out3, cap, d_counter,
                   d_overflow_flags, d_any_overflow, d_error, s);

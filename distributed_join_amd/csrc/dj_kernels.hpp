@@ -141,7 +141,9 @@ void lds_join(const longlong2* d_lrows, const int64_t* d_loff, const longlong2* 
               int64_t* d_out2, int64_t* d_out3, int64_t cap, int64_t* d_counter,
               uint32_t* d_overflow_flags, int* d_any_overflow, int* d_error, hipStream_t s);
 /* Same fused join over the slack bucket layout of bucket_partition2_slack:
- * bucket b = d_lrows[b*capL .. +d_llen[b]) x d_rrows[b*capR .. +d_rlen[b]). */
+ * bucket b = d_lrows[b*capL .. +d_llen[b]) x d_rrows[b*capR .. +d_rlen[b]).
+ * Contract: B must be a multiple of 4 (the in-kernel flush-group size) —
+ * pad d_llen/d_rlen with zero-length buckets; errors loudly otherwise. */
 void lds_join_slack(const longlong2* d_lrows, const uint32_t* d_llen, int64_t capL,
                     const longlong2* d_rrows, const uint32_t* d_rlen, int64_t capR, int B,
                     int table_slots, int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
