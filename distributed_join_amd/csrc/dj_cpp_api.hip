@@ -1637,7 +1637,7 @@ std::unique_ptr<cudf::table> distributed_inner_join(
   CommunicationGroup group(nvl, 1, communicator->mpi_rank);
   const int nparts = G * over_decom_factor;
   DJ_CHECK_ERROR(nparts <= dj::kMaxPartitions,
-                 "join-group size x over_decom_factor must be <= 64");
+                 "join-group size x over_decom_factor must be <= 1024");
 
   /* fused wire path for the hot shape (2 x INT64 columns, key at 0):
    * partitions carry the bucket grouping on the wire, removing the separate

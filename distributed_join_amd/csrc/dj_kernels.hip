@@ -163,6 +163,12 @@ __device__ __forceinline__ uint32_t part_of(int64_t key, int hash_fn, uint32_t s
   return dj_row_hash(key, hash_fn, seed) % (uint32_t)nparts;
 }
 
+/* NPL = ceil(nparts/64): lane q owns partitions q, q+64, ..., q+(NPL-1)*64.
+ * One ballot per partition id per 64-row batch keeps the scatter stable and
+ * atomic-free at any nparts <= 64*NPL (the reference's own tests drive
+ * nparts = ranks x over_decom up to 80, compare_against_single_gpu.cu:237);
+ * cost grows linearly in nparts, which only large worlds x od ever pay. */
+template <int NPL>
 __global__ void part_count_kernel(const int64_t* __restrict__ keys, int64_t n, int nparts,
                                   int hash_fn, uint32_t seed, int64_t rows_per_wave,
                                   int64_t* __restrict__ wave_counts)
@@ -171,17 +177,27 @@ __global__ void part_count_kernel(const int64_t* __restrict__ keys, int64_t n, i
   const int64_t w = (int64_t)blockIdx.x * WPB + (threadIdx.x >> 6);
   const int64_t start = w * rows_per_wave;
   const int64_t end = min(start + rows_per_wave, n);
-  int64_t my_count = 0;  // lane q counts partition q
+  int64_t my_count[NPL];  // lane q counts partition c*64 + q
+#pragma unroll
+  for (int c = 0; c < NPL; c++) my_count[c] = 0;
   for (int64_t base = start; base < end; base += WAVE) {
     int64_t i = base + lane;
     uint32_t p = 0xFFFFFFFFu;
     if (i < end) p = part_of(keys[i], hash_fn, seed, nparts);
-    for (int q = 0; q < nparts; q++) {
-      uint64_t m = __ballot(p == (uint32_t)q);
-      if (lane == q) my_count += __popcll(m);
+#pragma unroll
+    for (int c = 0; c < NPL; c++) {
+      const int qhi = min(WAVE, nparts - c * WAVE);  // uniform
+      for (int j = 0; j < qhi; j++) {
+        uint64_t m = __ballot(p == (uint32_t)(c * WAVE + j));
+        if (lane == j) my_count[c] += __popcll(m);
+      }
     }
   }
-  if (lane < nparts) wave_counts[w * nparts + lane] = my_count;
+#pragma unroll
+  for (int c = 0; c < NPL; c++) {
+    const int q = c * WAVE + lane;
+    if (q < nparts) wave_counts[w * nparts + q] = my_count[c];
+  }
 }
 
 /* grid = nparts blocks; block p scans column p over nwaves (exclusive),
@@ -226,6 +242,7 @@ __global__ void part_offsets_kernel(const int64_t* totals, int nparts, int64_t* 
   }
 }
 
+template <int NPL>
 __global__ void part_scatter_kernel(const int64_t* __restrict__ keys,
                                     const int64_t* __restrict__ pay, int64_t n, int nparts,
                                     int hash_fn, uint32_t seed, int64_t rows_per_wave,
@@ -238,9 +255,13 @@ __global__ void part_scatter_kernel(const int64_t* __restrict__ keys,
   const int64_t w = (int64_t)blockIdx.x * WPB + (threadIdx.x >> 6);
   const int64_t start = w * rows_per_wave;
   const int64_t end = min(start + rows_per_wave, n);
-  /* lane q holds the cursor for partition q */
-  int64_t cursor = 0;
-  if (lane < nparts) cursor = offsets[lane] + wave_prefix[w * nparts + lane];
+  /* lane q holds the cursor for partitions c*64 + q */
+  int64_t cursor[NPL];
+#pragma unroll
+  for (int c = 0; c < NPL; c++) {
+    const int q = c * WAVE + lane;
+    cursor[c] = (q < nparts) ? offsets[q] + wave_prefix[w * nparts + q] : 0;
+  }
   const uint64_t lt_mask = (lane == 0) ? 0ull : (~0ull >> (64 - lane));
   for (int64_t base = start; base < end; base += WAVE) {
     int64_t i = base + lane;
@@ -252,32 +273,60 @@ __global__ void part_scatter_kernel(const int64_t* __restrict__ keys,
       v = pay ? pay[i] : 0;
       p = part_of(k, hash_fn, seed, nparts);
     }
-    uint64_t my_q_mask = 0;  // lane q's ballot mask for partition q
-    for (int q = 0; q < nparts; q++) {
-      uint64_t m = __ballot(p == (uint32_t)q);
-      if (lane == q) my_q_mask = m;
+    uint64_t my_q_mask[NPL];  // lane q's ballot mask for partition c*64 + q
+#pragma unroll
+    for (int c = 0; c < NPL; c++) {
+      my_q_mask[c] = 0;
+      const int qhi = min(WAVE, nparts - c * WAVE);  // uniform
+      for (int j = 0; j < qhi; j++) {
+        uint64_t m = __ballot(p == (uint32_t)(c * WAVE + j));
+        if (lane == j) my_q_mask[c] = m;
+      }
     }
-    /* mask and cursor of MY partition, fetched from lane p */
-    uint64_t m_p = __shfl((unsigned long long)my_q_mask, (int)(valid ? p : 0));
-    int64_t base_dst = __shfl(cursor, (int)(valid ? p : 0));
+    /* mask and cursor of MY partition, fetched from lane p%64, slot p/64 */
+    const int src = (int)(valid ? (p & (WAVE - 1)) : 0);
+    uint64_t m_p = 0;
+    int64_t base_dst = 0;
+#pragma unroll
+    for (int c = 0; c < NPL; c++) {
+      uint64_t mc = __shfl((unsigned long long)my_q_mask[c], src);
+      int64_t bc = __shfl(cursor[c], src);
+      if (NPL == 1 || (valid && (p >> 6) == (uint32_t)c)) {
+        m_p = mc;
+        base_dst = bc;
+      }
+    }
     if (valid) {
       int rank = __popcll(m_p & lt_mask);
       int64_t dst = base_dst + rank;
       out_keys[dst] = k;
       if (out_pay) out_pay[dst] = v;
     }
-    cursor += __popcll(my_q_mask);
+#pragma unroll
+    for (int c = 0; c < NPL; c++) cursor[c] += __popcll(my_q_mask[c]);
   }
 }
 
 void partition_count(const int64_t* d_keys, int64_t n, int nparts, int hash_fn,
                      uint32_t hash_seed, void* d_scratch, hipStream_t s)
 {
-  DJ_CHECK_ERROR(nparts >= 1 && nparts <= kMaxPartitions, "nparts must be in [1,64]");
+  DJ_CHECK_ERROR(nparts >= 1 && nparts <= kMaxPartitions, "nparts must be in [1,1024]");
   if (n <= 0) return;
   PartGeom g = part_geom(n);
-  hipLaunchKernelGGL(part_count_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, n, nparts,
-                     hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, n, nparts, hash_fn,
+                       hash_seed, g.rows_per_wave, (int64_t*)d_scratch);
+  };
+  if (nparts <= 64)
+    launch(part_count_kernel<1>);
+  else if (nparts <= 128)
+    launch(part_count_kernel<2>);
+  else if (nparts <= 256)
+    launch(part_count_kernel<4>);
+  else if (nparts <= 512)
+    launch(part_count_kernel<8>);
+  else
+    launch(part_count_kernel<16>);
   DJ_HIP_CALL(hipGetLastError());
 }
 
@@ -303,9 +352,21 @@ void partition_scatter(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
 {
   if (n <= 0) return;
   PartGeom g = part_geom(n);
-  hipLaunchKernelGGL(part_scatter_kernel, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, d_pay, n,
-                     nparts, hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch,
-                     d_offsets, d_out_keys, d_out_pay);
+  auto launch = [&](auto kern) {
+    hipLaunchKernelGGL(kern, dim3(g.blocks), dim3(BLOCK), 0, s, d_keys, d_pay, n, nparts,
+                       hash_fn, hash_seed, g.rows_per_wave, (int64_t*)d_scratch, d_offsets,
+                       d_out_keys, d_out_pay);
+  };
+  if (nparts <= 64)
+    launch(part_scatter_kernel<1>);
+  else if (nparts <= 128)
+    launch(part_scatter_kernel<2>);
+  else if (nparts <= 256)
+    launch(part_scatter_kernel<4>);
+  else if (nparts <= 512)
+    launch(part_scatter_kernel<8>);
+  else
+    launch(part_scatter_kernel<16>);
   DJ_HIP_CALL(hipGetLastError());
 }
 

@@ -24,7 +24,11 @@ namespace dj {
  * rejected (loud error), see dj_kernels.hip. Set via 0xFF memset. */
 constexpr int64_t kEmptyKey = -1;
 
-constexpr int kMaxPartitions = 64;  // nparts = world_size x over_decom <= 64
+/* nparts = world_size x over_decom; the stable wave-ballot partition covers
+ * any nparts <= 1024 (lane q owns partitions q, q+64, ... — cost linear in
+ * nparts). The reference's own tests drive up to 80 (8 ranks x od 10,
+ * compare_against_single_gpu.cu:237-268). */
+constexpr int kMaxPartitions = 1024;
 
 /* ----- synthetic inputs (deterministic, spec in dj_rng.h) ----- */
 void generate_build(int64_t* d_keys, int64_t* d_pay, int64_t n_global, int64_t rand_max,

@@ -51,6 +51,10 @@ def test_generator_parity_slice(dj):
 @pytest.mark.parametrize("nparts,hash_fn,seed", [
     (1, 0, 12345678), (2, 0, 12345678), (8, 0, 12345678), (8, 0, 87654321),
     (13, 0, 0), (64, 0, 12345678), (8, 1, 0),
+    # beyond one wave's 64 lanes: the reference's own test envelope reaches
+    # 8 ranks x over_decom 10 = 80 (compare_against_single_gpu.cu:237-268);
+    # 100/256/1024 cover the multi-chunk dispatch tiers incl. the cap
+    (80, 0, 12345678), (100, 0, 0), (256, 0, 12345678), (1024, 1, 0),
 ])
 def test_partition_parity(dj, nparts, hash_fn, seed):
     n = 1_000_000
@@ -118,6 +122,20 @@ def test_join_parity_duplicates(dj):
     rk = rng.randint(0, 1000, n).astype(np.int64)
     rp = np.arange(n, dtype=np.int64)
     # ~n*n/1000 = 10M output rows
+    _join_parity(dj, lk, lp, rk, rp)
+
+
+def test_join_sparse_buckets_drain(dj):
+    # ~100 distinct keys over B=256 buckets: most buckets empty, so many
+    # 4-bucket flush groups end on an empty bucket while earlier buckets
+    # staged matches under the watermark — exercises the join kernel's
+    # goto-drain path (the exit-clean guarantee; see lds_join_kernel)
+    n = 2_000
+    rng = np.random.RandomState(11)
+    lk = rng.randint(0, 100, n).astype(np.int64)
+    lp = np.arange(n, dtype=np.int64)
+    rk = rng.randint(0, 100, n).astype(np.int64)
+    rp = np.arange(n, dtype=np.int64)
     _join_parity(dj, lk, lp, rk, rp)
 
 
