@@ -81,6 +81,7 @@ def lib():
             "dj_all_to_all_i64": ([vp, vp, vp, vp], None),
             "dj_exchange_sizes": ([vp, vp], None),
             "dj_rccl_selftest": ([i64], i32),
+            "dj_compress_roundtrip": ([vp, i64, i32, i32, i32, i32, vp], i64),
             "dj_cpp_comm_create": ([i32, i32, vp], vp),
             "dj_cpp_comm_destroy": ([vp], None),
             "dj_cpp_distributed_inner_join_i64": ([vp, vp, vp, i64, vp, vp, i64, i32, i32], vp),

@@ -604,4 +604,29 @@ void dj_exchange_sizes(const int64_t* h_send_counts, int64_t* h_recv_counts)
   dj_dfree(d_all);
 }
 
+/* cascaded codec roundtrip (test hook): compress d_in (count elements of
+ * elem_size 4/8) with the given cascaded passes, decompress into d_out, and
+ * return the wire size in bytes (header included). Pins the codec's
+ * compressed == uncompressed semantics per scheme (the reference pins
+ * nvcomp only through end-to-end equality, compare_against_analytical
+ * runs with compression; our wire format is parity-unpinned). */
+int64_t dj_compress_roundtrip(const void* d_in, int64_t count, int elem_size, int num_rles,
+                              int num_deltas, int use_bp, void* d_out)
+{
+  hipStream_t st = stream();
+  void* comp = dj_dmalloc((int64_t)dj::compress_bound(count, elem_size));
+  void* scratch = dj_dmalloc((int64_t)dj::compress_scratch_bytes(count));
+  dj::compress_slice_async(d_in, count, elem_size, num_rles, num_deltas, use_bp,
+                           (uint8_t*)comp, scratch, st);
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  dj::CompSliceHeader h;
+  DJ_HIP_CALL(hipMemcpy(&h, comp, sizeof(h), hipMemcpyDeviceToHost));
+  int64_t wire = (int64_t)dj::compressed_size_from_header(h, elem_size);
+  dj::decompress_slice_async((const uint8_t*)comp, h, elem_size, d_out, scratch, st);
+  DJ_HIP_CALL(hipStreamSynchronize(st));
+  dj_dfree(comp);
+  dj_dfree(scratch);
+  return wire;
+}
+
 }  // extern "C"

@@ -6,22 +6,32 @@
  * The compressed WIRE FORMAT is ours (the reference's nvcomp format is
  * parity-unpinned — no reference test fixes it):
  *
- *   slice := header(16 B) || payload
- *   header: u32 bits      — packed bit width (0..64); 0xFFFF = stored raw
- *           u32 scheme    — bit0: delta applied
- *           u64 count     — element count
- *   payload (bitpacked): groups of 32 zigzag(u64) values, each group packed
- *     into ceil(32*bits/32) u32 words (group-aligned for fully parallel
- *     pack/unpack); last group zero-padded.
- *   payload (raw): count * elem_size bytes verbatim (used when packing
- *     would not shrink the slice — the codec never expands beyond +16 B).
+ *   slice := header(32 B, CompSliceHeader) || payload
+ *   header: u32 bits      — value bit width (0..64); 0xFFFF = stored raw
+ *           u32 scheme    — bit0: delta applied, bit1: RLE applied
+ *           u64 count     — original element count
+ *           u64 nruns     — RLE only: number of runs
+ *           u32 len_bits  — RLE only: run-length bit width
+ *           u32 reserved
+ *   bitpacked subslice (count N, width W): groups of 32 zigzag(u64) values,
+ *     each group packed into ceil(32*W/32) u32 words (group-aligned for
+ *     fully parallel pack/unpack); last group zero-padded.
+ *   payload (no RLE, packed): one subslice over the (optionally delta'd)
+ *     values, N = count, W = bits.
+ *   payload (RLE): run-lengths subslice (N = nruns, W = len_bits) followed
+ *     by run-values subslice (N = nruns, W = bits; delta per scheme bit0 —
+ *     the nvcomp cascaded order: RLE first, delta on the run values).
+ *   payload (raw): count * elem_size bytes verbatim (used whenever the
+ *     encoded form would not shrink the slice — the codec never expands
+ *     beyond the +32 B header).
  *
  * Cascaded options mapping (ColumnCompressionOptions.cascaded_format):
- *   num_deltas ∈ {0,1}: delta-encode before packing (decode = inclusive
- *     scan); use_bp: bitpack. num_RLEs > 0 is not implemented (throws at
- *     option validation in dj_cpp_api.hip).
+ *   num_RLEs ∈ {0,1}, num_deltas ∈ {0,1}, use_bp ∈ {0,1}; >1 passes raise
+ *   at option validation (dj_cpp_api.hip). Without use_bp the subslices
+ *   pack at width 64 (the raw fallback then usually wins on-device).
  * INT32 inputs are widened to i64 values before zigzag (they pack to their
- * natural width anyway).
+ * natural width anyway). LZ4 is refused at validation: the wire targets
+ * xGMI, where cascaded already rarely pays.
  */
 #include "dj_error.hpp"
 #include "dj_kernels.hpp"
@@ -33,6 +43,37 @@ namespace dj {
 namespace {
 constexpr int CBLOCK = 256;
 constexpr uint32_t RAW_BITS = 0xFFFFu;
+constexpr int HDR = 32;  // sizeof(CompSliceHeader)
+
+/* scratch layout (compress_scratch_bytes): cells then three i64[n] arrays
+ * then scan partials */
+struct CompScratch {
+  unsigned long long* nruns;  // [0]
+  uint32_t* vbits;            // [8]
+  uint32_t* lbits;            // [12]
+  int64_t* vals;              // run values (RLE) / unpacked values (decomp)
+  int64_t* starts;            // run starts (RLE; starts[nruns] = count) /
+                              // run offsets (decomp)
+  int64_t* marks;             // marks -> inclusive run index (compress) /
+                              // unpacked lengths (decomp)
+  int64_t* partials;
+};
+
+constexpr int SCHUNK = 2048;
+
+__host__ __device__ inline CompScratch carve_comp_scratch(void* p, int64_t n)
+{
+  CompScratch s;
+  char* c = (char*)p;
+  s.nruns = (unsigned long long*)c;
+  s.vbits = (uint32_t*)(c + 8);
+  s.lbits = (uint32_t*)(c + 12);
+  s.vals = (int64_t*)(c + 32);
+  s.starts = s.vals + (n + 1);
+  s.marks = s.starts + (n + 1);
+  s.partials = s.marks + (n + 1);
+  return s;
+}
 
 int cgrid(int64_t n)
 {
@@ -55,6 +96,13 @@ template <typename T>
 __device__ __forceinline__ int64_t load_elem(const void* p, int64_t i)
 {
   return (int64_t)((const T*)p)[i];
+}
+
+__host__ __device__ inline size_t packed_bytes(int64_t n, uint32_t bits)
+{
+  int wpg = (32 * (int)bits + 31) / 32;
+  int64_t ngroups = (n + 31) / 32;
+  return (size_t)ngroups * wpg * 4;
 }
 
 }  // namespace
@@ -90,8 +138,7 @@ __global__ void comp_maxbits_kernel(const void* __restrict__ in, int64_t n, int 
   }
 }
 
-/* header writer: decides packed-vs-raw, writes the 16 B header, finalizes
- * *bits_p for the payload kernels and the host */
+/* header writer (no RLE): decides packed-vs-raw, writes the 32 B header */
 __global__ void comp_header_kernel(int64_t count, int elem_size, int delta, int use_bp,
                                    uint8_t* out, uint32_t* bits_p)
 {
@@ -99,12 +146,10 @@ __global__ void comp_header_kernel(int64_t count, int elem_size, int delta, int 
   uint32_t bits = RAW_BITS;
   if (use_bp) {
     bits = *bits_p;
-    if (bits == 0) bits = 1;  /* all-zero slices still need a width */
-    int wpg = (32 * (int)bits + 31) / 32;
-    int64_t ngroups = (count + 31) / 32;
-    if ((size_t)ngroups * wpg * 4 >= (size_t)count * elem_size) bits = RAW_BITS;
+    if (bits == 0) bits = 1; /* all-zero slices still need a width */
+    if (packed_bytes(count, bits) >= (size_t)count * elem_size) bits = RAW_BITS;
   }
-  CompSliceHeader h;
+  CompSliceHeader h{};
   h.bits = bits;
   h.scheme = delta ? 1u : 0u;
   h.count = (uint64_t)count;
@@ -112,28 +157,29 @@ __global__ void comp_header_kernel(int64_t count, int elem_size, int delta, int 
   *bits_p = bits;
 }
 
-/* raw fallback payload (guarded on the final bits decision) */
+/* raw fallback payload (guarded on the final header decision) */
 __global__ void comp_rawcopy_kernel(const void* __restrict__ in, int64_t count, int elem_size,
-                                    uint8_t* __restrict__ out, const uint32_t* bits_p)
+                                    uint8_t* __restrict__ out)
 {
-  if (*bits_p != RAW_BITS) return;
+  if (((const CompSliceHeader*)out)->bits != RAW_BITS) return;
   const uint8_t* src = (const uint8_t*)in;
-  uint8_t* dst = out + 16;
+  uint8_t* dst = out + HDR;
   int64_t total = count * elem_size;
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < total; i += stride) dst[i] = src[i];
 }
 
-/* packed payload (guarded) */
+/* packed payload for the plain (no-RLE) path (guarded) */
 __global__ void comp_pack_kernel(const void* __restrict__ in, int64_t count, int elem_size,
-                                 int delta, uint8_t* __restrict__ out, const uint32_t* bits_p)
+                                 int delta, uint8_t* __restrict__ out)
 {
-  const uint32_t bits = *bits_p;
-  if (bits == RAW_BITS || bits == 0) return;
+  const CompSliceHeader h = *(const CompSliceHeader*)out;
+  const uint32_t bits = h.bits;
+  if (bits == RAW_BITS || bits == 0 || (h.scheme & 2u)) return;
   const int wpg = (32 * (int)bits + 31) / 32;
   int64_t ngroups = (count + 31) / 32;
-  uint32_t* words = (uint32_t*)(out + 16);
+  uint32_t* words = (uint32_t*)(out + HDR);
   int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; g < ngroups; g += stride) {
@@ -160,7 +206,51 @@ __global__ void comp_pack_kernel(const void* __restrict__ in, int64_t count, int
   }
 }
 
-/* unpack into i64 deltas/values */
+/* generic bitpack of i64 values given by a functor-free source array,
+ * width from a header field, guarded by a predicate on the header */
+__global__ void comp_pack_i64_kernel(const int64_t* __restrict__ src, const uint8_t* hdr_out,
+                                     int64_t payload_off_mode /*0: lengths, 1: values*/,
+                                     int delta, uint8_t* __restrict__ out)
+{
+  const CompSliceHeader h = *(const CompSliceHeader*)hdr_out;
+  if (h.bits == RAW_BITS || !(h.scheme & 2u)) return;
+  const int64_t n = (int64_t)h.nruns;
+  const uint32_t bits = payload_off_mode == 0 ? h.len_bits : h.bits;
+  if (bits == 0) return;
+  size_t off = HDR;
+  if (payload_off_mode == 1) off += packed_bytes(n, h.len_bits);
+  const int wpg = (32 * (int)bits + 31) / 32;
+  int64_t ngroups = (n + 31) / 32;
+  uint32_t* words = (uint32_t*)(out + off);
+  int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; g < ngroups; g += stride) {
+    uint32_t* w = words + g * wpg;
+    for (int k = 0; k < wpg; k++) w[k] = 0;
+    for (int j = 0; j < 32; j++) {
+      int64_t i = g * 32 + j;
+      if (i >= n) break;
+      int64_t v = src[i];
+      if (payload_off_mode == 0) {
+        v = src[i + 1] - src[i];  // length r = starts[r+1] - starts[r]
+      } else if (delta && i > 0) {
+        v -= src[i - 1];
+      }
+      uint64_t z = zigzag(v);
+      int64_t bitpos = (int64_t)j * bits;
+      int word = (int)(bitpos >> 5);
+      int off2 = (int)(bitpos & 31);
+      w[word] |= (uint32_t)(z << off2);
+      if (off2 + (int)bits > 32) {
+        uint64_t rest = z >> (32 - off2);
+        w[word + 1] |= (uint32_t)rest;
+        if (off2 + (int)bits > 64) w[word + 2] |= (uint32_t)(rest >> 32);
+      }
+    }
+  }
+}
+
+/* unpack a bitpacked subslice into i64 values */
 __global__ void comp_unpack_kernel(const uint32_t* __restrict__ words, uint32_t bits,
                                    int64_t n, int64_t* __restrict__ out)
 {
@@ -186,7 +276,7 @@ __global__ void comp_unpack_kernel(const uint32_t* __restrict__ words, uint32_t 
   }
 }
 
-/* narrow i64 -> T with optional inclusive-scan-free direct store (no delta) */
+/* narrow i64 -> T direct store (no delta) */
 template <typename T>
 __global__ void comp_store_kernel(const int64_t* __restrict__ vals, int64_t n,
                                   void* __restrict__ out)
@@ -196,11 +286,117 @@ __global__ void comp_store_kernel(const int64_t* __restrict__ vals, int64_t n,
   for (; i < n; i += stride) ((T*)out)[i] = (T)vals[i];
 }
 
-/* ---- i64 inclusive scan (delta decode) — 3-kernel chunked scan ---- */
+/* ---------------- RLE compress kernels ---------------- */
 
-namespace {
-constexpr int SCHUNK = 2048;
+/* marks[i] = 1 iff row i starts a new run */
+template <typename T>
+__global__ void rle_marks_kernel(const void* __restrict__ in, int64_t n,
+                                 int64_t* __restrict__ marks)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride)
+    marks[i] = (i == 0 || load_elem<T>(in, i) != load_elem<T>(in, i - 1)) ? 1 : 0;
 }
+
+/* after the chunked exclusive scan of marks (partials holds chunk-exclusive
+ * bases): emit run values/starts, nruns, and starts[nruns] = n */
+template <typename T>
+__global__ void rle_emit_kernel(const void* __restrict__ in, int64_t n, int64_t nchunks,
+                                const int64_t* __restrict__ marks,
+                                const int64_t* __restrict__ partials,
+                                int64_t* __restrict__ vals, int64_t* __restrict__ starts,
+                                unsigned long long* __restrict__ nruns)
+{
+  int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; c < nchunks; c += stride) {
+    int64_t s = c * SCHUNK, e = min(s + (int64_t)SCHUNK, n);
+    int64_t acc = partials[c];
+    for (int64_t i = s; i < e; i++) {
+      if (marks[i]) {
+        vals[acc] = load_elem<T>(in, i);
+        starts[acc] = i;
+        acc++;
+      }
+      if (i == n - 1) {
+        *nruns = (unsigned long long)acc;
+        starts[acc] = n;
+      }
+    }
+  }
+}
+
+/* max zigzag width over run values (optionally delta'd) and run lengths;
+ * count read from d_nruns (device-only value) */
+__global__ void rle_maxbits_kernel(const int64_t* __restrict__ vals,
+                                   const int64_t* __restrict__ starts,
+                                   const unsigned long long* __restrict__ d_nruns, int delta,
+                                   uint32_t* __restrict__ vbits, uint32_t* __restrict__ lbits)
+{
+  __shared__ uint64_t redv[CBLOCK], redl[CBLOCK];
+  const int64_t n = (int64_t)*d_nruns;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  uint64_t mv = 0, ml = 0;
+  for (; i < n; i += stride) {
+    int64_t v = vals[i];
+    if (delta && i > 0) v -= vals[i - 1];
+    mv |= zigzag(v);
+    ml |= zigzag(starts[i + 1] - starts[i]);
+  }
+  redv[threadIdx.x] = mv;
+  redl[threadIdx.x] = ml;
+  __syncthreads();
+  for (int off = CBLOCK / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      redv[threadIdx.x] |= redv[threadIdx.x + off];
+      redl[threadIdx.x] |= redl[threadIdx.x + off];
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) {
+    uint64_t m = redv[0];
+    uint32_t b = 0;
+    while (m) {
+      b++;
+      m >>= 1;
+    }
+    atomicMax(vbits, b);
+    m = redl[0];
+    b = 0;
+    while (m) {
+      b++;
+      m >>= 1;
+    }
+    atomicMax(lbits, b);
+  }
+}
+
+/* header writer (RLE): decides encoded-vs-raw on device */
+__global__ void rle_header_kernel(int64_t count, int elem_size, int delta, int use_bp,
+                                  const unsigned long long* __restrict__ d_nruns,
+                                  const uint32_t* __restrict__ vbits_p,
+                                  const uint32_t* __restrict__ lbits_p, uint8_t* out)
+{
+  if (blockIdx.x != 0 || threadIdx.x != 0) return;
+  const int64_t nruns = (int64_t)*d_nruns;
+  uint32_t vbits = use_bp ? *vbits_p : 64;
+  uint32_t lbits = use_bp ? *lbits_p : 64;
+  if (vbits == 0) vbits = 1;
+  if (lbits == 0) lbits = 1;
+  CompSliceHeader h{};
+  h.count = (uint64_t)count;
+  h.scheme = 2u | (delta ? 1u : 0u);
+  h.nruns = (uint64_t)nruns;
+  h.len_bits = lbits;
+  h.bits = vbits;
+  size_t enc = packed_bytes(nruns, lbits) + packed_bytes(nruns, vbits);
+  if (enc >= (size_t)count * elem_size) h.bits = RAW_BITS;
+  *(CompSliceHeader*)out = h;
+}
+
+/* ---- i64 inclusive/exclusive scan helpers — 3-kernel chunked scan ---- */
 
 __global__ void scan64_partials_kernel(const int64_t* __restrict__ v, int64_t n,
                                        int64_t nchunks, int64_t* __restrict__ partials)
@@ -257,66 +453,152 @@ __global__ void scan64_finalize_kernel(const int64_t* __restrict__ v, int64_t n,
   }
 }
 
+/* exclusive-scan finalize: out[i] = sum(v[0..i)) as i64 (run offsets) */
+__global__ void scan64_excl_finalize_kernel(const int64_t* __restrict__ v, int64_t n,
+                                            int64_t nchunks,
+                                            const int64_t* __restrict__ partials,
+                                            int64_t* __restrict__ out)
+{
+  int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; c < nchunks; c += stride) {
+    int64_t s = c * SCHUNK, e = min(s + (int64_t)SCHUNK, n);
+    int64_t acc = partials[c];
+    for (int64_t i = s; i < e; i++) {
+      out[i] = acc;
+      acc += v[i];
+    }
+  }
+}
+
+/* expand runs: out[i] = vals[r] where offsets[r] <= i < offsets[r+1]
+ * (binary search; offsets has nruns entries, monotonically increasing) */
+template <typename T>
+__global__ void rle_expand_kernel(const int64_t* __restrict__ vals,
+                                  const int64_t* __restrict__ offsets, int64_t nruns,
+                                  int64_t n, void* __restrict__ out)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int64_t lo = 0, hi = nruns - 1;
+    while (lo < hi) {
+      int64_t mid = (lo + hi + 1) >> 1;
+      if (offsets[mid] <= i)
+        lo = mid;
+      else
+        hi = mid - 1;
+    }
+    ((T*)out)[i] = (T)vals[lo];
+  }
+}
+
 /* ---------------- host launchers ---------------- */
 
 size_t compress_bound(int64_t count, int elem_size)
 {
-  /* header + worst case raw */
-  return 16 + (size_t)count * elem_size + 16;
+  /* header + worst case raw (+ slack for the last zero-padded pack group) */
+  return HDR + (size_t)count * elem_size + 32;
 }
 
 size_t compress_scratch_bytes(int64_t count)
 {
-  int64_t nchunks = (count + SCHUNK - 1) / SCHUNK;
-  /* unpacked i64 values + scan partials */
-  return (size_t)count * 8 + (size_t)(nchunks > 0 ? nchunks : 1) * 8 + 64;
+  int64_t n = count > 0 ? count : 1;
+  int64_t nchunks = (n + SCHUNK - 1) / SCHUNK;
+  return 32 + 3 * (size_t)(n + 1) * 8 + (size_t)nchunks * 8 + 64;
 }
 
 /*
- * Compress one slice. d_out must hold compress_bound() bytes; d_bits_tmp is
- * a device u32 (zeroed by this call). Returns stream-ordered; the actual
- * compressed size is written to h_size AFTER the stream syncs (caller syncs
- * once for all slices, then reads).
+ * Compress one slice (stream-ordered; no host sync). d_out must hold
+ * compress_bound() bytes; d_scratch compress_scratch_bytes(count) bytes.
+ * After the caller syncs the stream, the 32 B CompSliceHeader at d_out is
+ * final and compressed_size_from_header() gives the wire size.
  */
-void compress_slice_async(const void* d_in, int64_t count, int elem_size, int num_deltas,
-                          int use_bp, uint8_t* d_out, uint32_t* d_bits_tmp,
+void compress_slice_async(const void* d_in, int64_t count, int elem_size, int num_rles,
+                          int num_deltas, int use_bp, uint8_t* d_out, void* d_scratch,
                           hipStream_t s)
 {
   DJ_CHECK_ERROR(elem_size == 4 || elem_size == 8, "cascaded: 4/8-byte elements only");
-  DJ_HIP_CALL(hipMemsetAsync(d_bits_tmp, 0, 4, s));
+  CompScratch scr = carve_comp_scratch(d_scratch, count);
+  DJ_HIP_CALL(hipMemsetAsync(d_scratch, 0, 16, s));
+  if (num_rles > 0 && count > 0) {
+    const int64_t nchunks = (count + SCHUNK - 1) / SCHUNK;
+    if (elem_size == 8)
+      hipLaunchKernelGGL((rle_marks_kernel<int64_t>), dim3(cgrid(count)), dim3(CBLOCK), 0, s,
+                         d_in, count, scr.marks);
+    else
+      hipLaunchKernelGGL((rle_marks_kernel<int32_t>), dim3(cgrid(count)), dim3(CBLOCK), 0, s,
+                         d_in, count, scr.marks);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(scan64_partials_kernel, dim3(cgrid(nchunks)), dim3(CBLOCK), 0, s,
+                       scr.marks, count, nchunks, scr.partials);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(scan64_exclusive_kernel, dim3(1), dim3(1024), 0, s, scr.partials,
+                       nchunks);
+    DJ_HIP_CALL(hipGetLastError());
+    if (elem_size == 8)
+      hipLaunchKernelGGL((rle_emit_kernel<int64_t>), dim3(cgrid(nchunks)), dim3(CBLOCK), 0, s,
+                         d_in, count, nchunks, scr.marks, scr.partials, scr.vals, scr.starts,
+                         scr.nruns);
+    else
+      hipLaunchKernelGGL((rle_emit_kernel<int32_t>), dim3(cgrid(nchunks)), dim3(CBLOCK), 0, s,
+                         d_in, count, nchunks, scr.marks, scr.partials, scr.vals, scr.starts,
+                         scr.nruns);
+    DJ_HIP_CALL(hipGetLastError());
+    if (use_bp) {
+      hipLaunchKernelGGL(rle_maxbits_kernel, dim3(cgrid(count)), dim3(CBLOCK), 0, s, scr.vals,
+                         scr.starts, scr.nruns, num_deltas, scr.vbits, scr.lbits);
+      DJ_HIP_CALL(hipGetLastError());
+    }
+    hipLaunchKernelGGL(rle_header_kernel, dim3(1), dim3(64), 0, s, count, elem_size,
+                       num_deltas, use_bp, scr.nruns, scr.vbits, scr.lbits, d_out);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(comp_rawcopy_kernel, dim3(cgrid(count * elem_size)), dim3(CBLOCK), 0,
+                       s, d_in, count, elem_size, d_out);
+    DJ_HIP_CALL(hipGetLastError());
+    /* grids sized by count (upper bound on nruns); kernels bound by nruns */
+    hipLaunchKernelGGL(comp_pack_i64_kernel, dim3(cgrid((count + 31) / 32)), dim3(CBLOCK), 0,
+                       s, scr.starts, d_out, 0, num_deltas, d_out);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(comp_pack_i64_kernel, dim3(cgrid((count + 31) / 32)), dim3(CBLOCK), 0,
+                       s, scr.vals, d_out, 1, num_deltas, d_out);
+    DJ_HIP_CALL(hipGetLastError());
+    return;
+  }
   if (count > 0 && use_bp) {
     if (elem_size == 8)
       hipLaunchKernelGGL((comp_maxbits_kernel<int64_t>), dim3(cgrid(count)), dim3(CBLOCK), 0,
-                         s, d_in, count, num_deltas, d_bits_tmp);
+                         s, d_in, count, num_deltas, scr.vbits);
     else
       hipLaunchKernelGGL((comp_maxbits_kernel<int32_t>), dim3(cgrid(count)), dim3(CBLOCK), 0,
-                         s, d_in, count, num_deltas, d_bits_tmp);
+                         s, d_in, count, num_deltas, scr.vbits);
     DJ_HIP_CALL(hipGetLastError());
   }
   hipLaunchKernelGGL(comp_header_kernel, dim3(1), dim3(64), 0, s, count, elem_size,
-                     num_deltas, use_bp, d_out, d_bits_tmp);
+                     num_deltas, use_bp, d_out, scr.vbits);
   DJ_HIP_CALL(hipGetLastError());
   if (count > 0) {
-    hipLaunchKernelGGL(comp_rawcopy_kernel, dim3(cgrid(count * elem_size)), dim3(CBLOCK), 0, s,
-                       d_in, count, elem_size, d_out, d_bits_tmp);
+    hipLaunchKernelGGL(comp_rawcopy_kernel, dim3(cgrid(count * elem_size)), dim3(CBLOCK), 0,
+                       s, d_in, count, elem_size, d_out);
     DJ_HIP_CALL(hipGetLastError());
     hipLaunchKernelGGL(comp_pack_kernel, dim3(cgrid((count + 31) / 32)), dim3(CBLOCK), 0, s,
-                       d_in, count, elem_size, num_deltas, d_out, d_bits_tmp);
+                       d_in, count, elem_size, num_deltas, d_out);
     DJ_HIP_CALL(hipGetLastError());
   }
 }
 
-/* compressed size from the (synced) bits value */
-size_t compressed_size_from_bits(int64_t count, int elem_size, uint32_t bits)
+/* wire size from the final (synced) header */
+size_t compressed_size_from_header(const CompSliceHeader& h, int elem_size)
 {
-  if (bits == RAW_BITS) return 16 + (size_t)count * elem_size;
-  int wpg = (32 * (int)bits + 31) / 32;
-  int64_t ngroups = (count + 31) / 32;
-  return 16 + (size_t)ngroups * wpg * 4;
+  if (h.bits == RAW_BITS) return HDR + (size_t)h.count * elem_size;
+  if (h.scheme & 2u)
+    return HDR + packed_bytes((int64_t)h.nruns, h.len_bits) +
+           packed_bytes((int64_t)h.nruns, h.bits);
+  return HDR + packed_bytes((int64_t)h.count, h.bits);
 }
 
 /*
- * Decompress one slice (header read on host from h_header after recv) into
+ * Decompress one slice (header read on host from the received bytes) into
  * d_out (count elements of elem_size). d_scratch: compress_scratch_bytes.
  */
 void decompress_slice_async(const uint8_t* d_comp, const CompSliceHeader& h, int elem_size,
@@ -325,15 +607,56 @@ void decompress_slice_async(const uint8_t* d_comp, const CompSliceHeader& h, int
   const int64_t n = (int64_t)h.count;
   if (n == 0) return;
   if (h.bits == RAW_BITS) {
-    DJ_HIP_CALL(hipMemcpyAsync(d_out, d_comp + 16, (size_t)n * elem_size,
+    DJ_HIP_CALL(hipMemcpyAsync(d_out, d_comp + HDR, (size_t)n * elem_size,
                                hipMemcpyDeviceToDevice, s));
     return;
   }
-  int64_t* vals = (int64_t*)d_scratch;
+  CompScratch scr = carve_comp_scratch(d_scratch, n);
+  if (h.scheme & 2u) {
+    /* RLE: unpack lengths -> offsets; unpack values (+delta scan); expand */
+    const int64_t r = (int64_t)h.nruns;
+    const int64_t rchunks = (r + SCHUNK - 1) / SCHUNK;
+    hipLaunchKernelGGL(comp_unpack_kernel, dim3(cgrid(r)), dim3(CBLOCK), 0, s,
+                       (const uint32_t*)(d_comp + HDR), h.len_bits, r, scr.marks);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(scan64_partials_kernel, dim3(cgrid(rchunks)), dim3(CBLOCK), 0, s,
+                       scr.marks, r, rchunks, scr.partials);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(scan64_exclusive_kernel, dim3(1), dim3(1024), 0, s, scr.partials,
+                       rchunks);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(scan64_excl_finalize_kernel, dim3(cgrid(rchunks)), dim3(CBLOCK), 0, s,
+                       scr.marks, r, rchunks, scr.partials, scr.starts);
+    DJ_HIP_CALL(hipGetLastError());
+    hipLaunchKernelGGL(comp_unpack_kernel, dim3(cgrid(r)), dim3(CBLOCK), 0, s,
+                       (const uint32_t*)(d_comp + HDR + packed_bytes(r, h.len_bits)), h.bits,
+                       r, scr.vals);
+    DJ_HIP_CALL(hipGetLastError());
+    if (h.scheme & 1u) {
+      hipLaunchKernelGGL(scan64_partials_kernel, dim3(cgrid(rchunks)), dim3(CBLOCK), 0, s,
+                         scr.vals, r, rchunks, scr.partials);
+      DJ_HIP_CALL(hipGetLastError());
+      hipLaunchKernelGGL(scan64_exclusive_kernel, dim3(1), dim3(1024), 0, s, scr.partials,
+                         rchunks);
+      DJ_HIP_CALL(hipGetLastError());
+      hipLaunchKernelGGL((scan64_finalize_kernel<int64_t>), dim3(cgrid(rchunks)),
+                         dim3(CBLOCK), 0, s, scr.vals, r, rchunks, scr.partials, scr.vals);
+      DJ_HIP_CALL(hipGetLastError());
+    }
+    if (elem_size == 8)
+      hipLaunchKernelGGL((rle_expand_kernel<int64_t>), dim3(cgrid(n)), dim3(CBLOCK), 0, s,
+                         scr.vals, scr.starts, r, n, d_out);
+    else
+      hipLaunchKernelGGL((rle_expand_kernel<int32_t>), dim3(cgrid(n)), dim3(CBLOCK), 0, s,
+                         scr.vals, scr.starts, r, n, d_out);
+    DJ_HIP_CALL(hipGetLastError());
+    return;
+  }
+  int64_t* vals = scr.vals;
   int64_t nchunks = (n + SCHUNK - 1) / SCHUNK;
-  int64_t* partials = vals + n;
+  int64_t* partials = scr.partials;
   hipLaunchKernelGGL(comp_unpack_kernel, dim3(cgrid(n)), dim3(CBLOCK), 0, s,
-                     (const uint32_t*)(d_comp + 16), h.bits, n, vals);
+                     (const uint32_t*)(d_comp + HDR), h.bits, n, vals);
   DJ_HIP_CALL(hipGetLastError());
   if (h.scheme & 1u) {
     /* delta decode: inclusive scan */
@@ -343,11 +666,11 @@ void decompress_slice_async(const uint8_t* d_comp, const CompSliceHeader& h, int
     hipLaunchKernelGGL(scan64_exclusive_kernel, dim3(1), dim3(1024), 0, s, partials, nchunks);
     DJ_HIP_CALL(hipGetLastError());
     if (elem_size == 8)
-      hipLaunchKernelGGL((scan64_finalize_kernel<int64_t>), dim3(cgrid(nchunks)), dim3(CBLOCK),
-                         0, s, vals, n, nchunks, partials, d_out);
+      hipLaunchKernelGGL((scan64_finalize_kernel<int64_t>), dim3(cgrid(nchunks)),
+                         dim3(CBLOCK), 0, s, vals, n, nchunks, partials, d_out);
     else
-      hipLaunchKernelGGL((scan64_finalize_kernel<int32_t>), dim3(cgrid(nchunks)), dim3(CBLOCK),
-                         0, s, vals, n, nchunks, partials, d_out);
+      hipLaunchKernelGGL((scan64_finalize_kernel<int32_t>), dim3(cgrid(nchunks)),
+                         dim3(CBLOCK), 0, s, vals, n, nchunks, partials, d_out);
     DJ_HIP_CALL(hipGetLastError());
   } else {
     if (elem_size == 8)

@@ -191,12 +191,14 @@ const int64_t* key_as_i64(cudf::column_view col, DBuf& tmp)
 void validate_compression(std::vector<ColumnCompressionOptions> const& opts)
 {
   for (auto& o : opts) {
+    if (!o.children_compression_options.empty())
+      validate_compression(o.children_compression_options);
     if (o.compression_method == CompressionMethod::none) continue;
     if (o.compression_method == CompressionMethod::lz4)
       throw std::runtime_error("lz4 compression is not implemented (cascaded or none)");
-    if (o.cascaded_format.num_RLEs > 0)
-      throw std::runtime_error(
-        "cascaded RLE passes are not implemented (delta/bitpack only; dj_compress.hip)");
+    if (o.cascaded_format.num_RLEs < 0 || o.cascaded_format.num_RLEs > 1)
+      throw std::runtime_error("cascaded num_RLEs must be 0 or 1 (one RLE pass; "
+                               "dj_compress.hip wire format)");
     if (o.cascaded_format.num_deltas < 0 || o.cascaded_format.num_deltas > 1)
       throw std::runtime_error("cascaded num_deltas must be 0 or 1");
   }
@@ -407,10 +409,10 @@ std::vector<ColumnCompressionOptions> generate_none_compression_options(cudf::ta
 
 namespace {
 
-/* sampling selector over our executable cascaded schemes ({0|1 deltas} +
- * bitpack; RLE passes are not implemented — dj_compress.hip header): picks
- * the scheme with the smaller packed-bit total on a host-side sample, in
- * the role of nvcomp's CascadedSelector (reference compression.hpp:253-292,
+/* sampling selector over our executable cascaded schemes ({0|1 RLE} x
+ * {0|1 deltas} + bitpack): picks the scheme with the smaller estimated
+ * packed total on a host-side sample, in the role of nvcomp's
+ * CascadedSelector (reference compression.hpp:253-292,
  * compression.cpp:36-69). Selection is data-dependent, not parity-pinned
  * (nvcomp is absent). */
 nvcompCascadedFormatOpts select_cascaded_on_sample(const void* d_data, int64_t n, int esize)
@@ -428,21 +430,49 @@ nvcompCascadedFormatOpts select_cascaded_on_sample(const void* d_data, int64_t n
     for (int64_t i = 0; i < sample; i++) h[(size_t)i] = h32[(size_t)i];
   }
   auto zigzag = [](int64_t v) { return ((uint64_t)v << 1) ^ (uint64_t)(v >> 63); };
-  auto packed_bits = [&](bool delta) {
+  auto packed_bits = [&](const std::vector<int64_t>& v, bool delta) {
     uint64_t total = 0;
-    for (int64_t g = 0; g < sample; g += 32) {
+    const int64_t m = (int64_t)v.size();
+    for (int64_t g = 0; g < m; g += 32) {
       int w = 0;
-      const int64_t e = std::min<int64_t>(g + 32, sample);
+      const int64_t e = std::min<int64_t>(g + 32, m);
       for (int64_t i = g; i < e; i++) {
-        int64_t v = delta ? (i ? h[(size_t)i] - h[(size_t)i - 1] : h[0]) : h[(size_t)i];
-        uint64_t z = zigzag(v) | 1;
+        int64_t d = delta ? (i ? v[(size_t)i] - v[(size_t)i - 1] : v[0]) : v[(size_t)i];
+        uint64_t z = zigzag(d) | 1;
         w = std::max(w, 64 - __builtin_clzll(z));
       }
       total += (uint64_t)w * 32;
     }
     return total;
   };
-  o.num_deltas = packed_bits(true) < packed_bits(false) ? 1 : 0;
+  /* RLE candidate: run values + run lengths on the sample */
+  std::vector<int64_t> rvals, rlens;
+  rvals.reserve((size_t)sample);
+  for (int64_t i = 0; i < sample; i++) {
+    if (i == 0 || h[(size_t)i] != h[(size_t)i - 1]) {
+      rvals.push_back(h[(size_t)i]);
+      rlens.push_back(1);
+    } else {
+      rlens.back()++;
+    }
+  }
+  struct Cand {
+    int rle, delta;
+    uint64_t bits;
+  };
+  std::vector<Cand> cands;
+  cands.push_back({0, 0, packed_bits(h, false)});
+  cands.push_back({0, 1, packed_bits(h, true)});
+  if ((int64_t)rvals.size() * 2 < sample) {  // only worth considering on runs
+    uint64_t lbits = packed_bits(rlens, false);
+    cands.push_back({1, 0, packed_bits(rvals, false) + lbits});
+    cands.push_back({1, 1, packed_bits(rvals, true) + lbits});
+  }
+  const Cand* best = &cands[0];
+  for (const auto& c : cands)
+    if (c.bits < best->bits) best = &c;
+  o.num_RLEs = best->rle;
+  o.num_deltas = best->delta;
   return o;
 }
 
@@ -862,11 +892,10 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
     const uint8_t* src;       // uncompressed source buffer
     uint8_t* dst;             // uncompressed destination
     int esize;
-    int delta, bp;
+    int rles, delta, bp;
     const std::vector<int64_t>* soff;  // element offsets (source)
     const std::vector<int64_t>* roff;  // element offsets (destination)
     DBuf comp;                         // compressed send slices (bound-sized slots)
-    DBuf bits;                         // u32 per peer
     std::vector<size_t> slot;          // slot byte offsets (G+1)
     std::vector<int64_t> csize;        // compressed bytes per peer
     std::vector<int64_t> csend_off;    // prefix of csize (G+1)
@@ -883,6 +912,7 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
       c.src = (const uint8_t*)src;
       c.dst = (uint8_t*)dst;
       c.esize = esize;
+      c.rles = opt.cascaded_format.num_RLEs;
       c.delta = opt.cascaded_format.num_deltas;
       c.bp = opt.cascaded_format.use_bp;
       c.soff = so;
@@ -902,10 +932,17 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
     auto in = input_table.column(c);
     auto out = communicated_table.column(c);
     if (in.type().id() == cudf::type_id::STRING) {
-      /* sizes on the wire (may be compressed); chars never compressed
-       * (reference policy, compression.cpp:44-60) */
+      /* sizes on the wire take the OFFSETS CHILD's options — the reference
+       * applies children[0] to the offsets buffer (all_to_all_comm.cpp:
+       * 269-279); generate_auto_select emits parent=none, child[0]=cascaded,
+       * so using the parent here would silently send sizes uncompressed.
+       * Chars (children[1]) are never compressed (compression.cpp:44-60). */
+      const ColumnCompressionOptions& sizes_opt =
+        compression_options[c].children_compression_options.empty()
+          ? compression_options[c]
+          : compression_options[c].children_compression_options[0];
       add_buffer(strings->sizes_to_send[c].p, strings->sizes_received[c].p, 4, rows_soff,
-                 &recv_offsets, compression_options[c]);
+                 &recv_offsets, sizes_opt);
       plains.push_back(Plain{(const int8_t*)in.chars(), (int8_t*)out.chars(), 1,
                              &strings->send_char_offsets[c], &strings->recv_char_offsets[c]});
     } else {
@@ -917,6 +954,11 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
   /* compress send slices (compute stream), then read final sizes */
   int64_t max_recv_count = 0;
   if (!comps.empty()) {
+    int64_t max_send_count = 1;
+    for (auto& c : comps)
+      for (int i = 0; i < G; i++)
+        max_send_count = std::max(max_send_count, (*c.soff)[i + 1] - (*c.soff)[i]);
+    DBuf cscratch(dj::compress_scratch_bytes(max_send_count));
     for (auto& c : comps) {
       c.slot.resize(G + 1);
       size_t acc = 0;
@@ -927,25 +969,25 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
       }
       c.slot[G] = acc;
       c.comp = DBuf(acc);
-      c.bits = DBuf((size_t)G * 4);
       for (int i = 0; i < G; i++) {
         if (i == me && !include_self) continue;
         int64_t cnt = (*c.soff)[i + 1] - (*c.soff)[i];
-        dj::compress_slice_async(c.src + (*c.soff)[i] * c.esize, cnt, c.esize, c.delta, c.bp,
-                                 (uint8_t*)c.comp.p + c.slot[i], (uint32_t*)c.bits.p + i, st);
+        dj::compress_slice_async(c.src + (*c.soff)[i] * c.esize, cnt, c.esize, c.rles,
+                                 c.delta, c.bp, (uint8_t*)c.comp.p + c.slot[i], cscratch.p,
+                                 st);
       }
     }
     DJ_HIP_CALL(hipStreamSynchronize(st));
     for (auto& c : comps) {
-      std::vector<uint32_t> bits(G, 0);
-      DJ_HIP_CALL(hipMemcpy(bits.data(), c.bits.p, (size_t)G * 4, hipMemcpyDeviceToHost));
       c.csize.assign(G, 0);
       c.csend_off.assign(G + 1, 0);
       for (int i = 0; i < G; i++) {
-        int64_t cnt = (*c.soff)[i + 1] - (*c.soff)[i];
-        c.csize[i] = (i == me && !include_self)
-                       ? 0
-                       : (int64_t)dj::compressed_size_from_bits(cnt, c.esize, bits[i]);
+        if (!(i == me && !include_self)) {
+          dj::CompSliceHeader h;
+          DJ_HIP_CALL(hipMemcpy(&h, (uint8_t*)c.comp.p + c.slot[i], sizeof(h),
+                                hipMemcpyDeviceToHost));
+          c.csize[i] = (int64_t)dj::compressed_size_from_header(h, c.esize);
+        }
         c.csend_off[i + 1] = c.csend_off[i] + c.csize[i];
       }
       /* exchange compressed byte counts (communicate_sizes over bytes) */
@@ -1003,7 +1045,7 @@ void AllToAllCommunicator::launch_communication(cudf::mutable_table_view communi
         int64_t rcount = (*c.roff)[i + 1] - (*c.roff)[i];
         if (rbytes == 0 || rcount == 0) continue;
         dj::CompSliceHeader h;
-        DJ_HIP_CALL(hipMemcpy(&h, (uint8_t*)c.comp_recv.p + c.crecv_off[i], 16,
+        DJ_HIP_CALL(hipMemcpy(&h, (uint8_t*)c.comp_recv.p + c.crecv_off[i], sizeof(h),
                               hipMemcpyDeviceToHost));
         DJ_CHECK_ERROR((int64_t)h.count == rcount,
                        "cascaded: received slice count mismatch");

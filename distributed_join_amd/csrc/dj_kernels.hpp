@@ -185,17 +185,28 @@ void subpart_lists(const int64_t* d_keys, const int64_t* d_pay, const int64_t* d
                    int nseg, int PA, int F, const int64_t* d_group_base,
                    longlong2* d_out_pairs, int64_t* d_bucket_offsets, hipStream_t s);
 
-/* ----- cascaded codec (dj_compress.hip; wire format in its header) ----- */
+/* ----- cascaded codec (dj_compress.hip; wire format in its header) -----
+ * 32-byte slice header. scheme bit0 = delta, bit1 = RLE (values subslice
+ * and run-lengths subslice follow, each bitpacked; nruns/len_bits valid
+ * iff RLE). bits == 0xFFFF means stored raw (scheme ignored). */
 struct CompSliceHeader {
-  uint32_t bits;
-  uint32_t scheme;
-  uint64_t count;
+  uint32_t bits;      // value bit width (0xFFFF = raw)
+  uint32_t scheme;    // bit0 delta, bit1 RLE
+  uint64_t count;     // original element count
+  uint64_t nruns;     // RLE only
+  uint32_t len_bits;  // RLE only: run-length bit width
+  uint32_t reserved;
 };
 size_t compress_bound(int64_t count, int elem_size);
 size_t compress_scratch_bytes(int64_t count);
-void compress_slice_async(const void* d_in, int64_t count, int elem_size, int num_deltas,
-                          int use_bp, uint8_t* d_out, uint32_t* d_bits_tmp, hipStream_t s);
-size_t compressed_size_from_bits(int64_t count, int elem_size, uint32_t bits);
+/* num_rles > 0 requires d_scratch (compress_scratch_bytes(count)); the
+ * packed-vs-raw decision happens on-device, fully stream-ordered. The
+ * final 32 B header at d_out carries everything needed for
+ * compressed_size_from_header after one sync. */
+void compress_slice_async(const void* d_in, int64_t count, int elem_size, int num_rles,
+                          int num_deltas, int use_bp, uint8_t* d_out, void* d_scratch,
+                          hipStream_t s);
+size_t compressed_size_from_header(const CompSliceHeader& h, int elem_size);
 void decompress_slice_async(const uint8_t* d_comp, const CompSliceHeader& h, int elem_size,
                             void* d_out, void* d_scratch, hipStream_t s);
 

@@ -155,6 +155,12 @@ void dj_exchange_sizes(const int64_t* h_send_counts, int64_t* h_recv_counts);
  * returns 0 on success, 1 on payload mismatch */
 int dj_rccl_selftest(int64_t n);
 
+/* cascaded codec roundtrip (test hook): compress count elements of
+ * elem_size (4/8) from d_in with {num_rles, num_deltas, use_bp} cascaded
+ * passes, decompress into d_out, return the wire size in bytes */
+int64_t dj_compress_roundtrip(const void* d_in, int64_t count, int elem_size, int num_rles,
+                              int num_deltas, int use_bp, void* d_out);
+
 #ifdef __cplusplus
 }
 #endif
