@@ -40,13 +40,19 @@ def synth_orders(n_orders, seed, row0, nrows):
     i = np.arange(row0, row0 + nrows, dtype=np.int64)
     keys = i * 4 + (i % 3)
     pr_idx = (np.abs(keys * 2654435761) % 5).astype(np.int64)
-    sizes = np.array([len(PRIORITIES[k]) for k in pr_idx], dtype=np.int32)
+    lens = np.array([len(p) for p in PRIORITIES], dtype=np.int32)
+    sizes = lens[pr_idx]
     offsets = np.zeros(nrows + 1, dtype=np.int32)
     np.cumsum(sizes, out=offsets[1:])
-    chars = np.empty(offsets[-1], dtype=np.uint8)
-    table = [np.frombuffer(p, dtype=np.uint8) for p in PRIORITIES]
-    for r in range(nrows):
-        chars[offsets[r]:offsets[r + 1]] = table[pr_idx[r]]
+    # vectorized chars: pad each priority string to maxlen, gather the
+    # padded rows, then drop the padding columns with a per-row length mask
+    maxlen = int(lens.max())
+    padded = np.zeros((len(PRIORITIES), maxlen), dtype=np.uint8)
+    for i, p in enumerate(PRIORITIES):
+        padded[i, :len(p)] = np.frombuffer(p, dtype=np.uint8)
+    rows = padded[pr_idx]                              # nrows x maxlen
+    mask = np.arange(maxlen, dtype=np.int32)[None, :] < sizes[:, None]
+    chars = rows[mask]
     return keys, offsets, chars
 
 

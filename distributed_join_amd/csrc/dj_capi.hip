@@ -337,11 +337,13 @@ int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
  * kernel and must be re-joined by the caller (dj_bucket_local_join does this
  * via the global-table path; the pipelined C++ orchestration redoes the
  * whole batch). */
-void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
-                                  const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
-                                  int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
-                                  int64_t* d_out3, int64_t cap, int64_t* d_counter,
-                                  int* d_error, int* d_any_overflow, void* d_scratch)
+static void bucket_local_join_enqueue_impl(const int64_t* d_lk, const int64_t* d_lp,
+                                           int64_t ln, const int64_t* d_rk,
+                                           const int64_t* d_rp, int64_t rn, int64_t* d_out0,
+                                           int64_t* d_out1, int64_t* d_out2, int64_t* d_out3,
+                                           int64_t cap, int64_t* d_counter, int* d_error,
+                                           int* d_any_overflow, void* d_scratch,
+                                           bool force_compact)
 {
   if (ln == 0 || rn == 0) {
     DJ_HIP_CALL(hipMemsetAsync(d_any_overflow, 0, sizeof(int), stream()));
@@ -357,7 +359,7 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
    * to the 4096-slot table (cap 3072, covers the 800M single-GPU shape)
    * instead of overflowing every bucket into the per-bucket fallback */
   const int slots = (maxn / B > 1300) ? 4096 : 2048;
-  if (bucket_slack_mode(ln, rn, B)) {
+  if (!force_compact && bucket_slack_mode(ln, rn, B)) {
     const int64_t capBl = dj::slack_capB(ln, B), capBr = dj::slack_capB(rn, B);
     /* pad B to a multiple of 4 with zero-length buckets (the lds_join_slack
      * KBUK contract; carve_bucket_scratch sized llen/rlen for it) */
@@ -393,6 +395,17 @@ void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int6
   }
 }
 
+void dj_bucket_local_join_enqueue(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                                  const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
+                                  int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
+                                  int64_t* d_out3, int64_t cap, int64_t* d_counter,
+                                  int* d_error, int* d_any_overflow, void* d_scratch)
+{
+  bucket_local_join_enqueue_impl(d_lk, d_lp, ln, d_rk, d_rp, rn, d_out0, d_out1, d_out2,
+                                 d_out3, cap, d_counter, d_error, d_any_overflow, d_scratch,
+                                 false);
+}
+
 void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
                           const int64_t* d_rk, const int64_t* d_rp, int64_t rn,
                           int64_t* d_out0, int64_t* d_out1, int64_t* d_out2,
@@ -409,10 +422,27 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
   int any = 0;
   DJ_HIP_CALL(hipMemcpyAsync(&any, s.any_overflow, sizeof(int), hipMemcpyDeviceToHost, st));
   DJ_HIP_CALL(hipStreamSynchronize(st));
+  bool used_slack = bucket_slack_mode(ln, rn, B);
   if (any & 2) {
-    /* slack-partition overflow (skew beyond the pass-A slack): the bucketed
-     * data is incomplete -- redo the WHOLE join via the global-table path
-     * on the original inputs */
+    /* slack overflow (pass-A capA or per-bucket capB blown — duplicate-heavy
+     * keys inflate bucket variance beyond the Poisson slack model): the
+     * bucketed data is incomplete. First retry with the EXACT counted
+     * two-pass partition (compact layout, ~2x the partition cost); only if
+     * that also overflows its pass-A slack fall back to the global-table
+     * join (the last-resort path — it cost 75 ms/step on the TPC-H shape
+     * before this retry existed, gpurun_out/r2_tpch_prof2). */
+    DJ_HIP_CALL(hipMemsetAsync(d_counter, 0, sizeof(int64_t), st));
+    DJ_HIP_CALL(hipMemsetAsync(d_error, 0, sizeof(int), st));
+    bucket_local_join_enqueue_impl(d_lk, d_lp, ln, d_rk, d_rp, rn, d_out0, d_out1, d_out2,
+                                   d_out3, cap, d_counter, d_error, s.any_overflow,
+                                   d_scratch, true);
+    used_slack = false;
+    any = 0;
+    DJ_HIP_CALL(hipMemcpyAsync(&any, s.any_overflow, sizeof(int), hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+  }
+  if (any & 2) {
+    /* compact pass-A slack also blown: global-table redo on the inputs */
     DJ_HIP_CALL(hipMemsetAsync(d_counter, 0, sizeof(int64_t), st));
     DJ_HIP_CALL(hipMemsetAsync(d_error, 0, sizeof(int), st));
     int64_t nslots = dj::join_table_slots(ln);
@@ -426,7 +456,7 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
     return;
   }
   if (any) {
-    const bool slack = bucket_slack_mode(ln, rn, B);
+    const bool slack = used_slack;
     const int64_t capBl = slack ? dj::slack_capB(ln, B) : 0;
     const int64_t capBr = slack ? dj::slack_capB(rn, B) : 0;
     std::vector<uint32_t> flags((size_t)B);

@@ -1100,10 +1100,22 @@ std::unique_ptr<cudf::column> gather_string_column(cudf::column_view src, const 
   dj::gather_sizes(src.head<int32_t>(), d_idx, n, (int32_t*)sizes.p, st);
   DBuf off(((size_t)n + 1) * 4);
   DBuf scan_scratch(dj::offsets_from_sizes_scratch_bytes(n));
+  DBuf total64(8);
+  DJ_HIP_CALL(hipMemsetAsync(total64.p, 0, 8, st));
   dj::offsets_from_sizes((const int32_t*)sizes.p, n, (int32_t*)off.p, scan_scratch.p, st);
+  dj::sum_sizes_i64((const int32_t*)sizes.p, n, total64.i64(), st);
   int32_t total = 0;
+  int64_t tot64 = 0;
   DJ_HIP_CALL(hipMemcpyAsync(&total, (int32_t*)off.p + n, 4, hipMemcpyDeviceToHost, st));
+  DJ_HIP_CALL(hipMemcpyAsync(&tot64, total64.p, 8, hipMemcpyDeviceToHost, st));
   DJ_HIP_CALL(hipStreamSynchronize(st));
+  /* int32 offsets are the cudf convention this boundary keeps (the
+   * reference's cuDF 0.19 has the same 2^31 chars-per-column cap) —
+   * overflow must refuse loudly, not scribble (a gathered join output
+   * easily exceeds it: 600 M rows x ~7.6 B chars did) */
+  DJ_CHECK_ERROR(tot64 <= INT32_MAX,
+                 "string gather: output chars exceed 2^31 (int32 offsets limit; shrink "
+                 "the per-batch output with over_decom or more ranks)");
   auto col = std::make_unique<cudf::column>((cudf::size_type)n, (int64_t)total);
   DJ_HIP_CALL(hipMemcpyAsync(col->head(), off.p, ((size_t)n + 1) * 4,
                              hipMemcpyDeviceToDevice, st));

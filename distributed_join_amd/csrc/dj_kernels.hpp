@@ -164,6 +164,9 @@ void join_probe_pairs(const longlong2* d_rows, int64_t rn, const int64_t* d_tabl
 /* ----- strings-column kernels (dj_strings.hip; see its header comment for
  * the reference helpers each replaces) ----- */
 void sizes_from_offsets(const int32_t* d_offsets, int64_t n, int32_t* d_sizes, hipStream_t s);
+/* true 64-bit total of int32 sizes -> d_total (one i64; caller zeroes) —
+ * overflow guard for the int32 offsets convention */
+void sum_sizes_i64(const int32_t* d_sizes, int64_t n, int64_t* d_total, hipStream_t s);
 size_t offsets_from_sizes_scratch_bytes(int64_t n);
 void offsets_from_sizes(const int32_t* d_sizes, int64_t n, int32_t* d_offsets, void* d_scratch,
                         hipStream_t s);

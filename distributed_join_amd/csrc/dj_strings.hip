@@ -48,6 +48,33 @@ __global__ void sizes_from_offsets_kernel(const int32_t* __restrict__ offsets, i
   for (; i < n; i += stride) sizes[i] = offsets[i + 1] - offsets[i];
 }
 
+__global__ void sum_sizes_i64_kernel(const int32_t* __restrict__ sizes, int64_t n,
+                                     int64_t* __restrict__ total)
+{
+  __shared__ int64_t red[256];
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t acc = 0;
+  for (; i < n; i += stride) acc += sizes[i];
+  red[threadIdx.x] = acc;
+  __syncthreads();
+  for (int off = 128; off > 0; off >>= 1) {
+    if (threadIdx.x < (unsigned)off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd((unsigned long long*)total, (unsigned long long)red[0]);
+}
+
+void sum_sizes_i64(const int32_t* d_sizes, int64_t n, int64_t* d_total, hipStream_t s)
+{
+  int64_t b = (n + 255) / 256;
+  if (b > 1024) b = 1024;
+  if (b < 1) b = 1;
+  hipLaunchKernelGGL(sum_sizes_i64_kernel, dim3((uint32_t)b), dim3(256), 0, s, d_sizes, n,
+                     d_total);
+  DJ_HIP_CALL(hipGetLastError());
+}
+
 void sizes_from_offsets(const int32_t* d_offsets, int64_t n, int32_t* d_sizes, hipStream_t s)
 {
   if (n <= 0) return;

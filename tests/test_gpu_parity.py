@@ -139,6 +139,21 @@ def test_join_sparse_buckets_drain(dj):
     _join_parity(dj, lk, lp, rk, rp)
 
 
+def test_join_duplicate_variance_slack_overflow(dj):
+    # the TPC-H lineitem shape: probe keys with multiplicity ~16 inflate
+    # per-bucket variance beyond the Poisson slack model, so pass B's capB
+    # overflows (bit 2) and the join must redo through the exact compact
+    # path (dj_capi compact retry), not the slow global-table fallback
+    rng = np.random.RandomState(23)
+    n_build, n_probe = 250_000, 4_000_000
+    lk = np.arange(n_build, dtype=np.int64) * 4 + 1
+    lp = np.arange(n_build, dtype=np.int64)
+    oi = rng.randint(0, n_build, n_probe).astype(np.int64)
+    rk = oi * 4 + 1
+    rp = np.arange(n_probe, dtype=np.int64)
+    _join_parity(dj, lk, lp, rk, rp)
+
+
 def test_join_parity_selectivity_1(dj):
     n = 200_000
     bk, bp = oracle.gen_build(n)
