@@ -249,12 +249,12 @@ def main():
     dom_ms = phases[dom]["ms"]
     achieved = (alg[dom] / 1e9) / (dom_ms / 1e3) if dom_ms > 0 else None
     # measured HBM traffic per step: PMC calibration committed under
-    # profiles/r01_pmc_traffic.json (rocprofv3 --pmc FETCH_SIZE / WRITE_SIZE
+    # profiles/r02_pmc_traffic.json (rocprofv3 --pmc FETCH_SIZE / WRITE_SIZE
     # passes; FETCH x2 gfx950 correction — see that file)
     traffic = None
     if N == 1:  # coefficients were calibrated on the N=1 kernel set only
         try:
-            with open(os.path.join(REPO, "profiles", "r01_pmc_traffic.json")) as fh:
+            with open(os.path.join(REPO, "profiles", "r02_pmc_traffic.json")) as fh:
                 coeff = json.load(fh)["phase_bytes_per_input_row_pair"]
             traffic = coeff[dom] * (lrows + rrows)
         except Exception:
