@@ -441,6 +441,20 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
     DJ_HIP_CALL(hipMemcpyAsync(&any, s.any_overflow, sizeof(int), hipMemcpyDeviceToHost, st));
     DJ_HIP_CALL(hipStreamSynchronize(st));
   }
+  /* sentinel (-1) keys: every join path skips them, setting the flag in
+   * d_error; join them out-of-band at the end (cross product of the two
+   * sides' -1 rows) and clear the flag — legal int64 data, not an error
+   * (the reference's cudf::inner_join joins -1 normally). */
+  auto neg1_fixup = [&] {
+    int saw = 0;
+    DJ_HIP_CALL(hipMemcpyAsync(&saw, d_error, sizeof(int), hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    if (saw) {
+      DJ_HIP_CALL(hipMemsetAsync(d_error, 0, sizeof(int), st));
+      dj::neg1_cross_join(d_lk, d_lp, ln, d_rk, d_rp, rn, d_out0, d_out1, d_out2, d_out3,
+                          cap, d_counter, st);
+    }
+  };
   if (any & 2) {
     /* compact pass-A slack also blown: global-table redo on the inputs */
     DJ_HIP_CALL(hipMemsetAsync(d_counter, 0, sizeof(int64_t), st));
@@ -453,6 +467,7 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
                   d_counter);
     DJ_HIP_CALL(hipStreamSynchronize(st));
     dj_dfree(d_table);
+    neg1_fixup();
     return;
   }
   if (any) {
@@ -489,6 +504,7 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
       dj_dfree(d_table);
     }
   }
+  neg1_fixup();
 }
 
 int64_t dj_local_inner_join_global(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,

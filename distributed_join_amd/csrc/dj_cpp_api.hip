@@ -1267,7 +1267,7 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
     DJ_HIP_CALL(hipMemcpyAsync(&nout, d_cnt.p, 8, hipMemcpyDeviceToHost, st));
     DJ_HIP_CALL(hipMemcpyAsync(&err, d_err.p, 4, hipMemcpyDeviceToHost, st));
     DJ_HIP_CALL(hipStreamSynchronize(st));
-    DJ_CHECK_ERROR(err == 0, "join build: key equal to the empty sentinel (-1) is unsupported");
+    DJ_CHECK_ERROR(err == 0, "join: sentinel flag not cleared by the bucket path");
     if (nout > cap) {
       cap = nout;
       continue;
@@ -1602,9 +1602,7 @@ std::unique_ptr<cudf::table> distributed_inner_join_fused(
       int any_overflow;
     } meta;
     DJ_HIP_CALL(hipMemcpy(&meta, bt.meta.p, 16, hipMemcpyDeviceToHost));
-    DJ_CHECK_ERROR(meta.error == 0,
-                   "join build: key equal to the empty sentinel (-1) is unsupported");
-    if (bt.lrows == 0 || bt.rrows == 0 || meta.count == 0) {
+    if (bt.lrows == 0 || bt.rrows == 0) {
       std::vector<std::unique_ptr<cudf::column>> cols;
       for (int c = 0; c < 4; c++)
         cols.push_back(std::make_unique<cudf::column>(data_type(type_id::INT64),
@@ -1612,7 +1610,9 @@ std::unique_ptr<cudf::table> distributed_inner_join_fused(
       results.push_back(std::make_unique<cudf::table>(std::move(cols)));
       continue;
     }
-    if (meta.any_overflow || meta.count > bt.cap) {
+    /* meta.error = sentinel (-1) keys seen: redo through local_inner_join,
+     * whose bucket path joins them out-of-band (neg1_cross_join) */
+    if (meta.error || meta.any_overflow || meta.count > bt.cap) {
       results.push_back(local_inner_join(bt.lrecv->view(), bt.rrecv->view(), 0, 0));
       continue;
     }
@@ -1853,8 +1853,13 @@ std::unique_ptr<cudf::table> distributed_inner_join(
       int any_overflow;
     } meta;
     DJ_HIP_CALL(hipMemcpy(&meta, bt.meta.p, 16, hipMemcpyDeviceToHost));
-    DJ_CHECK_ERROR(meta.error == 0,
-                   "join build: key equal to the empty sentinel (-1) is unsupported");
+    if (meta.error) {
+      /* sentinel (-1) keys seen: redo through local_inner_join, whose
+       * bucket path joins them out-of-band (neg1_cross_join) */
+      batch_results.push_back(
+        local_inner_join(bt.lrecv->view(), bt.rrecv->view(), left_on[0], right_on[0]));
+      continue;
+    }
     if (bt.ln == 0 || bt.rn == 0 || meta.count == 0) {
       /* empty batch result with the output schema */
       std::vector<std::unique_ptr<cudf::column>> cols;

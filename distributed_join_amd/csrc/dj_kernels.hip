@@ -64,6 +64,91 @@ static int grid_for(int64_t n)
   return (int)blocks;
 }
 
+/* ---- sentinel-key (-1) fixup: the hash tables reserve -1 as the empty
+ * marker, so build rows with key == -1 are skipped by every join path (they
+ * set the saw-sentinel flag) and probe rows with -1 never match. The host
+ * then joins them out-of-band: collect both sides' -1 payloads, emit the
+ * cross product. Rare path: runs only when the flag fired. */
+__global__ void collect_neg1_kernel(const int64_t* __restrict__ keys,
+                                    const int64_t* __restrict__ pay, int64_t n,
+                                    int64_t* __restrict__ out_pay,
+                                    unsigned long long* __restrict__ count, int64_t cap)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    if (keys[i] == kEmptyKey) {
+      unsigned long long idx = atomicAdd(count, 1ull);
+      if ((int64_t)idx < cap) out_pay[idx] = pay ? pay[i] : i;
+    }
+  }
+}
+
+__global__ void emit_neg1_cross_kernel(const int64_t* __restrict__ lpay, int64_t n1,
+                                       const int64_t* __restrict__ rpay, int64_t n2,
+                                       int64_t base, int64_t* __restrict__ out0,
+                                       int64_t* __restrict__ out1,
+                                       int64_t* __restrict__ out2,
+                                       int64_t* __restrict__ out3, int64_t cap)
+{
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t total = n1 * n2;
+  for (; t < total; t += stride) {
+    int64_t idx = base + t;
+    if (idx < cap) {
+      out0[idx] = kEmptyKey;
+      out1[idx] = lpay[t / n2];
+      out2[idx] = kEmptyKey;
+      out3[idx] = rpay[t % n2];
+    }
+  }
+}
+
+void neg1_cross_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                     const int64_t* d_rk, const int64_t* d_rp, int64_t rn, int64_t* d_out0,
+                     int64_t* d_out1, int64_t* d_out2, int64_t* d_out3, int64_t cap,
+                     int64_t* d_counter, hipStream_t s)
+{
+  unsigned long long* d_cnt = nullptr;
+  DJ_HIP_CALL(hipMalloc(&d_cnt, 16));
+  DJ_HIP_CALL(hipMemsetAsync(d_cnt, 0, 16, s));
+  /* pass 1: count (cap 0 => no payload writes) */
+  hipLaunchKernelGGL(collect_neg1_kernel, dim3(grid_for(ln)), dim3(BLOCK), 0, s, d_lk, d_lp,
+                     ln, nullptr, d_cnt, 0);
+  hipLaunchKernelGGL(collect_neg1_kernel, dim3(grid_for(rn)), dim3(BLOCK), 0, s, d_rk, d_rp,
+                     rn, nullptr, d_cnt + 1, 0);
+  DJ_HIP_CALL(hipGetLastError());
+  unsigned long long h_cnt[2] = {0, 0};
+  DJ_HIP_CALL(hipMemcpyAsync(h_cnt, d_cnt, 16, hipMemcpyDeviceToHost, s));
+  DJ_HIP_CALL(hipStreamSynchronize(s));
+  const int64_t n1 = (int64_t)h_cnt[0], n2 = (int64_t)h_cnt[1];
+  if (n1 > 0 && n2 > 0) {
+    int64_t *d_lpv = nullptr, *d_rpv = nullptr;
+    DJ_HIP_CALL(hipMalloc(&d_lpv, (size_t)n1 * 8));
+    DJ_HIP_CALL(hipMalloc(&d_rpv, (size_t)n2 * 8));
+    DJ_HIP_CALL(hipMemsetAsync(d_cnt, 0, 16, s));
+    hipLaunchKernelGGL(collect_neg1_kernel, dim3(grid_for(ln)), dim3(BLOCK), 0, s, d_lk,
+                       d_lp, ln, d_lpv, d_cnt, n1);
+    hipLaunchKernelGGL(collect_neg1_kernel, dim3(grid_for(rn)), dim3(BLOCK), 0, s, d_rk,
+                       d_rp, rn, d_rpv, d_cnt + 1, n2);
+    DJ_HIP_CALL(hipGetLastError());
+    int64_t base = 0;
+    DJ_HIP_CALL(hipMemcpyAsync(&base, d_counter, 8, hipMemcpyDeviceToHost, s));
+    DJ_HIP_CALL(hipStreamSynchronize(s));
+    const int64_t total = n1 * n2;
+    hipLaunchKernelGGL(emit_neg1_cross_kernel, dim3(grid_for(total)), dim3(BLOCK), 0, s,
+                       d_lpv, n1, d_rpv, n2, base, d_out0, d_out1, d_out2, d_out3, cap);
+    DJ_HIP_CALL(hipGetLastError());
+    const int64_t newcount = base + total;
+    DJ_HIP_CALL(hipMemcpyAsync(d_counter, &newcount, 8, hipMemcpyHostToDevice, s));
+    DJ_HIP_CALL(hipStreamSynchronize(s));
+    DJ_HIP_CALL(hipFree(d_lpv));
+    DJ_HIP_CALL(hipFree(d_rpv));
+  }
+  DJ_HIP_CALL(hipFree(d_cnt));
+}
+
 void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s)
 {
   if (n <= 0) return;

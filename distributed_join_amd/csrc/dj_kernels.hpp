@@ -215,5 +215,13 @@ void decompress_slice_async(const uint8_t* d_comp, const CompSliceHeader& h, int
 
 /* ----- small utilities ----- */
 void fill_i64(int64_t* d_dst, int64_t value, int64_t n, hipStream_t s);
+/* out-of-band join of rows whose key equals the reserved empty sentinel
+ * (-1): every table path skips them (setting the saw-sentinel flag); this
+ * collects both sides' -1 payloads and appends their cross product to the
+ * output, advancing *d_counter. Host-synchronous; rare path. */
+void neg1_cross_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
+                     const int64_t* d_rk, const int64_t* d_rp, int64_t rn, int64_t* d_out0,
+                     int64_t* d_out1, int64_t* d_out2, int64_t* d_out3, int64_t cap,
+                     int64_t* d_counter, hipStream_t s);
 
 }  // namespace dj

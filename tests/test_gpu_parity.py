@@ -139,6 +139,33 @@ def test_join_sparse_buckets_drain(dj):
     _join_parity(dj, lk, lp, rk, rp)
 
 
+def test_join_sentinel_minus_one_keys(dj):
+    # key == -1 is the hash tables' reserved empty marker; legal int64 data
+    # must still join (reference cudf::inner_join joins -1 normally) — the
+    # paths skip them in-table and append the -1 cross product out-of-band
+    rng = np.random.RandomState(31)
+    n = 100_000
+    lk = rng.randint(0, 50_000, n).astype(np.int64)
+    lk[rng.choice(n, 40, replace=False)] = -1
+    lp = np.arange(n, dtype=np.int64)
+    rk = rng.randint(0, 50_000, n).astype(np.int64)
+    rk[rng.choice(n, 25, replace=False)] = -1
+    rp = np.arange(n, dtype=np.int64)
+    # 40 x 25 = 1000 extra cross-product rows on top of the normal matches
+    _join_parity(dj, lk, lp, rk, rp)
+
+
+def test_join_sentinel_only_one_side(dj):
+    # -1 on the probe side only: no matches for those rows, no error
+    n = 50_000
+    lk = np.arange(n, dtype=np.int64)
+    lp = np.arange(n, dtype=np.int64)
+    rk = np.arange(n, dtype=np.int64)
+    rk[:100] = -1
+    rp = np.arange(n, dtype=np.int64)
+    _join_parity(dj, lk, lp, rk, rp)
+
+
 def test_join_duplicate_variance_slack_overflow(dj):
     # the TPC-H lineitem shape: probe keys with multiplicity ~16 inflate
     # per-bucket variance beyond the Poisson slack model, so pass B's capB
