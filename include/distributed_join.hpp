@@ -33,6 +33,10 @@
 #include <memory>
 #include <vector>
 
+/* RESTRICTION vs the reference signature below: this build joins on exactly
+ * ONE key column — left_on/right_on of size != 1 throw at runtime (the hot
+ * path the reference benchmarks is single-key). nparts = join-group size x
+ * over_decom_factor caps at 1024. See INTEGRATION.md "Known restrictions". */
 std::unique_ptr<cudf::table> distributed_inner_join(
   cudf::table_view left,
   cudf::table_view right,
