@@ -53,8 +53,14 @@ def main():
     L.dj_set_device(local_rank % max(L.dj_device_count(), 1))
 
     from bench import rccl_bootstrap
-    id_bytes = rccl_bootstrap(dist, rank, world, L) if world > 1 else None
-    comm = dj.CppCommunicator(rank, world, id_bytes)
+    if world > 1:
+        id_bytes = rccl_bootstrap(dist, rank, world, L)
+    else:
+        id_bytes = np.zeros(L.dj_rccl_unique_id_bytes(), dtype=np.uint8)
+        L.dj_rccl_get_unique_id(id_bytes.ctypes.data)
+    # dj_all_to_all_i64 drives the raw RCCL comm (dj_comm_init), the same
+    # grouped ncclSend/Recv the Communicator path issues per peer
+    L.dj_comm_init(rank, world, id_bytes.ctypes.data)
 
     def barrier():
         L.dj_sync()
@@ -107,7 +113,7 @@ def main():
         recv.free()
         size *= 2
 
-    comm.destroy()
+    L.dj_comm_finalize()
     if dist is not None:
         dist.destroy_process_group()
 
