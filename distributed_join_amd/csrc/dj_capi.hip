@@ -41,6 +41,21 @@ struct TimedSpan {
   hipEvent_t start, stop;
 };
 std::vector<TimedSpan> g_spans;
+std::vector<hipEvent_t> g_event_pool;  // hipEventCreate costs ~10-20 us;
+                                       // per-phase-per-step creation showed
+                                       // up in the step time — reuse
+
+hipEvent_t timing_event()
+{
+  if (!g_event_pool.empty()) {
+    hipEvent_t e = g_event_pool.back();
+    g_event_pool.pop_back();
+    return e;
+  }
+  hipEvent_t e;
+  DJ_HIP_CALL(hipEventCreate(&e));
+  return e;
+}
 
 using PhaseScope = dj_timing::Scope;
 
@@ -60,8 +75,8 @@ void record_begin(int phase, hipStream_t s)
 {
   TimedSpan t;
   t.phase = phase;
-  DJ_HIP_CALL(hipEventCreate(&t.start));
-  DJ_HIP_CALL(hipEventCreate(&t.stop));
+  t.start = timing_event();
+  t.stop = timing_event();
   DJ_HIP_CALL(hipEventRecord(t.start, s));
   g_spans.push_back(t);
 }
@@ -418,11 +433,14 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
   hipStream_t st = stream();
   dj_bucket_local_join_enqueue(d_lk, d_lp, ln, d_rk, d_rp, rn, d_out0, d_out1, d_out2,
                                d_out3, cap, d_counter, d_error, s.any_overflow, d_scratch);
-  /* skew fallback: buckets whose build side exceeded the LDS row cap */
-  int any = 0;
+  /* skew fallback: buckets whose build side exceeded the LDS row cap.
+   * The sentinel flag rides the same sync (one host round trip per step). */
+  int any = 0, saw_neg1 = 0;
   DJ_HIP_CALL(hipMemcpyAsync(&any, s.any_overflow, sizeof(int), hipMemcpyDeviceToHost, st));
+  DJ_HIP_CALL(hipMemcpyAsync(&saw_neg1, d_error, sizeof(int), hipMemcpyDeviceToHost, st));
   DJ_HIP_CALL(hipStreamSynchronize(st));
   bool used_slack = bucket_slack_mode(ln, rn, B);
+  bool retried = false;
   if (any & 2) {
     /* slack overflow (pass-A capA or per-bucket capB blown — duplicate-heavy
      * keys inflate bucket variance beyond the Poisson slack model): the
@@ -437,6 +455,7 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
                                    d_out3, cap, d_counter, d_error, s.any_overflow,
                                    d_scratch, true);
     used_slack = false;
+    retried = true;
     any = 0;
     DJ_HIP_CALL(hipMemcpyAsync(&any, s.any_overflow, sizeof(int), hipMemcpyDeviceToHost, st));
     DJ_HIP_CALL(hipStreamSynchronize(st));
@@ -445,10 +464,12 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
    * d_error; join them out-of-band at the end (cross product of the two
    * sides' -1 rows) and clear the flag — legal int64 data, not an error
    * (the reference's cudf::inner_join joins -1 normally). */
-  auto neg1_fixup = [&] {
-    int saw = 0;
-    DJ_HIP_CALL(hipMemcpyAsync(&saw, d_error, sizeof(int), hipMemcpyDeviceToHost, st));
-    DJ_HIP_CALL(hipStreamSynchronize(st));
+  auto neg1_fixup = [&](bool reread) {
+    int saw = saw_neg1;
+    if (reread) { /* a retry path re-ran the join and may have re-flagged */
+      DJ_HIP_CALL(hipMemcpyAsync(&saw, d_error, sizeof(int), hipMemcpyDeviceToHost, st));
+      DJ_HIP_CALL(hipStreamSynchronize(st));
+    }
     if (saw) {
       DJ_HIP_CALL(hipMemsetAsync(d_error, 0, sizeof(int), st));
       dj::neg1_cross_join(d_lk, d_lp, ln, d_rk, d_rp, rn, d_out0, d_out1, d_out2, d_out3,
@@ -467,7 +488,7 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
                   d_counter);
     DJ_HIP_CALL(hipStreamSynchronize(st));
     dj_dfree(d_table);
-    neg1_fixup();
+    neg1_fixup(true);
     return;
   }
   if (any) {
@@ -504,7 +525,7 @@ void dj_bucket_local_join(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
       dj_dfree(d_table);
     }
   }
-  neg1_fixup();
+  neg1_fixup(retried);
 }
 
 int64_t dj_local_inner_join_global(const int64_t* d_lk, const int64_t* d_lp, int64_t ln,
@@ -539,8 +560,8 @@ void dj_timing_enable(int on) { g_timing_on = on != 0; }
 void dj_timing_reset(void)
 {
   for (auto& t : g_spans) {
-    DJ_HIP_CALL(hipEventDestroy(t.start));
-    DJ_HIP_CALL(hipEventDestroy(t.stop));
+    g_event_pool.push_back(t.start);
+    g_event_pool.push_back(t.stop);
   }
   g_spans.clear();
 }
