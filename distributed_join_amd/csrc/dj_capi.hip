@@ -279,6 +279,7 @@ struct BucketScratch {
   longlong2 *lpairs, *rpairs;  // bucketed {key,payload} pairs (final; slack
                                // layout B*capB when bucket_slack_mode)
   longlong2* tmp_pairs;        // pass-A staging (size max(ln,rn))
+  longlong2* tmp_pairs2;       // second table's pass-A staging (slack mode)
   int64_t *loff, *roff;        // int64[B+1] (compact mode)
   uint32_t *llen, *rlen;       // u32[B] (slack mode)
   int64_t* segoff;             // int64[PA+1]
@@ -305,6 +306,8 @@ BucketScratch carve_bucket_scratch(void* base, int64_t ln, int64_t rn, int B)
   s.lpairs = (longlong2*)take(lrows * 16);
   s.rpairs = (longlong2*)take(rrows * 16);
   s.tmp_pairs = (longlong2*)take((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);
+  s.tmp_pairs2 = slack ? (longlong2*)take((size_t)(PA * dj::slack_capA(maxn, PA)) * 16)
+                       : nullptr;  // pair-launch partition needs both live
   s.loff = (int64_t*)take((size_t)(B + 1) * 8);
   s.roff = (int64_t*)take((size_t)(B + 1) * 8);
   const size_t Bpad = ((size_t)B + 3) & ~(size_t)3;  // lds_join_slack KBUK pad
@@ -333,6 +336,7 @@ int64_t dj_bucket_join_scratch_bytes(int64_t ln, int64_t rn)
   add(lrows * 16);
   add(rrows * 16);
   add((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);  // pass-A slack staging (>= maxn)
+  if (slack) add((size_t)(PA * dj::slack_capA(maxn, PA)) * 16);  // second table (pair launch)
   add((size_t)(B + 1) * 8);
   add((size_t)(B + 1) * 8);
   const size_t Bpad = ((size_t)B + 3) & ~(size_t)3;
@@ -385,10 +389,13 @@ static void bucket_local_join_enqueue_impl(const int64_t* d_lk, const int64_t* d
     }
     {
       PhaseScope t(DJ_PHASE_BUCKET_SCATTER, st);
-      dj::bucket_partition2_slack(d_lk, d_lp, ln, B, s.tmp_pairs, s.totals, capBl, s.lpairs,
-                                  s.llen, d_any_overflow, st);
-      dj::bucket_partition2_slack(d_rk, d_rp, rn, B, s.tmp_pairs, s.totals, capBr, s.rpairs,
-                                  s.rlen, d_any_overflow, st);
+      /* both tables in one pass-A and one pass-B launch: the second table's
+       * blocks fill the first's tail wave (s.counts is free in slack mode
+       * and serves as the second cursor array) */
+      dj::bucket_partition2_slack_pair(d_lk, d_lp, ln, s.tmp_pairs, s.totals, capBl,
+                                       s.lpairs, s.llen, d_rk, d_rp, rn, s.tmp_pairs2,
+                                       s.counts, capBr, s.rpairs, s.rlen, B, d_any_overflow,
+                                       st);
     }
     PhaseScope t(DJ_PHASE_JOIN_FUSED, st);
     dj::lds_join_slack(s.lpairs, s.llen, capBl, s.rpairs, s.rlen, capBr, Bpad, slots, d_out0,

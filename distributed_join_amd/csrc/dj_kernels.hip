@@ -1317,10 +1317,10 @@ __device__ __forceinline__ void wave_excl_scan(const uint32_t* hist, uint32_t* b
  * under the store burst (this kernel runs 1 block/CU, so barriered phases
  * cannot overlap across blocks; measured 0.90 -> ~0.86 ms/table with the
  * wave scans, experiments/join_v4). */
-__global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
+__device__ __forceinline__ void slackA_body(
   const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int P,
   int64_t capA, uint32_t* __restrict__ gcursor, int* __restrict__ any_overflow,
-  longlong2* __restrict__ out_pairs)
+  longlong2* __restrict__ out_pairs, int bid, int nblocks)
 {
   constexpr int VPT = SLACK_TILE / BUCKET_THREADS;
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -1336,8 +1336,8 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
   const int tid = threadIdx.x;
   if (tid == 0) s_ovf = 0;
   if (tid < P) hist[tid] = 0;
-  const int64_t chunk = (n + gridDim.x - 1) / gridDim.x;
-  const int64_t start = (int64_t)blockIdx.x * chunk;
+  const int64_t chunk = (n + nblocks - 1) / nblocks;
+  const int64_t start = (int64_t)bid * chunk;
   const int64_t end = min(start + chunk, n);
   if (start >= end) return;
   __syncthreads();
@@ -1409,6 +1409,33 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
   if (tid == 0 && s_ovf) atomicOr(any_overflow, 2);
 }
 
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_kernel(
+  const int64_t* __restrict__ keys, const int64_t* __restrict__ pay, int64_t n, int P,
+  int64_t capA, uint32_t* __restrict__ gcursor, int* __restrict__ any_overflow,
+  longlong2* __restrict__ out_pairs)
+{
+  slackA_body(keys, pay, n, P, capA, gcursor, any_overflow, out_pairs, blockIdx.x,
+              gridDim.x);
+}
+
+/* both tables in ONE launch (halves the pass-A launches and lets the second
+ * table's blocks fill the first's tail wave — 1 block/CU at 512 blocks means
+ * two sequential waves per launch otherwise) */
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_pair_kernel(
+  const int64_t* __restrict__ keys0, const int64_t* __restrict__ pay0, int64_t n0,
+  int64_t capA0, uint32_t* __restrict__ gcursor0, longlong2* __restrict__ out0,
+  const int64_t* __restrict__ keys1, const int64_t* __restrict__ pay1, int64_t n1,
+  int64_t capA1, uint32_t* __restrict__ gcursor1, longlong2* __restrict__ out1, int P,
+  int* __restrict__ any_overflow)
+{
+  const int half = gridDim.x / 2;
+  if (blockIdx.x < half)
+    slackA_body(keys0, pay0, n0, P, capA0, gcursor0, any_overflow, out0, blockIdx.x, half);
+  else
+    slackA_body(keys1, pay1, n1, P, capA1, gcursor1, any_overflow, out1, blockIdx.x - half,
+                half);
+}
+
 /* on slack overflow the atomic cursor kept counting skipped rows; clamp the
  * lengths to capA so pass B stays in bounds (results are discarded — the
  * caller redoes the join on bit 2) */
@@ -1468,10 +1495,10 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack_kernel(
  * 2-block tile, and per-bucket limits are precomputed in LDS (glim) so the
  * flush bound check pays no per-row 64-bit multiply (experiments/join_v7). */
 constexpr int BTILE = 8192;  // pass-B staging tile (128 KiB, 1 block/CU)
-__global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_kernel(
+__device__ __forceinline__ void slackB_body(
   const longlong2* __restrict__ in_pairs, const uint32_t* __restrict__ seg_len, int64_t capA,
   int F, int64_t capB, longlong2* __restrict__ out_pairs, uint32_t* __restrict__ lens,
-  int* __restrict__ any_overflow)
+  int* __restrict__ any_overflow, int a)
 {
   constexpr int VPT = BTILE / BUCKET_THREADS; /* 8 */
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -1483,7 +1510,6 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_kernel(
   uint32_t* partials = glim + F; /* 16 */
   __shared__ int s_ovf;
   const int tid = threadIdx.x;
-  const int a = blockIdx.x;
   const int64_t s0 = (int64_t)a * capA;
   const int64_t s1 = s0 + seg_len[a];
   if (tid == 0) s_ovf = 0;
@@ -1562,6 +1588,70 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_kernel(
     lens[b] = len > (uint32_t)capB ? (uint32_t)capB : len;
   }
   if (tid == 0 && s_ovf) atomicOr(any_overflow, 2);
+}
+
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_kernel(
+  const longlong2* __restrict__ in_pairs, const uint32_t* __restrict__ seg_len, int64_t capA,
+  int F, int64_t capB, longlong2* __restrict__ out_pairs, uint32_t* __restrict__ lens,
+  int* __restrict__ any_overflow)
+{
+  slackB_body(in_pairs, seg_len, capA, F, capB, out_pairs, lens, any_overflow, blockIdx.x);
+}
+
+/* both tables in ONE launch (grid = 2*PA; see bucket_scatter_slack_pair) */
+__global__ __launch_bounds__(BUCKET_THREADS) void bucket_subpart_slack2_pair_kernel(
+  const longlong2* __restrict__ in0, const uint32_t* __restrict__ seg0, int64_t capA0,
+  int64_t capB0, longlong2* __restrict__ out0, uint32_t* __restrict__ lens0,
+  const longlong2* __restrict__ in1, const uint32_t* __restrict__ seg1, int64_t capA1,
+  int64_t capB1, longlong2* __restrict__ out1, uint32_t* __restrict__ lens1, int F,
+  int* __restrict__ any_overflow)
+{
+  const int half = gridDim.x / 2;
+  if (blockIdx.x < half)
+    slackB_body(in0, seg0, capA0, F, capB0, out0, lens0, any_overflow, blockIdx.x);
+  else
+    slackB_body(in1, seg1, capA1, F, capB1, out1, lens1, any_overflow, blockIdx.x - half);
+}
+
+/* both tables through ONE pass-A launch and ONE pass-B launch (each table
+ * gets half the grid; the second table's blocks fill the first's tail wave
+ * — the sequential two-launch form leaves two wave boundaries idle) */
+void bucket_partition2_slack_pair(const int64_t* d_k0, const int64_t* d_p0, int64_t n0,
+                                  longlong2* d_tmp0, uint32_t* d_cur0, int64_t capB0,
+                                  longlong2* d_out0, uint32_t* d_len0, const int64_t* d_k1,
+                                  const int64_t* d_p1, int64_t n1, longlong2* d_tmp1,
+                                  uint32_t* d_cur1, int64_t capB1, longlong2* d_out1,
+                                  uint32_t* d_len1, int B, int* d_any_overflow, hipStream_t s)
+{
+  DJ_CHECK_ERROR(n0 < (int64_t)UINT32_MAX && n1 < (int64_t)UINT32_MAX,
+                 "bucket_partition_slack: n must be < 2^32");
+  const int PA = bucket_groups_for(B);
+  const int F = B / PA;
+  const int64_t capA0 = slack_capA(n0, PA), capA1 = slack_capA(n1, PA);
+  DJ_CHECK_ERROR(PA >= 2 && PA <= 1024 && F >= 1 && F <= 1024,
+                 "bucket_partition_slack: B out of range (PA must be >= 2)");
+  DJ_CHECK_ERROR((int64_t)PA * capA0 + n0 < (int64_t)UINT32_MAX &&
+                   (int64_t)PA * capA1 + n1 < (int64_t)UINT32_MAX &&
+                   (int64_t)B * capB0 < (int64_t)UINT32_MAX &&
+                   (int64_t)B * capB1 < (int64_t)UINT32_MAX,
+                 "bucket_partition_slack: slack layout exceeds u32 row index");
+  size_t scatter_lds = SLACK_TILE * sizeof(longlong2) + 4 * (size_t)PA * sizeof(uint32_t) + 64;
+  size_t sub_lds = BTILE * sizeof(longlong2) + 4 * (size_t)F * sizeof(uint32_t) + 64;
+  DJ_HIP_CALL(hipMemsetAsync(d_cur0, 0, (size_t)PA * 4, s));
+  DJ_HIP_CALL(hipMemsetAsync(d_cur1, 0, (size_t)PA * 4, s));
+  hipLaunchKernelGGL(bucket_scatter_slack_pair_kernel, dim3(2 * BUCKET_BLOCKS),
+                     dim3(BUCKET_THREADS), scatter_lds, s, d_k0, d_p0, n0, capA0, d_cur0,
+                     d_tmp0, d_k1, d_p1, n1, capA1, d_cur1, d_tmp1, PA, d_any_overflow);
+  DJ_HIP_CALL(hipGetLastError());
+  hipLaunchKernelGGL(clamp_seglen_kernel, dim3((PA + 255) / 256), dim3(256), 0, s, d_cur0,
+                     PA, (uint32_t)capA0);
+  hipLaunchKernelGGL(clamp_seglen_kernel, dim3((PA + 255) / 256), dim3(256), 0, s, d_cur1,
+                     PA, (uint32_t)capA1);
+  DJ_HIP_CALL(hipGetLastError());
+  hipLaunchKernelGGL(bucket_subpart_slack2_pair_kernel, dim3(2 * PA), dim3(BUCKET_THREADS),
+                     sub_lds, s, d_tmp0, d_cur0, capA0, capB0, d_out0, d_len0, d_tmp1,
+                     d_cur1, capA1, capB1, d_out1, d_len1, F, d_any_overflow);
+  DJ_HIP_CALL(hipGetLastError());
 }
 
 void bucket_partition2_slack(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int B,

@@ -20,8 +20,9 @@
 
 namespace dj {
 
-/* int64 value marking an empty hash-table slot; build rows with this key are
- * rejected (loud error), see dj_kernels.hip. Set via 0xFF memset. */
+/* int64 value marking an empty hash-table slot; build rows carrying it are
+ * skipped in-table and joined out-of-band (neg1_cross_join below). Set via
+ * 0xFF memset. */
 constexpr int64_t kEmptyKey = -1;
 
 /* nparts = world_size x over_decom; the stable wave-ballot partition covers
@@ -47,7 +48,7 @@ void partition_scatter(const int64_t* d_keys, const int64_t* d_pay, int64_t n, i
                        void* d_scratch, int64_t* d_out_keys, int64_t* d_out_pay,
                        hipStream_t s);
 /* d_offsets: device array of nparts+1 int64 partition offsets (offsets[0]=0).
- * Stable: rows keep input order inside each partition. nparts <= 64. */
+ * Stable: rows keep input order inside each partition. nparts <= 1024. */
 void hash_partition(const int64_t* d_keys, const int64_t* d_pay, int64_t n, int nparts,
                     int hash_fn, uint32_t hash_seed, int64_t* d_out_keys, int64_t* d_out_pay,
                     int64_t* d_offsets, void* d_scratch, hipStream_t s);
@@ -134,6 +135,15 @@ void bucket_partition2_slack(const int64_t* d_keys, const int64_t* d_pay, int64_
                              longlong2* d_tmp_pairs, uint32_t* d_cursors, int64_t capB,
                              longlong2* d_out_pairs, uint32_t* d_lens, int* d_any_overflow,
                              hipStream_t s);
+/* both tables through one pass-A and one pass-B launch (tail-wave fill);
+ * needs a private tmp/cursor set per table */
+void bucket_partition2_slack_pair(const int64_t* d_k0, const int64_t* d_p0, int64_t n0,
+                                  longlong2* d_tmp0, uint32_t* d_cur0, int64_t capB0,
+                                  longlong2* d_out0, uint32_t* d_len0, const int64_t* d_k1,
+                                  const int64_t* d_p1, int64_t n1, longlong2* d_tmp1,
+                                  uint32_t* d_cur1, int64_t capB1, longlong2* d_out1,
+                                  uint32_t* d_len1, int B, int* d_any_overflow,
+                                  hipStream_t s);
 /* Fused per-bucket LDS build+probe over bucketed pair tables. table_slots:
  * 2048 (2 blocks/CU, bucket cap 1536 build rows = kJoinBucketRowCap) or
  * 4096 (1 block/CU, cap 3072 — for the fused wire path when the PA*F

@@ -129,7 +129,9 @@ def test_multirank_loopback():
                  ["4", "1", "400000", "2", "0"],   # 2-level hierarchy
                  ["4", "1", "400000", "1", "0"],   # nvl=1: full shuffle + local join
                  ["2", "2", "400000", "2", "1"],   # compressed wire
-                 ["4", "2", "37", "4", "1"]):      # tiny: empty slices/buckets
+                 ["4", "2", "37", "4", "1"],       # tiny: empty slices/buckets
+                 ["8", "1", "800000", "8", "0"],   # the driver's config-3 shape
+                 ["8", "1", "800000", "1", "0"]):  # config 3, reference default mode
         r = subprocess.run([exe] + args, capture_output=True, text=True, timeout=240)
         assert r.returncode == 0, " ".join(args) + "\n" + r.stdout + r.stderr
         assert "MULTIRANK OK" in r.stdout
