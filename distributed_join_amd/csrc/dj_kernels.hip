@@ -1001,9 +1001,9 @@ void subpart_lists(const int64_t* d_keys, const int64_t* d_pay, const int64_t* d
  * against the constant KBUK-1: the b >= B tail break, a dynamic flush-slot
  * register, and an in-range guard each put the kernel back at ~2.35 ms
  * (experiments/join_v7 factorial; the goto-drain structure below measured
- * 1.40 ms). KBUK == 1 (used for the compact wire path, whose B is not
- * padded) flushes every bucket — the r1 behavior, ~0.3 ms slower per 100 M
- * rows but contract-free. */
+ * 1.40 ms). KBUK == 1 (only for a compact B that is not a multiple of 4 —
+ * no current caller produces one) flushes every bucket — the r1 behavior,
+ * ~0.3 ms slower per 100 M rows but contract-free. */
 template <int SLOTS2, bool SLACK, int KBUK>
 __global__ __launch_bounds__(BUCKET_THREADS) void lds_join_kernel(
   const longlong2* __restrict__ lrows, const int64_t* __restrict__ loff,
@@ -1754,10 +1754,12 @@ static void lds_join_launch(const longlong2* d_lrows, const int64_t* d_loff,
 {
   DJ_CHECK_ERROR(table_slots == 2048 || table_slots == 4096,
                  "lds_join: table_slots must be 2048 or 4096");
-  /* slack path: KBUK = 4 (B padded to a multiple of 4 by the caller);
-   * compact path: KBUK = 1 (no padding contract, flush per bucket) */
+  /* KBUK = 4 whenever B is a multiple of 4 — slack callers pad to it, and
+   * every compact caller's B already is (bucket_count_for powers of two,
+   * fused-wire PA*F). KBUK = 1 (flush per bucket, ~0.3 ms/100M slower)
+   * only for an unaligned compact B. */
   const bool slack = d_llen != nullptr;
-  const int kbuk = slack ? 4 : 1;
+  const int kbuk = (slack || B % 4 == 0) ? 4 : 1;
   int64_t groups = ((int64_t)B + kbuk - 1) / kbuk;
   int grid = (int)(groups < 8192 ? groups : 8192);
   size_t lds =
@@ -1768,10 +1770,19 @@ static void lds_join_launch(const longlong2* d_lrows, const int64_t* d_loff,
                        d_out2, d_out3, cap, (unsigned long long*)d_counter, d_overflow_flags,
                        d_any_overflow, d_error);
   };
-  if (table_slots == 4096)
-    slack ? launch(lds_join_kernel<4096, true, 4>) : launch(lds_join_kernel<4096, false, 1>);
-  else
-    slack ? launch(lds_join_kernel<2048, true, 4>) : launch(lds_join_kernel<2048, false, 1>);
+  if (table_slots == 4096) {
+    if (slack)
+      launch(lds_join_kernel<4096, true, 4>);
+    else
+      kbuk == 4 ? launch(lds_join_kernel<4096, false, 4>)
+                : launch(lds_join_kernel<4096, false, 1>);
+  } else {
+    if (slack)
+      launch(lds_join_kernel<2048, true, 4>);
+    else
+      kbuk == 4 ? launch(lds_join_kernel<2048, false, 4>)
+                : launch(lds_join_kernel<2048, false, 1>);
+  }
   DJ_HIP_CALL(hipGetLastError());
 }
 
