@@ -1209,6 +1209,14 @@ PartitionedTable partition_table(cudf::table_view in, cudf::size_type key_col, i
 
 /* local inner join of two tables via the bucketed-LDS engine; returns
  * left-cols + right-cols with keys duplicated, row order unspecified */
+__global__ void narrow_i64_to_i32_kernel(const int64_t* __restrict__ src, int64_t n,
+                                         int32_t* __restrict__ dst)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) dst[i] = (int32_t)src[i];
+}
+
 std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table_view right,
                                               cudf::size_type left_on, cudf::size_type right_on)
 {
@@ -1288,8 +1296,25 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
                                                     (cudf::size_type)nout, o3.p));
       o3.p = nullptr;
     } else {
-      /* general: o1/o3 hold source row indices; gather every column */
-      auto gather_col = [&](cudf::column_view src, DBuf& idx) {
+      /* general: o1/o3 hold source row indices; gather every column —
+       * EXCEPT the key columns, whose joined values already sit in o0/o2
+       * (each random-line gather of 240 M rows costs ~5 ms; adopting the
+       * key buffer is free, narrowing to INT32 is a streaming copy) */
+      auto gather_col = [&](cudf::column_view src, DBuf& idx, DBuf* key_vals) {
+        if (key_vals != nullptr && src.type().id() != cudf::type_id::STRING) {
+          if (cudf::is_rep_int64(src.type())) {
+            auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout,
+                                                      key_vals->p);
+            key_vals->p = nullptr;
+            return col;
+          }
+          auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
+          if (nout > 0)
+            hipLaunchKernelGGL(narrow_i64_to_i32_kernel, dim3(grid_for_n(nout)),
+                               dim3(kBlock), 0, st, key_vals->i64(), nout,
+                               (int32_t*)col->head());
+          return col;
+        }
         if (src.type().id() == cudf::type_id::STRING)
           return gather_string_column(src, idx.i64(), nout);
         auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
@@ -1303,8 +1328,10 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
         }
         return col;
       };
-      for (cudf::size_type c = 0; c < ncl; c++) cols.push_back(gather_col(left.column(c), o1));
-      for (cudf::size_type c = 0; c < ncr; c++) cols.push_back(gather_col(right.column(c), o3));
+      for (cudf::size_type c = 0; c < ncl; c++)
+        cols.push_back(gather_col(left.column(c), o1, c == left_on ? &o0 : nullptr));
+      for (cudf::size_type c = 0; c < ncr; c++)
+        cols.push_back(gather_col(right.column(c), o3, c == right_on ? &o2 : nullptr));
       DJ_HIP_CALL(hipStreamSynchronize(st));
     }
     return std::make_unique<cudf::table>(std::move(cols));
@@ -1896,7 +1923,21 @@ std::unique_ptr<cudf::table> distributed_inner_join(
       cols.push_back(adopt(bt.o2));
       cols.push_back(adopt(bt.o3));
     } else {
-      auto gather_col = [&](cudf::column_view src, DBuf& idx) {
+      /* key columns adopt/narrow o0/o2 (joined key values) instead of a
+       * random-line gather — see local_inner_join's assembly */
+      auto gather_col = [&](cudf::column_view src, DBuf& idx, DBuf* key_vals) {
+        if (key_vals != nullptr && src.type().id() != cudf::type_id::STRING) {
+          if (cudf::is_rep_int64(src.type())) {
+            auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout,
+                                                      key_vals->p);
+            key_vals->p = nullptr;
+            return col;
+          }
+          auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
+          hipLaunchKernelGGL(narrow_i64_to_i32_kernel, dim3(grid_for_n(nout)), dim3(kBlock),
+                             0, st, key_vals->i64(), nout, (int32_t*)col->head());
+          return col;
+        }
         if (src.type().id() == cudf::type_id::STRING)
           return gather_string_column(src, idx.i64(), nout);
         auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)nout);
@@ -1909,9 +1950,11 @@ std::unique_ptr<cudf::table> distributed_inner_join(
         return col;
       };
       for (cudf::size_type c = 0; c < left.num_columns(); c++)
-        cols.push_back(gather_col(bt.lrecv->view().column(c), bt.o1));
+        cols.push_back(
+          gather_col(bt.lrecv->view().column(c), bt.o1, c == left_on[0] ? &bt.o0 : nullptr));
       for (cudf::size_type c = 0; c < right.num_columns(); c++)
-        cols.push_back(gather_col(bt.rrecv->view().column(c), bt.o3));
+        cols.push_back(
+          gather_col(bt.rrecv->view().column(c), bt.o3, c == right_on[0] ? &bt.o2 : nullptr));
       DJ_HIP_CALL(hipStreamSynchronize(st));
     }
     batch_results.push_back(std::make_unique<cudf::table>(std::move(cols)));
