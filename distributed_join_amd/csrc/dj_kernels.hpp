@@ -95,16 +95,19 @@ inline int64_t slack_capA(int64_t n, int PA)
   const int64_t m = n / PA;
   return m + m / 16 + 1024;
 }
-/* per-bucket slack capacity for the count-free pass B (mean + ~7 sigma,
- * rounded to 8): Poisson bucket occupancy overflows this with probability
- * ~1e-12 per bucket; on overflow bit 2 of any_overflow is set and the caller
- * redoes the join exactly (same contract as slack_capA) */
+/* per-bucket slack capacity for the count-free pass B (mean + 16 sqrt(mean),
+ * rounded to 8). 7 sigma covers Poisson (unique-ish keys) but duplicate keys
+ * inflate bucket variance by ~(1 + multiplicity): TPC-H lineitem (m ~ 4)
+ * overflowed 7 sigma EVERY step, wasting the slack attempt before the exact
+ * compact retry. 16 sqrt(lam) covers multiplicity up to ~8 for ~25% more
+ * slack memory; beyond that bit 2 of any_overflow still routes to the
+ * compact retry (same contract as slack_capA). */
 inline int64_t slack_capB(int64_t n, int B)
 {
   const int64_t lam = n / B < 1 ? 1 : n / B;
   int64_t s = 1;
   while (s * s < lam) s++;  // ceil(sqrt(lam))
-  int64_t cap = lam + 7 * s + 8;
+  int64_t cap = lam + 16 * s + 16;
   return (cap + 7) & ~(int64_t)7;
 }
 /* Two-level non-stable partition into B buckets (B = PA*F per
