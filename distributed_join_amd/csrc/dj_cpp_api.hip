@@ -1097,7 +1097,9 @@ std::unique_ptr<cudf::column> gather_string_column(cudf::column_view src, const 
 {
   hipStream_t st = dj_rt_stream();
   DBuf sizes((size_t)(n > 0 ? n : 1) * 4);
-  dj::gather_sizes(src.head<int32_t>(), d_idx, n, (int32_t*)sizes.p, st);
+  DBuf starts((size_t)(n > 0 ? n : 1) * 4);
+  dj::gather_sizes_starts(src.head<int32_t>(), d_idx, n, (int32_t*)sizes.p,
+                          (int32_t*)starts.p, st);
   DBuf off(((size_t)n + 1) * 4);
   DBuf scan_scratch(dj::offsets_from_sizes_scratch_bytes(n));
   DBuf total64(8);
@@ -1119,8 +1121,8 @@ std::unique_ptr<cudf::column> gather_string_column(cudf::column_view src, const 
   auto col = std::make_unique<cudf::column>((cudf::size_type)n, (int64_t)total);
   DJ_HIP_CALL(hipMemcpyAsync(col->head(), off.p, ((size_t)n + 1) * 4,
                              hipMemcpyDeviceToDevice, st));
-  dj::gather_chars(src.head<int32_t>(), (const uint8_t*)src.chars(), d_idx, n,
-                   (const int32_t*)col->head(), (uint8_t*)col->chars(), st);
+  dj::gather_chars_from_starts((const uint8_t*)src.chars(), (const int32_t*)starts.p, n,
+                               (const int32_t*)col->head(), (uint8_t*)col->chars(), st);
   DJ_HIP_CALL(hipStreamSynchronize(st));
   return col;
 }

@@ -183,6 +183,10 @@ void sum_sizes_i64(const int32_t* d_sizes, int64_t n, int64_t* d_total, hipStrea
 size_t offsets_from_sizes_scratch_bytes(int64_t n);
 void offsets_from_sizes(const int32_t* d_sizes, int64_t n, int32_t* d_offsets, void* d_scratch,
                         hipStream_t s);
+void gather_sizes_starts(const int32_t* d_src_off, const int64_t* d_idx, int64_t n,
+                         int32_t* d_sizes, int32_t* d_starts, hipStream_t s);
+void gather_chars_from_starts(const uint8_t* d_src_chars, const int32_t* d_starts, int64_t n,
+                              const int32_t* d_dst_off, uint8_t* d_dst_chars, hipStream_t s);
 void gather_sizes(const int32_t* d_src_off, const int64_t* d_idx, int64_t n, int32_t* d_sizes,
                   hipStream_t s);
 void gather_chars(const int32_t* d_src_off, const uint8_t* d_src_chars, const int64_t* d_idx,

@@ -83,19 +83,87 @@ void sizes_from_offsets(const int32_t* d_offsets, int64_t n, int32_t* d_sizes, h
   DJ_HIP_CALL(hipGetLastError());
 }
 
-/* ---- exclusive scan of int32 sizes -> int32 offsets[n+1], offsets[0]=0 ---- */
+/* ---- exclusive scan of int32 sizes -> int32 offsets[n+1], offsets[0]=0 ----
+ * Tile-coalesced: a block owns a 4096-element tile; loads/stores are
+ * lane-contiguous, the per-thread serial work happens in LDS. (The first
+ * version gave each THREAD a serial chunk — lane-adjacent addresses 8 KB
+ * apart, 64 lines per load instruction: the scan alone cost 5 ms of the
+ * TPC-H step, gpurun_out/r2_tpch_prof3.) Whole-column totals stay < 2^31
+ * (the int32 offsets cap, enforced upstream), so in-tile prefixes fit i32. */
 
-__global__ void scan_partials_kernel(const int32_t* __restrict__ sizes, int64_t n,
-                                     int64_t nchunks, int64_t* __restrict__ partials)
+constexpr int SCAN_T = 4096;             // elements per block tile (16 KiB LDS)
+constexpr int SSEG = SCAN_T / SBLOCK;    // serial segment per thread (16)
+
+__global__ void scan_tile_partials_kernel(const int32_t* __restrict__ sizes, int64_t n,
+                                          int64_t ntiles, int64_t* __restrict__ partials)
 {
-  int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; c < nchunks; c += stride) {
-    int64_t start = c * CHUNK;
-    int64_t end = min(start + (int64_t)CHUNK, n);
+  __shared__ int64_t red[SBLOCK];
+  for (int64_t c = blockIdx.x; c < ntiles; c += gridDim.x) {
+    const int64_t base = c * SCAN_T;
     int64_t acc = 0;
-    for (int64_t i = start; i < end; i++) acc += sizes[i];
-    partials[c] = acc;
+#pragma unroll
+    for (int k = 0; k < SSEG; k++) {
+      int64_t i = base + (int64_t)k * SBLOCK + threadIdx.x;
+      if (i < n) acc += sizes[i];
+    }
+    red[threadIdx.x] = acc;
+    __syncthreads();
+    for (int off = SBLOCK / 2; off > 0; off >>= 1) {
+      if (threadIdx.x < (unsigned)off) red[threadIdx.x] += red[threadIdx.x + off];
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) partials[c] = red[0];
+    __syncthreads();
+  }
+}
+
+__global__ void scan_tile_finalize_kernel(const int32_t* __restrict__ sizes, int64_t n,
+                                          int64_t ntiles,
+                                          const int64_t* __restrict__ partials,
+                                          int32_t* __restrict__ offsets)
+{
+  __shared__ int32_t tile[SCAN_T];
+  __shared__ int64_t segsum[SBLOCK + 1];
+  for (int64_t c = blockIdx.x; c < ntiles; c += gridDim.x) {
+    const int64_t base = c * SCAN_T;
+#pragma unroll
+    for (int k = 0; k < SSEG; k++) {
+      int64_t i = base + (int64_t)k * SBLOCK + threadIdx.x;
+      tile[(int64_t)k * SBLOCK + threadIdx.x] = (i < n) ? sizes[i] : 0;
+    }
+    __syncthreads();
+    /* in-place exclusive scan of this thread's contiguous LDS segment */
+    int32_t acc = 0;
+#pragma unroll
+    for (int v = 0; v < SSEG; v++) {
+      int32_t x = tile[threadIdx.x * SSEG + v];
+      tile[threadIdx.x * SSEG + v] = acc;
+      acc += x;
+    }
+    segsum[threadIdx.x] = acc;
+    __syncthreads();
+    /* exclusive scan of the 256 segment sums (Hillis-Steele in LDS) */
+    for (int off = 1; off < SBLOCK; off <<= 1) {
+      int64_t add = (threadIdx.x >= (unsigned)off) ? segsum[threadIdx.x - off] : 0;
+      __syncthreads();
+      segsum[threadIdx.x] += add;
+      __syncthreads();
+    }
+    if (threadIdx.x == 0) segsum[SBLOCK] = segsum[SBLOCK - 1];  // tile total
+    __syncthreads();
+    const int64_t tbase = partials[c];
+#pragma unroll
+    for (int k = 0; k < SSEG; k++) {
+      int64_t i = base + (int64_t)k * SBLOCK + threadIdx.x;
+      if (i < n) {
+        int j = (int)((int64_t)k * SBLOCK + threadIdx.x);
+        int64_t seg_excl = (j >= SSEG) ? segsum[j / SSEG - 1] : 0;
+        offsets[i] = (int32_t)(tbase + seg_excl + tile[j]);
+      }
+    }
+    if (threadIdx.x == 0 && base + SCAN_T >= n)
+      offsets[n] = (int32_t)(tbase + segsum[SBLOCK]);
+    __syncthreads();
   }
 }
 
@@ -125,24 +193,6 @@ __global__ void scan_partials_exclusive_kernel(int64_t* partials, int64_t nchunk
   }
 }
 
-__global__ void scan_finalize_kernel(const int32_t* __restrict__ sizes, int64_t n,
-                                     int64_t nchunks, const int64_t* __restrict__ partials,
-                                     int32_t* __restrict__ offsets)
-{
-  int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; c < nchunks; c += stride) {
-    int64_t start = c * CHUNK;
-    int64_t end = min(start + (int64_t)CHUNK, n);
-    int64_t acc = partials[c];
-    for (int64_t i = start; i < end; i++) {
-      offsets[i] = (int32_t)acc;
-      acc += sizes[i];
-    }
-    if (end == n) offsets[n] = (int32_t)acc;
-  }
-}
-
 size_t offsets_from_sizes_scratch_bytes(int64_t n)
 {
   int64_t nchunks = (n + CHUNK - 1) / CHUNK;
@@ -156,16 +206,17 @@ void offsets_from_sizes(const int32_t* d_sizes, int64_t n, int32_t* d_offsets, v
     DJ_HIP_CALL(hipMemsetAsync(d_offsets, 0, sizeof(int32_t), s));
     return;
   }
-  int64_t nchunks = (n + CHUNK - 1) / CHUNK;
-  int64_t* partials = (int64_t*)d_scratch;
-  hipLaunchKernelGGL(scan_partials_kernel, dim3(sgrid(nchunks)), dim3(SBLOCK), 0, s, d_sizes, n,
-                     nchunks, partials);
+  int64_t* partials = (int64_t*)d_scratch;  // scratch sized by CHUNK < SCAN_T
+  int64_t ntiles = (n + SCAN_T - 1) / SCAN_T;
+  int tgrid = (int)(ntiles < 2048 ? ntiles : 2048);
+  hipLaunchKernelGGL(scan_tile_partials_kernel, dim3(tgrid), dim3(SBLOCK), 0, s, d_sizes, n,
+                     ntiles, partials);
   DJ_HIP_CALL(hipGetLastError());
   hipLaunchKernelGGL(scan_partials_exclusive_kernel, dim3(1), dim3(1024), 0, s, partials,
-                     nchunks);
+                     ntiles);
   DJ_HIP_CALL(hipGetLastError());
-  hipLaunchKernelGGL(scan_finalize_kernel, dim3(sgrid(nchunks)), dim3(SBLOCK), 0, s, d_sizes, n,
-                     nchunks, partials, d_offsets);
+  hipLaunchKernelGGL(scan_tile_finalize_kernel, dim3(tgrid), dim3(SBLOCK), 0, s, d_sizes, n,
+                     ntiles, partials, d_offsets);
   DJ_HIP_CALL(hipGetLastError());
 }
 
@@ -189,6 +240,57 @@ void gather_sizes(const int32_t* d_src_off, const int64_t* d_idx, int64_t n, int
   if (n <= 0) return;
   hipLaunchKernelGGL(gather_sizes_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s, d_src_off, d_idx,
                      n, d_sizes);
+  DJ_HIP_CALL(hipGetLastError());
+}
+
+/* sizes AND source char starts in one pass over the random offsets (the
+ * 8 B line serves both); gather_chars_from_starts then streams the staged
+ * starts sequentially instead of re-walking idx + src_off at random —
+ * cuts the chars gather's line traffic roughly in half at TPC-H scale */
+__global__ void gather_sizes_starts_kernel(const int32_t* __restrict__ src_off,
+                                           const int64_t* __restrict__ idx, int64_t n,
+                                           int32_t* __restrict__ sizes,
+                                           int32_t* __restrict__ starts)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int64_t j = idx[i];
+    int32_t s0 = src_off[j];
+    sizes[i] = src_off[j + 1] - s0;
+    starts[i] = s0;
+  }
+}
+
+void gather_sizes_starts(const int32_t* d_src_off, const int64_t* d_idx, int64_t n,
+                         int32_t* d_sizes, int32_t* d_starts, hipStream_t s)
+{
+  if (n <= 0) return;
+  hipLaunchKernelGGL(gather_sizes_starts_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s,
+                     d_src_off, d_idx, n, d_sizes, d_starts);
+  DJ_HIP_CALL(hipGetLastError());
+}
+
+__global__ void gather_chars_from_starts_kernel(const uint8_t* __restrict__ src_chars,
+                                                const int32_t* __restrict__ starts, int64_t n,
+                                                const int32_t* __restrict__ dst_off,
+                                                uint8_t* __restrict__ dst_chars)
+{
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    int32_t s0 = starts[i];
+    int32_t d0 = dst_off[i], d1 = dst_off[i + 1];
+    for (int32_t k = 0; k < d1 - d0; k++) dst_chars[d0 + k] = src_chars[s0 + k];
+  }
+}
+
+void gather_chars_from_starts(const uint8_t* d_src_chars, const int32_t* d_starts, int64_t n,
+                              const int32_t* d_dst_off, uint8_t* d_dst_chars, hipStream_t s)
+{
+  if (n <= 0) return;
+  hipLaunchKernelGGL(gather_chars_from_starts_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s,
+                     d_src_chars, d_starts, n, d_dst_off, d_dst_chars);
   DJ_HIP_CALL(hipGetLastError());
 }
 
