@@ -796,6 +796,11 @@ AllToAllCommunicator::AllToAllCommunicator(
     }
     std::vector<int64_t> roff;
     communicate_sizes(soff, roff, comm_group, communicator);
+    /* the received column keeps the int32 offsets convention: refuse loudly
+     * if the gathered slices' chars exceed it (same cap as the reference's
+     * cuDF 0.19; the receiver-side scan would otherwise wrap) */
+    DJ_CHECK_ERROR(roff.empty() || roff.back() <= (int64_t)INT32_MAX,
+                   "all-to-all: received string chars exceed 2^31 (int32 offsets limit)");
     DBuf sizes((size_t)(n > 0 ? n : 1) * 4);
     dj::sizes_from_offsets(col.head<int32_t>(), n, (int32_t*)sizes.p, st);
     strings->send_char_offsets.push_back(std::move(soff));
