@@ -53,6 +53,26 @@ def test_gen_test_strings_matches_formula(dj):
     L.dj_dfree(ch_p)
 
 
+@pytest.mark.parametrize("n", [1, 2, 4095, 4096, 4097, 8193, 100_000])
+def test_offsets_scan_tile_boundaries(dj, n):
+    # the tile-coalesced sizes->offsets scan (dj_strings.hip) must be exact
+    # at and around its 4096-element tile boundary; offsets are the prefix
+    # sums of the analytical sizes len(k) = k % 7 + 1
+    keys = (np.arange(n, dtype=np.int64) * 13 + 5)
+    dk = dj.DeviceArray.from_numpy(keys)
+    off_p, ch_p, nb = dj.gen_test_strings(dk, n)
+    L = dj.lib()
+    off = np.empty(n + 1, dtype=np.int32)
+    L.dj_memcpy_d2h(off.ctypes.data, off_p, (n + 1) * 4)
+    sizes = (keys % 7 + 1).astype(np.int64)
+    want = np.zeros(n + 1, dtype=np.int64)
+    np.cumsum(sizes, out=want[1:])
+    assert (off == want).all()
+    assert nb == int(want[-1])
+    L.dj_dfree(off_p)
+    L.dj_dfree(ch_p)
+
+
 @pytest.mark.parametrize("over_decom", [1, 4])
 def test_string_payload_join(dj, comm, over_decom):
     # reference KAT (string_payload.cu run_test): multiples of 3 x multiples
