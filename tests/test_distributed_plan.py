@@ -87,10 +87,12 @@ def _all_to_all(arr, send_offsets, recv_counts):
 
 
 def _distributed_join_plan(rank, world):
-    # local slices of the global 200k x 200k tables
+    # local slices of the global 200k x 200k tables (ragged for odd worlds:
+    # the first n_global % world ranks take one extra row)
     n_global = 200_000
-    rows = n_global // world
-    row0 = rank * rows
+    base, rem = divmod(n_global, world)
+    rows = base + (1 if rank < rem else 0)
+    row0 = rank * base + min(rank, rem)
     bk, bp = oracle.gen_build(n_global, row0=row0, nrows=rows)
     pk, pp = oracle.gen_probe(n_global, n_global, selectivity=0.3, row0=row0, nrows=rows)
 
@@ -112,8 +114,11 @@ def _distributed_join_plan(rank, world):
 
 
 @pytest.mark.timeout(300)
-def test_distributed_join_plan_matches_global_join():
-    results = launch(_distributed_join_plan)
+@pytest.mark.parametrize("world", [2, 3])
+def test_distributed_join_plan_matches_global_join(world):
+    # world 3 covers the non-power-of-two paths: murmur % 3 placement and
+    # ragged rank slices (the reference's tests run arbitrary mpi_size)
+    results = launch(_distributed_join_plan, world=world)
     got = np.concatenate([results[r] for r in sorted(results)], axis=1)
     got = np.stack(oracle.sort_rows(*[got[i] for i in range(4)]))
 
