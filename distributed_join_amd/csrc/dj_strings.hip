@@ -271,6 +271,13 @@ void gather_sizes_starts(const int32_t* d_src_off, const int64_t* d_idx, int64_t
   DJ_HIP_CALL(hipGetLastError());
 }
 
+/* per row: load the 1..3 aligned u64 words covering [s0, s0+len) once and
+ * extract bytes from registers — a byte-loop re-loads the same line len
+ * times (len ≈ 8 per row at TPC-H/config-4 string sizes, so this quarters
+ * the load instructions; the byte STORES stay — adjacent rows share dst
+ * words, so wide stores would race on the shared bytes). Safe to read the
+ * full aligned window: hipMalloc allocations are >= 4 KiB aligned, so the
+ * 8 B-aligned words of any valid [s0, s0+len) range stay in-allocation. */
 __global__ void gather_chars_from_starts_kernel(const uint8_t* __restrict__ src_chars,
                                                 const int32_t* __restrict__ starts, int64_t n,
                                                 const int32_t* __restrict__ dst_off,
@@ -279,9 +286,21 @@ __global__ void gather_chars_from_starts_kernel(const uint8_t* __restrict__ src_
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
-    int32_t s0 = starts[i];
-    int32_t d0 = dst_off[i], d1 = dst_off[i + 1];
-    for (int32_t k = 0; k < d1 - d0; k++) dst_chars[d0 + k] = src_chars[s0 + k];
+    const int64_t s0 = (int64_t)starts[i];
+    const int64_t d0 = dst_off[i];
+    const int32_t len = dst_off[i + 1] - (int32_t)d0;
+    const uint64_t* words = (const uint64_t*)(src_chars + (s0 & ~(int64_t)7));
+    int off = (int)(s0 & 7);  // first byte's position in words[0]
+    uint64_t w = words[0];
+    int wi = 0;
+    for (int32_t k = 0; k < len; k++) {
+      const int b = off + k;
+      if ((b >> 3) != wi) {
+        wi = b >> 3;
+        w = words[wi];
+      }
+      dst_chars[d0 + k] = (uint8_t)(w >> ((b & 7) * 8));
+    }
   }
 }
 
