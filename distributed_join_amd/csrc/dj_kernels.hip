@@ -659,8 +659,16 @@ constexpr int SUB_BUCKETS = 256;            // pass-B fanout (fixed)
 int bucket_count_for(int64_t ln, int64_t rn)
 {
   int64_t maxn = ln > rn ? ln : rn;
+  int64_t bmax = 1048576;
+  if (const char* e = getenv("DJ_MAX_B")) {
+    /* TEST HOOK: cap B to trade pass-B flush-run length (F = B/PA) against
+     * join table size — e.g. 800M at B=524288 runs lam~1526 on the
+     * 4096-slot join but doubles pass-B's per-bucket run length */
+    int64_t v = atoll(e);
+    if (v >= 256 && v <= 1048576 && (v & (v - 1)) == 0) bmax = v;
+  }
   int64_t B = 256;
-  while (B < 1048576 && maxn / B > 800) B <<= 1;
+  while (B < bmax && maxn / B > 800) B <<= 1;
   return (int)B;
 }
 
