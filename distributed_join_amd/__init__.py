@@ -98,6 +98,8 @@ def lib():
                                                       vp, vp, vp, i64, i64, i32, i32], vp),
             "dj_cpp_distributed_inner_join_cols": ([vp, vp, i32, i64, vp, i32, i64,
                                                     i32, i32, i32, i32], vp),
+            "dj_cpp_distributed_inner_join_cols_multi": ([vp, vp, i32, i64, vp, i32, i64,
+                                                          vp, vp, i32, i32, i32], vp),
             "dj_cpp_distribute_collect_roundtrip_i64": ([vp, vp, vp, i64], vp),
             "dj_table_column_type": ([vp, i32], i32),
             "dj_table_column_chars": ([vp, i32], vp),
@@ -344,6 +346,28 @@ def cpp_distributed_inner_join_cols(comm, lcols, ln, rcols, rn, key_l=0, key_r=0
     t = lib().dj_cpp_distributed_inner_join_cols(
         comm.ptr, ctypes.cast(la, ctypes.c_void_p), len(lcols), ln,
         ctypes.cast(ra, ctypes.c_void_p), len(rcols), rn, key_l, key_r, over_decom, 0)
+    return table_to_numpy(t)
+
+
+def cpp_distributed_inner_join_cols_multi(comm, lcols, ln, rcols, rn, left_on, right_on,
+                                          over_decom=1):
+    """Composite-key join over column descriptors (left_on/right_on are
+    lists of key column indices; the single-key restriction lifts here)."""
+    def pack(cols):
+        arr = (ColDesc * len(cols))()
+        for i, c in enumerate(cols):
+            arr[i].type_id = c[0]
+            arr[i].data = c[1]
+            arr[i].chars = c[2] if len(c) > 2 else None
+            arr[i].chars_bytes = c[3] if len(c) > 3 else 0
+        return arr
+    la, ra = pack(lcols), pack(rcols)
+    lon = np.asarray(left_on, dtype=np.int32)
+    ron = np.asarray(right_on, dtype=np.int32)
+    t = lib().dj_cpp_distributed_inner_join_cols_multi(
+        comm.ptr, ctypes.cast(la, ctypes.c_void_p), len(lcols), ln,
+        ctypes.cast(ra, ctypes.c_void_p), len(rcols), rn,
+        lon.ctypes.data, ron.ctypes.data, len(lon), over_decom, 0)
     return table_to_numpy(t)
 
 

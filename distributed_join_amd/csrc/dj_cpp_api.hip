@@ -177,6 +177,54 @@ void sync_streams()
 
 /* int64 view of a key column (widening INT32; spec: dj_hash.h operates on
  * int64 keys — INT32 keys are widened, identity placement k % G preserved) */
+/* fused multi-column key: h_i = chain of mix64 over the key tuple — the
+ * single-key engine then partitions/joins on h as if it were the key, and
+ * the caller filters hash-collision false positives against the real
+ * columns afterwards (local_inner_join_multi). DJ_TEST_WEAK_FUSE collapses
+ * h to 4 bits so tests can exercise that filter deterministically. */
+__global__ void fuse_keys_kernel(const int64_t* __restrict__ c0, const int64_t* __restrict__ c1,
+                                 const int64_t* __restrict__ c2, const int64_t* __restrict__ c3,
+                                 uint32_t wide_mask /*bit c: col c is int64*/, int ncols,
+                                 int64_t n, int weak, int64_t* __restrict__ out)
+{
+  const int64_t* cols[4] = {c0, c1, c2, c3};
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    uint64_t h = 0x9e3779b97f4a7c15ull;
+    for (int c = 0; c < ncols; c++) {
+      int64_t v = (wide_mask >> c) & 1 ? cols[c][i] : (int64_t)((const int32_t*)cols[c])[i];
+      h = dj_mix64(h ^ (uint64_t)v);
+    }
+    out[i] = weak ? (int64_t)(h & 0xF) : (int64_t)h;
+  }
+}
+
+/* -> DBuf of n fused int64 keys for the given key columns (<= 4, int-rep) */
+DBuf fuse_keys(cudf::table_view t, std::vector<cudf::size_type> const& on, hipStream_t st)
+{
+  const int64_t n = t.num_rows();
+  DJ_CHECK_ERROR(on.size() >= 1 && on.size() <= 4,
+                 "multi-column join keys: 1..4 key columns supported");
+  const int64_t* ptrs[4] = {nullptr, nullptr, nullptr, nullptr};
+  uint32_t wide = 0;
+  for (size_t c = 0; c < on.size(); c++) {
+    cudf::column_view col = t.column(on[c]);
+    DJ_CHECK_ERROR(cudf::is_rep_int64(col.type()) || cudf::is_rep_int32(col.type()),
+                   "multi-column join keys must be integer-rep columns");
+    ptrs[c] = col.head<int64_t>();
+    if (cudf::is_rep_int64(col.type())) wide |= 1u << c;
+  }
+  static const int weak = getenv("DJ_TEST_WEAK_FUSE") ? 1 : 0;
+  DBuf out((size_t)(n > 0 ? n : 1) * 8);
+  if (n > 0) {
+    hipLaunchKernelGGL(fuse_keys_kernel, dim3(grid_for_n(n)), dim3(kBlock), 0, st, ptrs[0],
+                       ptrs[1], ptrs[2], ptrs[3], wide, (int)on.size(), n, weak, out.i64());
+    DJ_HIP_CALL(hipGetLastError());
+  }
+  return out;
+}
+
 const int64_t* key_as_i64(cudf::column_view col, DBuf& tmp)
 {
   if (cudf::is_rep_int64(col.type())) return col.head<int64_t>();
@@ -1141,17 +1189,20 @@ struct PartitionedTable {
   std::vector<cudf::size_type> offsets;  // nparts+1
 };
 
+/* ext_keys: when non-null, placement uses this caller-provided key array
+ * (fused multi-column keys) instead of column key_col */
 PartitionedTable partition_table(cudf::table_view in, cudf::size_type key_col, int nparts,
-                                 int hash_fn, uint32_t seed)
+                                 int hash_fn, uint32_t seed,
+                                 const int64_t* ext_keys = nullptr)
 {
   hipStream_t st = dj_rt_stream();
   const int64_t n = in.num_rows();
   DBuf key_tmp;
-  const int64_t* keys = key_as_i64(in.column(key_col), key_tmp);
+  const int64_t* keys = ext_keys ? ext_keys : key_as_i64(in.column(key_col), key_tmp);
   DBuf scratch(dj::hash_partition_scratch_bytes(n, nparts));
   DBuf d_off((size_t)(nparts + 1) * 8);
 
-  const bool fast2 = in.num_columns() == 2 &&
+  const bool fast2 = ext_keys == nullptr && in.num_columns() == 2 &&
                      in.column(0).type().id() == cudf::type_id::INT64 &&
                      in.column(1).type().id() == cudf::type_id::INT64;
 
@@ -1222,6 +1273,42 @@ __global__ void narrow_i64_to_i32_kernel(const int64_t* __restrict__ src, int64_
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (; i < n; i += stride) dst[i] = (int32_t)src[i];
+}
+
+/* multi-key collision filter: keep (li, ri) pairs whose REAL key tuples are
+ * equal (the bucket join matched on the fused hash; unequal tuples sharing a
+ * fused value are hash collisions to drop). Output order is unspecified
+ * (atomic append), as the join's own output order already is. */
+__global__ void filter_tuple_matches_kernel(
+  const int64_t* __restrict__ lc0, const int64_t* __restrict__ lc1,
+  const int64_t* __restrict__ lc2, const int64_t* __restrict__ lc3,
+  const int64_t* __restrict__ rc0, const int64_t* __restrict__ rc1,
+  const int64_t* __restrict__ rc2, const int64_t* __restrict__ rc3, uint32_t lwide,
+  uint32_t rwide, int nk, const int64_t* __restrict__ li, const int64_t* __restrict__ ri,
+  int64_t n, int64_t* __restrict__ out_li, int64_t* __restrict__ out_ri,
+  unsigned long long* __restrict__ count)
+{
+  const int64_t* lc[4] = {lc0, lc1, lc2, lc3};
+  const int64_t* rc[4] = {rc0, rc1, rc2, rc3};
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    const int64_t a = li[i], b = ri[i];
+    bool eq = true;
+    for (int c = 0; c < nk; c++) {
+      int64_t lv = (lwide >> c) & 1 ? lc[c][a] : (int64_t)((const int32_t*)lc[c])[a];
+      int64_t rv = (rwide >> c) & 1 ? rc[c][b] : (int64_t)((const int32_t*)rc[c])[b];
+      if (lv != rv) {
+        eq = false;
+        break;
+      }
+    }
+    if (eq) {
+      unsigned long long pos = atomicAdd(count, 1ull);
+      out_li[pos] = a;
+      out_ri[pos] = b;
+    }
+  }
 }
 
 std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table_view right,
@@ -1341,6 +1428,110 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
         cols.push_back(gather_col(right.column(c), o3, c == right_on ? &o2 : nullptr));
       DJ_HIP_CALL(hipStreamSynchronize(st));
     }
+    return std::make_unique<cudf::table>(std::move(cols));
+  }
+}
+
+/* multi-column-key local inner join: bucket-join on the fused key chain
+ * (fuse_keys) with row-index payloads, drop fused-hash collisions against
+ * the real columns, then gather all output columns — the single-key engine
+ * does the heavy lifting; only placement/equality semantics widen.
+ * (Reference: cudf::inner_join on arbitrary left_on/right_on,
+ * distributed_join.cpp:71-132.) */
+std::unique_ptr<cudf::table> local_inner_join_multi(cudf::table_view left,
+                                                    cudf::table_view right,
+                                                    std::vector<cudf::size_type> const& lon,
+                                                    std::vector<cudf::size_type> const& ron)
+{
+  DJ_CHECK_ERROR(lon.size() == ron.size(), "left_on/right_on must have equal length");
+  if (lon.size() == 1) return local_inner_join(left, right, lon[0], ron[0]);
+  hipStream_t st = dj_rt_stream();
+  const int64_t ln = left.num_rows(), rn = right.num_rows();
+  const cudf::size_type ncl = left.num_columns(), ncr = right.num_columns();
+
+  auto empty_col = [&](cudf::column_view v) {
+    if (v.type().id() == cudf::type_id::STRING)
+      return std::make_unique<cudf::column>((cudf::size_type)0, (int64_t)0);
+    return std::make_unique<cudf::column>(v.type(), (cudf::size_type)0);
+  };
+  auto make_empty = [&]() {
+    std::vector<std::unique_ptr<cudf::column>> cols;
+    for (cudf::size_type c = 0; c < ncl; c++) cols.push_back(empty_col(left.column(c)));
+    for (cudf::size_type c = 0; c < ncr; c++) cols.push_back(empty_col(right.column(c)));
+    return std::make_unique<cudf::table>(std::move(cols));
+  };
+  if (ln == 0 || rn == 0) return make_empty();
+
+  DBuf lfused = fuse_keys(left, lon, st);
+  DBuf rfused = fuse_keys(right, ron, st);
+  DBuf liota((size_t)ln * 8), riota((size_t)rn * 8);
+  hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(ln)), dim3(kBlock), 0, st, liota.i64(),
+                     ln);
+  hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(rn)), dim3(kBlock), 0, st, riota.i64(),
+                     rn);
+
+  DBuf scratch((size_t)dj_bucket_join_scratch_bytes(ln, rn));
+  DBuf d_err(16), d_cnt(16);
+  int64_t cap = std::max<int64_t>(rn + (rn >> 3), 1024);
+  for (;;) {
+    DBuf o0((size_t)cap * 8), o1((size_t)cap * 8), o2((size_t)cap * 8), o3((size_t)cap * 8);
+    DJ_HIP_CALL(hipMemsetAsync(d_err.p, 0, 4, st));
+    DJ_HIP_CALL(hipMemsetAsync(d_cnt.p, 0, 8, st));
+    dj_bucket_local_join(lfused.i64(), liota.i64(), ln, rfused.i64(), riota.i64(), rn,
+                         o0.i64(), o1.i64(), o2.i64(), o3.i64(), cap, d_cnt.i64(),
+                         (int*)d_err.p, scratch.p);
+    int64_t nout = 0;
+    int err = 0;
+    DJ_HIP_CALL(hipMemcpyAsync(&nout, d_cnt.p, 8, hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipMemcpyAsync(&err, d_err.p, 4, hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    DJ_CHECK_ERROR(err == 0, "join: sentinel flag not cleared by the bucket path");
+    if (nout > cap) {
+      cap = nout;
+      continue;
+    }
+    /* drop fused-hash collisions against the real key columns */
+    const int64_t* lptr[4] = {nullptr, nullptr, nullptr, nullptr};
+    const int64_t* rptr[4] = {nullptr, nullptr, nullptr, nullptr};
+    uint32_t lwide = 0, rwide = 0;
+    for (size_t c = 0; c < lon.size(); c++) {
+      lptr[c] = left.column(lon[c]).head<int64_t>();
+      rptr[c] = right.column(ron[c]).head<int64_t>();
+      if (cudf::is_rep_int64(left.column(lon[c]).type())) lwide |= 1u << c;
+      if (cudf::is_rep_int64(right.column(ron[c]).type())) rwide |= 1u << c;
+    }
+    DBuf li2((size_t)std::max<int64_t>(nout, 1) * 8);
+    DBuf ri2((size_t)std::max<int64_t>(nout, 1) * 8);
+    DJ_HIP_CALL(hipMemsetAsync(d_cnt.p, 0, 8, st));
+    if (nout > 0) {
+      hipLaunchKernelGGL(filter_tuple_matches_kernel, dim3(grid_for_n(nout)), dim3(kBlock),
+                         0, st, lptr[0], lptr[1], lptr[2], lptr[3], rptr[0], rptr[1],
+                         rptr[2], rptr[3], lwide, rwide, (int)lon.size(), o1.i64(), o3.i64(),
+                         nout, li2.i64(), ri2.i64(), (unsigned long long*)d_cnt.p);
+      DJ_HIP_CALL(hipGetLastError());
+    }
+    int64_t m = 0;
+    DJ_HIP_CALL(hipMemcpyAsync(&m, d_cnt.p, 8, hipMemcpyDeviceToHost, st));
+    DJ_HIP_CALL(hipStreamSynchronize(st));
+    if (m == 0) return make_empty();
+
+    std::vector<std::unique_ptr<cudf::column>> cols;
+    auto gather_col = [&](cudf::column_view src, DBuf& idx) {
+      if (src.type().id() == cudf::type_id::STRING)
+        return gather_string_column(src, idx.i64(), m);
+      auto col = std::make_unique<cudf::column>(src.type(), (cudf::size_type)m);
+      if (cudf::is_rep_int64(src.type()))
+        hipLaunchKernelGGL(gather_i64_kernel, dim3(grid_for_n(m)), dim3(kBlock), 0, st,
+                           src.head<int64_t>(), idx.i64(), m, (int64_t*)col->head());
+      else
+        hipLaunchKernelGGL(gather_i32_kernel, dim3(grid_for_n(m)), dim3(kBlock), 0, st,
+                           src.head<int32_t>(), idx.i64(), m, (int32_t*)col->head());
+      return col;
+    };
+    for (cudf::size_type c = 0; c < ncl; c++) cols.push_back(gather_col(left.column(c), li2));
+    for (cudf::size_type c = 0; c < ncr; c++)
+      cols.push_back(gather_col(right.column(c), ri2));
+    DJ_HIP_CALL(hipStreamSynchronize(st));
     return std::make_unique<cudf::table>(std::move(cols));
   }
 }
@@ -1694,8 +1885,9 @@ std::unique_ptr<cudf::table> distributed_inner_join(
   void* preallocated_pinned_buffer,
   int nvlink_domain_size)
 {
-  DJ_CHECK_ERROR(left_on.size() == 1 && right_on.size() == 1,
-                 "this build joins on a single key column (the hot-path shape)");
+  DJ_CHECK_ERROR(left_on.size() == right_on.size() && !left_on.empty(),
+                 "left_on/right_on must name the same number of key columns");
+  DJ_CHECK_ERROR(left_on.size() <= 4, "up to 4 join key columns supported");
   validate_compression(left_compression_options);
   validate_compression(right_compression_options);
 
@@ -1719,7 +1911,25 @@ std::unique_ptr<cudf::table> distributed_inner_join(
     left = l_ib->view();
     right = r_ib->view();
   }
-  if (nvl == 1) return local_inner_join(left, right, left_on[0], right_on[0]);
+  if (nvl == 1) {
+    if (left_on.size() > 1) return local_inner_join_multi(left, right, left_on, right_on);
+    return local_inner_join(left, right, left_on[0], right_on[0]);
+  }
+
+  if (left_on.size() > 1) {
+    /* multi-column keys take the shuffle + local-join route (placement on
+     * the fused key chain both sides, collisions filtered in the local
+     * join); the batched od pipeline stays single-key — the hot shape the
+     * reference benchmarks */
+    CommunicationGroup g2(nvl, 1, communicator->mpi_rank);
+    auto ls = shuffle_on(left, left_on, g2, communicator, left_compression_options,
+                         cudf::hash_id::HASH_MURMUR3, DJ_SEED_INTRA, report_timing,
+                         preallocated_pinned_buffer);
+    auto rs = shuffle_on(right, right_on, g2, communicator, right_compression_options,
+                         cudf::hash_id::HASH_MURMUR3, DJ_SEED_INTRA, report_timing,
+                         preallocated_pinned_buffer);
+    return local_inner_join_multi(ls->view(), rs->view(), left_on, right_on);
+  }
 
   const int G = nvl;
   CommunicationGroup group(nvl, 1, communicator->mpi_rank);
@@ -1986,11 +2196,19 @@ std::unique_ptr<cudf::table> shuffle_on(cudf::table_view const& input,
                                         bool report_timing,
                                         void* preallocated_pinned_buffer)
 {
-  DJ_CHECK_ERROR(on_columns.size() == 1, "this build shuffles on a single key column");
+  DJ_CHECK_ERROR(on_columns.size() >= 1, "shuffle_on: at least one key column");
   const int hash_fn =
     hash_function == cudf::hash_id::HASH_IDENTITY ? DJ_HASH_IDENTITY : DJ_HASH_MURMUR3;
-  PartitionedTable part =
-    partition_table(input, on_columns[0], comm_group.size(), hash_fn, hash_seed);
+  /* multi-column keys: placement on the fused key chain (our spec —
+   * MurmurHash3 placement is parity-unpinned) */
+  DBuf fused;
+  if (on_columns.size() > 1) {
+    DJ_CHECK_ERROR(hash_fn != DJ_HASH_IDENTITY,
+                   "identity-hash shuffle requires a single key column");
+    fused = fuse_keys(input, on_columns, dj_rt_stream());
+  }
+  PartitionedTable part = partition_table(input, on_columns[0], comm_group.size(), hash_fn,
+                                          hash_seed, fused.p ? fused.i64() : nullptr);
   AllToAllCommunicator atoa(part.tbl->view(), part.offsets, comm_group, communicator,
                             compression_options, false);
   auto out = atoa.allocate_communicated_table();
@@ -2388,6 +2606,38 @@ void* dj_cpp_distributed_inner_join_cols(void* comm, const dj_col_desc* lcols, i
   auto ropts = generate_compression_options_distributed(right, false);
   auto result = distributed_inner_join(left, right, {(cudf::size_type)key_l},
                                        {(cudf::size_type)key_r}, (Communicator*)comm, lopts,
+                                       ropts, over_decom, report_timing != 0, nullptr, 1);
+  return result.release();
+}
+
+/* multi-column-key variant: lon/ron are int32 arrays of nkeys column
+ * indices (distributed_inner_join with composite left_on/right_on) */
+void* dj_cpp_distributed_inner_join_cols_multi(void* comm, const dj_col_desc* lcols, int nl,
+                                               int64_t ln, const dj_col_desc* rcols, int nr,
+                                               int64_t rn, const int32_t* lon,
+                                               const int32_t* ron, int nkeys, int over_decom,
+                                               int report_timing)
+{
+  using cudf::column_view;
+  using cudf::data_type;
+  using cudf::type_id;
+  auto mk = [](const dj_col_desc* cols, int nc, int64_t n) {
+    std::vector<column_view> v;
+    for (int c = 0; c < nc; c++) {
+      if ((type_id)cols[c].type_id == type_id::STRING)
+        v.emplace_back(data_type(type_id::STRING), (cudf::size_type)n, cols[c].data,
+                       cols[c].chars, cols[c].chars_bytes);
+      else
+        v.emplace_back(data_type((type_id)cols[c].type_id), (cudf::size_type)n, cols[c].data);
+    }
+    return cudf::table_view(v);
+  };
+  auto left = mk(lcols, nl, ln);
+  auto right = mk(rcols, nr, rn);
+  std::vector<cudf::size_type> lo(lon, lon + nkeys), ro(ron, ron + nkeys);
+  auto lopts = generate_compression_options_distributed(left, false);
+  auto ropts = generate_compression_options_distributed(right, false);
+  auto result = distributed_inner_join(left, right, lo, ro, (Communicator*)comm, lopts,
                                        ropts, over_decom, report_timing != 0, nullptr, 1);
   return result.release();
 }
