@@ -33,10 +33,11 @@
 #include <memory>
 #include <vector>
 
-/* RESTRICTION vs the reference signature below: this build joins on exactly
- * ONE key column — left_on/right_on of size != 1 throw at runtime (the hot
- * path the reference benchmarks is single-key). nparts = join-group size x
- * over_decom_factor caps at 1024. See INTEGRATION.md "Known restrictions". */
+/* vs the reference signature below: up to 4 integer-rep key columns are
+ * supported (composite keys join on a fused key chain with a post-join
+ * collision filter and take the shuffle + local-join route); STRING key
+ * columns throw. nparts = join-group size x over_decom_factor caps at
+ * 1024. See INTEGRATION.md "Known restrictions". */
 std::unique_ptr<cudf::table> distributed_inner_join(
   cudf::table_view left,
   cudf::table_view right,
