@@ -1342,17 +1342,15 @@ std::unique_ptr<cudf::table> local_inner_join(cudf::table_view left, cudf::table
   const int64_t* rk = key_as_i64(right.column(right_on), rkey_tmp);
   const int64_t* lp;
   const int64_t* rp;
-  DBuf liota, riota;
   if (fast2) {
     lp = left.column(1).head<int64_t>();
     rp = right.column(1).head<int64_t>();
   } else {
-    liota = DBuf((size_t)ln * 8);
-    riota = DBuf((size_t)rn * 8);
-    hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(ln)), dim3(kBlock), 0, st, liota.i64(), ln);
-    hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(rn)), dim3(kBlock), 0, st, riota.i64(), rn);
-    lp = liota.i64();
-    rp = riota.i64();
+    /* general path payload = row index; the partition kernels synthesize it
+     * (pay == nullptr -> i), so no iota arrays to materialize or re-read
+     * (2.4 GB/step at the TPC-H shape) */
+    lp = nullptr;
+    rp = nullptr;
   }
 
   DBuf scratch((size_t)dj_bucket_join_scratch_bytes(ln, rn));
@@ -1464,11 +1462,6 @@ std::unique_ptr<cudf::table> local_inner_join_multi(cudf::table_view left,
 
   DBuf lfused = fuse_keys(left, lon, st);
   DBuf rfused = fuse_keys(right, ron, st);
-  DBuf liota((size_t)ln * 8), riota((size_t)rn * 8);
-  hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(ln)), dim3(kBlock), 0, st, liota.i64(),
-                     ln);
-  hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(rn)), dim3(kBlock), 0, st, riota.i64(),
-                     rn);
 
   DBuf scratch((size_t)dj_bucket_join_scratch_bytes(ln, rn));
   DBuf d_err(16), d_cnt(16);
@@ -1477,9 +1470,9 @@ std::unique_ptr<cudf::table> local_inner_join_multi(cudf::table_view left,
     DBuf o0((size_t)cap * 8), o1((size_t)cap * 8), o2((size_t)cap * 8), o3((size_t)cap * 8);
     DJ_HIP_CALL(hipMemsetAsync(d_err.p, 0, 4, st));
     DJ_HIP_CALL(hipMemsetAsync(d_cnt.p, 0, 8, st));
-    dj_bucket_local_join(lfused.i64(), liota.i64(), ln, rfused.i64(), riota.i64(), rn,
-                         o0.i64(), o1.i64(), o2.i64(), o3.i64(), cap, d_cnt.i64(),
-                         (int*)d_err.p, scratch.p);
+    dj_bucket_local_join(lfused.i64(), nullptr, ln, rfused.i64(), nullptr, rn, o0.i64(),
+                         o1.i64(), o2.i64(), o3.i64(), cap, d_cnt.i64(), (int*)d_err.p,
+                         scratch.p);
     int64_t nout = 0;
     int err = 0;
     DJ_HIP_CALL(hipMemcpyAsync(&nout, d_cnt.p, 8, hipMemcpyDeviceToHost, st));
@@ -2022,8 +2015,6 @@ std::unique_ptr<cudf::table> distributed_inner_join(
         bt.lkw = DBuf((size_t)bt.ln * 8);
       if (bt.rn && cudf::is_rep_int32(right.column(right_on[0]).type()))
         bt.rkw = DBuf((size_t)bt.rn * 8);
-      bt.liota = DBuf((size_t)std::max<int64_t>(bt.ln, 1) * 8);
-      bt.riota = DBuf((size_t)std::max<int64_t>(bt.rn, 1) * 8);
     }
   }
   DBuf scratch((size_t)dj_bucket_join_scratch_bytes(max_ln, max_rn));
@@ -2067,12 +2058,8 @@ std::unique_ptr<cudf::table> distributed_inner_join(
       };
       lk = widen_or_use(bt.lrecv->view().column(left_on[0]), bt.lkw);
       rk = widen_or_use(bt.rrecv->view().column(right_on[0]), bt.rkw);
-      hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(bt.ln)), dim3(kBlock), 0, st,
-                         bt.liota.i64(), bt.ln);
-      hipLaunchKernelGGL(iota_i64_kernel, dim3(grid_for_n(bt.rn)), dim3(kBlock), 0, st,
-                         bt.riota.i64(), bt.rn);
-      lp = bt.liota.i64();
-      rp = bt.riota.i64();
+      lp = nullptr;  // partition kernels synthesize the row-index payload
+      rp = nullptr;
     }
     DJ_HIP_CALL(hipMemsetAsync(bt.meta.p, 0, 16, st));
     dj_bucket_local_join_enqueue(lk, lp, bt.ln, rk, rp, bt.rn, bt.o0.i64(), bt.o1.i64(),
