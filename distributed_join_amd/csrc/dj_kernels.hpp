@@ -189,8 +189,6 @@ void gather_chars_from_starts(const uint8_t* d_src_chars, const int32_t* d_start
                               const int32_t* d_dst_off, uint8_t* d_dst_chars, hipStream_t s);
 void gather_sizes(const int32_t* d_src_off, const int64_t* d_idx, int64_t n, int32_t* d_sizes,
                   hipStream_t s);
-void gather_chars(const int32_t* d_src_off, const uint8_t* d_src_chars, const int64_t* d_idx,
-                  int64_t n, const int32_t* d_dst_off, uint8_t* d_dst_chars, hipStream_t s);
 void make_test_string_sizes(const int64_t* d_keys, int64_t n, int32_t* d_sizes, hipStream_t s);
 void fill_test_strings(const int64_t* d_keys, int64_t n, const int32_t* d_offsets,
                        uint8_t* d_chars, hipStream_t s);

@@ -313,31 +313,6 @@ void gather_chars_from_starts(const uint8_t* d_src_chars, const int32_t* d_start
   DJ_HIP_CALL(hipGetLastError());
 }
 
-__global__ void gather_chars_kernel(const int32_t* __restrict__ src_off,
-                                    const uint8_t* __restrict__ src_chars,
-                                    const int64_t* __restrict__ idx, int64_t n,
-                                    const int32_t* __restrict__ dst_off,
-                                    uint8_t* __restrict__ dst_chars)
-{
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (; i < n; i += stride) {
-    int64_t j = idx[i];
-    int32_t s0 = src_off[j], s1 = src_off[j + 1];
-    int32_t d0 = dst_off[i];
-    for (int32_t k = 0; k < s1 - s0; k++) dst_chars[d0 + k] = src_chars[s0 + k];
-  }
-}
-
-void gather_chars(const int32_t* d_src_off, const uint8_t* d_src_chars, const int64_t* d_idx,
-                  int64_t n, const int32_t* d_dst_off, uint8_t* d_dst_chars, hipStream_t s)
-{
-  if (n <= 0) return;
-  hipLaunchKernelGGL(gather_chars_kernel, dim3(sgrid(n)), dim3(SBLOCK), 0, s, d_src_off,
-                     d_src_chars, d_idx, n, d_dst_off, d_dst_chars);
-  DJ_HIP_CALL(hipGetLastError());
-}
-
 /* ---- deterministic test/bench string payload (string_payload.cu:50-94) ---- */
 
 __global__ void test_string_sizes_kernel(const int64_t* __restrict__ keys, int64_t n,
