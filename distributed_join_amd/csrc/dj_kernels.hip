@@ -1434,14 +1434,15 @@ __global__ __launch_bounds__(BUCKET_THREADS) void bucket_scatter_slack_pair_kern
   int64_t capA0, uint32_t* __restrict__ gcursor0, longlong2* __restrict__ out0,
   const int64_t* __restrict__ keys1, const int64_t* __restrict__ pay1, int64_t n1,
   int64_t capA1, uint32_t* __restrict__ gcursor1, longlong2* __restrict__ out1, int P,
+  int blocks0 /* grid split point, proportional to n0/(n0+n1) */,
   int* __restrict__ any_overflow)
 {
-  const int half = gridDim.x / 2;
-  if (blockIdx.x < half)
-    slackA_body(keys0, pay0, n0, P, capA0, gcursor0, any_overflow, out0, blockIdx.x, half);
+  if ((int)blockIdx.x < blocks0)
+    slackA_body(keys0, pay0, n0, P, capA0, gcursor0, any_overflow, out0, blockIdx.x,
+                blocks0);
   else
-    slackA_body(keys1, pay1, n1, P, capA1, gcursor1, any_overflow, out1, blockIdx.x - half,
-                half);
+    slackA_body(keys1, pay1, n1, P, capA1, gcursor1, any_overflow, out1,
+                blockIdx.x - blocks0, gridDim.x - blocks0);
 }
 
 /* on slack overflow the atomic cursor kept counting skipped rows; clamp the
@@ -1647,9 +1648,15 @@ void bucket_partition2_slack_pair(const int64_t* d_k0, const int64_t* d_p0, int6
   size_t sub_lds = BTILE * sizeof(longlong2) + 4 * (size_t)F * sizeof(uint32_t) + 64;
   DJ_HIP_CALL(hipMemsetAsync(d_cur0, 0, (size_t)PA * 4, s));
   DJ_HIP_CALL(hipMemsetAsync(d_cur1, 0, (size_t)PA * 4, s));
+  /* grid split proportional to table sizes (TPC-H joins 60M x 240M; an
+   * even split leaves the small half's chunks 4x lighter) */
+  int blocks0 = (int)(2.0 * BUCKET_BLOCKS * (double)n0 / (double)(n0 + n1) + 0.5);
+  if (blocks0 < 1) blocks0 = 1;
+  if (blocks0 > 2 * BUCKET_BLOCKS - 1) blocks0 = 2 * BUCKET_BLOCKS - 1;
   hipLaunchKernelGGL(bucket_scatter_slack_pair_kernel, dim3(2 * BUCKET_BLOCKS),
                      dim3(BUCKET_THREADS), scatter_lds, s, d_k0, d_p0, n0, capA0, d_cur0,
-                     d_tmp0, d_k1, d_p1, n1, capA1, d_cur1, d_tmp1, PA, d_any_overflow);
+                     d_tmp0, d_k1, d_p1, n1, capA1, d_cur1, d_tmp1, PA, blocks0,
+                     d_any_overflow);
   DJ_HIP_CALL(hipGetLastError());
   hipLaunchKernelGGL(clamp_seglen_kernel, dim3((PA + 255) / 256), dim3(256), 0, s, d_cur0,
                      PA, (uint32_t)capA0);
