@@ -147,6 +147,43 @@ print("WEAK_FUSE_OK", len(g[0]))
     assert "WEAK_FUSE_OK" in r.stdout
 
 
+def test_multikey_with_string_payload(dj, comm):
+    # composite keys with a STRING payload column on the left: the output
+    # assembly gathers the strings through the collision-filtered indices
+    rng = np.random.RandomState(9)
+    n = 50_000
+    lk0 = rng.randint(0, 2000, n).astype(np.int64)
+    lk1 = rng.randint(0, 2000, n).astype(np.int64)
+    rk0 = rng.randint(0, 2000, n).astype(np.int64)
+    rk1 = rng.randint(0, 2000, n).astype(np.int64)
+    rp = np.arange(n, dtype=np.int64)
+    L = dj.lib()
+    d = [dj.DeviceArray.from_numpy(a) for a in (lk0, lk1, rk0, rk1, rp)]
+    # string payload derived from lk0 (len = k%7+1, char = 'a'+k%26)
+    off_p, ch_p, nb = dj.gen_test_strings(d[0], n)
+    lcols = [(dj.TYPE_INT64, d[0].ptr), (dj.TYPE_INT64, d[1].ptr),
+             (dj.TYPE_STRING, off_p, ch_p, nb)]
+    rcols = [(dj.TYPE_INT64, d[2].ptr), (dj.TYPE_INT64, d[3].ptr),
+             (dj.TYPE_INT64, d[4].ptr)]
+    got = dj.cpp_distributed_inner_join_cols_multi(comm, lcols, n, rcols, n, [0, 1], [0, 1])
+    L.dj_dfree(off_p)
+    L.dj_dfree(ch_p)
+    # expected sizes + invariants: every output string must be the formula
+    # string of its left k0
+    comb_l = lk0 * (1 << 32) + lk1
+    comb_r = rk0 * (1 << 32) + rk1
+    lp = np.arange(n, dtype=np.int64)
+    c0, c1, c2, c3 = oracle.inner_join(comb_l, lp, comb_r, rp)
+    assert len(got[0]) == len(c0)
+    gk0 = np.asarray(got[0], dtype=np.int64)
+    goff, gch = got[2]
+    m = len(gk0)
+    for i in (0, 1, m // 2, m - 1) if m else ():
+        k = int(gk0[i])
+        want = bytes([ord('a') + k % 26]) * (k % 7 + 1)
+        assert gch[goff[i]:goff[i + 1]].tobytes() == want
+
+
 def test_multikey_shuffle_on(dj, comm):
     """shuffle_on with composite on_columns places by the fused key chain —
     at world 1 the shuffled table is a permutation of the input."""
